@@ -1,0 +1,972 @@
+"""The user-facing API: ``App``, ``@app.function``, Cls lifecycle, invocation verbs.
+
+This is the local MI355X-node implementation of the decorator/API surface the
+reference example corpus exercises.  Anchors per feature:
+
+- ``App``/``@app.function``/``@app.local_entrypoint`` — 01_getting_started/hello_world.py:18-57
+- ``.local/.remote/.map/.starmap/.spawn/.remote_gen/.for_each`` + ``.aio`` —
+  verb counts at SURVEY.md §1 L5; 08_advanced/parallel_execution.py:41
+- ``@app.cls`` + ``@modal.enter/@modal.exit/@modal.method`` —
+  06_gpu_and_ml/stable_diffusion/text_to_image.py:92-137
+- ``modal.parameter()`` — stable_diffusion/flux.py:126-128
+- ``@modal.batched`` — 03_scaling_out/dynamic_batching.py:29,57
+- ``@modal.concurrent`` — llm-serving/sglang_snapshot.py:260-261
+- ``with_options`` — 03_scaling_out/cls_with_options.py:57
+- ``modal.Retries`` — 06_gpu_and_ml/long-training.py:114-122
+- ``FunctionCall.from_id/gather`` — 08_advanced/poll_delayed_result.py:43-60
+"""
+from __future__ import annotations
+
+import asyncio
+import inspect
+import itertools
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Any, Callable, Dict, List, Optional
+
+from . import config
+from .exception import InvalidError, NotFoundError
+from .gpu.device_pool import DevicePool, parse_gpu
+from .runtime import ipc, store
+from .runtime.batching import Batcher
+from .runtime.pool import Call, Pool, RetryPolicy
+
+
+# ---------------------------------------------------------------- scheduling
+
+
+class Retries:
+    def __init__(self, max_retries: int = 2, backoff_coefficient: float = 2.0,
+                 initial_delay: float = 1.0, max_delay: float = 60.0):
+        self.policy = RetryPolicy(max_retries, initial_delay, backoff_coefficient, max_delay)
+
+
+class Period:
+    """``modal.Period(seconds=..., minutes=..., hours=..., days=...)``
+    (reference: 05_scheduling/schedule_simple.py:27)."""
+
+    def __init__(self, seconds: float = 0, minutes: float = 0, hours: float = 0,
+                 days: float = 0, weeks: float = 0):
+        self.total_seconds = seconds + 60 * minutes + 3600 * hours + 86400 * days + 604800 * weeks
+        if self.total_seconds <= 0:
+            raise InvalidError("Period must be positive")
+
+
+class Cron:
+    """5-field cron expression (reference: 05_scheduling/schedule_simple.py:34)."""
+
+    def __init__(self, expr: str):
+        parts = expr.split()
+        if len(parts) != 5:
+            raise InvalidError(f"cron expression needs 5 fields: {expr!r}")
+        self.expr = expr
+        self.fields = parts
+
+    def matches(self, t: time.struct_time) -> bool:
+        vals = [t.tm_min, t.tm_hour, t.tm_mday, t.tm_mon, t.tm_wday]
+        # cron dow: 0=Sunday; struct_time: 0=Monday
+        vals[4] = (t.tm_wday + 1) % 7
+        for spec, v in zip(self.fields, vals):
+            if not _cron_field_matches(spec, v):
+                return False
+        return True
+
+
+def _cron_field_matches(spec: str, v: int) -> bool:
+    for part in spec.split(","):
+        step = 1
+        if "/" in part:
+            part, _, s = part.partition("/")
+            step = int(s)
+        if part in ("*", ""):
+            if v % step == 0 or step == 1:
+                return True
+            continue
+        if "-" in part:
+            lo, _, hi = part.partition("-")
+            if int(lo) <= v <= int(hi) and (v - int(lo)) % step == 0:
+                return True
+        elif int(part) == v:
+            return True
+    return False
+
+
+# ---------------------------------------------------------------- runtime singleton
+
+
+class _Runtime:
+    """Process-wide scheduler state: the GPU pool + all function pools."""
+
+    _instance: Optional["_Runtime"] = None
+    _lock = threading.Lock()
+
+    def __init__(self):
+        self.device_pool = DevicePool()
+        self.pools: List[Pool] = []
+        self.device_pool.register_reclaim_hook(self._reclaim_idle)
+
+    @classmethod
+    def get(cls) -> "_Runtime":
+        with cls._lock:
+            if cls._instance is None:
+                cls._instance = cls()
+            return cls._instance
+
+    def _reclaim_idle(self):
+        for p in self.pools:
+            p.reap_idle(force=True)
+
+    def register(self, pool: Pool):
+        self.pools.append(pool)
+
+    def shutdown(self):
+        for p in self.pools:
+            try:
+                p.shutdown()
+            except Exception:
+                pass
+        self.pools.clear()
+
+    @classmethod
+    def reset(cls):
+        with cls._lock:
+            if cls._instance is not None:
+                cls._instance.shutdown()
+            cls._instance = None
+
+
+# ---------------------------------------------------------------- decorators (module level)
+
+
+def _set_flag(fn, **flags):
+    d = getattr(fn, "_modal_flags", None)
+    if d is None:
+        d = {}
+        fn._modal_flags = d
+    d.update(flags)
+    return fn
+
+
+def method(*, is_generator: bool = False):
+    def deco(fn):
+        return _set_flag(fn, method=True, is_generator=is_generator)
+    return deco
+
+
+def enter(*, snap: bool = False):
+    def deco(fn):
+        return _set_flag(fn, enter=True, snap=snap)
+    return deco
+
+
+def exit():  # noqa: A001 — mirrors modal.exit
+    def deco(fn):
+        return _set_flag(fn, exit=True)
+    return deco
+
+
+def batched(*, max_batch_size: int, wait_ms: int):
+    def deco(fn):
+        return _set_flag(fn, batched=True, max_batch_size=max_batch_size, wait_ms=wait_ms)
+    return deco
+
+
+def concurrent(*, max_inputs: int, target_inputs: Optional[int] = None):
+    def deco(obj):
+        return _set_flag(obj, concurrent=True, max_inputs=max_inputs,
+                         target_inputs=target_inputs or max_inputs)
+    return deco
+
+
+class _Parameter:
+    _counter = itertools.count()
+
+    def __init__(self, default=None, init: bool = True):
+        self.default = default
+        self.order = next(self._counter)
+
+
+def parameter(*, default=None, init: bool = True) -> Any:
+    """``modal.parameter()`` class-attribute marker (flux.py:126-128)."""
+    return _Parameter(default=default, init=init)
+
+
+def _iter_parameters(cls):
+    for name in dir(cls):
+        v = inspect.getattr_static(cls, name, None)
+        if isinstance(v, _Parameter):
+            yield name, v.default
+
+
+# ---------------------------------------------------------------- FunctionCall
+
+
+class FunctionCall:
+    """Handle to a spawned call; pollable by id across processes
+    (08_advanced/poll_delayed_result.py:43-60)."""
+
+    _live: Dict[str, Call] = {}
+
+    def __init__(self, call: Optional[Call] = None, object_id: Optional[str] = None):
+        self._call = call
+        self.object_id = object_id or (call.id if call else None)
+        if call is not None:
+            FunctionCall._live[call.id] = call
+
+    def get(self, timeout: Optional[float] = None):
+        if self._call is not None:
+            try:
+                return self._call.wait(timeout)
+            except TimeoutError:
+                raise TimeoutError(f"{self.object_id} still running")
+        found, ok, value, tb = store.get_result(self.object_id)
+        if not found:
+            if timeout == 0:
+                raise TimeoutError(f"{self.object_id} still running")
+            deadline = None if timeout is None else time.monotonic() + timeout
+            while True:
+                time.sleep(0.05)
+                found, ok, value, tb = store.get_result(self.object_id)
+                if found:
+                    break
+                if deadline is not None and time.monotonic() > deadline:
+                    raise TimeoutError(f"{self.object_id} still running")
+        if not ok:
+            raise value if isinstance(value, BaseException) else RuntimeError(str(value))
+        return value
+
+    async def get_aio(self, timeout: Optional[float] = None):
+        return await asyncio.to_thread(self.get, timeout)
+
+    def cancel(self):
+        pass  # local calls are not cancellable mid-flight yet
+
+    @staticmethod
+    def from_id(object_id: str) -> "FunctionCall":
+        call = FunctionCall._live.get(object_id)
+        return FunctionCall(call=call, object_id=object_id)
+
+    @staticmethod
+    def gather(*calls: "FunctionCall"):
+        return [c.get() for c in calls]
+
+
+def gather(*calls: FunctionCall):
+    """modal.functions.gather analog (08_advanced/parallel_execution.py:41)."""
+    return FunctionCall.gather(*calls)
+
+
+# ---------------------------------------------------------------- verbs
+
+
+class _Verb:
+    """Callable verb with an ``.aio`` async twin (``await f.remote.aio(x)``)."""
+
+    def __init__(self, sync_fn, aio_fn=None):
+        self._sync = sync_fn
+        self._aio = aio_fn
+
+    def __call__(self, *args, **kwargs):
+        return self._sync(*args, **kwargs)
+
+    @property
+    def aio(self):
+        if self._aio is not None:
+            return self._aio
+
+        async def default_aio(*args, **kwargs):
+            return await asyncio.to_thread(self._sync, *args, **kwargs)
+
+        return default_aio
+
+
+def _agen_from_sync(make_gen):
+    """Wrap a sync generator factory as an async generator (for map.aio)."""
+
+    async def agen(*args, **kwargs):
+        g = make_gen(*args, **kwargs)
+        sentinel = object()
+        while True:
+            item = await asyncio.to_thread(next, g, sentinel)
+            if item is sentinel:
+                return
+            yield item
+
+    return agen
+
+
+# ---------------------------------------------------------------- Function
+
+
+@dataclass
+class FunctionOptions:
+    gpu: Any = None
+    image: Any = None
+    volumes: dict = field(default_factory=dict)
+    secrets: list = field(default_factory=list)
+    timeout: Optional[float] = None
+    retries: Any = None
+    min_containers: int = 0
+    max_containers: Optional[int] = None
+    buffer_containers: int = 0
+    scaledown_window: Optional[float] = None
+    single_use_containers: bool = False
+    schedule: Any = None
+    enable_memory_snapshot: bool = False
+    experimental_options: dict = field(default_factory=dict)
+    region: Any = None
+    cpu: Any = None
+    memory: Any = None
+    ephemeral_disk: Any = None
+    cloud: Any = None
+    serialized: bool = False
+    name: Optional[str] = None
+
+    def merged(self, **overrides) -> "FunctionOptions":
+        import copy
+
+        out = copy.copy(self)
+        for k, v in overrides.items():
+            if v is not None:
+                setattr(out, k, v)
+        return out
+
+
+class Function:
+    """A decorated function bound to an autoscaling worker pool."""
+
+    def __init__(self, app: "App", raw: Callable, opts: FunctionOptions):
+        self.app = app
+        self.raw = raw
+        self.opts = opts
+        self.name = opts.name or getattr(raw, "__name__", "fn")
+        self._pool: Optional[Pool] = None
+        self._batcher: Optional[Batcher] = None
+        self._lock = threading.Lock()
+        flags = getattr(raw, "_modal_flags", {})
+        self.is_generator = inspect.isgeneratorfunction(raw)
+        self.is_batched = bool(flags.get("batched"))
+        self.max_inputs = int(flags.get("max_inputs", 1)) if flags.get("concurrent") else 1
+        self._flags = flags
+        # verbs
+        self.remote = _Verb(self._remote)
+        self.local = _Verb(self._local)
+        self.spawn = _Verb(self._spawn)
+        self.map = _Verb(self._map, _agen_from_sync(self._map))
+        self.starmap = _Verb(self._starmap, _agen_from_sync(self._starmap))
+        self.for_each = _Verb(self._for_each)
+        self.remote_gen = _Verb(self._remote_gen, _agen_from_sync(self._remote_gen))
+
+    def __call__(self, *args, **kwargs):
+        return self.raw(*args, **kwargs)
+
+    # -- infra --
+
+    def _get_pool(self) -> Pool:
+        with self._lock:
+            if self._pool is None:
+                rt = _Runtime.get()
+                gpu_count = parse_gpu(self.opts.gpu)
+                raw = self.raw
+                env = {}
+                for s in self.opts.secrets or []:
+                    env.update(getattr(s, "env", {}))
+                vols = {p: getattr(v, "name", str(v)) for p, v in (self.opts.volumes or {}).items()}
+                fn_name = self.name
+                app_name = self.app.name
+                max_inputs = self.max_inputs
+                target_blob = ipc.dumps(raw)
+                mem_snap = bool(self.opts.enable_memory_snapshot)
+                gpu_snap = bool((self.opts.experimental_options or {}).get("enable_gpu_snapshot"))
+
+                def make_spec(devices):
+                    return ipc.ServiceSpec(
+                        app_name=app_name,
+                        name=fn_name,
+                        target_blob=target_blob,
+                        max_inputs=max_inputs,
+                        gpu_devices=devices,
+                        env=dict(env),
+                        volumes=vols,
+                        enable_memory_snapshot=mem_snap,
+                        enable_gpu_snapshot=gpu_snap,
+                    )
+
+                retries = self.opts.retries
+                if isinstance(retries, Retries):
+                    policy = retries.policy
+                elif isinstance(retries, int):
+                    policy = RetryPolicy(max_retries=retries, initial_delay=1.0)
+                else:
+                    policy = RetryPolicy()
+                default_max = 16 if gpu_count == 0 else max(1, _Runtime.get().device_pool.n // max(1, gpu_count))
+                self._pool = Pool(
+                    name=f"{self.app.name}.{self.name}",
+                    make_spec=make_spec,
+                    gpu_count=gpu_count,
+                    min_containers=self.opts.min_containers,
+                    max_containers=self.opts.max_containers or default_max,
+                    buffer_containers=self.opts.buffer_containers,
+                    scaledown_window=self.opts.scaledown_window,
+                    timeout=self.opts.timeout,
+                    retries=policy,
+                    single_use_containers=self.opts.single_use_containers,
+                    max_inputs_per_worker=max_inputs,
+                    device_pool=rt.device_pool,
+                )
+                self._pool.on_spawned_result = _persist_spawned
+                rt.register(self._pool)
+                if self.is_batched:
+                    self._batcher = Batcher(
+                        self._pool,
+                        int(self._flags["max_batch_size"]),
+                        float(self._flags["wait_ms"]),
+                    )
+            return self._pool
+
+    def keep_warm(self, n: int):
+        self._get_pool().warm(n)
+
+    def update_autoscaler(self, min_containers=None, max_containers=None, buffer_containers=None):
+        p = self._get_pool()
+        if min_containers is not None:
+            p.min_containers = min_containers
+            p.warm(min_containers)
+        if max_containers is not None:
+            p.max_containers = max_containers
+        if buffer_containers is not None:
+            p.buffer_containers = buffer_containers
+
+    def with_options(self, **overrides) -> "Function":
+        return Function(self.app, self.raw, self.opts.merged(**overrides))
+
+    def get_web_url(self):
+        from .web.ingress import web_url_for
+
+        return web_url_for(self)
+
+    @property
+    def web_url(self):
+        return self.get_web_url()
+
+    # -- verbs --
+
+    def _submit(self, args, kwargs, is_gen=False, spawned=False) -> Call:
+        pool = self._get_pool()
+        if self.is_batched:
+            return self._batcher.enqueue(args, kwargs)
+        return pool.submit("", args, kwargs, is_gen=is_gen, spawned=spawned)
+
+    def _remote(self, *args, **kwargs):
+        if self.is_generator:
+            raise InvalidError(f"{self.name} is a generator; use .remote_gen()")
+        return self._submit(args, kwargs).wait()
+
+    def _remote_gen(self, *args, **kwargs):
+        call = self._submit(args, kwargs, is_gen=True)
+        while True:
+            alive, item = call.gen_q.get()
+            if not alive:
+                break
+            yield item
+        if call.exc is not None:
+            raise call.exc
+
+    def _local(self, *args, **kwargs):
+        return self.raw(*args, **kwargs)
+
+    def _spawn(self, *args, **kwargs) -> FunctionCall:
+        call = self._submit(args, kwargs, spawned=True)
+        return FunctionCall(call=call)
+
+    def _map(self, *input_iterators, order_outputs: bool = True,
+             return_exceptions: bool = False, ignore_exceptions: bool = False,
+             wrap_returned_exceptions: bool = False):
+        if len(input_iterators) == 1:
+            items = ((x,) for x in input_iterators[0])
+        else:
+            items = zip(*input_iterators)
+        yield from self._run_map(items, order_outputs, return_exceptions, ignore_exceptions)
+
+    def _starmap(self, input_iterator, order_outputs: bool = True,
+                 return_exceptions: bool = False, ignore_exceptions: bool = False):
+        items = (tuple(x) for x in input_iterator)
+        yield from self._run_map(items, order_outputs, return_exceptions, ignore_exceptions)
+
+    def _for_each(self, *input_iterators, ignore_exceptions: bool = False):
+        for _ in self._map(*input_iterators, order_outputs=False,
+                           ignore_exceptions=ignore_exceptions,
+                           return_exceptions=ignore_exceptions):
+            pass
+
+    def _run_map(self, items, order_outputs, return_exceptions, ignore_exceptions):
+        """Windowed fan-out: keeps ~2× pool capacity in flight for backpressure."""
+        pool = self._get_pool()
+        window = max(8, 2 * pool.max_containers * pool.max_inputs_per_worker)
+        inflight: List[Call] = []
+        items = iter(items)
+        exhausted = False
+
+        def fill():
+            nonlocal exhausted
+            while not exhausted and len(inflight) < window:
+                try:
+                    args = next(items)
+                except StopIteration:
+                    exhausted = True
+                    return
+                inflight.append(self._submit(args, {}))
+
+        fill()
+        while inflight:
+            if order_outputs:
+                call = inflight.pop(0)
+                call.event.wait()
+            else:
+                call = None
+                while call is None:
+                    for c in inflight:
+                        if c.done:
+                            call = c
+                            break
+                    if call is None:
+                        time.sleep(0.002)
+                inflight.remove(call)
+            fill()
+            if call.exc is not None:
+                if ignore_exceptions:
+                    continue
+                if return_exceptions:
+                    yield call.exc
+                    continue
+                raise call.exc
+            yield call.value
+
+
+def _persist_spawned(call: Call):
+    if call.exc is not None:
+        store.put_result(call.id, False, call.exc,
+                         getattr(call.exc, "remote_traceback", ""))
+    else:
+        store.put_result(call.id, True, call.value)
+
+
+# ---------------------------------------------------------------- Cls
+
+
+class Cls:
+    """``@app.cls`` wrapper. Calling it binds parameters → an _Obj whose methods
+    carry the invocation verbs.  (text_to_image.py:92-137 is the canonical use.)"""
+
+    def __init__(self, app: "App", user_cls: type, opts: FunctionOptions):
+        self.app = app
+        self.user_cls = user_cls
+        self.opts = opts
+        self.name = opts.name or user_cls.__name__
+        self._instances: Dict[frozenset, "_Obj"] = {}
+        cls_flags = getattr(user_cls, "_modal_flags", {})
+        self.cls_max_inputs = int(cls_flags.get("max_inputs", 1)) if cls_flags.get("concurrent") else 1
+
+    def __call__(self, **params) -> "_Obj":
+        key = frozenset(params.items())
+        inst = self._instances.get(key)
+        if inst is None:
+            inst = _Obj(self, params)
+            self._instances[key] = inst
+        return inst
+
+    def with_options(self, **overrides) -> "Cls":
+        return Cls(self.app, self.user_cls, self.opts.merged(**overrides))
+
+    @staticmethod
+    def from_name(app_name: str, name: str) -> "Cls":
+        c = App.registry_lookup_cls(app_name, name)
+        if c is None:
+            raise NotFoundError(f"class {app_name}/{name} not deployed locally")
+        return c
+
+
+class _Obj:
+    """A Cls instance (= parameter binding = its own autoscale pool,
+    hp_sweep_gpt.py:438-511)."""
+
+    def __init__(self, cls: Cls, params: dict):
+        object.__setattr__(self, "_cls", cls)
+        object.__setattr__(self, "_params", params)
+        object.__setattr__(self, "_pool", None)
+        object.__setattr__(self, "_methods", {})
+        object.__setattr__(self, "_local_obj", None)
+        object.__setattr__(self, "_lock", threading.Lock())
+
+    def _get_pool(self) -> Pool:
+        with self._lock:
+            if self._pool is None:
+                cls = self._cls
+                rt = _Runtime.get()
+                gpu_count = parse_gpu(cls.opts.gpu)
+                env = {}
+                for s in cls.opts.secrets or []:
+                    env.update(getattr(s, "env", {}))
+                target_blob = ipc.dumps(cls.user_cls)
+                params = dict(self._params)
+                # per-method concurrency: class-level @modal.concurrent
+                max_inputs = cls.cls_max_inputs
+                name = f"{cls.name}({','.join(f'{k}={v}' for k, v in sorted(params.items()))})" if params else cls.name
+                mem_snap = bool(cls.opts.enable_memory_snapshot)
+                gpu_snap = bool((cls.opts.experimental_options or {}).get("enable_gpu_snapshot"))
+
+                def make_spec(devices):
+                    return ipc.ServiceSpec(
+                        app_name=cls.app.name,
+                        name=name,
+                        target_blob=target_blob,
+                        is_cls=True,
+                        cls_params=params,
+                        max_inputs=max_inputs,
+                        gpu_devices=devices,
+                        env=dict(env),
+                        enable_memory_snapshot=mem_snap,
+                        enable_gpu_snapshot=gpu_snap,
+                    )
+
+                retries = cls.opts.retries
+                policy = retries.policy if isinstance(retries, Retries) else (
+                    RetryPolicy(max_retries=retries) if isinstance(retries, int) else RetryPolicy())
+                default_max = 16 if gpu_count == 0 else max(1, rt.device_pool.n // max(1, gpu_count))
+                pool = Pool(
+                    name=f"{cls.app.name}.{name}",
+                    make_spec=make_spec,
+                    gpu_count=gpu_count,
+                    min_containers=cls.opts.min_containers,
+                    max_containers=cls.opts.max_containers or default_max,
+                    buffer_containers=cls.opts.buffer_containers,
+                    scaledown_window=cls.opts.scaledown_window,
+                    timeout=cls.opts.timeout,
+                    retries=policy,
+                    single_use_containers=cls.opts.single_use_containers,
+                    max_inputs_per_worker=max_inputs,
+                    device_pool=rt.device_pool,
+                )
+                pool.on_spawned_result = _persist_spawned
+                rt.register(pool)
+                object.__setattr__(self, "_pool", pool)
+            return self._pool
+
+    def _local_instance(self):
+        if self._local_obj is None:
+            cls = self._cls.user_cls
+            obj = cls()
+            for pname, default in _iter_parameters(cls):
+                setattr(obj, pname, self._params.get(pname, default))
+            for name in dir(cls):
+                fn = getattr(cls, name, None)
+                flags = getattr(fn, "_modal_flags", None)
+                if flags and flags.get("enter"):
+                    getattr(obj, name)()
+            object.__setattr__(self, "_local_obj", obj)
+        return self._local_obj
+
+    def __getattr__(self, item):
+        cls = self._cls
+        fn = getattr(cls.user_cls, item, None)
+        if fn is None:
+            raise AttributeError(item)
+        flags = getattr(fn, "_modal_flags", None)
+        if not flags:
+            raise AttributeError(f"{item} is not a @modal.method")
+        m = self._methods.get(item)
+        if m is None:
+            m = _BoundMethod(self, item, fn, flags)
+            self._methods[item] = m
+        return m
+
+    def keep_warm(self, n: int):
+        self._get_pool().warm(n)
+
+    def update_autoscaler(self, min_containers=None, max_containers=None, buffer_containers=None):
+        p = self._get_pool()
+        if min_containers is not None:
+            p.min_containers = min_containers
+            p.warm(min_containers)
+        if max_containers is not None:
+            p.max_containers = max_containers
+        if buffer_containers is not None:
+            p.buffer_containers = buffer_containers
+
+
+class _BoundMethod:
+    def __init__(self, obj: _Obj, name: str, raw, flags):
+        self.obj = obj
+        self.name = name
+        self.raw = raw
+        self.flags = flags
+        self.is_generator = inspect.isgeneratorfunction(raw) or flags.get("is_generator")
+        self.is_batched = bool(flags.get("batched"))
+        self._batcher = None
+        self.remote = _Verb(self._remote)
+        self.local = _Verb(self._local)
+        self.spawn = _Verb(self._spawn)
+        self.map = _Verb(self._map, _agen_from_sync(self._map))
+        self.starmap = _Verb(self._starmap, _agen_from_sync(self._starmap))
+        self.for_each = _Verb(self._for_each)
+        self.remote_gen = _Verb(self._remote_gen, _agen_from_sync(self._remote_gen))
+
+    def _submit(self, args, kwargs, is_gen=False, spawned=False) -> Call:
+        pool = self.obj._get_pool()
+        if self.is_batched:
+            if self._batcher is None:
+                self._batcher = _MethodBatcher(pool, self.name,
+                                               int(self.flags["max_batch_size"]),
+                                               float(self.flags["wait_ms"]))
+            return self._batcher.enqueue(args, kwargs)
+        return pool.submit(self.name, args, kwargs, is_gen=is_gen, spawned=spawned)
+
+    def _remote(self, *args, **kwargs):
+        if self.is_generator:
+            return self._remote_gen(*args, **kwargs)
+        return self._submit(args, kwargs).wait()
+
+    def _remote_gen(self, *args, **kwargs):
+        call = self._submit(args, kwargs, is_gen=True)
+        while True:
+            alive, item = call.gen_q.get()
+            if not alive:
+                break
+            yield item
+        if call.exc is not None:
+            raise call.exc
+
+    def _local(self, *args, **kwargs):
+        obj = self.obj._local_instance()
+        return getattr(obj, self.name)(*args, **kwargs)
+
+    def _spawn(self, *args, **kwargs) -> FunctionCall:
+        return FunctionCall(call=self._submit(args, kwargs, spawned=True))
+
+    def _map(self, *input_iterators, order_outputs=True, return_exceptions=False,
+             ignore_exceptions=False):
+        if len(input_iterators) == 1:
+            items = ((x,) for x in input_iterators[0])
+        else:
+            items = zip(*input_iterators)
+        yield from self._run_map(items, order_outputs, return_exceptions, ignore_exceptions)
+
+    def _starmap(self, input_iterator, order_outputs=True, return_exceptions=False,
+                 ignore_exceptions=False):
+        items = (tuple(x) for x in input_iterator)
+        yield from self._run_map(items, order_outputs, return_exceptions, ignore_exceptions)
+
+    def _for_each(self, *input_iterators, ignore_exceptions=False):
+        for _ in self._map(*input_iterators, order_outputs=False,
+                           return_exceptions=ignore_exceptions,
+                           ignore_exceptions=ignore_exceptions):
+            pass
+
+    def _run_map(self, items, order_outputs, return_exceptions, ignore_exceptions):
+        pool = self.obj._get_pool()
+        window = max(8, 2 * pool.max_containers * pool.max_inputs_per_worker)
+        inflight: List[Call] = []
+        items = iter(items)
+        exhausted = False
+
+        def fill():
+            nonlocal exhausted
+            while not exhausted and len(inflight) < window:
+                try:
+                    args = next(items)
+                except StopIteration:
+                    exhausted = True
+                    return
+                inflight.append(self._submit(args, {}))
+
+        fill()
+        while inflight:
+            if order_outputs:
+                call = inflight.pop(0)
+                call.event.wait()
+            else:
+                call = None
+                while call is None:
+                    for c in inflight:
+                        if c.done:
+                            call = c
+                            break
+                    if call is None:
+                        time.sleep(0.002)
+                inflight.remove(call)
+            fill()
+            if call.exc is not None:
+                if ignore_exceptions:
+                    continue
+                if return_exceptions:
+                    yield call.exc
+                    continue
+                raise call.exc
+            yield call.value
+
+
+class _MethodBatcher(Batcher):
+    def __init__(self, pool, method_name, max_batch_size, wait_ms):
+        super().__init__(pool, max_batch_size, wait_ms)
+        self.method_name = method_name
+
+    def _dispatch(self, batch):
+        items = [(a, k) for a, k, _, _ in batch]
+        calls = [Call(self.method_name, ipc.dumps(it), False, self.pool.timeout, RetryPolicy())
+                 for it in items]
+        with self.pool.lock:
+            self.pool._ensure_threads()
+            for c in calls:
+                self.pool.calls[c.id] = c
+            w = self.pool._pick_worker() or self.pool._maybe_scale_up() or self.pool._wait_for_worker()
+            for c in calls:
+                c.worker_id = w.id
+                c.deadline = time.monotonic() + c.timeout
+                w.inflight.add(c.id)
+            w.last_active = time.monotonic()
+            w.task_q.put((ipc.T_BATCH, [c.id for c in calls], self.method_name,
+                          [c.args_blob for c in calls]))
+        for (a, k, ev, slot), call in zip(batch, calls):
+            slot["call"] = call
+            ev.set()
+
+
+# ---------------------------------------------------------------- App
+
+
+class App:
+    """Application: a named registry of functions/classes + entrypoints.
+    (reference: ~191 ``modal.App`` uses; hello_world.py:18)."""
+
+    _registry: Dict[str, "App"] = {}
+
+    def __init__(self, name: str = "app", *, image=None, secrets=None, volumes=None,
+                 include_source: bool = True):
+        self.name = name
+        self.default_image = image
+        self.default_secrets = secrets or []
+        self.default_volumes = volumes or {}
+        self.functions: Dict[str, Function] = {}
+        self.classes: Dict[str, Cls] = {}
+        self.entrypoints: Dict[str, Callable] = {}
+        self.web_endpoints: Dict[str, Function] = {}
+        self._servers: Dict[str, Any] = {}
+        App._registry[name] = self
+
+    # decorators
+
+    def function(self, _fn=None, **opts):
+        def deco(fn):
+            o = self._make_opts(opts)
+            f = Function(self, fn, o)
+            self.functions[f.name] = f
+            flags = getattr(fn, "_modal_flags", {})
+            if flags.get("web"):
+                self.web_endpoints[f.name] = f
+            return f
+
+        if _fn is not None:
+            return deco(_fn)
+        return deco
+
+    def cls(self, _cls=None, **opts):
+        def deco(user_cls):
+            o = self._make_opts(opts)
+            c = Cls(self, user_cls, o)
+            self.classes[c.name] = c
+            return c
+
+        if _cls is not None:
+            return deco(_cls)
+        return deco
+
+    def local_entrypoint(self, _fn=None, name: Optional[str] = None):
+        def deco(fn):
+            self.entrypoints[name or fn.__name__] = fn
+            return fn
+
+        if _fn is not None:
+            return deco(_fn)
+        return deco
+
+    def server(self, *, port: int, **opts):
+        """``@app.server`` raw-port serving (07_web/server.py:55; vllm_inference.py:139)."""
+
+        def deco(obj):
+            _set_flag(obj, web=True, web_kind="server", port=port)
+            o = self._make_opts(opts)
+            if inspect.isclass(obj):
+                c = Cls(self, obj, o)
+                self.classes[c.name] = c
+                return c
+            f = Function(self, obj, o)
+            self.functions[f.name] = f
+            self.web_endpoints[f.name] = f
+            return f
+
+        return deco
+
+    def _make_opts(self, opts: dict) -> FunctionOptions:
+        known = {f for f in FunctionOptions.__dataclass_fields__}
+        clean = {k: v for k, v in opts.items() if k in known}
+        o = FunctionOptions(**clean)
+        if o.image is None:
+            o.image = self.default_image
+        if not o.secrets:
+            o.secrets = list(self.default_secrets)
+        if not o.volumes:
+            o.volumes = dict(self.default_volumes)
+        return o
+
+    # lifecycle
+
+    def run(self, **kwargs):
+        return _AppRun(self)
+
+    def deploy(self, name: Optional[str] = None):
+        from .runtime.cron import start_schedules
+
+        start_schedules(self)
+        return self
+
+    @staticmethod
+    def registry_lookup_cls(app_name, cls_name):
+        app = App._registry.get(app_name)
+        return app.classes.get(cls_name) if app else None
+
+    @staticmethod
+    def lookup(name: str, create_if_missing: bool = False) -> "App":
+        app = App._registry.get(name)
+        if app is None:
+            if not create_if_missing:
+                raise NotFoundError(f"app {name} not found")
+            app = App(name)
+        return app
+
+
+class _AppRun:
+    def __init__(self, app: App):
+        self.app = app
+
+    def __enter__(self):
+        return self.app
+
+    def __exit__(self, *exc):
+        return False
+
+
+def enable_output():
+    class _Ctx:
+        def __enter__(self):
+            return self
+
+        def __exit__(self, *a):
+            return False
+
+    return _Ctx()
+
+
+# Stub alias used by examples that declare `app = modal.App(...)`; `Stub` was
+# the old name in the reference corpus history.
+Stub = App

@@ -1,0 +1,140 @@
+"""Volume: shared directory with commit/reload semantics.
+
+Reference usage: 118 ``Volume.from_name`` sites; commit/reload at
+06_gpu_and_ml/dreambooth/diffusers_lora_finetune.py:343,367.  Locally a volume
+is a directory under the state dir, shared by all worker processes (same host),
+so ``commit``/``reload`` are consistency barriers: commit fsyncs, reload is a
+no-op that revalidates existence.  Workers mount volumes by symlinking the
+mount path to the volume directory (they run as root on this node).
+"""
+from __future__ import annotations
+
+import os
+from pathlib import Path
+from typing import Optional
+
+from .. import config
+from ..exception import NotFoundError
+
+
+class Volume:
+    def __init__(self, name: str, _create: bool = True, version: int = 2):
+        self.name = name
+        self.version = version
+        p = self.path
+        if _create:
+            p.mkdir(parents=True, exist_ok=True)
+        elif not p.exists():
+            raise NotFoundError(f"volume {name!r} does not exist")
+
+    @property
+    def path(self) -> Path:
+        return config.state_dir() / "volumes" / self.name
+
+    @staticmethod
+    def from_name(name: str, create_if_missing: bool = False, version: int = 2) -> "Volume":
+        return Volume(name, _create=create_if_missing or True, version=version)
+
+    @staticmethod
+    def ephemeral():
+        import contextlib
+        import uuid
+
+        @contextlib.contextmanager
+        def ctx():
+            v = Volume(f"ephemeral-{uuid.uuid4().hex[:8]}")
+            try:
+                yield v
+            finally:
+                import shutil
+
+                shutil.rmtree(v.path, ignore_errors=True)
+
+        return ctx()
+
+    def commit(self):
+        """Flush writes so other workers observe them (fsync the tree)."""
+        d = os.open(self.path, os.O_RDONLY)
+        try:
+            os.fsync(d)
+        finally:
+            os.close(d)
+
+    def reload(self):
+        if not self.path.exists():
+            raise NotFoundError(f"volume {self.name!r} vanished")
+
+    def listdir(self, path: str = "/", recursive: bool = False):
+        base = self.path / path.lstrip("/")
+        if recursive:
+            out = []
+            for root, _dirs, files in os.walk(base):
+                rel = Path(root).relative_to(self.path)
+                out.extend(str(rel / f) for f in files)
+            return out
+        return [p.name for p in base.iterdir()]
+
+    def iterdir(self, path: str = "/"):
+        yield from self.listdir(path)
+
+    def read_file(self, path: str) -> bytes:
+        return (self.path / path.lstrip("/")).read_bytes()
+
+    def remove_file(self, path: str, recursive: bool = False):
+        p = self.path / path.lstrip("/")
+        if recursive:
+            import shutil
+
+            shutil.rmtree(p)
+        else:
+            p.unlink()
+
+    @staticmethod
+    def delete(name: str):
+        import shutil
+
+        shutil.rmtree(config.state_dir() / "volumes" / name, ignore_errors=True)
+
+    def batch_upload(self):
+        vol = self
+
+        class _Batch:
+            def __enter__(self):
+                return self
+
+            def __exit__(self, *a):
+                vol.commit()
+                return False
+
+            def put_file(self, local, remote):
+                import shutil
+
+                dst = vol.path / str(remote).lstrip("/")
+                dst.parent.mkdir(parents=True, exist_ok=True)
+                shutil.copy2(local, dst)
+
+            def put_directory(self, local, remote):
+                import shutil
+
+                dst = vol.path / str(remote).lstrip("/")
+                shutil.copytree(local, dst, dirs_exist_ok=True)
+
+        return _Batch()
+
+
+class CloudBucketMount:
+    """S3/GCS bucket mount shim (10_integrations/s3_bucket_mount.py:66): no
+    network locally, so this maps to a named local directory that tests can
+    pre-populate."""
+
+    def __init__(self, bucket_name: str, secret=None, read_only: bool = False,
+                 key_prefix: Optional[str] = None, **kw):
+        self.bucket_name = bucket_name
+        self.read_only = read_only
+        self.key_prefix = key_prefix or ""
+        self.path = config.state_dir() / "buckets" / bucket_name
+        self.path.mkdir(parents=True, exist_ok=True)
+
+    @property
+    def name(self) -> str:
+        return f"bucket:{self.bucket_name}"

@@ -1,0 +1,162 @@
+"""Worker-process main loop.
+
+One worker process ≈ one Modal "container" (reference semantics at SURVEY.md §2.2:
+container≈worker-process lifecycle).  A worker:
+
+  1. pins its assigned GPUs via ``HIP_VISIBLE_DEVICES`` *before* any torch import,
+  2. unpickles its service target (function or class),
+  3. runs ``@enter`` lifecycle hooks (snap=True phase first, then snap=False —
+     mirroring the phase split at reference 06_gpu_and_ml/gpu_snapshot.py:41-53),
+  4. serves tasks from its task queue with ``max_inputs`` concurrent slots
+     (``@modal.concurrent``, reference 06_gpu_and_ml/llm-serving/sglang_snapshot.py:260),
+  5. on shutdown runs ``@exit`` hooks (reference lifecycle decorator counts: §1 L5).
+"""
+from __future__ import annotations
+
+import os
+import queue as _queue
+import sys
+import threading
+import traceback
+from concurrent.futures import ThreadPoolExecutor
+
+from . import ipc
+
+
+def _apply_env(spec_env: dict, gpu_devices: tuple) -> None:
+    if gpu_devices:
+        vis = ",".join(str(d) for d in gpu_devices)
+        os.environ["HIP_VISIBLE_DEVICES"] = vis
+        os.environ["CUDA_VISIBLE_DEVICES"] = vis
+    os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    for k, v in spec_env.items():
+        os.environ[k] = str(v)
+
+
+def _resolve_target(spec: "ipc.ServiceSpec"):
+    """Returns (callable_map, exit_hooks). callable_map: method_name -> callable."""
+    target = ipc.loads(spec.target_blob)
+    if not spec.is_cls:
+        return {"": target}, []
+
+    cls = target
+    obj = cls()
+    # Bind modal.parameter() fields (reference: stable_diffusion/flux.py:126-128)
+    from ..app import _iter_parameters
+
+    for pname, default in _iter_parameters(cls):
+        setattr(obj, pname, spec.cls_params.get(pname, default))
+
+    # Lifecycle hooks, snapshot phase first (gpu_snapshot.py:41-53 ordering)
+    enter_hooks = []
+    exit_hooks = []
+    methods = {}
+    for name in dir(cls):
+        fn = getattr(cls, name, None)
+        flags = getattr(fn, "_modal_flags", None)
+        if not flags:
+            continue
+        bound = getattr(obj, name)
+        if flags.get("enter"):
+            enter_hooks.append((bool(flags.get("snap")), bound))
+        if flags.get("exit"):
+            exit_hooks.append(bound)
+        if flags.get("method") or flags.get("web") or flags.get("batched"):
+            methods[name] = bound
+    for snap_phase in (True, False):
+        for is_snap, hook in enter_hooks:
+            if is_snap == snap_phase:
+                hook()
+    methods.setdefault("", None)
+    return methods, exit_hooks
+
+
+def worker_main(worker_id: int, spec_blob: bytes, task_q, result_q) -> None:
+    spec: ipc.ServiceSpec = ipc.loads(spec_blob)
+    _apply_env(spec.env, spec.gpu_devices)
+    os.environ["MODAL_TASK_ID"] = f"ta-local-{spec.name}-{worker_id}"
+
+    def post(kind, call_id=None, payload=None, text=""):
+        result_q.put(ipc.WorkerMsg(worker_id, kind, call_id, payload, text))
+
+    try:
+        methods, exit_hooks = _resolve_target(spec)
+    except BaseException as e:  # noqa: BLE001
+        post(ipc.ERROR, None, ipc.dumps(e), traceback.format_exc())
+        return
+    post(ipc.READY)
+
+    import inspect
+
+    def run_one(call_id: str, method_name: str, args_blob: bytes) -> None:
+        try:
+            args, kwargs = ipc.loads(args_blob)
+            fn = methods.get(method_name)
+            if fn is None:
+                raise RuntimeError(f"no method {method_name!r} on service {spec.name}")
+            raw = getattr(fn, "_modal_raw", fn)
+            if inspect.isgeneratorfunction(raw):
+                for item in fn(*args, **kwargs):
+                    post(ipc.YIELD, call_id, ipc.dumps(item))
+                post(ipc.GEN_END, call_id)
+            else:
+                out = fn(*args, **kwargs)
+                post(ipc.RESULT, call_id, ipc.dumps(out))
+        except BaseException as e:  # noqa: BLE001
+            try:
+                blob = ipc.dumps(e)
+            except Exception:
+                blob = ipc.dumps(RuntimeError(repr(e)))
+            post(ipc.ERROR, call_id, blob, traceback.format_exc())
+
+    def run_batch(call_ids, method_name: str, args_blobs) -> None:
+        """@modal.batched: collect single-input calls into one list-shaped call.
+
+        The wrapped function receives lists and must return a parallel list
+        (reference: 03_scaling_out/dynamic_batching.py:29-45).
+        """
+        try:
+            unpacked = [ipc.loads(b) for b in args_blobs]
+            fn = methods.get(method_name)
+            # each call's args: (args_tuple, kwargs). Batched fns take positional
+            # lists: transpose the per-call positional args into per-arg lists.
+            nargs = max(len(a) for a, _ in unpacked)
+            cols = [[a[i] for a, _ in unpacked] for i in range(nargs)]
+            outs = fn(*cols)
+            if len(outs) != len(call_ids):
+                raise RuntimeError(
+                    f"batched function returned {len(outs)} outputs for {len(call_ids)} inputs"
+                )
+            for cid, out in zip(call_ids, outs):
+                post(ipc.RESULT, cid, ipc.dumps(out))
+        except BaseException as e:  # noqa: BLE001
+            tb = traceback.format_exc()
+            for cid in call_ids:
+                post(ipc.ERROR, cid, ipc.dumps(e), tb)
+
+    pool = ThreadPoolExecutor(max_workers=max(1, spec.max_inputs))
+    while True:
+        try:
+            task = task_q.get(timeout=3600.0)
+        except _queue.Empty:
+            continue
+        kind = task[0]
+        if kind == ipc.T_SHUTDOWN:
+            break
+        if kind == ipc.T_CALL:
+            _, call_id, method_name, args_blob = task
+            if spec.max_inputs > 1:
+                pool.submit(run_one, call_id, method_name, args_blob)
+            else:
+                run_one(call_id, method_name, args_blob)
+        elif kind == ipc.T_BATCH:
+            _, call_ids, method_name, args_blobs = task
+            run_batch(call_ids, method_name, args_blobs)
+
+    pool.shutdown(wait=True)
+    for hook in exit_hooks:
+        try:
+            hook()
+        except Exception:  # noqa: BLE001
+            traceback.print_exc()
+    post(ipc.EXITED)

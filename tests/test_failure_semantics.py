@@ -1,0 +1,94 @@
+"""Retries, timeouts, single-use containers, crash recovery.
+
+Reference behavior spec: 06_gpu_and_ml/long-training.py:108-137 (Retries +
+timeout interruption + resume), doc_ocr_jobs.py:89 (retries=3),
+gpu_fallbacks.py:22 (single_use_containers)."""
+import os
+import time
+
+import pytest
+
+import modal_examples_amd as modal
+
+app = modal.App("test-fail")
+
+_attempt_dict_name = "test-fail-attempts"
+
+
+@app.function(retries=modal.Retries(max_retries=3, initial_delay=0.05, backoff_coefficient=1.0))
+def flaky(key):
+    d = modal.Dict.from_name(_attempt_dict_name)
+    n = d.get(key, 0) + 1
+    d[key] = n
+    if n < 3:
+        raise RuntimeError(f"transient {n}")
+    return n
+
+
+@app.function(timeout=1.5, retries=modal.Retries(max_retries=0, initial_delay=0.0))
+def sleeper(t):
+    time.sleep(t)
+    return "done"
+
+
+@app.function(timeout=2.0, retries=modal.Retries(max_retries=2, initial_delay=0.05))
+def checkpointed(key):
+    """Simulates long-training.py: times out, retried, resumes from 'checkpoint'."""
+    d = modal.Dict.from_name(_attempt_dict_name)
+    step = d.get(key, 0)
+    if step < 1:
+        d[key] = step + 1
+        time.sleep(30)  # will be killed by timeout
+    return f"resumed-from-{d.get(key)}"
+
+
+@app.function()
+def crasher():
+    os._exit(17)  # hard worker death
+
+
+@app.function(single_use_containers=True)
+def single_use():
+    return os.getpid()
+
+
+def test_retries_until_success():
+    modal.Dict.from_name(_attempt_dict_name).clear()
+    assert flaky.remote("k1") == 3
+
+
+def test_retries_exhausted():
+    modal.Dict.from_name(_attempt_dict_name).clear()
+
+    @app.function(retries=modal.Retries(max_retries=1, initial_delay=0.05))
+    def always_fails():
+        raise ValueError("nope")
+
+    with pytest.raises(ValueError):
+        always_fails.remote()
+
+
+def test_timeout_raises():
+    with pytest.raises(modal.FunctionTimeoutError):
+        sleeper.remote(30)
+
+
+def test_timeout_fast_function_ok():
+    assert sleeper.remote(0.01) == "done"
+
+
+def test_timeout_then_retry_resumes():
+    modal.Dict.from_name(_attempt_dict_name).clear()
+    assert checkpointed.remote("ck") == "resumed-from-1"
+
+
+def test_worker_crash_surfaces_error():
+    from modal_examples_amd.exception import ExecutionError
+
+    with pytest.raises(ExecutionError):
+        crasher.remote()
+
+
+def test_single_use_containers_fresh_pid():
+    pids = {single_use.remote() for _ in range(3)}
+    assert len(pids) == 3, "single_use_containers must not reuse workers"

@@ -1,0 +1,157 @@
+"""Volume / Dict / Queue / Secret / Image / Sandbox semantics.
+
+Reference behavior spec: diffusers_lora_finetune.py:343,367 (volume commit/
+reload), dicts_and_queues.py:72-95, sandbox_pool.py:80-292 (queue partitions),
+safe_code_execution.py:28-46 (sandbox exec)."""
+import pytest
+
+import modal_examples_amd as modal
+
+
+def test_volume_write_read_across_processes():
+    vol = modal.Volume.from_name("test-vol", create_if_missing=True)
+    app = modal.App("test-vol-app")
+
+    @app.function()
+    def writer(text):
+        v = modal.Volume.from_name("test-vol")
+        (v.path / "data.txt").write_text(text)
+        v.commit()
+        return str(v.path)
+
+    writer.remote("hello-volume")
+    vol.reload()
+    assert vol.read_file("data.txt") == b"hello-volume"
+    assert "data.txt" in vol.listdir("/")
+    modal.Volume.delete("test-vol")
+
+
+def test_volume_ephemeral():
+    with modal.Volume.ephemeral() as v:
+        (v.path / "x").write_text("1")
+        assert v.listdir("/") == ["x"]
+
+
+def test_dict_ops():
+    with modal.Dict.ephemeral() as d:
+        d["a"] = 1
+        d.put("b", {"nested": [1, 2]})
+        assert d["a"] == 1
+        assert d.get("b")["nested"] == [1, 2]
+        assert "a" in d
+        assert len(d) == 2
+        assert sorted(d.keys()) == ["a", "b"]
+        assert d.pop("a") == 1
+        assert len(d) == 1
+        with pytest.raises(KeyError):
+            d["missing"]
+
+
+def test_dict_shared_across_workers():
+    d = modal.Dict.from_name("test-shared-dict")
+    d.clear()
+    app = modal.App("test-dict-app")
+
+    @app.function()
+    def put(k, v):
+        modal.Dict.from_name("test-shared-dict")[k] = v
+
+    put.remote("from-worker", 42)
+    assert d["from-worker"] == 42
+    modal.Dict.delete("test-shared-dict")
+
+
+def test_queue_fifo_and_partitions():
+    with modal.Queue.ephemeral() as q:
+        q.put(1)
+        q.put_many([2, 3])
+        q.put("p1", partition="other")
+        assert q.get() == 1
+        assert q.get_many(2) == [2, 3]
+        assert q.len() == 0
+        assert q.len(partition="other") == 1
+        assert q.get(partition="other") == "p1"
+
+
+def test_queue_get_nonblocking_empty():
+    with modal.Queue.ephemeral() as q:
+        assert q.get(block=False) is None
+        assert q.get_many(3, block=False) == []
+
+
+def test_secret_from_dict_injects_env():
+    app = modal.App("test-secret-app")
+    s = modal.Secret.from_dict({"MY_TEST_KEY": "sekrit"})
+
+    @app.function(secrets=[s])
+    def read_env():
+        import os
+
+        return os.environ.get("MY_TEST_KEY")
+
+    assert read_env.remote() == "sekrit"
+
+
+def test_secret_required_keys_missing():
+    with pytest.raises(modal.NotFoundError):
+        modal.Secret.from_name("nonexistent", required_keys=["NOPE_NOT_SET_XYZ"])
+
+
+def test_image_chain_and_hash():
+    img = (
+        modal.Image.debian_slim(python_version="3.12")
+        .apt_install("git")
+        .uv_pip_install("torch")
+        .env({"HELLO": "world"})
+    )
+    assert img.content_hash()
+    assert img.build_env == {"HELLO": "world"}
+    img2 = img.env({"X": "1"})
+    assert img2.content_hash() != img.content_hash()
+
+
+def test_image_imports_suppresses():
+    img = modal.Image.debian_slim()
+    with img.imports():
+        import nonexistent_module_xyz  # noqa: F401
+
+
+def test_sandbox_exec():
+    sb = modal.Sandbox.create(app=None, timeout=30)
+    p = sb.exec("echo", "hello-sandbox")
+    assert p.wait() == 0
+    assert "hello-sandbox" in p.stdout.read()
+    sb.terminate()
+
+
+def test_sandbox_entrypoint_and_poll():
+    sb = modal.Sandbox.create("python3", "-c", "print(6*7)", timeout=30)
+    assert sb.wait() == 0
+    assert "42" in sb.stdout.read()
+    sb.terminate()
+
+
+def test_cron_matching():
+    c = modal.Cron("*/5 14 * * *")
+    import time as _t
+
+    t = _t.struct_time((2026, 9, 11, 14, 10, 0, 4, 254, 0))
+    assert c.matches(t)
+    t2 = _t.struct_time((2026, 9, 11, 15, 10, 0, 4, 254, 0))
+    assert not c.matches(t2)
+
+
+def test_period_validation():
+    assert modal.Period(minutes=5).total_seconds == 300
+    with pytest.raises(modal.InvalidError):
+        modal.Period()
+
+
+def test_gpu_string_parsing():
+    from modal_examples_amd.gpu.device_pool import parse_gpu
+
+    assert parse_gpu(None) == 0
+    assert parse_gpu("mi355x") == 1
+    assert parse_gpu("MI355X:4") == 4
+    assert parse_gpu("H100!") == 1
+    assert parse_gpu(["H100:2", "A100"]) == 2

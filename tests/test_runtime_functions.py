@@ -1,0 +1,159 @@
+"""Core Function verb semantics — the hello_world tier of the reference
+(reference behavior spec: 01_getting_started/hello_world.py:34-70,
+generators.py:13-21, 08_advanced/parallel_execution.py:41-48)."""
+import time
+
+import pytest
+
+import modal_examples_amd as modal
+
+app = modal.App("test-fn")
+
+
+@app.function()
+def square(x: int) -> int:
+    return x * x
+
+
+@app.function()
+def fail_on_odd(x: int) -> int:
+    if x % 2:
+        raise ValueError(f"odd: {x}")
+    return x
+
+
+@app.function()
+def gen_fn(n: int):
+    for i in range(n):
+        yield i * 10
+
+
+@app.function()
+def add(a: int, b: int) -> int:
+    return a + b
+
+
+@app.function()
+def slow_echo(x):
+    time.sleep(0.2)
+    return x
+
+
+def test_local():
+    assert square.local(7) == 49
+
+
+def test_call_direct():
+    assert square(6) == 36
+
+
+def test_remote():
+    assert square.remote(9) == 81
+
+
+def test_remote_kwargs():
+    assert add.remote(2, b=3) == 5
+
+
+def test_map_ordered():
+    assert list(square.map(range(10))) == [x * x for x in range(10)]
+
+
+def test_map_unordered():
+    out = list(square.map(range(12), order_outputs=False))
+    assert sorted(out) == sorted(x * x for x in range(12))
+
+
+def test_map_multiple_iterators():
+    assert list(add.map(range(4), range(4, 8))) == [4, 6, 8, 10]
+
+
+def test_starmap():
+    assert list(add.starmap([(1, 2), (3, 4)])) == [3, 7]
+
+
+def test_map_exceptions_raise():
+    with pytest.raises(ValueError):
+        list(fail_on_odd.map(range(4)))
+
+
+def test_map_return_exceptions():
+    out = list(fail_on_odd.map(range(4), return_exceptions=True))
+    assert out[0] == 0 and out[2] == 2
+    assert isinstance(out[1], ValueError) and isinstance(out[3], ValueError)
+
+
+def test_map_ignore_exceptions():
+    out = list(fail_on_odd.map(range(6), ignore_exceptions=True))
+    assert out == [0, 2, 4]
+
+
+def test_for_each():
+    fail_on_odd.for_each(range(0, 6, 2))
+
+
+def test_spawn_and_get():
+    fc = square.spawn(12)
+    assert fc.get() == 144
+    assert fc.object_id.startswith("fc-")
+
+
+def test_spawn_gather():
+    calls = [square.spawn(i) for i in range(5)]
+    assert modal.functions.gather(*calls) == [0, 1, 4, 9, 16]
+
+
+def test_spawn_exception_propagates():
+    fc = fail_on_odd.spawn(3)
+    with pytest.raises(ValueError, match="odd: 3"):
+        fc.get()
+
+
+def test_function_call_from_id():
+    fc = square.spawn(5)
+    assert fc.get() == 25
+    time.sleep(0.2)  # allow durable store write
+    fc2 = modal.FunctionCall.from_id(fc.object_id)
+    assert fc2.get(timeout=5) == 25
+
+
+def test_get_timeout_zero_raises():
+    fc = slow_echo.spawn("hi")
+    with pytest.raises(TimeoutError):
+        fc.get(timeout=0)
+    assert fc.get(timeout=10) == "hi"
+
+
+def test_remote_gen():
+    assert list(gen_fn.remote_gen(4)) == [0, 10, 20, 30]
+
+
+def test_remote_exception_has_traceback():
+    try:
+        fail_on_odd.remote(1)
+    except ValueError as e:
+        assert "odd: 1" in str(e)
+        assert getattr(e, "remote_traceback", "")
+    else:
+        pytest.fail("no exception")
+
+
+def test_aio_remote():
+    import asyncio
+
+    async def go():
+        return await square.remote.aio(4)
+
+    assert asyncio.run(go()) == 16
+
+
+def test_aio_map():
+    import asyncio
+
+    async def go():
+        out = []
+        async for r in square.map.aio(range(5)):
+            out.append(r)
+        return out
+
+    assert asyncio.run(go()) == [0, 1, 4, 9, 16]
