@@ -1,0 +1,168 @@
+// Paged-KV decode attention (single query token per sequence), bf16, gfx950.
+//
+// Serves K6: the vLLM/SGLang-style PagedAttention that the
+// reference's serving examples lean on (llm-serving/vllm_inference.py:158-209).
+//
+// MI355X-first design: decode is HBM-bound (reads the whole KV cache once per
+// step), so the kernel is organized around 16-byte-per-lane vector loads and
+// GQA amortization — each workgroup loads the KV rows of ONE kv-head and scores
+// all G = Hq/Hkv query heads of its group against them, so KV bytes are read
+// once per group instead of once per query head.
+//
+//   grid = (Hkv, B), block = 256 (4 waves)
+//   lane split: D/8 lanes cover one KV row (16B each) → ROWS=64/(D/8) rows
+//   per wave step; each row-group keeps an online-softmax partial (m, l, acc)
+//   per query head; partials merge through LDS at the end (flash-decoding
+//   style merge with max realignment).
+//
+// KV cache layout: [num_blocks, Hkv, block_size, D] (block_table int32
+// [B, max_blocks]); pass block_table = nullptr for a contiguous [B, Hkv, S, D]
+// cache (then block_size must be >= max seq len).
+#include "common.h"
+
+#include <cstdio>
+
+#define DEC_THREADS 256
+#define DEC_WAVES 4
+#define MAX_G 8
+
+template <int D>
+__global__ __launch_bounds__(DEC_THREADS) void paged_decode_kernel(
+    const short* __restrict__ Q,       // [B, Hq, D]
+    const short* __restrict__ Kc,      // cache, layout above
+    const short* __restrict__ Vc,
+    const int* __restrict__ block_table,  // [B, max_blocks] or nullptr
+    const int* __restrict__ seq_lens,     // [B]
+    short* __restrict__ O,             // [B, Hq, D]
+    int B, int Hq, int Hkv, int block_size, int max_blocks, float scale) {
+  constexpr int LPR = D / 8;          // lanes per KV row (16B chunks)
+  constexpr int ROWS = WAVE / LPR;    // KV rows per wave step
+  constexpr int NPART = DEC_WAVES * ROWS;  // softmax partials to merge
+
+  const int hkv = blockIdx.x;
+  const int b = blockIdx.y;
+  const int G = Hq / Hkv;
+  const int S = seq_lens[b];
+
+  const int tid = threadIdx.x;
+  const int w = tid / WAVE;
+  const int l = tid % WAVE;
+  const int slot = l % LPR;           // which 16B chunk of the row
+  const int rg = l / LPR;             // row-group within the wave
+  const int part = w * ROWS + rg;     // global partial index
+
+  // Q fragments: per head g, this lane's 8 bf16 of q (its slot), as f32
+  float qf[MAX_G][8];
+#pragma unroll
+  for (int g = 0; g < MAX_G; ++g) {
+    if (g < G) {
+      const short* qp = &Q[(((long long)b * Hq) + hkv * G + g) * D + slot * 8];
+      bf16x8 q8 = *(const bf16x8*)qp;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) qf[g][j] = bf2f(q8[j]);
+    }
+  }
+
+  float m_run[MAX_G], l_run[MAX_G], accv[MAX_G][8];
+#pragma unroll
+  for (int g = 0; g < MAX_G; ++g) {
+    m_run[g] = -1e30f;
+    l_run[g] = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) accv[g][j] = 0.f;
+  }
+
+  // ---- sweep the KV rows of this (b, hkv) ----
+  for (int kv = part; kv < S; kv += NPART) {
+    long long row_off;
+    if (block_table != nullptr) {
+      int blk = block_table[(long long)b * max_blocks + kv / block_size];
+      row_off = (((long long)blk * Hkv + hkv) * block_size + kv % block_size) * D;
+    } else {
+      row_off = (((long long)b * Hkv + hkv) * (long long)block_size + kv) * D;
+    }
+    bf16x8 k8 = *(const bf16x8*)&Kc[row_off + slot * 8];
+    bf16x8 v8 = *(const bf16x8*)&Vc[row_off + slot * 8];
+    float kfl[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) kfl[j] = bf2f(k8[j]);
+
+#pragma unroll
+    for (int g = 0; g < MAX_G; ++g) {
+      if (g >= G) break;
+      float d = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) d += qf[g][j] * kfl[j];
+      // reduce across the LPR lanes of this row
+#pragma unroll
+      for (int s_ = 1; s_ < LPR; s_ <<= 1) d += __shfl_xor(d, s_, WAVE);
+      d *= scale;
+      float m_new = fmaxf(m_run[g], d);
+      float rs = __expf(m_run[g] - m_new);
+      float p = __expf(d - m_new);
+      l_run[g] = l_run[g] * rs + p;
+      m_run[g] = m_new;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) accv[g][j] = accv[g][j] * rs + p * bf2f(v8[j]);
+    }
+  }
+
+  // ---- merge partials through LDS ----
+  __shared__ float sm[MAX_G][NPART];
+  __shared__ float sl[MAX_G][NPART];
+  __shared__ float sacc[MAX_G][NPART][D];
+
+#pragma unroll
+  for (int g = 0; g < MAX_G; ++g) {
+    if (g >= G) break;
+    if (slot == 0) {
+      sm[g][part] = m_run[g];
+      sl[g][part] = l_run[g];
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) sacc[g][part][slot * 8 + j] = accv[g][j];
+  }
+  __syncthreads();
+
+  // threads cover (g, d) output elements
+  for (int gd = tid; gd < G * D; gd += DEC_THREADS) {
+    int g = gd / D, d = gd % D;
+    float m_star = -1e30f;
+    for (int p_ = 0; p_ < NPART; ++p_) m_star = fmaxf(m_star, sm[g][p_]);
+    float num = 0.f, den = 0.f;
+    for (int p_ = 0; p_ < NPART; ++p_) {
+      float f = __expf(sm[g][p_] - m_star);
+      num += f * sacc[g][p_][d];
+      den += f * sl[g][p_];
+    }
+    O[(((long long)b * Hq) + hkv * G + g) * D + d] =
+        f2bf(den > 0.f ? num / den : 0.f);
+  }
+}
+
+extern "C" void paged_decode_bf16(const void* q, const void* kc, const void* vc,
+                                  const int* block_table, const int* seq_lens,
+                                  void* o, int B, int Hq, int Hkv, int D,
+                                  int block_size, int max_blocks, float scale,
+                                  hipStream_t stream) {
+  dim3 grid(Hkv, B);
+  dim3 block(DEC_THREADS);
+  if (Hq / Hkv > MAX_G) {
+    fprintf(stderr, "paged_decode_bf16: GQA group %d > %d\n", Hq / Hkv, MAX_G);
+    abort();
+  }
+  if (D == 64) {
+    hipLaunchKernelGGL((paged_decode_kernel<64>), grid, block, 0, stream,
+                       (const short*)q, (const short*)kc, (const short*)vc,
+                       block_table, seq_lens, (short*)o, B, Hq, Hkv, block_size,
+                       max_blocks, scale);
+  } else if (D == 128) {
+    hipLaunchKernelGGL((paged_decode_kernel<128>), grid, block, 0, stream,
+                       (const short*)q, (const short*)kc, (const short*)vc,
+                       block_table, seq_lens, (short*)o, B, Hq, Hkv, block_size,
+                       max_blocks, scale);
+  } else {
+    fprintf(stderr, "paged_decode_bf16: unsupported head_dim %d\n", D);
+    abort();
+  }
+}

@@ -1,0 +1,204 @@
+// Fused elementwise kernels, bf16, gfx950.  All HBM-bound → bf16x8 vector
+// access, grid-stride, capped grid (guide §6 G11/G13).
+//
+// Serves: K4 sampler step + CFG combine (the per-step
+// elementwise work inside the reference's diffusion pipelines,
+// text_to_image.py:114-120), SiLU-mul / GEGLU for transformer FFNs (K2 fusion
+// family), residual adds, RoPE with host-precomputed cos/sin tables
+// (guide Appendix B: never compute trig on-device).
+#include "common.h"
+
+#define EW_BLOCK 256
+
+// ---------------------------------------------------------------- CFG + Euler step
+// Classifier-free guidance combine and ancestral-free Euler update in one pass:
+//   eps = eps_u + g*(eps_c - eps_u);  x_next = x + (sig_next - sig) * eps
+// (one kernel per denoise step instead of 4 reads + 3 writes).
+
+__global__ __launch_bounds__(EW_BLOCK) void cfg_euler_kernel(
+    const short* __restrict__ Xt, const short* __restrict__ EpsC,
+    const short* __restrict__ EpsU, short* __restrict__ Xn, float guidance,
+    float dsigma, long long n) {
+  long long stride = (long long)gridDim.x * EW_BLOCK * 8;
+  for (long long i = ((long long)blockIdx.x * EW_BLOCK + threadIdx.x) * 8;
+       i + 8 <= n; i += stride) {
+    bf16x8 x = *(const bf16x8*)&Xt[i];
+    bf16x8 ec = *(const bf16x8*)&EpsC[i];
+    bf16x8 o;
+    if (EpsU != nullptr) {
+      bf16x8 eu = *(const bf16x8*)&EpsU[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float e = bf2f(eu[j]) + guidance * (bf2f(ec[j]) - bf2f(eu[j]));
+        o[j] = f2bf(bf2f(x[j]) + dsigma * e);
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o[j] = f2bf(bf2f(x[j]) + dsigma * bf2f(ec[j]));
+    }
+    *(bf16x8*)&Xn[i] = o;
+  }
+  // tail
+  long long full = (n / 8) * 8;
+  for (long long i = full + (long long)blockIdx.x * EW_BLOCK + threadIdx.x;
+       i < n; i += (long long)gridDim.x * EW_BLOCK) {
+    float e = bf2f(EpsC[i]);
+    if (EpsU != nullptr) {
+      float eu = bf2f(EpsU[i]);
+      e = eu + guidance * (e - eu);
+    }
+    Xn[i] = f2bf(bf2f(Xt[i]) + dsigma * e);
+  }
+}
+
+extern "C" void cfg_euler_bf16(const void* xt, const void* eps_c,
+                               const void* eps_u, void* xn, float guidance,
+                               float dsigma, long long n, hipStream_t stream) {
+  int grid = elementwise_grid(n, EW_BLOCK);
+  hipLaunchKernelGGL(cfg_euler_kernel, dim3(grid), dim3(EW_BLOCK), 0, stream,
+                     (const short*)xt, (const short*)eps_c,
+                     (const short*)eps_u, (short*)xn, guidance, dsigma, n);
+}
+
+// ---------------------------------------------------------------- SiLU-mul (SwiGLU)
+// y = silu(a) * b  — Llama FFN gate; a,b are the two halves of the gate_up proj.
+
+__global__ __launch_bounds__(EW_BLOCK) void silu_mul_kernel(
+    const short* __restrict__ A, const short* __restrict__ Bv,
+    short* __restrict__ Y, long long n) {
+  long long stride = (long long)gridDim.x * EW_BLOCK * 8;
+  for (long long i = ((long long)blockIdx.x * EW_BLOCK + threadIdx.x) * 8;
+       i + 8 <= n; i += stride) {
+    bf16x8 a = *(const bf16x8*)&A[i];
+    bf16x8 b = *(const bf16x8*)&Bv[i];
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bf2f(a[j]);
+      o[j] = f2bf(f / (1.f + __expf(-f)) * bf2f(b[j]));
+    }
+    *(bf16x8*)&Y[i] = o;
+  }
+  long long full = (n / 8) * 8;
+  for (long long i = full + (long long)blockIdx.x * EW_BLOCK + threadIdx.x;
+       i < n; i += (long long)gridDim.x * EW_BLOCK) {
+    float f = bf2f(A[i]);
+    Y[i] = f2bf(f / (1.f + __expf(-f)) * bf2f(Bv[i]));
+  }
+}
+
+extern "C" void silu_mul_bf16(const void* a, const void* b, void* y,
+                              long long n, hipStream_t stream) {
+  int grid = elementwise_grid(n, EW_BLOCK);
+  hipLaunchKernelGGL(silu_mul_kernel, dim3(grid), dim3(EW_BLOCK), 0, stream,
+                     (const short*)a, (const short*)b, (short*)y, n);
+}
+
+// ---------------------------------------------------------------- GEGLU
+// y = gelu(a) * b — SDXL transformer FFN (diffusers GEGLU). tanh approximation.
+
+__global__ __launch_bounds__(EW_BLOCK) void geglu_kernel(
+    const short* __restrict__ A, const short* __restrict__ Bv,
+    short* __restrict__ Y, long long n) {
+  long long stride = (long long)gridDim.x * EW_BLOCK * 8;
+  const float k0 = 0.7978845608028654f, k1 = 0.044715f;
+  for (long long i = ((long long)blockIdx.x * EW_BLOCK + threadIdx.x) * 8;
+       i + 8 <= n; i += stride) {
+    bf16x8 a = *(const bf16x8*)&A[i];
+    bf16x8 b = *(const bf16x8*)&Bv[i];
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bf2f(a[j]);
+      float g = 0.5f * f * (1.f + tanhf(k0 * (f + k1 * f * f * f)));
+      o[j] = f2bf(g * bf2f(b[j]));
+    }
+    *(bf16x8*)&Y[i] = o;
+  }
+  long long full = (n / 8) * 8;
+  for (long long i = full + (long long)blockIdx.x * EW_BLOCK + threadIdx.x;
+       i < n; i += (long long)gridDim.x * EW_BLOCK) {
+    float f = bf2f(A[i]);
+    float g = 0.5f * f * (1.f + tanhf(k0 * (f + k1 * f * f * f)));
+    Y[i] = f2bf(g * bf2f(Bv[i]));
+  }
+}
+
+extern "C" void geglu_bf16(const void* a, const void* b, void* y, long long n,
+                           hipStream_t stream) {
+  int grid = elementwise_grid(n, EW_BLOCK);
+  hipLaunchKernelGGL(geglu_kernel, dim3(grid), dim3(EW_BLOCK), 0, stream,
+                     (const short*)a, (const short*)b, (short*)y, n);
+}
+
+// ---------------------------------------------------------------- residual add
+
+__global__ __launch_bounds__(EW_BLOCK) void add_kernel(
+    const short* __restrict__ A, const short* __restrict__ Bv,
+    short* __restrict__ Y, long long n) {
+  long long stride = (long long)gridDim.x * EW_BLOCK * 8;
+  for (long long i = ((long long)blockIdx.x * EW_BLOCK + threadIdx.x) * 8;
+       i + 8 <= n; i += stride) {
+    bf16x8 a = *(const bf16x8*)&A[i];
+    bf16x8 b = *(const bf16x8*)&Bv[i];
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = f2bf(bf2f(a[j]) + bf2f(b[j]));
+    *(bf16x8*)&Y[i] = o;
+  }
+  long long full = (n / 8) * 8;
+  for (long long i = full + (long long)blockIdx.x * EW_BLOCK + threadIdx.x;
+       i < n; i += (long long)gridDim.x * EW_BLOCK)
+    Y[i] = f2bf(bf2f(A[i]) + bf2f(Bv[i]));
+}
+
+extern "C" void add_bf16(const void* a, const void* b, void* y, long long n,
+                         hipStream_t stream) {
+  int grid = elementwise_grid(n, EW_BLOCK);
+  hipLaunchKernelGGL(add_kernel, dim3(grid), dim3(EW_BLOCK), 0, stream,
+                     (const short*)a, (const short*)b, (short*)y, n);
+}
+
+// ---------------------------------------------------------------- RoPE
+// q/k: [B, H, S, D] bf16; cos/sin: [S, D/2] f32 host-precomputed (G13/App.B:
+// on-device trig turns this memory-bound op VALU-bound). Llama rotate-half
+// convention: (x1, x2) = (x[..D/2], x[D/2..]); x1' = x1*cos - x2*sin, etc.
+// In-place over q and k in one launch.
+
+__global__ __launch_bounds__(EW_BLOCK) void rope_kernel(
+    short* __restrict__ Qk, const float* __restrict__ Cos,
+    const float* __restrict__ Sin, long long BH, int S, int D,
+    const int* __restrict__ positions /*nullable [S]*/) {
+  // one wave handles one (bh, s) row; lanes cover D/2 rotation pairs
+  long long rows = BH * S;
+  int l = threadIdx.x % WAVE;
+  for (long long row = blockIdx.x * (EW_BLOCK / WAVE) + threadIdx.x / WAVE;
+       row < rows; row += (long long)gridDim.x * (EW_BLOCK / WAVE)) {
+    int s = (int)(row % S);
+    int pos = positions != nullptr ? positions[s] : s;
+    short* x = Qk + row * D;
+    for (int i = l * 2; i < D / 2; i += WAVE * 2) {
+      float c0 = Cos[(long long)pos * (D / 2) + i];
+      float s0 = Sin[(long long)pos * (D / 2) + i];
+      float c1 = Cos[(long long)pos * (D / 2) + i + 1];
+      float s1 = Sin[(long long)pos * (D / 2) + i + 1];
+      float x0 = bf2f(x[i]), x2 = bf2f(x[i + D / 2]);
+      float x1 = bf2f(x[i + 1]), x3 = bf2f(x[i + 1 + D / 2]);
+      x[i] = f2bf(x0 * c0 - x2 * s0);
+      x[i + D / 2] = f2bf(x2 * c0 + x0 * s0);
+      x[i + 1] = f2bf(x1 * c1 - x3 * s1);
+      x[i + 1 + D / 2] = f2bf(x3 * c1 + x1 * s1);
+    }
+  }
+}
+
+extern "C" void rope_bf16(void* qk, const float* cosv, const float* sinv,
+                          long long BH, int S, int D, const int* positions,
+                          hipStream_t stream) {
+  long long rows = BH * S;
+  int grid = (int)((rows + 3) / 4);
+  if (grid > 4096) grid = 4096;
+  hipLaunchKernelGGL(rope_kernel, dim3(grid), dim3(EW_BLOCK), 0, stream,
+                     (short*)qk, cosv, sinv, BH, S, D, positions);
+}

@@ -1,0 +1,179 @@
+// Normalization kernels, bf16 in/out, f32 compute, gfx950.
+//
+// All are HBM-bound: per guide Appendix B, loads are vectorized bf16x8
+// (16 B/lane) and each op is fused with its adjacent elementwise work so the
+// tensor is read once (SiLU into GroupNorm for the SDXL UNet/VAE resnets,
+// affine into LayerNorm/RMSNorm).
+//
+// Serves: GroupNorm+SiLU — K3 VAE/UNet resnet blocks
+// (reference trigger: diffusers pipelines, text_to_image.py:114-120);
+// LayerNorm — Whisper/GPT blocks (hp_sweep_gpt src/model.py); RMSNorm — Llama.
+#include "common.h"
+
+// ---------------------------------------------------------------- GroupNorm(+SiLU)
+// x: [N, C, H*W] contiguous (NCHW). groups divide C. One workgroup per (n, g):
+// the group's slab is C/G * HW contiguous elements — two-pass (sum/sqsum, then
+// normalize+affine+optional SiLU).
+
+__global__ __launch_bounds__(256) void groupnorm_silu_kernel(
+    const short* __restrict__ X, short* __restrict__ Y,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    int N, int C, long long HW, int G, float eps, int do_silu) {
+  const int n = blockIdx.x / G;
+  const int g = blockIdx.x % G;
+  const int cpg = C / G;
+  const long long slab = (long long)cpg * HW;
+  const short* x = X + ((long long)n * C + (long long)g * cpg) * HW;
+  short* y = Y + ((long long)n * C + (long long)g * cpg) * HW;
+
+  const long long full = (slab / 8) * 8;
+  float s = 0.f, ss = 0.f;
+  for (long long i = (long long)threadIdx.x * 8; i + 8 <= slab; i += 256 * 8) {
+    bf16x8 v = *(const bf16x8*)&x[i];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bf2f(v[j]);
+      s += f;
+      ss += f * f;
+    }
+  }
+  for (long long i = full + threadIdx.x; i < slab; i += 256) {
+    float f = bf2f(x[i]);
+    s += f;
+    ss += f * f;
+  }
+  // block reduce
+  __shared__ float red[2][4];
+  s = wave_sum(s);
+  ss = wave_sum(ss);
+  int w = threadIdx.x / WAVE, l = threadIdx.x % WAVE;
+  if (l == 0) {
+    red[0][w] = s;
+    red[1][w] = ss;
+  }
+  __syncthreads();
+  s = red[0][0] + red[0][1] + red[0][2] + red[0][3];
+  ss = red[1][0] + red[1][1] + red[1][2] + red[1][3];
+  float mean = s / (float)slab;
+  float var = ss / (float)slab - mean * mean;
+  float rstd = rsqrtf(var + eps);
+
+  for (long long i = (long long)threadIdx.x * 8; i + 8 <= slab; i += 256 * 8) {
+    bf16x8 v = *(const bf16x8*)&x[i];
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      long long c = g * cpg + (i + j) / HW;
+      float f = (bf2f(v[j]) - mean) * rstd * gamma[c] + beta[c];
+      if (do_silu) f = f / (1.f + __expf(-f));
+      o[j] = f2bf(f);
+    }
+    *(bf16x8*)&y[i] = o;
+  }
+  for (long long i = full + threadIdx.x; i < slab; i += 256) {
+    long long c = g * cpg + i / HW;
+    float f = (bf2f(x[i]) - mean) * rstd * gamma[c] + beta[c];
+    if (do_silu) f = f / (1.f + __expf(-f));
+    y[i] = f2bf(f);
+  }
+}
+
+extern "C" void groupnorm_silu_bf16(const void* x, void* y, const float* gamma,
+                                    const float* beta, int N, int C,
+                                    long long HW, int G, float eps, int do_silu,
+                                    hipStream_t stream) {
+  dim3 grid(N * G);
+  hipLaunchKernelGGL(groupnorm_silu_kernel, grid, dim3(256), 0, stream,
+                     (const short*)x, (short*)y, gamma, beta, N, C, HW, G, eps,
+                     do_silu);
+}
+
+// ---------------------------------------------------------------- LayerNorm
+// x: [rows, D]; one wave per row for D<=4096 (bf16x8 loads), block = 4 rows.
+
+__global__ __launch_bounds__(256) void layernorm_kernel(
+    const short* __restrict__ X, short* __restrict__ Y,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    long long rows, int D, float eps) {
+  int w = threadIdx.x / WAVE, l = threadIdx.x % WAVE;
+  long long row = (long long)blockIdx.x * 4 + w;
+  if (row >= rows) return;
+  const short* x = X + row * D;
+  short* y = Y + row * D;
+  float s = 0.f, ss = 0.f;
+  for (int i = l * 8; i + 8 <= D; i += WAVE * 8) {
+    bf16x8 v = *(const bf16x8*)&x[i];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bf2f(v[j]);
+      s += f;
+      ss += f * f;
+    }
+  }
+  for (int i = (D / 8) * 8 + l; i < D; i += WAVE) {
+    float f = bf2f(x[i]);
+    s += f;
+    ss += f * f;
+  }
+  s = wave_sum(s);
+  ss = wave_sum(ss);
+  float mean = s / D, var = ss / D - mean * mean;
+  float rstd = rsqrtf(var + eps);
+  for (int i = l * 8; i + 8 <= D; i += WAVE * 8) {
+    bf16x8 v = *(const bf16x8*)&x[i];
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = f2bf((bf2f(v[j]) - mean) * rstd * gamma[i + j] + beta[i + j]);
+    *(bf16x8*)&y[i] = o;
+  }
+  for (int i = (D / 8) * 8 + l; i < D; i += WAVE)
+    y[i] = f2bf((bf2f(x[i]) - mean) * rstd * gamma[i] + beta[i]);
+}
+
+extern "C" void layernorm_bf16(const void* x, void* y, const float* gamma,
+                               const float* beta, long long rows, int D,
+                               float eps, hipStream_t stream) {
+  long long blocks = (rows + 3) / 4;
+  hipLaunchKernelGGL(layernorm_kernel, dim3((unsigned)blocks), dim3(256), 0,
+                     stream, (const short*)x, (short*)y, gamma, beta, rows, D,
+                     eps);
+}
+
+// ---------------------------------------------------------------- RMSNorm
+
+__global__ __launch_bounds__(256) void rmsnorm_kernel(
+    const short* __restrict__ X, short* __restrict__ Y,
+    const float* __restrict__ gamma, long long rows, int D, float eps) {
+  int w = threadIdx.x / WAVE, l = threadIdx.x % WAVE;
+  long long row = (long long)blockIdx.x * 4 + w;
+  if (row >= rows) return;
+  const short* x = X + row * D;
+  short* y = Y + row * D;
+  float ss = 0.f;
+  for (int i = l * 8; i + 8 <= D; i += WAVE * 8) {
+    bf16x8 v = *(const bf16x8*)&x[i];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bf2f(v[j]);
+      ss += f * f;
+    }
+  }
+  ss = wave_sum(ss);
+  float rstd = rsqrtf(ss / D + eps);
+  for (int i = l * 8; i + 8 <= D; i += WAVE * 8) {
+    bf16x8 v = *(const bf16x8*)&x[i];
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = f2bf(bf2f(v[j]) * rstd * gamma[i + j]);
+    *(bf16x8*)&y[i] = o;
+  }
+}
+
+extern "C" void rmsnorm_bf16(const void* x, void* y, const float* gamma,
+                             long long rows, int D, float eps,
+                             hipStream_t stream) {
+  long long blocks = (rows + 3) / 4;
+  hipLaunchKernelGGL(rmsnorm_kernel, dim3((unsigned)blocks), dim3(256), 0,
+                     stream, (const short*)x, (short*)y, gamma, rows, D, eps);
+}

@@ -1,0 +1,149 @@
+"""Public op API: gfx950 HIP kernels on GPU, fp32 torch references on CPU.
+
+Dispatch contract: on a GPU tensor the hand-written kernel MUST run — if the
+in-tree extension is missing the call raises rather than silently falling back
+(build contract: "make your ops fail loudly if their extension is missing on a
+GPU box").  CPU tensors use the fp32 reference path so the whole model stack is
+testable without hardware.
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+
+from . import reference as ref
+from ._build import get_ext
+
+
+def _ext_for(t: torch.Tensor):
+    if t.is_cuda:
+        return get_ext(required=True)
+    return None
+
+
+def attention(q, k, v, causal: bool = False, scale: Optional[float] = None):
+    """Flash attention fwd (K1/K5/K7). q [B,Hq,Sq,D], k/v [B,Hkv,Sk,D] bf16."""
+    scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    ext = _ext_for(q)
+    if ext is None:
+        return ref.attention_ref(q, k, v, causal, scale)
+    return ext.attention(q.contiguous(), k.contiguous(), v.contiguous(), causal, scale)
+
+
+def paged_decode(q, k_cache, v_cache, block_table, seq_lens, block_size: int = 0,
+                 scale: Optional[float] = None):
+    """Decode attention over a (paged) KV cache (K6). q [B,Hq,D]."""
+    scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    ext = _ext_for(q)
+    if ext is None:
+        return ref.paged_decode_ref(q, k_cache, v_cache, block_table, seq_lens,
+                                    block_size, scale)
+    return ext.paged_decode(q.contiguous(), k_cache, v_cache, block_table,
+                            seq_lens.int(), block_size, scale)
+
+
+def groupnorm_silu(x, gamma, beta, groups: int = 32, eps: float = 1e-5,
+                   do_silu: bool = True):
+    ext = _ext_for(x)
+    if ext is None:
+        return ref.groupnorm_silu_ref(x, gamma, beta, groups, eps, do_silu)
+    return ext.groupnorm_silu(x.contiguous(), gamma.float().contiguous(),
+                              beta.float().contiguous(), groups, eps, do_silu)
+
+
+def layernorm(x, gamma, beta, eps: float = 1e-5):
+    ext = _ext_for(x)
+    if ext is None:
+        return ref.layernorm_ref(x, gamma, beta, eps)
+    return ext.layernorm(x.contiguous(), gamma.float().contiguous(),
+                         beta.float().contiguous(), eps)
+
+
+def rmsnorm(x, gamma, eps: float = 1e-6):
+    ext = _ext_for(x)
+    if ext is None:
+        return ref.rmsnorm_ref(x, gamma, eps)
+    return ext.rmsnorm(x.contiguous(), gamma.float().contiguous(), eps)
+
+
+def cfg_euler(x_t, eps_c, eps_u, guidance: float, dsigma: float):
+    """Fused CFG combine + Euler update (K4)."""
+    ext = _ext_for(x_t)
+    if ext is None:
+        return ref.cfg_euler_ref(x_t, eps_c, eps_u, guidance, dsigma)
+    return ext.cfg_euler(x_t.contiguous(), eps_c.contiguous(),
+                         eps_u.contiguous() if eps_u is not None else None,
+                         guidance, dsigma)
+
+
+def silu_mul(a, b):
+    ext = _ext_for(a)
+    if ext is None:
+        return ref.silu_mul_ref(a, b)
+    return ext.silu_mul(a.contiguous(), b.contiguous())
+
+
+def geglu(a, b):
+    ext = _ext_for(a)
+    if ext is None:
+        return ref.geglu_ref(a, b)
+    return ext.geglu(a.contiguous(), b.contiguous())
+
+
+def add_residual(a, b):
+    ext = _ext_for(a)
+    if ext is None:
+        return (a.float() + b.float()).to(a.dtype)
+    return ext.add_residual(a.contiguous(), b.contiguous())
+
+
+def rope(x, cos, sin, positions=None, inplace: bool = False):
+    """RoPE with host-precomputed tables. x [B,H,S,D] bf16."""
+    ext = _ext_for(x)
+    if ext is None:
+        return ref.rope_ref(x, cos, sin, positions)
+    y = x.contiguous() if not (inplace and x.is_contiguous()) else x
+    if y.data_ptr() == x.data_ptr() and not inplace:
+        y = x.clone()
+    ext.rope_(y, cos.contiguous(), sin.contiguous(),
+              positions.int() if positions is not None else None)
+    return y
+
+
+def rope_tables(max_seq: int, dim: int, base: float = 10000.0, device="cpu"):
+    """Host-side cos/sin tables (guide App.B: no on-device trig)."""
+    inv = 1.0 / (base ** (torch.arange(0, dim, 2, dtype=torch.float64) / dim))
+    t = torch.arange(max_seq, dtype=torch.float64)
+    freqs = torch.outer(t, inv)
+    return (freqs.cos().float().to(device), freqs.sin().float().to(device))
+
+
+def adamw_step(p, g, m, v, lr, beta1=0.9, beta2=0.999, eps=1e-8, wd=0.01, step=1):
+    """Fused AdamW (K9); in-place on p/m/v."""
+    ext = _ext_for(p)
+    if ext is None:
+        return ref.adamw_ref(p, g, m, v, lr, beta1, beta2, eps, wd, step)
+    ext.adamw_(p, g.contiguous(), m, v, lr, beta1, beta2, eps, wd, step)
+    return p
+
+
+def sample(logits, temperature: float = 1.0, seed: int = 0):
+    """Fused sampling (K8): gumbel-max over softmax(logits/T); T=0 → argmax."""
+    ext = _ext_for(logits)
+    if ext is None:
+        if temperature <= 0:
+            return logits.argmax(-1).int()
+        g = torch.Generator(device="cpu").manual_seed(seed or 1)
+        u = torch.rand(logits.shape, generator=g).clamp_min(1e-10)
+        gumbel = -(-u.log()).log()
+        return (logits.float() / temperature + gumbel).argmax(-1).int()
+    return ext.sample_gumbel(logits.float().contiguous(), temperature, seed)
+
+
+def softmax(x):
+    ext = _ext_for(x)
+    if ext is None:
+        return torch.softmax(x.float(), dim=-1)
+    return ext.softmax_fwd(x.float().contiguous())

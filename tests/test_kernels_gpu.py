@@ -1,0 +1,240 @@
+"""Numerics: every gfx950 HIP kernel vs its plain-PyTorch fp32 reference.
+
+Run on a real MI355X via:  gpurun -- 'python -m pytest tests -m gpu -x -q'
+Tolerances reflect bf16 I/O with f32 accumulation.
+"""
+import math
+
+import pytest
+import torch
+
+import modal_examples_amd.ops.functional as F
+import modal_examples_amd.ops.reference as ref
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(), reason="no GPU")
+
+
+def _close(a, b, atol=2e-2, rtol=2e-2, frac=0.999):
+    """bf16 comparison: allow a tiny fraction of stragglers (rounding at tile
+    boundaries) but require the bulk to match tightly."""
+    a, b = a.float(), b.float()
+    ok = (a - b).abs() <= (atol + rtol * b.abs())
+    good = ok.float().mean().item()
+    assert good >= frac, f"only {good:.5f} of elements within tol; max err {(a-b).abs().max().item():.4f}"
+
+
+@requires_gpu
+@pytest.mark.parametrize("D", [64, 128])
+@pytest.mark.parametrize("causal", [False, True])
+def test_attention_self(D, causal):
+    torch.manual_seed(0)
+    B, H, S = 2, 4, 512
+    q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+    out = F.attention(q, k, v, causal=causal)
+    exp = ref.attention_ref(q, k, v, causal=causal)
+    _close(out, exp)
+
+
+@requires_gpu
+def test_attention_asymmetric_detects_transpose():
+    """Asymmetric low-rank inputs catch silent output transposes (guide G9)."""
+    torch.manual_seed(1)
+    B, H, S, D = 1, 2, 256, 64
+    q = torch.zeros(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+    q[..., 0] = torch.linspace(-2, 2, S, device="cuda").to(torch.bfloat16)
+    k = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+    _close(F.attention(q, k, v), ref.attention_ref(q, k, v))
+
+
+@requires_gpu
+def test_attention_cross_short_kv():
+    """SDXL cross-attention: Sk=77 text tokens (non multiple of 32)."""
+    torch.manual_seed(2)
+    q = torch.randn(2, 10, 1024, 64, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(2, 10, 77, 64, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(2, 10, 77, 64, device="cuda", dtype=torch.bfloat16)
+    _close(F.attention(q, k, v), ref.attention_ref(q, k, v))
+
+
+@requires_gpu
+def test_attention_gqa_causal():
+    """Llama-3 shape: 32 q heads, 8 kv heads, D=128."""
+    torch.manual_seed(3)
+    q = torch.randn(1, 32, 384, 128, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(1, 8, 384, 128, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(1, 8, 384, 128, device="cuda", dtype=torch.bfloat16)
+    _close(F.attention(q, k, v, causal=True), ref.attention_ref(q, k, v, causal=True))
+
+
+@requires_gpu
+def test_attention_ragged_seq():
+    torch.manual_seed(4)
+    q = torch.randn(1, 2, 200, 64, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(1, 2, 200, 64, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(1, 2, 200, 64, device="cuda", dtype=torch.bfloat16)
+    _close(F.attention(q, k, v, causal=True), ref.attention_ref(q, k, v, causal=True))
+
+
+@requires_gpu
+@pytest.mark.parametrize("D", [64, 128])
+def test_paged_decode_contiguous(D):
+    torch.manual_seed(5)
+    B, Hq, Hkv, S = 3, 8, 2, 300
+    q = torch.randn(B, Hq, D, device="cuda", dtype=torch.bfloat16)
+    kc = torch.randn(B, Hkv, S, D, device="cuda", dtype=torch.bfloat16)
+    vc = torch.randn(B, Hkv, S, D, device="cuda", dtype=torch.bfloat16)
+    lens = torch.tensor([300, 127, 64], device="cuda", dtype=torch.int32)
+    out = F.paged_decode(q, kc, vc, None, lens)
+    exp = ref.paged_decode_ref(q, kc, vc, None, lens.cpu(), S)
+    _close(out, exp)
+
+
+@requires_gpu
+def test_paged_decode_block_table():
+    torch.manual_seed(6)
+    B, Hq, Hkv, D, BS = 2, 32, 8, 128, 16
+    nblocks, max_blocks = 64, 20
+    q = torch.randn(B, Hq, D, device="cuda", dtype=torch.bfloat16)
+    kc = torch.randn(nblocks, Hkv, BS, D, device="cuda", dtype=torch.bfloat16)
+    vc = torch.randn(nblocks, Hkv, BS, D, device="cuda", dtype=torch.bfloat16)
+    bt = torch.randperm(nblocks, device="cuda")[: B * max_blocks].view(B, max_blocks).int()
+    lens = torch.tensor([310, 77], device="cuda", dtype=torch.int32)
+    out = F.paged_decode(q, kc, vc, bt, lens, block_size=BS)
+    exp = ref.paged_decode_ref(q, kc, vc, bt.cpu(), lens.cpu(), BS)
+    _close(out, exp)
+
+
+@requires_gpu
+def test_groupnorm_silu():
+    torch.manual_seed(7)
+    x = torch.randn(2, 64, 32, 32, device="cuda", dtype=torch.bfloat16)
+    g = torch.randn(64, device="cuda")
+    b = torch.randn(64, device="cuda")
+    _close(F.groupnorm_silu(x, g, b, 32), ref.groupnorm_silu_ref(x, g, b, 32))
+    _close(F.groupnorm_silu(x, g, b, 32, do_silu=False),
+           ref.groupnorm_silu_ref(x, g, b, 32, do_silu=False))
+
+
+@requires_gpu
+def test_layernorm():
+    torch.manual_seed(8)
+    x = torch.randn(512, 1280, device="cuda", dtype=torch.bfloat16)
+    g = torch.randn(1280, device="cuda")
+    b = torch.randn(1280, device="cuda")
+    _close(F.layernorm(x, g, b), ref.layernorm_ref(x, g, b))
+
+
+@requires_gpu
+def test_rmsnorm():
+    torch.manual_seed(9)
+    x = torch.randn(1024, 4096, device="cuda", dtype=torch.bfloat16)
+    g = torch.randn(4096, device="cuda")
+    _close(F.rmsnorm(x, g), ref.rmsnorm_ref(x, g))
+
+
+@requires_gpu
+def test_cfg_euler():
+    torch.manual_seed(10)
+    n = 2 * 4 * 128 * 128 + 5  # ragged tail
+    x = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    ec = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    eu = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    _close(F.cfg_euler(x, ec, eu, 7.5, -0.3), ref.cfg_euler_ref(x, ec, eu, 7.5, -0.3))
+    _close(F.cfg_euler(x, ec, None, 0.0, -0.3), ref.cfg_euler_ref(x, ec, None, 0.0, -0.3))
+
+
+@requires_gpu
+def test_silu_mul_geglu_add():
+    torch.manual_seed(11)
+    a = torch.randn(4097 * 8, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(4097 * 8, device="cuda", dtype=torch.bfloat16)
+    _close(F.silu_mul(a, b), ref.silu_mul_ref(a, b))
+    _close(F.geglu(a, b), ref.geglu_ref(a, b))
+    _close(F.add_residual(a, b), (a.float() + b.float()).to(torch.bfloat16))
+
+
+@requires_gpu
+def test_rope():
+    torch.manual_seed(12)
+    B, H, S, D = 2, 8, 128, 128
+    x = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+    cos, sin = F.rope_tables(256, D, device="cuda")
+    _close(F.rope(x, cos, sin), ref.rope_ref(x, cos, sin))
+    pos = torch.arange(64, 64 + S, device="cuda").int()
+    _close(F.rope(x, cos, sin, positions=pos), ref.rope_ref(x, cos, sin, pos.long()))
+
+
+@requires_gpu
+def test_adamw():
+    torch.manual_seed(13)
+    n = 10_001
+    p = torch.randn(n, device="cuda")
+    g = torch.randn(n, device="cuda")
+    m = torch.zeros(n, device="cuda")
+    v = torch.zeros(n, device="cuda")
+    p2, m2, v2 = p.clone(), m.clone(), v.clone()
+    for step in (1, 2, 3):
+        F.adamw_step(p, g, m, v, lr=1e-2, step=step)
+        ref.adamw_ref(p2, g, m2, v2, 1e-2, 0.9, 0.999, 1e-8, 0.01, step)
+    assert torch.allclose(p, p2, atol=1e-5, rtol=1e-4), (p - p2).abs().max()
+
+
+@requires_gpu
+def test_adamw_bf16_params():
+    torch.manual_seed(14)
+    n = 4096
+    p = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    g = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    m = torch.zeros(n, device="cuda")
+    v = torch.zeros(n, device="cuda")
+    p2, m2, v2 = p.clone(), m.clone(), v.clone()
+    F.adamw_step(p, g, m, v, lr=1e-2, step=1)
+    ref.adamw_ref(p2, g, m2, v2, 1e-2, 0.9, 0.999, 1e-8, 0.01, 1)
+    _close(p, p2, atol=1e-2)
+
+
+@requires_gpu
+def test_sample_greedy_and_distribution():
+    torch.manual_seed(15)
+    logits = torch.randn(4, 32000, device="cuda")
+    greedy = F.sample(logits, temperature=0.0)
+    assert torch.equal(greedy.long(), logits.argmax(-1))
+    # distribution sanity: a strongly-peaked row should sample its peak mostly
+    peaked = torch.zeros(1, 1000, device="cuda")
+    peaked[0, 123] = 10.0
+    hits = 0
+    for s in range(50):
+        tok = F.sample(peaked, temperature=1.0, seed=1000 + s)
+        hits += int(tok.item() == 123)
+    assert hits >= 45, f"peak sampled only {hits}/50"
+
+
+@requires_gpu
+def test_softmax_rows():
+    torch.manual_seed(16)
+    x = torch.randn(8, 128000, device="cuda")
+    _close(F.softmax(x), torch.softmax(x, -1), atol=1e-6, rtol=1e-4, frac=1.0)
+
+
+@requires_gpu
+def test_snapshot_roundtrip():
+    from modal_examples_amd.gpu.snapshot import WeightSnapshot
+
+    torch.manual_seed(17)
+    tensors = {
+        "a": torch.randn(1024, 1024, device="cuda", dtype=torch.bfloat16),
+        "b": torch.randn(333, device="cuda"),
+    }
+    snap = WeightSnapshot.capture(tensors)
+    originals = {k: t.clone() for k, t in tensors.items()}
+    for t in tensors.values():
+        t.zero_()
+    snap.restore(tensors)
+    for k in tensors:
+        assert torch.equal(tensors[k], originals[k]), k
+    snap.close()

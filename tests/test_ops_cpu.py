@@ -1,0 +1,106 @@
+"""CPU-tier op checks: the fp32 reference implementations are themselves
+validated against independent torch formulations, so the GPU numerics tests
+compare kernels against a trusted baseline."""
+import math
+
+import torch
+
+import modal_examples_amd.ops.functional as F
+import modal_examples_amd.ops.reference as ref
+
+
+def test_attention_ref_matches_sdpa():
+    torch.manual_seed(0)
+    q = torch.randn(2, 4, 64, 64)
+    k = torch.randn(2, 4, 64, 64)
+    v = torch.randn(2, 4, 64, 64)
+    for causal in (False, True):
+        exp = torch.nn.functional.scaled_dot_product_attention(q, k, v, is_causal=causal)
+        got = ref.attention_ref(q, k, v, causal=causal)
+        assert torch.allclose(got, exp, atol=1e-5), causal
+
+
+def test_attention_ref_gqa():
+    torch.manual_seed(1)
+    q = torch.randn(1, 8, 32, 64)
+    k = torch.randn(1, 2, 32, 64)
+    v = torch.randn(1, 2, 32, 64)
+    out = ref.attention_ref(q, k, v)
+    exp = torch.nn.functional.scaled_dot_product_attention(
+        q, k.repeat_interleave(4, 1), v.repeat_interleave(4, 1)
+    )
+    assert torch.allclose(out, exp, atol=1e-5)
+
+
+def test_paged_decode_ref_matches_attention_ref():
+    torch.manual_seed(2)
+    B, Hq, Hkv, S, D = 2, 4, 2, 40, 64
+    q = torch.randn(B, Hq, 1, D)
+    k = torch.randn(B, Hkv, S, D)
+    v = torch.randn(B, Hkv, S, D)
+    lens = torch.tensor([S, S])
+    dec = ref.paged_decode_ref(q[:, :, 0], k, v, None, lens, S)
+    att = ref.attention_ref(q, k, v)[:, :, 0]
+    assert torch.allclose(dec, att, atol=1e-5)
+
+
+def test_norm_refs():
+    torch.manual_seed(3)
+    x = torch.randn(4, 32, 8, 8)
+    g, b = torch.randn(32), torch.randn(32)
+    gn = ref.groupnorm_silu_ref(x, g, b, 8, do_silu=False)
+    exp = torch.nn.functional.group_norm(x, 8, g, b)
+    assert torch.allclose(gn, exp, atol=1e-5)
+
+    x2 = torch.randn(16, 256)
+    g2, b2 = torch.randn(256), torch.randn(256)
+    assert torch.allclose(
+        ref.layernorm_ref(x2, g2, b2),
+        torch.nn.functional.layer_norm(x2, (256,), g2, b2),
+        atol=1e-5,
+    )
+
+
+def test_rope_ref_rotation_property():
+    """RoPE preserves norms and inner products depend only on relative pos."""
+    torch.manual_seed(4)
+    D = 64
+    cos, sin = F.rope_tables(128, D)
+    x = torch.randn(1, 1, 128, D)
+    y = ref.rope_ref(x, cos, sin)
+    assert torch.allclose(x.norm(dim=-1), y.norm(dim=-1), atol=1e-4)
+    # relative property: <R_m q, R_n k> == <R_{m-n} q, k>
+    q = torch.randn(1, 1, 1, D)
+    k = torch.randn(1, 1, 1, D)
+    qk = torch.cat([q, k], dim=2)
+
+    def rot(vec, pos):
+        return ref.rope_ref(vec, cos, sin, positions=torch.tensor([pos]))
+
+    lhs = (rot(q, 7) * rot(k, 3)).sum()
+    rhs = (rot(q, 4) * rot(k, 0)).sum()
+    assert torch.allclose(lhs, rhs, atol=1e-3)
+
+
+def test_adamw_ref_matches_torch_optim():
+    torch.manual_seed(5)
+    p = torch.randn(100)
+    g = torch.randn(100)
+    pt = p.clone().requires_grad_(True)
+    opt = torch.optim.AdamW([pt], lr=1e-2, betas=(0.9, 0.999), eps=1e-8, weight_decay=0.01)
+    pt.grad = g.clone()
+    opt.step()
+    m = torch.zeros(100)
+    v = torch.zeros(100)
+    ref.adamw_ref(p, g, m, v, 1e-2, 0.9, 0.999, 1e-8, 0.01, 1)
+    assert torch.allclose(p, pt.detach(), atol=1e-6)
+
+
+def test_functional_cpu_paths():
+    """CPU dispatch goes through references without an extension."""
+    x = torch.randn(4, 8, 16, 64, dtype=torch.bfloat16)
+    out = F.attention(x, x, x, causal=True)
+    assert out.shape == x.shape and out.dtype == torch.bfloat16
+    assert F.sample(torch.tensor([[0.0, 100.0, 0.0]]), temperature=0.0).item() == 1
+    t = F.sample(torch.tensor([[0.0, 100.0, 0.0]]), temperature=1.0, seed=7)
+    assert t.item() == 1
