@@ -1,0 +1,142 @@
+"""Shared MI355X-native building blocks for the diffusion models.
+
+Every hot op routes to the gfx950 kernels in ops/functional.py; plain GEMMs and
+convolutions go through torch (hipBLASLt / MIOpen per the library-GEMM rule).
+QKV projections are single fused GEMMs (K2 fusion, reference trigger:
+stable_diffusion/flux.py:236-237 `fuse_qkv_projections`).
+"""
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn as nn
+
+from ...ops import functional as OF
+
+
+class LayerNormK(nn.Module):
+    """LayerNorm over last dim via the gfx950 kernel (f32 affine params)."""
+
+    def __init__(self, dim: int, eps: float = 1e-5):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(dim))
+        self.bias = nn.Parameter(torch.zeros(dim))
+        self.eps = eps
+
+    def forward(self, x):
+        return OF.layernorm(x, self.weight.float(), self.bias.float(), self.eps)
+
+
+class RMSNormK(nn.Module):
+    def __init__(self, dim: int, eps: float = 1e-6):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(dim))
+        self.eps = eps
+
+    def forward(self, x):
+        return OF.rmsnorm(x, self.weight.float(), self.eps)
+
+
+class GroupNormSiLU(nn.Module):
+    """Fused GroupNorm(+SiLU) NCHW via the gfx950 kernel."""
+
+    def __init__(self, channels: int, groups: int = 32, eps: float = 1e-5,
+                 silu: bool = True):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(channels))
+        self.bias = nn.Parameter(torch.zeros(channels))
+        self.groups = groups
+        self.eps = eps
+        self.silu = silu
+
+    def forward(self, x):
+        return OF.groupnorm_silu(x, self.weight.float(), self.bias.float(),
+                                 self.groups, self.eps, self.silu)
+
+
+class SelfAttention(nn.Module):
+    """Fused-QKV self-attention over [B, S, C] → gfx950 flash kernel."""
+
+    def __init__(self, dim: int, head_dim: int = 64):
+        super().__init__()
+        assert dim % head_dim == 0
+        self.heads = dim // head_dim
+        self.head_dim = head_dim
+        self.qkv = nn.Linear(dim, 3 * dim, bias=False)
+        self.out = nn.Linear(dim, dim, bias=True)
+
+    def forward(self, x):
+        B, S, C = x.shape
+        qkv = self.qkv(x).view(B, S, 3, self.heads, self.head_dim)
+        q, k, v = (qkv[:, :, i].transpose(1, 2).contiguous() for i in range(3))
+        o = OF.attention(q, k, v, causal=False)
+        return self.out(o.transpose(1, 2).reshape(B, S, C))
+
+
+class CrossAttention(nn.Module):
+    """Cross-attention: q from x, fused kv from context (text conditioning)."""
+
+    def __init__(self, dim: int, ctx_dim: int, head_dim: int = 64):
+        super().__init__()
+        self.heads = dim // head_dim
+        self.head_dim = head_dim
+        self.q = nn.Linear(dim, dim, bias=False)
+        self.kv = nn.Linear(ctx_dim, 2 * dim, bias=False)
+        self.out = nn.Linear(dim, dim, bias=True)
+
+    def forward(self, x, ctx):
+        B, S, C = x.shape
+        Sk = ctx.shape[1]
+        q = self.q(x).view(B, S, self.heads, self.head_dim).transpose(1, 2).contiguous()
+        kv = self.kv(ctx).view(B, Sk, 2, self.heads, self.head_dim)
+        k = kv[:, :, 0].transpose(1, 2).contiguous()
+        v = kv[:, :, 1].transpose(1, 2).contiguous()
+        o = OF.attention(q, k, v, causal=False)
+        return self.out(o.transpose(1, 2).reshape(B, S, C))
+
+
+class GEGLUFeedForward(nn.Module):
+    """x → Linear(2*4c) → gelu(a)*b (fused kernel) → Linear(c)."""
+
+    def __init__(self, dim: int, mult: int = 4):
+        super().__init__()
+        inner = dim * mult
+        self.proj_in = nn.Linear(dim, 2 * inner, bias=True)
+        self.proj_out = nn.Linear(inner, dim, bias=True)
+        self.inner = inner
+
+    def forward(self, x):
+        ab = self.proj_in(x)
+        a, b = ab[..., : self.inner], ab[..., self.inner:]
+        return self.proj_out(OF.geglu(a.contiguous(), b.contiguous()))
+
+
+class TransformerBlock(nn.Module):
+    """ln→self-attn→ln→cross-attn→ln→GEGLU-FF, pre-norm residuals (the
+    diffusers BasicTransformerBlock shape exercised at text_to_image.py:114)."""
+
+    def __init__(self, dim: int, ctx_dim: int, head_dim: int = 64):
+        super().__init__()
+        self.norm1 = LayerNormK(dim)
+        self.attn1 = SelfAttention(dim, head_dim)
+        self.norm2 = LayerNormK(dim)
+        self.attn2 = CrossAttention(dim, ctx_dim, head_dim)
+        self.norm3 = LayerNormK(dim)
+        self.ff = GEGLUFeedForward(dim)
+
+    def forward(self, x, ctx):
+        x = x + self.attn1(self.norm1(x))
+        x = x + self.attn2(self.norm2(x), ctx)
+        x = x + self.ff(self.norm3(x))
+        return x
+
+
+def timestep_embedding(t: torch.Tensor, dim: int, max_period: float = 10000.0):
+    """Sinusoidal timestep embedding, f32 (host-side trig, tiny)."""
+    half = dim // 2
+    freqs = torch.exp(
+        -math.log(max_period) * torch.arange(half, dtype=torch.float32, device=t.device) / half
+    )
+    args = t.float()[:, None] * freqs[None]
+    return torch.cat([torch.cos(args), torch.sin(args)], dim=-1)

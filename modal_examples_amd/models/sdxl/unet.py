@@ -1,0 +1,219 @@
+"""SDXL-class UNet, built MI355X-first.
+
+Architecture follows the SDXL-base UNet the reference's canonical image
+pipeline loads (reference trigger: 06_gpu_and_ml/stable_diffusion/
+text_to_image.py:92-120 — `StableDiffusion3Pipeline`-class workloads; the
+BASELINE.json headline names SDXL 1024px):
+
+  channels (320, 640, 1280), transformer depths (0, 2, 10), head_dim 64,
+  context dim 2048 (dual-CLIP concat), addition embedding 2816
+  (pooled 1280 + 6×256 Fourier time-ids), latent 4ch at 128×128 for 1024px.
+
+Hot ops → gfx950 kernels: attention (K1), GEGLU/GN+SiLU fusions, LayerNorm;
+GEMMs/convs → hipBLASLt/MIOpen.  ~2.6B params, bf16.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+
+from .layers import (
+    GroupNormSiLU,
+    TransformerBlock,
+    timestep_embedding,
+)
+
+
+@dataclass
+class UNetConfig:
+    in_channels: int = 4
+    out_channels: int = 4
+    channels: tuple = (320, 640, 1280)
+    layers_per_block: int = 2
+    transformer_depth: tuple = (0, 2, 10)
+    ctx_dim: int = 2048
+    head_dim: int = 64
+    time_embed_dim: int = 1280
+    pooled_dim: int = 1280
+    fourier_dim: int = 128  # per-value fourier half-dim; 6 values * 2*fdim
+    addition_dim: int = 2816  # 1280 pooled + 6 * 256 fourier time-ids
+
+    @staticmethod
+    def sdxl() -> "UNetConfig":
+        return UNetConfig()
+
+    @staticmethod
+    def small() -> "UNetConfig":
+        """Test-sized variant (same topology, ~1/100 params)."""
+        return UNetConfig(channels=(64, 128, 256), transformer_depth=(0, 1, 2),
+                          ctx_dim=256, head_dim=64, time_embed_dim=256,
+                          pooled_dim=256, fourier_dim=16,
+                          addition_dim=256 + 6 * 32)
+
+
+class ResnetBlock(nn.Module):
+    def __init__(self, c_in: int, c_out: int, temb_dim: int):
+        super().__init__()
+        self.norm1 = GroupNormSiLU(c_in)
+        self.conv1 = nn.Conv2d(c_in, c_out, 3, padding=1)
+        self.temb_proj = nn.Linear(temb_dim, c_out)
+        self.norm2 = GroupNormSiLU(c_out)
+        self.conv2 = nn.Conv2d(c_out, c_out, 3, padding=1)
+        self.skip = nn.Conv2d(c_in, c_out, 1) if c_in != c_out else nn.Identity()
+
+    def forward(self, x, temb):
+        h = self.conv1(self.norm1(x))
+        h = h + self.temb_proj(torch.nn.functional.silu(temb))[:, :, None, None]
+        h = self.conv2(self.norm2(h))
+        return self.skip(x) + h
+
+
+class SpatialTransformer(nn.Module):
+    """GN → 1x1 proj_in → depth× TransformerBlock over flattened HW → proj_out."""
+
+    def __init__(self, channels: int, depth: int, ctx_dim: int, head_dim: int):
+        super().__init__()
+        self.norm = GroupNormSiLU(channels, silu=False)
+        self.proj_in = nn.Linear(channels, channels)
+        self.blocks = nn.ModuleList(
+            [TransformerBlock(channels, ctx_dim, head_dim) for _ in range(depth)]
+        )
+        self.proj_out = nn.Linear(channels, channels)
+
+    def forward(self, x, ctx):
+        B, C, H, W = x.shape
+        res = x
+        h = self.norm(x).permute(0, 2, 3, 1).reshape(B, H * W, C)
+        h = self.proj_in(h)
+        for blk in self.blocks:
+            h = blk(h, ctx)
+        h = self.proj_out(h)
+        return res + h.reshape(B, H, W, C).permute(0, 3, 1, 2)
+
+
+class Downsample(nn.Module):
+    def __init__(self, c):
+        super().__init__()
+        self.conv = nn.Conv2d(c, c, 3, stride=2, padding=1)
+
+    def forward(self, x):
+        return self.conv(x)
+
+
+class Upsample(nn.Module):
+    def __init__(self, c):
+        super().__init__()
+        self.conv = nn.Conv2d(c, c, 3, padding=1)
+
+    def forward(self, x):
+        return self.conv(torch.nn.functional.interpolate(x, scale_factor=2.0, mode="nearest"))
+
+
+class UNetXL(nn.Module):
+    def __init__(self, cfg: UNetConfig = None):
+        super().__init__()
+        cfg = cfg or UNetConfig.sdxl()
+        self.cfg = cfg
+        ch = cfg.channels
+        ted = cfg.time_embed_dim
+
+        self.conv_in = nn.Conv2d(cfg.in_channels, ch[0], 3, padding=1)
+        self.time_mlp = nn.Sequential(
+            nn.Linear(ch[0], ted), nn.SiLU(), nn.Linear(ted, ted)
+        )
+        self.add_mlp = nn.Sequential(
+            nn.Linear(cfg.addition_dim, ted), nn.SiLU(), nn.Linear(ted, ted)
+        )
+
+        # down
+        self.down_blocks = nn.ModuleList()
+        self.downsamplers = nn.ModuleList()
+        skip_chs = [ch[0]]
+        c_prev = ch[0]
+        for lvl, c in enumerate(ch):
+            blocks = nn.ModuleList()
+            for _ in range(cfg.layers_per_block):
+                entry = nn.ModuleDict({"resnet": ResnetBlock(c_prev, c, ted)})
+                if cfg.transformer_depth[lvl] > 0:
+                    entry["attn"] = SpatialTransformer(
+                        c, cfg.transformer_depth[lvl], cfg.ctx_dim, cfg.head_dim
+                    )
+                blocks.append(entry)
+                c_prev = c
+                skip_chs.append(c)
+            self.down_blocks.append(blocks)
+            if lvl < len(ch) - 1:
+                self.downsamplers.append(Downsample(c))
+                skip_chs.append(c)
+            else:
+                self.downsamplers.append(nn.Identity())
+
+        # mid
+        top = ch[-1]
+        self.mid_res1 = ResnetBlock(top, top, ted)
+        self.mid_attn = SpatialTransformer(top, cfg.transformer_depth[-1],
+                                           cfg.ctx_dim, cfg.head_dim)
+        self.mid_res2 = ResnetBlock(top, top, ted)
+
+        # up (reverse, layers_per_block+1 resnets with skip concat)
+        self.up_blocks = nn.ModuleList()
+        self.upsamplers = nn.ModuleList()
+        c_prev = top
+        for lvl in reversed(range(len(ch))):
+            c = ch[lvl]
+            blocks = nn.ModuleList()
+            for _ in range(cfg.layers_per_block + 1):
+                skip_c = skip_chs.pop()
+                entry = nn.ModuleDict(
+                    {"resnet": ResnetBlock(c_prev + skip_c, c, ted)}
+                )
+                if cfg.transformer_depth[lvl] > 0:
+                    entry["attn"] = SpatialTransformer(
+                        c, cfg.transformer_depth[lvl], cfg.ctx_dim, cfg.head_dim
+                    )
+                blocks.append(entry)
+                c_prev = c
+            self.up_blocks.append(blocks)
+            if lvl > 0:
+                self.upsamplers.append(Upsample(c))
+            else:
+                self.upsamplers.append(nn.Identity())
+
+        self.norm_out = GroupNormSiLU(ch[0])
+        self.conv_out = nn.Conv2d(ch[0], cfg.out_channels, 3, padding=1)
+
+    def forward(self, x, t, ctx, pooled_add):
+        """x [B,4,H,W] bf16, t [B] f32 timesteps, ctx [B,77,ctx_dim] bf16,
+        pooled_add [B, addition_dim] bf16 (pooled text + fourier time-ids)."""
+        dt = x.dtype
+        temb = self.time_mlp(timestep_embedding(t, self.cfg.channels[0]).to(dt))
+        temb = temb + self.add_mlp(pooled_add)
+
+        h = self.conv_in(x)
+        skips = [h]
+        for lvl, blocks in enumerate(self.down_blocks):
+            for entry in blocks:
+                h = entry["resnet"](h, temb)
+                if "attn" in entry:
+                    h = entry["attn"](h, ctx)
+                skips.append(h)
+            if not isinstance(self.downsamplers[lvl], nn.Identity):
+                h = self.downsamplers[lvl](h)
+                skips.append(h)
+
+        h = self.mid_res1(h, temb)
+        h = self.mid_attn(h, ctx)
+        h = self.mid_res2(h, temb)
+
+        for i, blocks in enumerate(self.up_blocks):
+            for entry in blocks:
+                h = torch.cat([h, skips.pop()], dim=1)
+                h = entry["resnet"](h, temb)
+                if "attn" in entry:
+                    h = entry["attn"](h, ctx)
+            if not isinstance(self.upsamplers[i], nn.Identity):
+                h = self.upsamplers[i](h)
+
+        return self.conv_out(self.norm_out(h))

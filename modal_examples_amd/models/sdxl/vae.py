@@ -1,0 +1,96 @@
+"""SDXL-class VAE decoder (K3): latent [B,4,128,128] → RGB [B,3,1024,1024].
+
+Architecture mirrors the SDXL AutoencoderKL decoder the reference's pipelines
+call (trigger: text_to_image.py:114 `pipe(...)` → vae.decode; flux.py:259-261
+compiles exactly this module).  Channels (512,512,256,128), 3 resnets per
+level, nearest-2x upsample, mid-block single-head attention over 128² tokens.
+
+MI355X mapping: GroupNorm+SiLU → fused gfx950 kernel; convs → MIOpen;
+mid attention (1 head × 512 dim) → chunked hipBLASLt matmul+softmax (head_dim
+512 is outside the MFMA flash kernel's D∈{64,128}; one layer, ~0.5 GFLOP-ms).
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+from .layers import GroupNormSiLU
+
+
+class VAEResnet(nn.Module):
+    def __init__(self, c_in, c_out):
+        super().__init__()
+        self.norm1 = GroupNormSiLU(c_in)
+        self.conv1 = nn.Conv2d(c_in, c_out, 3, padding=1)
+        self.norm2 = GroupNormSiLU(c_out)
+        self.conv2 = nn.Conv2d(c_out, c_out, 3, padding=1)
+        self.skip = nn.Conv2d(c_in, c_out, 1) if c_in != c_out else nn.Identity()
+
+    def forward(self, x):
+        h = self.conv2(self.norm2(self.conv1(self.norm1(x))))
+        return self.skip(x) + h
+
+
+class VAEMidAttention(nn.Module):
+    """Single-head 512-dim attention, chunked over queries."""
+
+    def __init__(self, c, chunk=2048):
+        super().__init__()
+        self.norm = GroupNormSiLU(c, silu=False)
+        self.q = nn.Linear(c, c)
+        self.k = nn.Linear(c, c)
+        self.v = nn.Linear(c, c)
+        self.out = nn.Linear(c, c)
+        self.chunk = chunk
+        self.scale = c ** -0.5
+
+    def forward(self, x):
+        B, C, H, W = x.shape
+        h = self.norm(x).permute(0, 2, 3, 1).reshape(B, H * W, C)
+        q, k, v = self.q(h), self.k(h), self.v(h)
+        outs = []
+        for i in range(0, q.shape[1], self.chunk):
+            s = torch.matmul(q[:, i:i + self.chunk], k.transpose(1, 2)) * self.scale
+            p = torch.softmax(s.float(), dim=-1).to(v.dtype)
+            outs.append(torch.matmul(p, v))
+        o = self.out(torch.cat(outs, dim=1))
+        return x + o.reshape(B, H, W, C).permute(0, 3, 1, 2)
+
+
+class VAEDecoder(nn.Module):
+    def __init__(self, latent_channels=4, channels=(512, 512, 256, 128),
+                 out_channels=3, resnets_per_level=3, scaling_factor=0.13025):
+        super().__init__()
+        self.scaling_factor = scaling_factor
+        c0 = channels[0]
+        self.conv_in = nn.Conv2d(latent_channels, c0, 3, padding=1)
+        self.mid_res1 = VAEResnet(c0, c0)
+        self.mid_attn = VAEMidAttention(c0)
+        self.mid_res2 = VAEResnet(c0, c0)
+        self.levels = nn.ModuleList()
+        c_prev = c0
+        for li, c in enumerate(channels):
+            blocks = nn.ModuleList()
+            for _ in range(resnets_per_level):
+                blocks.append(VAEResnet(c_prev, c))
+                c_prev = c
+            up = nn.Conv2d(c, c, 3, padding=1) if li < len(channels) - 1 else None
+            self.levels.append(nn.ModuleList([blocks, nn.ModuleList([up] if up else [])]))
+        self.norm_out = GroupNormSiLU(channels[-1])
+        self.conv_out = nn.Conv2d(channels[-1], out_channels, 3, padding=1)
+
+    def forward(self, z):
+        h = self.conv_in(z / self.scaling_factor)
+        h = self.mid_res2(self.mid_attn(self.mid_res1(h)))
+        for blocks, ups in self.levels:
+            for blk in blocks:
+                h = blk(h)
+            if len(ups):
+                h = torch.nn.functional.interpolate(h, scale_factor=2.0, mode="nearest")
+                h = ups[0](h)
+        return self.conv_out(self.norm_out(h))
+
+
+class VAEDecoderSmall(VAEDecoder):
+    def __init__(self):
+        super().__init__(channels=(64, 64, 32, 32))

@@ -37,6 +37,12 @@ void adamw_step(void*, const void*, float*, float*, long long, float, float,
 void gumbel_sample(const float*, int*, int, int, float, unsigned long long,
                    hipStream_t);
 void softmax_rows(const float*, float*, int, int, hipStream_t);
+void scale_in_dev_bf16(const void*, void*, const float*, const long long*,
+                       long long, hipStream_t);
+void cfg_euler_dev_bf16(const void*, const void*, const void*, void*,
+                        const float*, const long long*, float, long long,
+                        hipStream_t);
+void advance_step(long long*, hipStream_t);
 }
 
 namespace {
@@ -210,6 +216,28 @@ torch::Tensor softmax_fwd(torch::Tensor x) {
   return y;
 }
 
+// graph-capturable denoise-step pieces: sigma schedule + step counter live on
+// device so one hipGraph capture serves every step.
+void scale_in_dev(torch::Tensor x, torch::Tensor y, torch::Tensor sigmas,
+                  torch::Tensor step) {
+  scale_in_dev_bf16(x.data_ptr(), y.data_ptr(), sigmas.data_ptr<float>(),
+                    (const long long*)step.data_ptr<int64_t>(), x.numel(), cur_stream());
+}
+
+void cfg_euler_dev(torch::Tensor xt, torch::Tensor eps_c,
+                   c10::optional<torch::Tensor> eps_u, torch::Tensor xn,
+                   torch::Tensor sigmas, torch::Tensor step, double guidance) {
+  cfg_euler_dev_bf16(xt.data_ptr(), eps_c.data_ptr(),
+                     eps_u.has_value() ? eps_u->data_ptr() : nullptr,
+                     xn.data_ptr(), sigmas.data_ptr<float>(),
+                     (const long long*)step.data_ptr<int64_t>(), (float)guidance, xt.numel(),
+                     cur_stream());
+}
+
+void step_advance(torch::Tensor step) {
+  advance_step((long long*)step.data_ptr<int64_t>(), cur_stream());
+}
+
 // ---------------------------------------------------------------- snapshot engine
 // Pinned-host weight snapshots: one hipHostMalloc region per snapshot, copies
 // overlapped across dedicated streams so restore saturates the PCIe Gen5 link
@@ -285,6 +313,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw_", &adamw_, "fused AdamW step (K9)");
   m.def("sample_gumbel", &sample_gumbel, "fused sampling (K8)");
   m.def("softmax_fwd", &softmax_fwd);
+  m.def("scale_in_dev", &scale_in_dev, "x * 1/sqrt(sigma^2+1), device sigma");
+  m.def("cfg_euler_dev", &cfg_euler_dev, "graph-capturable CFG+Euler step");
+  m.def("step_advance", &step_advance, "increment device step counter");
   m.def("snap_create", &snap_create, "pinned-host snapshot region (K13)");
   m.def("snap_save", &snap_save);
   m.def("snap_restore", &snap_restore);

@@ -202,3 +202,97 @@ extern "C" void rope_bf16(void* qk, const float* cosv, const float* sinv,
   hipLaunchKernelGGL(rope_kernel, dim3(grid), dim3(EW_BLOCK), 0, stream,
                      (short*)qk, cosv, sinv, BH, S, D, positions);
 }
+
+// ---------------------------------------------------------------- graph-capturable
+// Device-scalar variants for hipGraph capture of the denoise loop: the sigma
+// schedule lives in device memory and a device step counter advances INSIDE
+// the captured graph, so one capture serves every step (SURVEY.md §7 phase 3:
+// "hipGraph capture of the denoise loop").
+
+__global__ __launch_bounds__(EW_BLOCK) void scale_in_dev_kernel(
+    const short* __restrict__ X, short* __restrict__ Y,
+    const float* __restrict__ sigmas, const long long* __restrict__ step,
+    long long n) {
+  float s = sigmas[(int)*step];
+  float c = rsqrtf(s * s + 1.f);
+  long long stride = (long long)gridDim.x * EW_BLOCK * 8;
+  for (long long i = ((long long)blockIdx.x * EW_BLOCK + threadIdx.x) * 8;
+       i + 8 <= n; i += stride) {
+    bf16x8 x = *(const bf16x8*)&X[i];
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = f2bf(bf2f(x[j]) * c);
+    *(bf16x8*)&Y[i] = o;
+  }
+  long long full = (n / 8) * 8;
+  for (long long i = full + (long long)blockIdx.x * EW_BLOCK + threadIdx.x;
+       i < n; i += (long long)gridDim.x * EW_BLOCK)
+    Y[i] = f2bf(bf2f(X[i]) * c);
+}
+
+__global__ __launch_bounds__(EW_BLOCK) void cfg_euler_dev_kernel(
+    const short* __restrict__ Xt, const short* __restrict__ EpsC,
+    const short* __restrict__ EpsU, short* __restrict__ Xn,
+    const float* __restrict__ sigmas, const long long* __restrict__ step,
+    float guidance, long long n) {
+  int st = (int)*step;
+  float dsigma = sigmas[st + 1] - sigmas[st];
+  long long stride = (long long)gridDim.x * EW_BLOCK * 8;
+  for (long long i = ((long long)blockIdx.x * EW_BLOCK + threadIdx.x) * 8;
+       i + 8 <= n; i += stride) {
+    bf16x8 x = *(const bf16x8*)&Xt[i];
+    bf16x8 ec = *(const bf16x8*)&EpsC[i];
+    bf16x8 o;
+    if (EpsU != nullptr) {
+      bf16x8 eu = *(const bf16x8*)&EpsU[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float e = bf2f(eu[j]) + guidance * (bf2f(ec[j]) - bf2f(eu[j]));
+        o[j] = f2bf(bf2f(x[j]) + dsigma * e);
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o[j] = f2bf(bf2f(x[j]) + dsigma * bf2f(ec[j]));
+    }
+    *(bf16x8*)&Xn[i] = o;
+  }
+  long long full = (n / 8) * 8;
+  for (long long i = full + (long long)blockIdx.x * EW_BLOCK + threadIdx.x;
+       i < n; i += (long long)gridDim.x * EW_BLOCK) {
+    float e = bf2f(EpsC[i]);
+    if (EpsU != nullptr) {
+      float eu = bf2f(EpsU[i]);
+      e = eu + guidance * (e - eu);
+    }
+    Xn[i] = f2bf(bf2f(Xt[i]) + dsigma * e);
+  }
+}
+
+__global__ void advance_step_kernel(long long* step) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) *step += 1;
+}
+
+extern "C" void scale_in_dev_bf16(const void* x, void* y, const float* sigmas,
+                                  const long long* step, long long n,
+                                  hipStream_t stream) {
+  int grid = elementwise_grid(n, EW_BLOCK);
+  hipLaunchKernelGGL(scale_in_dev_kernel, dim3(grid), dim3(EW_BLOCK), 0,
+                     stream, (const short*)x, (short*)y, sigmas, step, n);
+}
+
+extern "C" void cfg_euler_dev_bf16(const void* xt, const void* eps_c,
+                                   const void* eps_u, void* xn,
+                                   const float* sigmas, const long long* step,
+                                   float guidance, long long n,
+                                   hipStream_t stream) {
+  int grid = elementwise_grid(n, EW_BLOCK);
+  hipLaunchKernelGGL(cfg_euler_dev_kernel, dim3(grid), dim3(EW_BLOCK), 0,
+                     stream, (const short*)xt, (const short*)eps_c,
+                     (const short*)eps_u, (short*)xn, sigmas, step, guidance,
+                     n);
+}
+
+extern "C" void advance_step(long long* step, hipStream_t stream) {
+  hipLaunchKernelGGL(advance_step_kernel, dim3(1), dim3(64), 0, stream, step);
+}

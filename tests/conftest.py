@@ -14,6 +14,7 @@ import pytest  # noqa: E402
 
 def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: needs a real MI355X (run via gpurun)")
+    config.addinivalue_line("markers", "slow: long-running CPU test")
 
 
 @pytest.fixture(autouse=True, scope="session")
