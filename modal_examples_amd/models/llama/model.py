@@ -1,0 +1,127 @@
+"""Llama-3-class decoder-only transformer, MI355X-first.
+
+The model family behind the reference's canonical LLM-serving examples
+(trigger: 06_gpu_and_ml/llm-serving/vllm_inference.py:158-209 — Llama/Qwen
+class models under vLLM; trtllm_latency.py:10-21 LLaMA-3-8B).
+
+MI355X mapping:
+  K7 prefill  → gfx950 flash attention (causal, GQA, D=128)
+  K6 decode   → gfx950 paged decode kernel over the paged KV cache
+  RMSNorm / RoPE / SwiGLU → fused gfx950 kernels
+  QKV and gate_up projections → single fused GEMMs (hipBLASLt)
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+
+from ...ops import functional as OF
+from ..sdxl.layers import RMSNormK
+
+
+@dataclass
+class LlamaConfig:
+    vocab_size: int = 128256
+    dim: int = 4096
+    n_layers: int = 32
+    n_heads: int = 32
+    n_kv_heads: int = 8
+    head_dim: int = 128
+    ffn_dim: int = 14336
+    rope_base: float = 500000.0
+    norm_eps: float = 1e-5
+    max_seq: int = 8192
+
+    @staticmethod
+    def llama3_8b() -> "LlamaConfig":
+        return LlamaConfig()
+
+    @staticmethod
+    def small() -> "LlamaConfig":
+        return LlamaConfig(vocab_size=1024, dim=256, n_layers=2, n_heads=4,
+                           n_kv_heads=2, head_dim=64, ffn_dim=512,
+                           rope_base=10000.0, max_seq=512)
+
+
+class LlamaBlock(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        d, hd = cfg.dim, cfg.head_dim
+        self.nq, self.nkv = cfg.n_heads, cfg.n_kv_heads
+        self.attn_norm = RMSNormK(d, cfg.norm_eps)
+        self.qkv = nn.Linear(d, (self.nq + 2 * self.nkv) * hd, bias=False)
+        self.o_proj = nn.Linear(self.nq * hd, d, bias=False)
+        self.ffn_norm = RMSNormK(d, cfg.norm_eps)
+        self.gate_up = nn.Linear(d, 2 * cfg.ffn_dim, bias=False)
+        self.down = nn.Linear(cfg.ffn_dim, d, bias=False)
+        self.hd = hd
+        self.ffn_dim = cfg.ffn_dim
+
+    def project_qkv(self, x):
+        """x [B,S,d] → q [B,nq,S,hd], k/v [B,nkv,S,hd]."""
+        B, S, _ = x.shape
+        qkv = self.qkv(x)
+        q, k, v = qkv.split(
+            [self.nq * self.hd, self.nkv * self.hd, self.nkv * self.hd], dim=-1
+        )
+        q = q.view(B, S, self.nq, self.hd).transpose(1, 2).contiguous()
+        k = k.view(B, S, self.nkv, self.hd).transpose(1, 2).contiguous()
+        v = v.view(B, S, self.nkv, self.hd).transpose(1, 2).contiguous()
+        return q, k, v
+
+    def ffn(self, x):
+        gu = self.gate_up(x)
+        g, u = gu[..., : self.ffn_dim], gu[..., self.ffn_dim:]
+        return self.down(OF.silu_mul(g.contiguous(), u.contiguous()))
+
+
+class LlamaModel(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.embed = nn.Embedding(cfg.vocab_size, cfg.dim)
+        self.blocks = nn.ModuleList([LlamaBlock(cfg) for _ in range(cfg.n_layers)])
+        self.norm = RMSNormK(cfg.dim, cfg.norm_eps)
+        self.lm_head = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
+        cos, sin = OF.rope_tables(cfg.max_seq, cfg.head_dim, cfg.rope_base)
+        self.register_buffer("rope_cos", cos, persistent=False)
+        self.register_buffer("rope_sin", sin, persistent=False)
+
+    @torch.no_grad()
+    def prefill(self, tokens: torch.Tensor, kv_writer=None):
+        """tokens [B,S] → logits [B, vocab] (last position only).
+        kv_writer(layer_idx, k, v): callback storing [B,nkv,S,hd] into cache."""
+        x = self.embed(tokens)
+        for li, blk in enumerate(self.blocks):
+            h = blk.attn_norm(x)
+            q, k, v = blk.project_qkv(h)
+            q = OF.rope(q, self.rope_cos, self.rope_sin)
+            k = OF.rope(k, self.rope_cos, self.rope_sin)
+            if kv_writer is not None:
+                kv_writer(li, k, v)
+            o = OF.attention(q, k, v, causal=True)
+            B, _, S, _ = o.shape
+            x = x + blk.o_proj(o.transpose(1, 2).reshape(B, S, -1))
+            x = x + blk.ffn(blk.ffn_norm(x))
+        x = self.norm(x[:, -1:])
+        return self.lm_head(x)[:, 0].float()
+
+    @torch.no_grad()
+    def decode_step(self, tokens, positions, kv_append, kv_attend):
+        """One token per sequence. tokens [B], positions [B] int32;
+        kv_append(li, k, v): store [B,nkv,1,hd] at per-seq positions;
+        kv_attend(li, q): paged attention of q [B,nq,hd] vs the cache."""
+        x = self.embed(tokens).unsqueeze(1)  # [B,1,d]
+        for li, blk in enumerate(self.blocks):
+            h = blk.attn_norm(x)
+            q, k, v = blk.project_qkv(h)  # [B,h,1,hd]
+            q = OF.rope(q, self.rope_cos, self.rope_sin, positions=positions)
+            k = OF.rope(k, self.rope_cos, self.rope_sin, positions=positions)
+            kv_append(li, k, v)
+            o = kv_attend(li, q[:, :, 0])  # [B,nq,hd]
+            x = x + blk.o_proj(o.reshape(o.shape[0], 1, -1))
+            x = x + blk.ffn(blk.ffn_norm(x))
+        x = self.norm(x[:, -1:])
+        return self.lm_head(x)[:, 0].float()

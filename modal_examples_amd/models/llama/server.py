@@ -1,0 +1,205 @@
+"""OpenAI-compatible serving layer over LlamaEngine.
+
+Mirrors the API surface the reference's serving examples expose and probe
+(vllm_inference.py:139-213: /v1/chat/completions + /v1/models + health URL;
+SSE streaming client at :263-345).  Tokenization is synthetic (hash-based) —
+the benchmark contract is random-init weights, so token text is `t<id>`.
+"""
+from __future__ import annotations
+
+import asyncio
+import hashlib
+import json
+import threading
+import time
+import uuid
+from typing import List, Optional
+
+from .engine import LlamaEngine
+
+
+class SyntheticTokenizer:
+    """Deterministic text↔ids without checkpoint files."""
+
+    def __init__(self, vocab_size: int):
+        self.vocab_size = vocab_size
+        self.bos, self.eos = 1, 2
+
+    def encode(self, text: str) -> List[int]:
+        ids = [self.bos]
+        for w in text.split():
+            h = int.from_bytes(hashlib.md5(w.encode()).digest()[:4], "little")
+            ids.append(10 + h % (self.vocab_size - 10))
+        return ids
+
+    def decode(self, ids: List[int]) -> str:
+        return " ".join(f"t{i}" for i in ids if i > 2)
+
+
+class LLMServer:
+    """Engine + background scheduler thread + request futures."""
+
+    def __init__(self, engine: LlamaEngine, model_name: str = "llama-3-8b"):
+        self.engine = engine
+        self.model_name = model_name
+        self.tok = SyntheticTokenizer(engine.cfg.vocab_size)
+        self._lock = threading.Lock()
+        self._wake = threading.Event()
+        self._events: dict = {}
+        self._stop = False
+        self._t = threading.Thread(target=self._loop, daemon=True)
+        self._t.start()
+
+    def _loop(self):
+        while not self._stop:
+            with self._lock:
+                has = self.engine.has_work
+            if not has:
+                self._wake.wait(timeout=0.02)
+                self._wake.clear()
+                continue
+            with self._lock:
+                done = self.engine.step()
+            for r in done:
+                ev = self._events.pop(r.req_id, None)
+                if ev is not None:
+                    ev.set()
+
+    def submit(self, prompt: str, max_tokens: int = 64, temperature: float = 0.0,
+               stream_cb=None) -> int:
+        ids = self.tok.encode(prompt)
+        with self._lock:
+            rid = self.engine.add_request(ids, max_tokens, temperature,
+                                          stream_cb=stream_cb)
+            self._events[rid] = threading.Event()
+        self._wake.set()
+        return rid
+
+    def wait(self, rid: int, timeout: float = 300.0):
+        ev = self._events.get(rid)
+        if ev is not None:
+            ev.wait(timeout)
+        r = self.engine.finished.get(rid)
+        if r is None:
+            raise TimeoutError(f"request {rid} did not finish")
+        return r
+
+    def generate(self, prompt: str, max_tokens: int = 64,
+                 temperature: float = 0.0) -> str:
+        rid = self.submit(prompt, max_tokens, temperature)
+        r = self.wait(rid)
+        return self.tok.decode(r.out_tokens)
+
+    def shutdown(self):
+        self._stop = True
+        self._wake.set()
+
+
+def create_openai_app(server: LLMServer):
+    """FastAPI app with the OpenAI-compatible routes the reference clients use."""
+    from fastapi import FastAPI
+    from fastapi.responses import JSONResponse, StreamingResponse
+
+    app = FastAPI(title="modal_examples_amd LLM server")
+
+    @app.get("/health")
+    async def health():
+        return {"status": "ok"}
+
+    @app.get("/v1/models")
+    async def models():
+        return {"object": "list",
+                "data": [{"id": server.model_name, "object": "model"}]}
+
+    async def _run(prompt: str, max_tokens: int, temperature: float,
+                   stream: bool, chat: bool):
+        created = int(time.time())
+        rid_str = f"cmpl-{uuid.uuid4().hex[:12]}"
+        if not stream:
+            loop = asyncio.get_running_loop()
+            text = await loop.run_in_executor(
+                None, server.generate, prompt, max_tokens, temperature)
+            usage = {"prompt_tokens": len(server.tok.encode(prompt)),
+                     "completion_tokens": len(text.split()),
+                     "total_tokens": len(server.tok.encode(prompt)) + len(text.split())}
+            if chat:
+                return JSONResponse({
+                    "id": rid_str, "object": "chat.completion", "created": created,
+                    "model": server.model_name,
+                    "choices": [{"index": 0, "message": {"role": "assistant", "content": text},
+                                 "finish_reason": "stop"}],
+                    "usage": usage,
+                })
+            return JSONResponse({
+                "id": rid_str, "object": "text_completion", "created": created,
+                "model": server.model_name,
+                "choices": [{"index": 0, "text": text, "finish_reason": "stop"}],
+                "usage": usage,
+            })
+
+        loop = asyncio.get_running_loop()
+        q: asyncio.Queue = asyncio.Queue()
+
+        def cb(tok_id):
+            loop.call_soon_threadsafe(q.put_nowait, tok_id)
+
+        rid = server.submit(prompt, max_tokens, temperature, stream_cb=cb)
+
+        async def gen():
+            sent = 0
+            while True:
+                try:
+                    tok_id = await asyncio.wait_for(q.get(), timeout=120)
+                except asyncio.TimeoutError:
+                    break
+                sent += 1
+                piece = f"t{tok_id} "
+                if chat:
+                    payload = {"id": rid_str, "object": "chat.completion.chunk",
+                               "created": created, "model": server.model_name,
+                               "choices": [{"index": 0, "delta": {"content": piece},
+                                            "finish_reason": None}]}
+                else:
+                    payload = {"id": rid_str, "object": "text_completion",
+                               "created": created, "model": server.model_name,
+                               "choices": [{"index": 0, "text": piece,
+                                            "finish_reason": None}]}
+                yield f"data: {json.dumps(payload)}\n\n"
+                if tok_id == server.engine.eos_id or sent >= max_tokens:
+                    break
+            yield "data: [DONE]\n\n"
+
+        return StreamingResponse(gen(), media_type="text/event-stream")
+
+    @app.post("/v1/completions")
+    async def completions(body: dict):
+        return await _run(body.get("prompt", ""), int(body.get("max_tokens", 64)),
+                          float(body.get("temperature", 0.0)),
+                          bool(body.get("stream", False)), chat=False)
+
+    @app.post("/v1/chat/completions")
+    async def chat_completions(body: dict):
+        msgs = body.get("messages", [])
+        prompt = "\n".join(m.get("content", "") for m in msgs)
+        return await _run(prompt, int(body.get("max_tokens", 64)),
+                          float(body.get("temperature", 0.0)),
+                          bool(body.get("stream", False)), chat=True)
+
+    return app
+
+
+def serve_openai(server: LLMServer, port: int = 8000, block: bool = True):
+    import uvicorn
+
+    app = create_openai_app(server)
+    cfg = uvicorn.Config(app, host="127.0.0.1", port=port, log_level="warning")
+    srv = uvicorn.Server(cfg)
+    if block:
+        srv.run()
+        return srv
+    t = threading.Thread(target=srv.run, daemon=True)
+    t.start()
+    deadline = time.monotonic() + 20
+    while not srv.started and time.monotonic() < deadline:
+        time.sleep(0.05)
+    return srv

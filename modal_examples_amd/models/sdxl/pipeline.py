@@ -46,6 +46,9 @@ class SDXLPipeline:
         self.dtype = dtype
         self.latent = latent_size
         self.use_graph = use_graph and self.device.type == "cuda"
+        if self.device.type == "cuda":
+            # MIOpen: benchmark-pick conv algos (with workspace) once per shape
+            torch.backends.cudnn.benchmark = True
         torch.manual_seed(seed)
         with torch.device(self.device):
             self.unet = UNetXL(self.cfg).to(self.device, dtype)

@@ -31,7 +31,7 @@ void silu_mul_bf16(const void*, const void*, void*, long long, hipStream_t);
 void geglu_bf16(const void*, const void*, void*, long long, hipStream_t);
 void add_bf16(const void*, const void*, void*, long long, hipStream_t);
 void rope_bf16(void*, const float*, const float*, long long, int, int,
-               const int*, hipStream_t);
+               int, const int*, hipStream_t);
 void adamw_step(void*, const void*, float*, float*, long long, float, float,
                 float, float, float, int, int, hipStream_t);
 void gumbel_sample(const float*, int*, int, int, float, unsigned long long,
@@ -177,10 +177,14 @@ void rope_(torch::Tensor qk, torch::Tensor cosv, torch::Tensor sinv,
            c10::optional<torch::Tensor> positions) {
   check_bf16(qk, "qk");
   TORCH_CHECK(qk.dim() == 4, "qk must be [B,H,S,D]");
-  long long BH = (long long)qk.size(0) * qk.size(1);
-  int S = qk.size(2), D = qk.size(3);
-  rope_bf16(qk.data_ptr(), cosv.data_ptr<float>(), sinv.data_ptr<float>(), BH,
-            S, D, positions.has_value() ? positions->data_ptr<int>() : nullptr,
+  long long B = qk.size(0);
+  int H = qk.size(1), S = qk.size(2), D = qk.size(3);
+  if (positions.has_value())
+    TORCH_CHECK(positions->numel() == B * S || positions->numel() == S,
+                "positions must be [B*S] (or [S] with B==1)");
+  rope_bf16(qk.data_ptr(), cosv.data_ptr<float>(), sinv.data_ptr<float>(), B,
+            H, S, D,
+            positions.has_value() ? positions->data_ptr<int>() : nullptr,
             cur_stream());
 }
 

@@ -168,15 +168,16 @@ extern "C" void add_bf16(const void* a, const void* b, void* y, long long n,
 
 __global__ __launch_bounds__(EW_BLOCK) void rope_kernel(
     short* __restrict__ Qk, const float* __restrict__ Cos,
-    const float* __restrict__ Sin, long long BH, int S, int D,
-    const int* __restrict__ positions /*nullable [S]*/) {
-  // one wave handles one (bh, s) row; lanes cover D/2 rotation pairs
-  long long rows = BH * S;
+    const float* __restrict__ Sin, long long B, int H, int S, int D,
+    const int* __restrict__ positions /*nullable [B*S], per-seq offsets*/) {
+  // one wave handles one (b, h, s) row; lanes cover D/2 rotation pairs
+  long long rows = B * H * S;
   int l = threadIdx.x % WAVE;
   for (long long row = blockIdx.x * (EW_BLOCK / WAVE) + threadIdx.x / WAVE;
        row < rows; row += (long long)gridDim.x * (EW_BLOCK / WAVE)) {
     int s = (int)(row % S);
-    int pos = positions != nullptr ? positions[s] : s;
+    long long b = row / ((long long)H * S);
+    int pos = positions != nullptr ? positions[b * S + s] : s;
     short* x = Qk + row * D;
     for (int i = l * 2; i < D / 2; i += WAVE * 2) {
       float c0 = Cos[(long long)pos * (D / 2) + i];
@@ -194,13 +195,13 @@ __global__ __launch_bounds__(EW_BLOCK) void rope_kernel(
 }
 
 extern "C" void rope_bf16(void* qk, const float* cosv, const float* sinv,
-                          long long BH, int S, int D, const int* positions,
-                          hipStream_t stream) {
-  long long rows = BH * S;
+                          long long B, int H, int S, int D,
+                          const int* positions, hipStream_t stream) {
+  long long rows = B * (long long)H * S;
   int grid = (int)((rows + 3) / 4);
   if (grid > 4096) grid = 4096;
   hipLaunchKernelGGL(rope_kernel, dim3(grid), dim3(EW_BLOCK), 0, stream,
-                     (short*)qk, cosv, sinv, BH, S, D, positions);
+                     (short*)qk, cosv, sinv, B, H, S, D, positions);
 }
 
 // ---------------------------------------------------------------- graph-capturable

@@ -100,15 +100,27 @@ def add_residual(a, b):
 
 
 def rope(x, cos, sin, positions=None, inplace: bool = False):
-    """RoPE with host-precomputed tables. x [B,H,S,D] bf16."""
+    """RoPE with host-precomputed tables. x [B,H,S,D] bf16.
+    positions: None (0..S-1), [S] (shared across batch) or [B*S] (per-seq —
+    the batched-decode case where every sequence sits at its own offset)."""
     ext = _ext_for(x)
     if ext is None:
-        return ref.rope_ref(x, cos, sin, positions)
+        if positions is not None and positions.numel() == x.shape[0] * x.shape[2] \
+                and x.shape[0] > 1:
+            B, H, S, D = x.shape
+            pos = positions.view(B, S).long()
+            outs = [ref.rope_ref(x[b:b + 1], cos, sin, pos[b]) for b in range(B)]
+            return torch.cat(outs, 0)
+        return ref.rope_ref(x, cos, sin, positions.long() if positions is not None else None)
     y = x.contiguous() if not (inplace and x.is_contiguous()) else x
     if y.data_ptr() == x.data_ptr() and not inplace:
         y = x.clone()
-    ext.rope_(y, cos.contiguous(), sin.contiguous(),
-              positions.int() if positions is not None else None)
+    pos = positions
+    if pos is not None:
+        pos = pos.int()
+        if pos.numel() == x.shape[2] and x.shape[0] > 1:
+            pos = pos.repeat(x.shape[0])
+    ext.rope_(y, cos.contiguous(), sin.contiguous(), pos)
     return y
 
 

@@ -1,0 +1,163 @@
+"""Llama engine: prefill/decode equivalence, continuous batching, paged cache,
+OpenAI server — CPU tier with the small config (reference op paths)."""
+import threading
+import time
+
+import pytest
+import torch
+
+from modal_examples_amd.models.llama.engine import BLOCK, LlamaEngine
+from modal_examples_amd.models.llama.model import LlamaConfig, LlamaModel
+from modal_examples_amd.models.llama.server import (
+    LLMServer,
+    SyntheticTokenizer,
+    create_openai_app,
+)
+
+
+def make_engine(**kw):
+    torch.manual_seed(0)
+    return LlamaEngine(LlamaConfig.small(), device="cpu", dtype=torch.float32,
+                       use_graph=False, kv_blocks=128, **kw)
+
+
+def test_decode_matches_full_prefill():
+    """Greedy decode via the paged cache must equal rerunning full prefill."""
+    eng = make_engine()
+    prompt = [1, 17, 99, 250, 31]
+    rid = eng.add_request(prompt, max_new_tokens=6, temperature=0.0)
+    eng.run_until_done()
+    out = eng.finished[rid].out_tokens
+
+    # reference: iteratively re-prefill the growing sequence (no cache)
+    seq = list(prompt)
+    ref_out = []
+    model = eng.model
+    for _ in range(6):
+        logits = model.prefill(torch.tensor([seq]))
+        tok = int(logits.argmax(-1)[0])
+        ref_out.append(tok)
+        seq.append(tok)
+        if tok == eng.eos_id:
+            break
+    assert out == ref_out, f"{out} vs {ref_out}"
+
+
+def test_continuous_batching_multiple_requests():
+    eng = make_engine()
+    rids = [eng.add_request([1, 10 + i, 20 + i], max_new_tokens=5) for i in range(4)]
+    eng.run_until_done()
+    assert all(r in eng.finished for r in rids)
+    outs = [eng.finished[r].out_tokens for r in rids]
+    assert all(1 <= len(o) <= 5 for o in outs)
+    # isolation: single-request run gives identical output for request 0
+    eng2 = make_engine()
+    r0 = eng2.add_request([1, 10, 20], max_new_tokens=5)
+    eng2.run_until_done()
+    assert eng2.finished[r0].out_tokens == outs[0]
+
+
+def test_blocks_freed_after_completion():
+    eng = make_engine()
+    free0 = len(eng.free_blocks)
+    rid = eng.add_request(list(range(1, BLOCK * 2 + 3)), max_new_tokens=4)
+    eng.run_until_done()
+    assert rid in eng.finished
+    assert len(eng.free_blocks) == free0
+
+
+def test_admission_blocks_when_cache_full():
+    eng = make_engine()
+    eng.free_blocks = eng.free_blocks[:2]  # starve the pool
+    long_prompt = list(range(1, BLOCK * 4))
+    eng.add_request(long_prompt, max_new_tokens=3)
+    eng.step()
+    assert len(eng.waiting) == 1  # could not admit; still queued, no crash
+
+
+def test_temperature_sampling_differs_by_seed():
+    eng = make_engine()
+    r1 = eng.add_request([1, 5, 9], max_new_tokens=8, temperature=1.5)
+    eng.run_until_done()
+    out1 = eng.finished[r1].out_tokens
+    assert len(out1) >= 1
+
+
+def test_synthetic_tokenizer_roundtrip():
+    tok = SyntheticTokenizer(1024)
+    ids = tok.encode("hello world hello")
+    assert ids[0] == tok.bos and len(ids) == 4
+    assert ids[1] == ids[3]  # same word, same id
+    assert tok.decode([1, 50, 60]) == "t50 t60"
+
+
+def test_llm_server_generate():
+    eng = make_engine()
+    srv = LLMServer(eng, "test-model")
+    try:
+        text = srv.generate("hello world", max_tokens=4)
+        assert isinstance(text, str) and len(text.split()) >= 1
+    finally:
+        srv.shutdown()
+
+
+def test_openai_routes():
+    import httpx
+
+    eng = make_engine()
+    srv = LLMServer(eng, "test-model")
+    app = create_openai_app(srv)
+    try:
+        transport = httpx.ASGITransport(app=app)
+        import asyncio
+
+        async def go():
+            async with httpx.AsyncClient(transport=transport, base_url="http://t") as c:
+                h = await c.get("/health")
+                assert h.json()["status"] == "ok"
+                m = await c.get("/v1/models")
+                assert m.json()["data"][0]["id"] == "test-model"
+                r = await c.post("/v1/chat/completions", json={
+                    "model": "test-model",
+                    "messages": [{"role": "user", "content": "hi there"}],
+                    "max_tokens": 4,
+                })
+                body = r.json()
+                assert body["object"] == "chat.completion"
+                assert body["choices"][0]["message"]["role"] == "assistant"
+                r2 = await c.post("/v1/completions", json={
+                    "prompt": "abc def", "max_tokens": 3})
+                assert r2.json()["object"] == "text_completion"
+
+        asyncio.run(go())
+    finally:
+        srv.shutdown()
+
+
+def test_openai_streaming():
+    import asyncio
+
+    import httpx
+
+    eng = make_engine()
+    srv = LLMServer(eng, "test-model")
+    try:
+        app = create_openai_app(srv)
+        transport = httpx.ASGITransport(app=app)
+
+        async def go():
+            async with httpx.AsyncClient(transport=transport, base_url="http://t") as c:
+                async with c.stream("POST", "/v1/chat/completions", json={
+                    "messages": [{"role": "user", "content": "stream me"}],
+                    "max_tokens": 4, "stream": True,
+                }) as r:
+                    chunks = []
+                    async for line in r.aiter_lines():
+                        if line.startswith("data: "):
+                            chunks.append(line[6:])
+                    assert chunks[-1] == "[DONE]"
+                    assert len(chunks) >= 2
+
+        asyncio.run(go())
+    finally:
+        srv.shutdown()
