@@ -1,0 +1,108 @@
+"""Whisper transcription pipeline: log-mel frontend + batched greedy decode.
+
+Mirrors the reference's batched_whisper flow (batched_whisper.py:127-138:
+``@modal.batched(max_batch_size=64)`` feeding ``pipeline(audio, batch_size)``)
+— here the batch feeds the gfx950 encoder attention in one launch set and the
+decoder loop runs all sequences in lockstep over contiguous KV caches.
+"""
+from __future__ import annotations
+
+import math
+from typing import List, Optional
+
+import torch
+
+from .model import WhisperConfig, WhisperModel
+
+SAMPLE_RATE = 16000
+HOP = 160
+N_FFT = 400
+
+
+def log_mel_spectrogram(audio: torch.Tensor, n_mels: int, n_frames: int) -> torch.Tensor:
+    """audio [B, samples] f32 → log-mel [B, n_mels, n_frames*2] (Whisper DSP:
+    400-pt STFT, hop 160, mel filterbank, log10, max-normalized)."""
+    device = audio.device
+    window = torch.hann_window(N_FFT, device=device)
+    stft = torch.stft(audio, N_FFT, HOP, window=window, return_complex=True)
+    mag = stft.abs() ** 2  # [B, n_fft/2+1, T]
+    # triangular mel filterbank
+    n_freqs = N_FFT // 2 + 1
+    mel_min, mel_max = 0.0, 2595.0 * math.log10(1 + (SAMPLE_RATE / 2) / 700.0)
+    mel_pts = torch.linspace(mel_min, mel_max, n_mels + 2, device=device)
+    hz_pts = 700.0 * (10 ** (mel_pts / 2595.0) - 1)
+    bins = (hz_pts / (SAMPLE_RATE / 2) * (n_freqs - 1)).long()
+    fb = torch.zeros(n_mels, n_freqs, device=device)
+    for m in range(n_mels):
+        lo, c, hi = bins[m], bins[m + 1], bins[m + 2]
+        if c > lo:
+            fb[m, lo:c] = (torch.arange(lo, c, device=device) - lo) / max(1, (c - lo))
+        if hi > c:
+            fb[m, c:hi] = (hi - torch.arange(c, hi, device=device)) / max(1, (hi - c))
+    mel = fb @ mag
+    logmel = torch.clamp(mel, min=1e-10).log10()
+    logmel = torch.maximum(logmel, logmel.amax(dim=(1, 2), keepdim=True) - 8.0)
+    logmel = (logmel + 4.0) / 4.0
+    T = n_frames * 2
+    if logmel.shape[-1] < T:
+        logmel = torch.nn.functional.pad(logmel, (0, T - logmel.shape[-1]))
+    return logmel[..., :T]
+
+
+class WhisperPipeline:
+    def __init__(self, cfg: Optional[WhisperConfig] = None, device: str = "cuda",
+                 dtype=torch.bfloat16, seed: int = 0):
+        self.cfg = cfg or WhisperConfig.large_v3()
+        self.device = torch.device(device)
+        self.dtype = dtype
+        torch.manual_seed(seed)
+        with torch.device(self.device):
+            self.model = WhisperModel(self.cfg).to(self.device, dtype)
+        self.model.eval()
+        self.sot, self.eot = 1, 2  # synthetic special tokens
+
+    @torch.no_grad()
+    def transcribe(self, audio_batch: List[torch.Tensor], max_tokens: int = 32
+                   ) -> List[List[int]]:
+        """Batch of mono 16 kHz waveforms → token id sequences (greedy)."""
+        B = len(audio_batch)
+        cfg = self.cfg
+        maxlen = cfg.n_audio_ctx * 2 * HOP
+        padded = torch.zeros(B, maxlen, device=self.device)
+        for i, a in enumerate(audio_batch):
+            a = a.to(self.device).float()[:maxlen]
+            padded[i, : a.numel()] = a
+        mel = log_mel_spectrogram(padded, cfg.n_mels, cfg.n_audio_ctx).to(self.dtype)
+        audio = self.model.encode(mel)
+
+        H, D = cfg.n_head, cfg.n_state // cfg.n_head
+        caches = [
+            {
+                "k": torch.zeros(B, H, cfg.n_text_ctx, D, device=self.device, dtype=self.dtype),
+                "v": torch.zeros(B, H, cfg.n_text_ctx, D, device=self.device, dtype=self.dtype),
+            }
+            for _ in range(cfg.n_text_layer)
+        ]
+        tokens = torch.full((B, 1), self.sot, device=self.device, dtype=torch.long)
+        logits = self.model.decode_prefill(tokens, audio, caches)
+        outs = [[] for _ in range(B)]
+        alive = torch.ones(B, dtype=torch.bool)
+        cur = logits.argmax(-1)
+        for i in range(B):
+            outs[i].append(int(cur[i]))
+        for pos in range(1, min(max_tokens, cfg.n_text_ctx - 1)):
+            logits = self.model.decode_step(cur, pos, caches, cfg.n_audio_ctx)
+            cur = logits.argmax(-1)
+            for i in range(B):
+                if alive[i]:
+                    t = int(cur[i])
+                    outs[i].append(t)
+                    if t == self.eot:
+                        alive[i] = False
+            if not alive.any():
+                break
+        return outs
+
+    def transcribe_text(self, audio_batch, max_tokens: int = 32) -> List[str]:
+        return [" ".join(f"t{t}" for t in seq if t > 2)
+                for seq in self.transcribe(audio_batch, max_tokens)]
