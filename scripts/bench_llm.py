@@ -1,0 +1,73 @@
+#!/usr/bin/env python3
+"""LLM serving throughput bench (config 4): Llama-3-8B continuous batching.
+
+Measures offline throughput the way the reference quotes vLLM's
+(vllm_throughput.py:27-40: ~30k input tok/s + ~2k output tok/s per H100):
+submit N requests (prompt P tokens, generate G tokens each), run the engine
+to completion, report input/output tok/s.
+"""
+import argparse
+import json
+import os
+import time
+
+os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--requests", type=int, default=64)
+    ap.add_argument("--prompt-len", type=int, default=128)
+    ap.add_argument("--gen-len", type=int, default=128)
+    ap.add_argument("--layers", type=int, default=0, help="0 = full 32")
+    ap.add_argument("--no-graph", action="store_true")
+    ap.add_argument("--small", action="store_true")
+    args = ap.parse_args()
+
+    import torch
+
+    from modal_examples_amd.models.llama.engine import LlamaEngine
+    from modal_examples_amd.models.llama.model import LlamaConfig
+
+    cfg = LlamaConfig.small() if args.small else LlamaConfig.llama3_8b()
+    if args.layers:
+        cfg.n_layers = args.layers
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+    dtype = torch.bfloat16 if device == "cuda" else torch.float32
+    t0 = time.perf_counter()
+    eng = LlamaEngine(cfg, device=device, dtype=dtype,
+                      use_graph=not args.no_graph and device == "cuda",
+                      max_batch=args.requests)
+    init_s = time.perf_counter() - t0
+
+    prompt = [(i % 1000) + 10 for i in range(args.prompt_len)]
+    for _ in range(args.requests):
+        eng.add_request(list(prompt), max_new_tokens=args.gen_len,
+                        temperature=0.0)
+    if device == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    eng.run_until_done()
+    if device == "cuda":
+        torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+
+    out_toks = sum(len(r.out_tokens) for r in eng.finished.values())
+    in_toks = args.requests * args.prompt_len
+    print(json.dumps({
+        "metric": "llama decode throughput",
+        "model": f"llama3-8b-class ({cfg.n_layers}L)",
+        "requests": args.requests,
+        "prompt_len": args.prompt_len,
+        "gen_len": args.gen_len,
+        "elapsed_s": round(dt, 3),
+        "init_s": round(init_s, 2),
+        "input_tok_per_s": round(in_toks / dt, 1),
+        "output_tok_per_s": round(out_toks / dt, 1),
+        "kv_blocks": eng.num_blocks,
+        "hipgraph": eng.use_graph,
+    }), flush=True)
+
+
+if __name__ == "__main__":
+    main()
