@@ -9,8 +9,10 @@ to completion, report input/output tok/s.
 import argparse
 import json
 import os
+import sys
 import time
 
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
 
 
