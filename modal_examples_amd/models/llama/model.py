@@ -85,20 +85,28 @@ class LlamaModel(nn.Module):
         self.blocks = nn.ModuleList([LlamaBlock(cfg) for _ in range(cfg.n_layers)])
         self.norm = RMSNormK(cfg.dim, cfg.norm_eps)
         self.lm_head = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
+        # NOT buffers: module .to(bf16) must not downcast the f32 trig tables
         cos, sin = OF.rope_tables(cfg.max_seq, cfg.head_dim, cfg.rope_base)
-        self.register_buffer("rope_cos", cos, persistent=False)
-        self.register_buffer("rope_sin", sin, persistent=False)
+        self._rope_cpu = (cos, sin)
+        self._rope_cache = {}
+
+    def _rope_tables(self, device) -> tuple:
+        key = str(device)
+        if key not in self._rope_cache:
+            self._rope_cache[key] = tuple(t.to(device) for t in self._rope_cpu)
+        return self._rope_cache[key]
 
     @torch.no_grad()
     def prefill(self, tokens: torch.Tensor, kv_writer=None):
         """tokens [B,S] → logits [B, vocab] (last position only).
         kv_writer(layer_idx, k, v): callback storing [B,nkv,S,hd] into cache."""
         x = self.embed(tokens)
+        rc, rs = self._rope_tables(x.device)
         for li, blk in enumerate(self.blocks):
             h = blk.attn_norm(x)
             q, k, v = blk.project_qkv(h)
-            q = OF.rope(q, self.rope_cos, self.rope_sin)
-            k = OF.rope(k, self.rope_cos, self.rope_sin)
+            q = OF.rope(q, rc, rs)
+            k = OF.rope(k, rc, rs)
             if kv_writer is not None:
                 kv_writer(li, k, v)
             o = OF.attention(q, k, v, causal=True)
@@ -114,11 +122,12 @@ class LlamaModel(nn.Module):
         kv_append(li, k, v): store [B,nkv,1,hd] at per-seq positions;
         kv_attend(li, q): paged attention of q [B,nq,hd] vs the cache."""
         x = self.embed(tokens).unsqueeze(1)  # [B,1,d]
+        rc, rs = self._rope_tables(x.device)
         for li, blk in enumerate(self.blocks):
             h = blk.attn_norm(x)
             q, k, v = blk.project_qkv(h)  # [B,h,1,hd]
-            q = OF.rope(q, self.rope_cos, self.rope_sin, positions=positions)
-            k = OF.rope(k, self.rope_cos, self.rope_sin, positions=positions)
+            q = OF.rope(q, rc, rs, positions=positions)
+            k = OF.rope(k, rc, rs, positions=positions)
             kv_append(li, k, v)
             o = kv_attend(li, q[:, :, 0])  # [B,nq,hd]
             x = x + blk.o_proj(o.reshape(o.shape[0], 1, -1))

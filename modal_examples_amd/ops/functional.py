@@ -120,7 +120,7 @@ def rope(x, cos, sin, positions=None, inplace: bool = False):
         pos = pos.int()
         if pos.numel() == x.shape[2] and x.shape[0] > 1:
             pos = pos.repeat(x.shape[0])
-    ext.rope_(y, cos.contiguous(), sin.contiguous(), pos)
+    ext.rope_(y, cos.float().contiguous(), sin.float().contiguous(), pos)
     return y
 
 
