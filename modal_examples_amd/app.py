@@ -368,6 +368,9 @@ class Function:
             if self._pool is None:
                 rt = _Runtime.get()
                 gpu_count = parse_gpu(self.opts.gpu)
+                # degrade gracefully on smaller pools (CPU CI, 1-GPU boxes):
+                # a request larger than the node clamps to what exists
+                gpu_count = min(gpu_count, rt.device_pool.n)
                 raw = self.raw
                 env = {}
                 for s in self.opts.secrets or []:
@@ -604,7 +607,7 @@ class _Obj:
             if self._pool is None:
                 cls = self._cls
                 rt = _Runtime.get()
-                gpu_count = parse_gpu(cls.opts.gpu)
+                gpu_count = min(parse_gpu(cls.opts.gpu), rt.device_pool.n)
                 env = {}
                 for s in cls.opts.secrets or []:
                     env.update(getattr(s, "env", {}))

@@ -1,0 +1,160 @@
+"""CLI: ``run`` / ``serve`` / ``deploy`` / ``shell`` over example files.
+
+Mirrors the reference's invocation surface (hello_world.py:73 ``modal run``,
+text_to_image.py:157-163 auto-derived entrypoint flags incl. ``--flag/--no-flag``
+booleans, flux.py:170-176).  Usage::
+
+    python -m modal_examples_amd run examples/01_getting_started/hello_world.py
+    python -m modal_examples_amd run file.py::entry --arg 3
+    python -m modal_examples_amd serve examples/07_web/basic_web.py
+    python -m modal_examples_amd deploy examples/05_scheduling/schedule_simple.py
+"""
+from __future__ import annotations
+
+import argparse
+import importlib.util
+import inspect
+import sys
+import time
+from pathlib import Path
+
+
+def load_module(path: str):
+    p = Path(path)
+    spec = importlib.util.spec_from_file_location(p.stem.replace("-", "_"), p)
+    mod = importlib.util.module_from_spec(spec)
+    sys.modules[spec.name] = mod
+    spec.loader.exec_module(mod)
+    # file-loaded modules are not importable by worker processes — ship their
+    # functions (and module globals they close over) by value instead
+    import cloudpickle
+
+    cloudpickle.register_pickle_by_value(mod)
+    return mod
+
+
+def find_app(mod):
+    from .app import App
+
+    for v in vars(mod).values():
+        if isinstance(v, App):
+            return v
+    raise SystemExit(f"no modal App found in {mod.__name__}")
+
+
+def _add_args_from_signature(parser: argparse.ArgumentParser, fn):
+    for name, p in inspect.signature(fn).parameters.items():
+        flag = "--" + name.replace("_", "-")
+        ann = p.annotation if p.annotation is not inspect.Parameter.empty else str
+        default = None if p.default is inspect.Parameter.empty else p.default
+        required = p.default is inspect.Parameter.empty
+        if ann is bool or isinstance(default, bool):
+            group = parser.add_mutually_exclusive_group(required=False)
+            group.add_argument(flag, dest=name, action="store_true")
+            group.add_argument("--no-" + name.replace("_", "-"), dest=name,
+                               action="store_false")
+            parser.set_defaults(**{name: bool(default)})
+        else:
+            typ = ann if ann in (int, float, str) else str
+            parser.add_argument(flag, dest=name, type=typ, default=default,
+                                required=required)
+
+
+def cmd_run(target: str, extra_args):
+    path, _, entry = target.partition("::")
+    mod = load_module(path)
+    app = find_app(mod)
+    if not app.entrypoints:
+        raise SystemExit(f"app {app.name!r} has no @app.local_entrypoint")
+    name = entry or next(iter(app.entrypoints))
+    fn = app.entrypoints[name]
+    parser = argparse.ArgumentParser(prog=f"run {path}::{name}")
+    _add_args_from_signature(parser, fn)
+    ns = parser.parse_args(extra_args)
+    from .app import _Runtime
+
+    try:
+        fn(**vars(ns))
+    finally:
+        _Runtime.reset()
+
+
+def cmd_serve(target: str, extra_args):
+    parser = argparse.ArgumentParser(prog="serve")
+    parser.add_argument("--port", type=int, default=8787)
+    parser.add_argument("--timeout", type=float, default=0,
+                        help="exit after N seconds (MODAL_SERVE_TIMEOUT analog)")
+    ns = parser.parse_args(extra_args)
+    import os
+
+    timeout = ns.timeout or float(os.environ.get("MODAL_SERVE_TIMEOUT", 0))
+    mod = load_module(target)
+    app = find_app(mod)
+    from .web.ingress import serve, stop_serving
+
+    url = serve(app, port=ns.port, block=False)
+    print(f"serving {app.name} at {url}")
+    for f in list(app.web_endpoints.values()):
+        print(f"  → {url}/{getattr(f.raw, '_modal_flags', {}).get('label') or f.name}")
+    try:
+        if timeout:
+            time.sleep(timeout)
+        else:
+            while True:
+                time.sleep(3600)
+    except KeyboardInterrupt:
+        pass
+    finally:
+        stop_serving()
+        from .app import _Runtime
+
+        _Runtime.reset()
+
+
+def cmd_deploy(target: str, extra_args):
+    mod = load_module(target)
+    app = find_app(mod)
+    app.deploy()
+    from . import config
+
+    rec = config.state_dir() / "deployments.txt"
+    with open(rec, "a") as f:
+        f.write(f"{time.strftime('%F %T')} {app.name} {target}\n")
+    scheduled = [n for n, fn in app.functions.items() if fn.opts.schedule]
+    print(f"deployed {app.name} ({len(app.functions)} functions, "
+          f"{len(app.classes)} classes)")
+    if scheduled:
+        print(f"schedules active for: {', '.join(scheduled)} — keeping process alive")
+        try:
+            while True:
+                time.sleep(3600)
+        except KeyboardInterrupt:
+            pass
+
+
+def cmd_shell(target: str, extra_args):
+    import code
+
+    mod = load_module(target) if target else None
+    banner = "modal_examples_amd shell"
+    ns = dict(vars(mod)) if mod else {}
+    code.interact(banner=banner, local=ns)
+
+
+def main(argv=None):
+    argv = list(sys.argv[1:] if argv is None else argv)
+    if not argv or argv[0] in ("-h", "--help"):
+        print(__doc__)
+        return 0
+    cmd, *rest = argv
+    if cmd == "run":
+        cmd_run(rest[0], rest[1:])
+    elif cmd == "serve":
+        cmd_serve(rest[0], rest[1:])
+    elif cmd == "deploy":
+        cmd_deploy(rest[0], rest[1:])
+    elif cmd == "shell":
+        cmd_shell(rest[0] if rest else None, rest[1:])
+    else:
+        raise SystemExit(f"unknown command {cmd!r}; use run/serve/deploy/shell")
+    return 0
