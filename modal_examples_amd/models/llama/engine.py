@@ -2,14 +2,17 @@
 
 The MI355X-native replacement for the vLLM/SGLang serving engines the
 reference wraps (vllm_inference.py:139-213, sglang_snapshot.py:176-218):
+
   - paged KV cache (16-token blocks, [nblocks, Hkv, block, D] layout feeding
-    the K6 decode kernel directly),
-  - continuous batching: new requests prefill (K7 flash kernel) and join the
-    running decode batch between steps; finished sequences free their blocks,
-  - steady-state decode captured per batch-size bucket into hipGraphs
-    (block tables / positions / lengths are device-resident static buffers —
-    the TRT-LLM-engine role),
-  - fused sampling (K8) outside the graph so seeds/temps stay dynamic.
+    the K6 decode kernel directly), block allocator, continuous batching,
+  - ONE hipGraph captured at the full slot width: at small batch the decode
+    GEMMs are weight-read-bound (16 GB of bf16 weights per step vs KBs of
+    activations), so padding inactive slots costs ~nothing on MI355X while
+    keeping every step a single graph replay,
+  - ALL per-step state (tokens, positions, block tables, lengths, temps) is
+    persistent and device-resident; the host loop does zero tensor staging —
+    one small D2H per step to read the sampled tokens,
+  - fused sampling (K8) outside the graph so seeds stay dynamic.
 
 Sized for 288 GB HBM3E: KV pool defaults to 60% of free memory after weights
 (Llama-3-8B: ~128 KiB/token → ~1.3M cached tokens).
@@ -26,7 +29,6 @@ from ...ops import functional as OF
 from .model import LlamaConfig, LlamaModel
 
 BLOCK = 16
-BUCKETS = (1, 2, 4, 8, 16, 32, 64, 128)
 
 
 @dataclass
@@ -38,6 +40,7 @@ class Request:
     out_tokens: List[int] = field(default_factory=list)
     blocks: List[int] = field(default_factory=list)
     pos: int = 0  # tokens stored in cache
+    slot: int = -1
     done: bool = False
     t_arrive: float = field(default_factory=time.monotonic)
     t_first_token: Optional[float] = None
@@ -79,7 +82,18 @@ class LlamaEngine:
         self.finished: Dict[int, Request] = {}
         self._next_id = 1
         self._step_count = 0
-        self._graphs = {}
+        self._graph = None
+        self._slots: List[Optional[Request]] = [None] * max_batch
+        # persistent device-side slot state (double as the graph's static bufs)
+        dev = self.device
+        self.toks_d = torch.zeros(max_batch, dtype=torch.long, device=dev)
+        self.pos_d = torch.zeros(max_batch, dtype=torch.int32, device=dev)
+        self.bt_d = torch.zeros(max_batch, self.max_blocks_per_seq,
+                                dtype=torch.int32, device=dev)
+        self.lens_d = torch.ones(max_batch, dtype=torch.int32, device=dev)
+        self.active_d = torch.zeros(max_batch, dtype=torch.int32, device=dev)
+        self.temps_d = torch.zeros(max_batch, dtype=torch.float32, device=dev)
+        self.logits_d = None
 
     # ------------------------------------------------ request lifecycle
 
@@ -96,35 +110,62 @@ class LlamaEngine:
             return None
         return [self.free_blocks.pop() for _ in range(n)]
 
-    def _free_seq(self, r: Request):
+    def _take_slot(self, r: Request) -> bool:
+        for i, s in enumerate(self._slots):
+            if s is None:
+                self._slots[i] = r
+                r.slot = i
+                return True
+        return False
+
+    def _release(self, r: Request):
         self.free_blocks.extend(r.blocks)
         r.blocks = []
+        if r.slot >= 0:
+            i = r.slot
+            self._slots[i] = None
+            self.pos_d[i] = 0
+            self.lens_d[i] = 1
+            self.active_d[i] = 0
+            self.bt_d[i].zero_()
+            r.slot = -1
 
     # ------------------------------------------------ prefill
 
     @torch.no_grad()
-    def _prefill(self, r: Request):
+    def _prefill(self, r: Request) -> bool:
         L = len(r.prompt)
-        nblk = (L + BLOCK) // BLOCK + 1  # prompt + headroom for decode
+        nblk = (L + BLOCK) // BLOCK + 1
         blocks = self._alloc_blocks(nblk)
         if blocks is None:
+            return False
+        if not self._take_slot(r):
+            self.free_blocks.extend(blocks)
             return False
         r.blocks = blocks
         toks = torch.tensor([r.prompt], device=self.device)
         pos = torch.arange(L, device=self.device)
-        blks = torch.tensor(r.blocks, device=self.device)[pos // BLOCK]
+        blk_t = torch.tensor(r.blocks, device=self.device)
+        blks = blk_t[pos // BLOCK]
         offs = pos % BLOCK
 
         def kv_writer(li, k, v):
-            # k/v [1, nkv, L, hd] → cache[li][blk, :, off] = [L, nkv, hd]
             self.cache_k[li][blks, :, offs] = k[0].permute(1, 0, 2)
             self.cache_v[li][blks, :, offs] = v[0].permute(1, 0, 2)
 
         logits = self.model.prefill(toks, kv_writer)
         r.pos = L
-        tok = self._sample(logits, torch.tensor([r.temperature]))
+        tok = self._sample_rows(logits, torch.tensor([r.temperature]))
         self._append_token(r, int(tok[0]))
         r.t_first_token = time.monotonic()
+        # stage slot state (once per request, not per step)
+        i = r.slot
+        self.toks_d[i] = int(r.out_tokens[-1])
+        self.pos_d[i] = r.pos
+        self.lens_d[i] = r.pos + 1
+        self.bt_d[i, : len(r.blocks)] = blk_t.int()
+        self.active_d[i] = 1
+        self.temps_d[i] = r.temperature
         return True
 
     def _append_token(self, r: Request, tok: int):
@@ -137,9 +178,9 @@ class LlamaEngine:
         if tok == self.eos_id or len(r.out_tokens) >= r.max_new_tokens:
             r.done = True
 
-    # ------------------------------------------------ decode
+    # ------------------------------------------------ sampling
 
-    def _sample(self, logits: torch.Tensor, temps: torch.Tensor) -> torch.Tensor:
+    def _sample_rows(self, logits: torch.Tensor, temps: torch.Tensor) -> torch.Tensor:
         temps = temps.to(logits.device)
         self._step_count += 1
         greedy = logits.argmax(-1).int()
@@ -149,6 +190,61 @@ class LlamaEngine:
         sampled = OF.sample(scaled, 1.0, seed=0x5EED + self._step_count)
         return torch.where(temps <= 0, greedy, sampled.to(greedy.device))
 
+    # ------------------------------------------------ decode
+
+    def _run_decode(self, limit: Optional[int] = None):
+        """The captured computation: one token for every slot (graph mode
+        always runs the full width; eager mode slices to the live prefix)."""
+        lim = self.max_batch if limit is None else limit
+        toks, pos = self.toks_d[:lim], self.pos_d[:lim]
+        bt, lens = self.bt_d[:lim], self.lens_d[:lim]
+        blks = bt.gather(1, (pos // BLOCK).long().unsqueeze(1))[:, 0].long()
+        offs = (pos % BLOCK).long()
+
+        def kv_append(li, k, v):
+            self.cache_k[li][blks, :, offs] = k[:, :, 0]
+            self.cache_v[li][blks, :, offs] = v[:, :, 0]
+
+        def kv_attend(li, q):
+            return OF.paged_decode(q, self.cache_k[li], self.cache_v[li],
+                                   bt, lens, BLOCK)
+
+        return self.model.decode_step(toks, pos, kv_append, kv_attend)
+
+    def _ensure_graph(self):
+        if self._graph is not None:
+            return
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s), torch.no_grad():
+            for _ in range(2):
+                self._run_decode()
+        torch.cuda.current_stream().wait_stream(s)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g), torch.no_grad():
+            self.logits_d = self._run_decode()
+        self._graph = g
+
+    @torch.no_grad()
+    def _decode_batch(self):
+        if self.use_graph:
+            self._ensure_graph()
+            self._graph.replay()
+            logits = self.logits_d
+            lim = self.max_batch
+        else:
+            lim = max(r.slot for r in self.running) + 1
+            logits = self._run_decode(lim)
+        new_toks = self._sample_rows(logits, self.temps_d[:lim])
+        # advance device state without host staging
+        self.toks_d[:lim].copy_(new_toks.long())
+        self.pos_d.add_(self.active_d)
+        self.lens_d.copy_(self.pos_d + 1)
+        toks_host = new_toks.cpu()  # the one D2H sync per step
+        for r in list(self.running):
+            r.pos += 1
+            self._append_token(r, int(toks_host[r.slot]))
+
     def _ensure_blocks(self, r: Request) -> bool:
         need = (r.pos + 1 + BLOCK - 1) // BLOCK
         while len(r.blocks) < need:
@@ -156,98 +252,8 @@ class LlamaEngine:
             if got is None:
                 return False
             r.blocks.extend(got)
+            self.bt_d[r.slot, len(r.blocks) - 1] = got[0]
         return True
-
-    @torch.no_grad()
-    def _decode_batch(self, batch: List[Request]):
-        B = len(batch)
-        toks = torch.tensor([r.out_tokens[-1] for r in batch], device=self.device)
-        positions = torch.tensor([r.pos for r in batch], device=self.device,
-                                 dtype=torch.int32)
-        bt = torch.zeros(B, self.max_blocks_per_seq, device=self.device,
-                         dtype=torch.int32)
-        for i, r in enumerate(batch):
-            bt[i, : len(r.blocks)] = torch.tensor(r.blocks, device=self.device,
-                                                  dtype=torch.int32)
-        lens = positions + 1  # after append
-
-        if self.use_graph:
-            logits = self._decode_graph(B, toks, positions, bt, lens)
-        else:
-            logits = self._decode_eager(toks, positions, bt, lens)
-
-        temps = torch.tensor([r.temperature for r in batch])
-        new_toks = self._sample(logits, temps).cpu()
-        for i, r in enumerate(batch):
-            r.pos += 1
-            self._append_token(r, int(new_toks[i]))
-
-    def _decode_eager(self, toks, positions, bt, lens):
-        blks = bt.gather(1, (positions // BLOCK).long().unsqueeze(1))[:, 0].long()
-        offs = (positions % BLOCK).long()
-
-        def kv_append(li, k, v):
-            self.cache_k[li][blks, :, offs] = k[:, :, 0]
-            self.cache_v[li][blks, :, offs] = v[:, :, 0]
-
-        def kv_attend(li, q):
-            return OF.paged_decode(q, self.cache_k[li], self.cache_v[li], bt,
-                                   lens, BLOCK)
-
-        return self.model.decode_step(toks, positions, kv_append, kv_attend)
-
-    def _decode_graph(self, B, toks, positions, bt, lens):
-        bucket = next(b for b in BUCKETS if b >= B)
-        st = self._graphs.get(bucket)
-        if st is None:
-            st = self._capture(bucket)
-        # stage inputs (pad rows attend block 0 / len 1)
-        st["toks"].zero_()
-        st["toks"][:B].copy_(toks)
-        st["pos"].zero_()
-        st["pos"][:B].copy_(positions)
-        st["bt"].zero_()
-        st["bt"][:B].copy_(bt)
-        st["lens"].fill_(1)
-        st["lens"][:B].copy_(lens)
-        st["graph"].replay()
-        return st["logits"][:B].clone()
-
-    def _capture(self, bucket: int):
-        dev = self.device
-        st = {
-            "toks": torch.zeros(bucket, dtype=torch.long, device=dev),
-            "pos": torch.zeros(bucket, dtype=torch.int32, device=dev),
-            "bt": torch.zeros(bucket, self.max_blocks_per_seq, dtype=torch.int32, device=dev),
-            "lens": torch.ones(bucket, dtype=torch.int32, device=dev),
-        }
-
-        def run():
-            blks = st["bt"].gather(1, (st["pos"] // BLOCK).long().unsqueeze(1))[:, 0].long()
-            offs = (st["pos"] % BLOCK).long()
-
-            def kv_append(li, k, v):
-                self.cache_k[li][blks, :, offs] = k[:, :, 0]
-                self.cache_v[li][blks, :, offs] = v[:, :, 0]
-
-            def kv_attend(li, q):
-                return OF.paged_decode(q, self.cache_k[li], self.cache_v[li],
-                                       st["bt"], st["lens"], BLOCK)
-
-            return self.model.decode_step(st["toks"], st["pos"], kv_append, kv_attend)
-
-        s = torch.cuda.Stream()
-        s.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(s), torch.no_grad():
-            for _ in range(2):
-                run()
-        torch.cuda.current_stream().wait_stream(s)
-        g = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g), torch.no_grad():
-            st["logits"] = run()
-        st["graph"] = g
-        self._graphs[bucket] = st
-        return st
 
     # ------------------------------------------------ scheduler
 
@@ -257,7 +263,7 @@ class LlamaEngine:
         while self.waiting and len(self.running) < self.max_batch:
             r = self.waiting[0]
             if not self._prefill(r):
-                break  # no KV blocks free — keep waiting
+                break  # no KV blocks / slots free — keep waiting
             self.waiting.pop(0)
             if r.done:
                 self._retire(r)
@@ -269,7 +275,7 @@ class LlamaEngine:
             for r in self.running:
                 if not self._ensure_blocks(r):
                     r.done = True  # out of memory: finish it
-            self._decode_batch(self.running)
+            self._decode_batch()
             still = []
             for r in self.running:
                 if r.done:
@@ -281,7 +287,7 @@ class LlamaEngine:
         return done_now
 
     def _retire(self, r: Request):
-        self._free_seq(r)
+        self._release(r)
         self.finished[r.req_id] = r
 
     def run_until_done(self, max_steps: int = 100000):

@@ -47,7 +47,12 @@ class SDXLPipeline:
         self.latent = latent_size
         self.use_graph = use_graph and self.device.type == "cuda"
         if self.device.type == "cuda":
-            # MIOpen: benchmark-pick conv algos (with workspace) once per shape
+            # MIOpen: benchmark-pick conv algos once per shape; restore the
+            # repo-cached find-db first so fresh boxes skip the minutes-long
+            # tuning pass (gpu/kernel_cache.py)
+            from ...gpu import kernel_cache
+
+            kernel_cache.restore()
             torch.backends.cudnn.benchmark = True
         torch.manual_seed(seed)
         with torch.device(self.device):
