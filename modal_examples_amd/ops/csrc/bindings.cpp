@@ -34,8 +34,8 @@ void rope_bf16(void*, const float*, const float*, long long, int, int,
                int, const int*, hipStream_t);
 void adamw_step(void*, const void*, float*, float*, long long, float, float,
                 float, float, float, int, int, hipStream_t);
-void gumbel_sample(const float*, int*, int, int, float, unsigned long long,
-                   hipStream_t);
+void gumbel_sample(const float*, unsigned long long*, int*, int, int,
+                   float, unsigned long long, hipStream_t);
 void softmax_rows(const float*, float*, int, int, hipStream_t);
 void scale_in_dev_bf16(const void*, void*, const float*, const long long*,
                        long long, hipStream_t);
@@ -205,8 +205,11 @@ torch::Tensor sample_gumbel(torch::Tensor logits, double temperature,
               "logits must be f32 GPU");
   int rows = logits.size(0), V = logits.size(1);
   auto out = torch::empty({rows}, logits.options().dtype(torch::kInt32));
-  gumbel_sample(logits.data_ptr<float>(), out.data_ptr<int>(), rows, V,
-                (float)temperature, (unsigned long long)seed, cur_stream());
+  auto keys = torch::zeros({rows}, logits.options().dtype(torch::kInt64));
+  gumbel_sample(logits.data_ptr<float>(),
+                (unsigned long long*)keys.data_ptr<int64_t>(),
+                out.data_ptr<int>(), rows, V, (float)temperature,
+                (unsigned long long)seed, cur_stream());
   return out;
 }
 

@@ -26,21 +26,25 @@ __global__ __launch_bounds__(256) void groupnorm_silu_kernel(
   const short* x = X + ((long long)n * C + (long long)g * cpg) * HW;
   short* y = Y + ((long long)n * C + (long long)g * cpg) * HW;
 
-  const long long full = (slab / 8) * 8;
+  // channel-major iteration: no per-element division, gamma/beta hoisted
+  const long long hw_full = (HW / 8) * 8;
   float s = 0.f, ss = 0.f;
-  for (long long i = (long long)threadIdx.x * 8; i + 8 <= slab; i += 256 * 8) {
-    bf16x8 v = *(const bf16x8*)&x[i];
+  for (int c = 0; c < cpg; ++c) {
+    const short* xc = x + (long long)c * HW;
+    for (long long i = (long long)threadIdx.x * 8; i + 8 <= HW; i += 256 * 8) {
+      bf16x8 v = *(const bf16x8*)&xc[i];
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      float f = bf2f(v[j]);
+      for (int j = 0; j < 8; ++j) {
+        float f = bf2f(v[j]);
+        s += f;
+        ss += f * f;
+      }
+    }
+    for (long long i = hw_full + threadIdx.x; i < HW; i += 256) {
+      float f = bf2f(xc[i]);
       s += f;
       ss += f * f;
     }
-  }
-  for (long long i = full + threadIdx.x; i < slab; i += 256) {
-    float f = bf2f(x[i]);
-    s += f;
-    ss += f * f;
   }
   // block reduce
   __shared__ float red[2][4];
@@ -58,23 +62,27 @@ __global__ __launch_bounds__(256) void groupnorm_silu_kernel(
   float var = ss / (float)slab - mean * mean;
   float rstd = rsqrtf(var + eps);
 
-  for (long long i = (long long)threadIdx.x * 8; i + 8 <= slab; i += 256 * 8) {
-    bf16x8 v = *(const bf16x8*)&x[i];
-    bf16x8 o;
+  for (int c = 0; c < cpg; ++c) {
+    const short* xc = x + (long long)c * HW;
+    short* yc = y + (long long)c * HW;
+    float gam = gamma[g * cpg + c] * rstd;
+    float bet = beta[g * cpg + c] - mean * gam;
+    for (long long i = (long long)threadIdx.x * 8; i + 8 <= HW; i += 256 * 8) {
+      bf16x8 v = *(const bf16x8*)&xc[i];
+      bf16x8 o;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      long long c = g * cpg + (i + j) / HW;
-      float f = (bf2f(v[j]) - mean) * rstd * gamma[c] + beta[c];
-      if (do_silu) f = f / (1.f + __expf(-f));
-      o[j] = f2bf(f);
+      for (int j = 0; j < 8; ++j) {
+        float f = bf2f(v[j]) * gam + bet;
+        if (do_silu) f = f / (1.f + __expf(-f));
+        o[j] = f2bf(f);
+      }
+      *(bf16x8*)&yc[i] = o;
     }
-    *(bf16x8*)&y[i] = o;
-  }
-  for (long long i = full + threadIdx.x; i < slab; i += 256) {
-    long long c = g * cpg + i / HW;
-    float f = (bf2f(x[i]) - mean) * rstd * gamma[c] + beta[c];
-    if (do_silu) f = f / (1.f + __expf(-f));
-    y[i] = f2bf(f);
+    for (long long i = hw_full + threadIdx.x; i < HW; i += 256) {
+      float f = bf2f(xc[i]) * gam + bet;
+      if (do_silu) f = f / (1.f + __expf(-f));
+      yc[i] = f2bf(f);
+    }
   }
 }
 

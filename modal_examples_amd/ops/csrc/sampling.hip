@@ -24,28 +24,28 @@ DEV_INLINE unsigned int hash3(unsigned int a, unsigned int b, unsigned int c) {
 
 __global__ __launch_bounds__(SMP_BLOCK) void gumbel_sample_kernel(
     const float* __restrict__ Logits,  // [rows, V] f32 (final-layer output)
-    int* __restrict__ Out, int rows, int V, float inv_temp,
-    unsigned long long seed) {
-  int row = blockIdx.x;
+    unsigned long long* __restrict__ Keys,  // [rows] packed argmax workspace
+    int rows, int V, int splits, float inv_temp, unsigned long long seed) {
+  int row = blockIdx.x / splits;
+  int split = blockIdx.x % splits;
   if (row >= rows) return;
   const float* lg = Logits + (long long)row * V;
+  int chunk = (V + splits - 1) / splits;
+  int lo = split * chunk, hi = min(V, lo + chunk);
   float best = -1e30f;
-  int best_i = 0;
-  for (int i = threadIdx.x; i < V; i += SMP_BLOCK) {
+  int best_i = lo;
+  for (int i = lo + threadIdx.x; i < hi; i += SMP_BLOCK) {
     float s = lg[i] * inv_temp;
     if (inv_temp != 0.f && seed != 0ull) {
       unsigned int u = hash3((unsigned int)seed, (unsigned int)(seed >> 32) ^ row, i);
       float uf = (u >> 8) * (1.f / 16777216.f) + 1e-10f;
       s += -__logf(-__logf(uf));
     }
-    if (s > best) {
+    if (s > best || (s == best && i < best_i)) {
       best = s;
       best_i = i;
     }
   }
-  // block argmax reduce via LDS
-  __shared__ float sv[SMP_BLOCK / WAVE];
-  __shared__ int si[SMP_BLOCK / WAVE];
 #pragma unroll
   for (int s_ = 1; s_ < WAVE; s_ <<= 1) {
     float ov = __shfl_xor(best, s_, WAVE);
@@ -55,6 +55,8 @@ __global__ __launch_bounds__(SMP_BLOCK) void gumbel_sample_kernel(
       best_i = oi;
     }
   }
+  __shared__ float sv[SMP_BLOCK / WAVE];
+  __shared__ int si[SMP_BLOCK / WAVE];
   int w = threadIdx.x / WAVE;
   if (threadIdx.x % WAVE == 0) {
     sv[w] = best;
@@ -67,18 +69,34 @@ __global__ __launch_bounds__(SMP_BLOCK) void gumbel_sample_kernel(
         best = sv[j];
         best_i = si[j];
       }
-    Out[row] = best_i;
+    // pack (orderable float, ~idx) so atomicMax picks max value, min index
+    unsigned int ub = __float_as_uint(best);
+    ub = (ub & 0x80000000u) ? ~ub : (ub | 0x80000000u);
+    unsigned long long key =
+        ((unsigned long long)ub << 32) | (unsigned int)(~best_i);
+    atomicMax(Keys + row, key);
   }
 }
 
-extern "C" void gumbel_sample(const float* logits, int* out, int rows, int V,
-                              float temperature, unsigned long long seed,
-                              hipStream_t stream) {
+__global__ void gumbel_unpack_kernel(const unsigned long long* __restrict__ Keys,
+                                     int* __restrict__ Out, int rows) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < rows) Out[i] = (int)(~(unsigned int)(Keys[i] & 0xFFFFFFFFull));
+}
+
+extern "C" void gumbel_sample(const float* logits, unsigned long long* keys,
+                              int* out, int rows, int V, float temperature,
+                              unsigned long long seed, hipStream_t stream) {
   float inv_t = temperature > 0.f ? 1.f / temperature : 0.f;
   if (temperature <= 0.f) seed = 0ull;  // greedy
-  hipLaunchKernelGGL(gumbel_sample_kernel, dim3(rows), dim3(SMP_BLOCK), 0,
-                     stream, logits, out, rows, V, inv_t == 0.f ? 1.f : inv_t,
-                     seed);
+  // fill the chip: aim for >=512 blocks (256 CUs), split each row's vocab
+  int splits = 1;
+  while (rows * splits < 512 && splits * SMP_BLOCK * 4 < V) splits *= 2;
+  hipLaunchKernelGGL(gumbel_sample_kernel, dim3(rows * splits),
+                     dim3(SMP_BLOCK), 0, stream, logits, keys, rows, V, splits,
+                     inv_t == 0.f ? 1.f : inv_t, seed);
+  hipLaunchKernelGGL(gumbel_unpack_kernel, dim3((rows + 255) / 256), dim3(256),
+                     0, stream, keys, out, rows);
 }
 
 // ---------------------------------------------------------------- row softmax
