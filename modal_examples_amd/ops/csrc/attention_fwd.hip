@@ -1,54 +1,55 @@
-// Flash-attention forward v2, bf16, gfx950 (CDNA4 MFMA 16x16x32).
+// Flash-attention forward v3, bf16, gfx950 (CDNA4 MFMA 16x16x32).
 //
 // Serves: K1 diffusion self/cross-attention (D=64), K5
 // Whisper encoder attention, K7 LLM prefill (D=128, causal, GQA).
 //
-// Structure (one workgroup = 4 waves, each wave owns 32 query rows):
+// v3 structure (one workgroup = 4 waves, MT m-tiles × 16 q-rows per wave):
 //   per 64-wide KV block:
-//     stage K and V row-major into LDS with vector 16B writes (no transpose
-//     pass) using an XOR row swizzle to spread the 16-lane fragment reads
-//     across banks (guide §6 G4),
-//     QK^T: A=Q (registers), B=K (b128 LDS reads, hoisted across m-tiles),
-//     online softmax in the MFMA C-layout (16-lane-group shuffle reductions),
+//     K staged row-major (vector 16B writes, XOR row swizzle — guide §6 G4),
+//     V staged into [kv/4][d/16][4][16] subtiles (still pure 16B writes),
+//     QK^T: A=Q (registers), B=K (b128 reads hoisted across m-tiles),
+//     online softmax in MFMA C-layout (16-lane-group shuffle reductions),
 //     P staged through per-wave LDS,
-//     PV computed TRANSPOSED — O^T[d][q] = V^T·P^T — so the V operand reads
-//     row-major V directly (strided u16) and the P operand reads row-major P
-//     contiguously; no V^T staging exists at all.  Per-row softmax factors
-//     reach the transposed accumulator via a 4-shuffle lane broadcast.
+//     PV computed TRANSPOSED: O^T = V^T·P^T, where the V^T fragment comes
+//     from ds_read_b64_tr_b16 hardware transpose reads of the subtiled V
+//     (guide §5.5 T10) — 2 tr-reads replace 8 scalar u16 reads — and the
+//     P^T fragment is a contiguous b128 read.  Per-row softmax factors reach
+//     the transposed accumulator via a 4-shuffle lane broadcast.
 //
 // MFMA fragment layouts (cdna_hip_programming.md §3):
 //   A[l&15][(l>>4)*8+j], B[(l>>4)*8+j][l&15], C[(l>>4)*4+r][l&15].
+// ds_read_b64_tr_b16 semantics (guide §2, m156/m162): lane with halfword
+// address a receives halfwords a + j*16, j=0..3 — i.e. column (a mod 16) of
+// the [4][16] halfword tile at a&~15.
 #include "common.h"
 
 #include <cstdio>
 
-#define QROWS_PER_WAVE 32
-#define KVBLK 64
 #define NWAVES 4
-#define QBLK (QROWS_PER_WAVE * NWAVES)  // 128 query rows per workgroup
+#define KVBLK 64
 #define PPAD 8
 
-// XOR row swizzle for the shared K/V tiles: flips 16B units by kv bits 3..4,
-// separating the four 16-lane groups' bank footprints. Applied identically on
-// the staging writes and every read.
+typedef __attribute__((ext_vector_type(4))) short bf16x4_t;
+typedef __attribute__((address_space(3))) bf16x4_t* lds_tr_ptr;
+
 DEV_INLINE int kv_swz(int row, int byte_off) {
   return byte_off ^ (((row >> 3) & 3) << 4);
 }
 
-template <int D, bool CAUSAL>
+template <int D, int MT, bool CAUSAL>
 __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
-    const short* __restrict__ Q,  // [B, Hq, Sq, D] bf16
-    const short* __restrict__ K,  // [B, Hkv, Sk, D]
-    const short* __restrict__ V,  // [B, Hkv, Sk, D]
-    short* __restrict__ O,        // [B, Hq, Sq, D]
+    const short* __restrict__ Q, const short* __restrict__ K,
+    const short* __restrict__ V, short* __restrict__ O,
     int B, int Hq, int Hkv, int Sq, int Sk, float scale) {
-  constexpr int DCH = D / 32;   // QK^T k-chunks
-  constexpr int DT = D / 16;    // d tiles
-  constexpr int KROW = D + PPAD;  // LDS row pitch (halfwords)
+  constexpr int DCH = D / 32;     // QK^T k-chunks
+  constexpr int DT = D / 16;      // d tiles
+  constexpr int KROW = D + PPAD;  // K row pitch (halfwords)
+  constexpr int QBLK = MT * 16 * NWAVES;
 
   __shared__ alignas(16) short Ks[KVBLK][KROW];
-  __shared__ alignas(16) short Vs[KVBLK][KROW];
-  __shared__ alignas(16) short Ps[NWAVES][QROWS_PER_WAVE][KVBLK + PPAD];
+  // V subtiles: [KVBLK/4][DT][4][16] halfwords
+  __shared__ alignas(16) short Vst[(KVBLK / 4) * DT * 64];
+  __shared__ alignas(16) short Ps[NWAVES][MT * 16][KVBLK + PPAD];
 
   const int tid = threadIdx.x;
   const int w = tid / WAVE;
@@ -63,13 +64,12 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
 
   const long long qbase = (((long long)b * Hq + h) * Sq) * D;
   const long long kbase = (((long long)b * Hkv + hkv) * Sk) * D;
-  const int q0 = qblk * QBLK + w * QROWS_PER_WAVE;
+  const int q0 = qblk * QBLK + w * MT * 16;
   const int causal_off = Sk - Sq;
 
-  // ---- Q fragments in registers ----
-  bf16x8 qf[2][DCH];
+  bf16x8 qf[MT][DCH];
 #pragma unroll
-  for (int mt = 0; mt < 2; ++mt) {
+  for (int mt = 0; mt < MT; ++mt) {
     int qr = q0 + mt * 16 + lr;
 #pragma unroll
     for (int kc = 0; kc < DCH; ++kc) {
@@ -79,15 +79,14 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
     }
   }
 
-  // O^T accumulator: acc[dt][mt] rows = d (lg*4+r), cols = q (lr)
-  f32x4 acc[DT][2];
-  float m_run[2][4], l_run[2][4];
+  f32x4 acc[DT][MT];
+  float m_run[MT][4], l_run[MT][4];
 #pragma unroll
   for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
-    for (int mt = 0; mt < 2; ++mt) acc[dt][mt] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int mt = 0; mt < MT; ++mt) acc[dt][mt] = f32x4{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-  for (int mt = 0; mt < 2; ++mt)
+  for (int mt = 0; mt < MT; ++mt)
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       m_run[mt][r] = -1e30f;
@@ -103,9 +102,8 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
 
   for (int kb = 0; kb < nkb; ++kb) {
     __syncthreads();
-    // ---- stage K and V (vector 16B writes, swizzled rows) ----
     {
-      constexpr int CPR = D / 8;
+      constexpr int CPR = D / 8;  // 16B chunks per row
       constexpr int NCH = KVBLK * CPR;
       for (int ci = tid; ci < NCH; ci += NWAVES * WAVE) {
         int row = ci / CPR, c8 = ci % CPR;
@@ -116,17 +114,19 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
           kv8 = *(const bf16x8*)&K[kbase + (long long)kvp * D + c8 * 8];
           vv8 = *(const bf16x8*)&V[kbase + (long long)kvp * D + c8 * 8];
         }
-        int boff = kv_swz(row, c8 * 16);
-        *(bf16x8*)((char*)&Ks[row][0] + boff) = kv8;
-        *(bf16x8*)((char*)&Vs[row][0] + boff) = vv8;
+        *(bf16x8*)((char*)&Ks[row][0] + kv_swz(row, c8 * 16)) = kv8;
+        // V subtile slot: ((row/4)*DT + c8/2)*64 + (row%4)*16 + (c8%2)*8
+        int vh = (((row >> 2) * DT + (c8 >> 1)) << 6) + ((row & 3) << 4) +
+                 ((c8 & 1) << 3);
+        *(bf16x8*)&Vst[vh] = vv8;
       }
     }
     __syncthreads();
 
-    // ---- S = scale * Q K^T : s[mt][nt], K fragments hoisted over mt ----
-    f32x4 s[2][4];
+    // ---- S = scale * Q K^T ----
+    f32x4 s[MT][4];
 #pragma unroll
-    for (int mt = 0; mt < 2; ++mt)
+    for (int mt = 0; mt < MT; ++mt)
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) s[mt][nt] = f32x4{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
@@ -137,18 +137,18 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
         bf16x8 kfrag = *(const bf16x8*)(
             (char*)&Ks[krow][0] + kv_swz(krow, (kc * 32 + lg * 8) * 2));
         __builtin_amdgcn_s_setprio(1);
-        s[0][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[0][kc], kfrag,
-                                                           s[0][nt], 0, 0, 0);
-        s[1][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[1][kc], kfrag,
-                                                           s[1][nt], 0, 0, 0);
+#pragma unroll
+        for (int mt = 0; mt < MT; ++mt)
+          s[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              qf[mt][kc], kfrag, s[mt][nt], 0, 0, 0);
         __builtin_amdgcn_s_setprio(0);
       }
     }
 
-    // ---- mask + online softmax (C layout: row q = lg*4+r, col kv = lr) ----
-    float fac[2][4];  // exp rescale per (mt, r)
+    // ---- mask + online softmax ----
+    float fac[MT][4];
 #pragma unroll
-    for (int mt = 0; mt < 2; ++mt) {
+    for (int mt = 0; mt < MT; ++mt) {
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
         int col = kb * KVBLK + nt * 16 + lr;
@@ -177,7 +177,6 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
         m_run[mt][r] = m_new;
         fac[mt][r] = rs;
       }
-      // P tile → per-wave LDS (row-major [32][KVBLK+8])
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt)
 #pragma unroll
@@ -185,9 +184,9 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
           Ps[w][mt * 16 + lg * 4 + r][nt * 16 + lr] = f2bf(s[mt][nt][r]);
     }
 
-    // ---- rescale O^T: factor for column q = lr via 4-shuffle broadcast ----
+    // ---- rescale O^T (factor for column q = lr via 4-shuffle broadcast) ----
 #pragma unroll
-    for (int mt = 0; mt < 2; ++mt) {
+    for (int mt = 0; mt < MT; ++mt) {
       int src = ((lr >> 2) << 4) | (l & 15);
       float f0 = __shfl(fac[mt][0], src, WAVE);
       float f1 = __shfl(fac[mt][1], src, WAVE);
@@ -201,36 +200,40 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
         for (int r = 0; r < 4; ++r) acc[dt][mt][r] *= ft;
     }
 
-    // ---- O^T += V^T P^T : A = V^T (strided u16 reads), B = P^T (b128) ----
+    // ---- O^T += V^T P^T (V^T via hardware transpose reads) ----
 #pragma unroll
     for (int kc2 = 0; kc2 < 2; ++kc2) {
-      bf16x8 pfrag[2];
+      bf16x8 pfrag[MT];
 #pragma unroll
-      for (int mt = 0; mt < 2; ++mt)
+      for (int mt = 0; mt < MT; ++mt)
         pfrag[mt] = *(const bf16x8*)&Ps[w][mt * 16 + lr][kc2 * 32 + lg * 8];
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
+        // lane halfword addr: subtile (kc2*8 + lg*2 + h, dt), column lr
+        int sub0 = (kc2 * 8 + lg * 2) * DT + dt;
+        bf16x4_t lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_tr_ptr)&Vst[(sub0 << 6) + lr]);
+        bf16x4_t hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_tr_ptr)&Vst[((sub0 + DT) << 6) + lr]);
         bf16x8 vfrag;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          int vrow = kc2 * 32 + lg * 8 + j;
-          vfrag[j] = *(const short*)(
-              (char*)&Vs[vrow][0] + kv_swz(vrow, (dt * 16 + lr) * 2));
+        for (int j = 0; j < 4; ++j) {
+          vfrag[j] = lo[j];
+          vfrag[j + 4] = hi[j];
         }
         __builtin_amdgcn_s_setprio(1);
-        acc[dt][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vfrag, pfrag[0],
-                                                             acc[dt][0], 0, 0, 0);
-        acc[dt][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vfrag, pfrag[1],
-                                                             acc[dt][1], 0, 0, 0);
+#pragma unroll
+        for (int mt = 0; mt < MT; ++mt)
+          acc[dt][mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              vfrag, pfrag[mt], acc[dt][mt], 0, 0, 0);
         __builtin_amdgcn_s_setprio(0);
       }
     }
   }
 
   // ---- epilogue: O[q][d] = O^T[d][q] / l[q] ----
-  float inv[2];
 #pragma unroll
-  for (int mt = 0; mt < 2; ++mt) {
+  for (int mt = 0; mt < MT; ++mt) {
     int src = ((lr >> 2) << 4) | (l & 15);
     float f0 = __shfl(l_run[mt][0], src, WAVE);
     float f1 = __shfl(l_run[mt][1], src, WAVE);
@@ -238,40 +241,66 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
     float f3 = __shfl(l_run[mt][3], src, WAVE);
     int rsel = lr & 3;
     float lv = rsel == 0 ? f0 : rsel == 1 ? f1 : rsel == 2 ? f2 : f3;
-    inv[mt] = lv > 0.f ? 1.f / lv : 0.f;
-  }
-#pragma unroll
-  for (int mt = 0; mt < 2; ++mt) {
-    int row = q0 + mt * 16 + lr;  // q index (column of O^T)
+    float inv = lv > 0.f ? 1.f / lv : 0.f;
+    int row = q0 + mt * 16 + lr;
     if (row >= Sq) continue;
 #pragma unroll
     for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
       for (int r = 0; r < 4; ++r)
         O[qbase + (long long)row * D + dt * 16 + lg * 4 + r] =
-            f2bf(acc[dt][mt][r] * inv[mt]);
+            f2bf(acc[dt][mt][r] * inv);
   }
 }
 
 extern "C" void fa_fwd_bf16(const void* q, const void* k, const void* v, void* o,
                             int B, int Hq, int Hkv, int Sq, int Sk, int D,
                             float scale, int causal, hipStream_t stream) {
-  dim3 grid((Sq + QBLK - 1) / QBLK, Hq, B);
-  dim3 block(NWAVES * WAVE);
   const short* Qp = (const short*)q;
   const short* Kp = (const short*)k;
   const short* Vp = (const short*)v;
   short* Op = (short*)o;
-#define LAUNCH(DD, CC)                                                        \
-  hipLaunchKernelGGL((fa_fwd_kernel<DD, CC>), grid, block, 0, stream, Qp, Kp, \
-                     Vp, Op, B, Hq, Hkv, Sq, Sk, scale)
+#define LAUNCH(DD, MM, CC)                                                    \
+  do {                                                                        \
+    dim3 grid((Sq + (MM * 16 * NWAVES) - 1) / (MM * 16 * NWAVES), Hq, B);     \
+    hipLaunchKernelGGL((fa_fwd_kernel<DD, MM, CC>), grid,                     \
+                       dim3(NWAVES * WAVE), 0, stream, Qp, Kp, Vp, Op, B, Hq, \
+                       Hkv, Sq, Sk, scale);                                   \
+  } while (0)
   if (D == 64) {
-    if (causal) LAUNCH(64, true); else LAUNCH(64, false);
+    if (causal) LAUNCH(64, 2, true); else LAUNCH(64, 2, false);
   } else if (D == 128) {
-    if (causal) LAUNCH(128, true); else LAUNCH(128, false);
+    // MT=1 (16 q-rows/wave) keeps D=128 at >=2 waves/SIMD occupancy
+    static int mt128 = -1;
+    if (mt128 < 0) {
+      const char* e = getenv("MODAL_AMD_FA_MT128");
+      mt128 = e ? atoi(e) : 1;
+    }
+    if (mt128 == 2) {
+      if (causal) LAUNCH(128, 2, true); else LAUNCH(128, 2, false);
+    } else {
+      if (causal) LAUNCH(128, 1, true); else LAUNCH(128, 1, false);
+    }
   } else {
     fprintf(stderr, "fa_fwd_bf16: unsupported head_dim %d (need 64 or 128)\n", D);
     abort();
   }
 #undef LAUNCH
+}
+
+// ---- semantics probe for ds_read_b64_tr_b16 (debug aid; see tests) ----
+__global__ void tr16_probe_kernel(short* out) {
+  __shared__ short lds[256];
+  for (int i = threadIdx.x; i < 256; i += blockDim.x) lds[i] = (short)i;
+  __syncthreads();
+  int l = threadIdx.x;
+  if (l < 64) {
+    bf16x4_t v = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds_tr_ptr)&lds[l]);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) out[l * 4 + j] = v[j];
+  }
+}
+
+extern "C" void tr16_probe(short* out, hipStream_t stream) {
+  hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(64), 0, stream, out);
 }

@@ -37,6 +37,7 @@ void adamw_step(void*, const void*, float*, float*, long long, float, float,
 void gumbel_sample(const float*, unsigned long long*, int*, int, int,
                    float, unsigned long long, hipStream_t);
 void softmax_rows(const float*, float*, int, int, hipStream_t);
+void tr16_probe(short*, hipStream_t);
 void scale_in_dev_bf16(const void*, void*, const float*, const long long*,
                        long long, hipStream_t);
 void cfg_euler_dev_bf16(const void*, const void*, const void*, void*,
@@ -320,6 +321,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw_", &adamw_, "fused AdamW step (K9)");
   m.def("sample_gumbel", &sample_gumbel, "fused sampling (K8)");
   m.def("softmax_fwd", &softmax_fwd);
+  m.def("tr16_probe", [](torch::Tensor out) {
+    tr16_probe((short*)out.data_ptr(), cur_stream());
+  }, "debug: ds_read_b64_tr_b16 lane/element semantics probe");
   m.def("scale_in_dev", &scale_in_dev, "x * 1/sqrt(sigma^2+1), device sigma");
   m.def("cfg_euler_dev", &cfg_euler_dev, "graph-capturable CFG+Euler step");
   m.def("step_advance", &step_advance, "increment device step counter");
