@@ -381,6 +381,25 @@ class Pool:
         call.value = value
         call.exc = exc
         call.done = True
+        try:
+            from ..observability import metrics, tracing
+
+            dur = time.monotonic() - call.t_submit
+            metrics.inc("calls_total", 1, {"fn": self.name,
+                                           "status": "error" if exc else "ok"})
+            metrics.observe("call_duration_s", dur, {"fn": self.name})
+            if tracing.enabled():
+                tracing._emit({
+                    "trace_id": call.id, "span_id": call.id[-16:],
+                    "parent_id": None, "name": f"call:{self.name}",
+                    "start": time.time() - dur, "end": time.time(),
+                    "duration_ms": round(dur * 1000, 3),
+                    "attrs": {"attempt": call.attempt,
+                              "worker": call.worker_id,
+                              "error": repr(exc) if exc else None},
+                })
+        except Exception:
+            pass
         if call.gen_q is not None:
             call.gen_q.close()
         call.event.set()
