@@ -32,9 +32,8 @@ def crawl_worker(worker_id: int) -> int:
             break
         if page is None:
             break
-        if d.contains(page):
-            continue
-        d[page] = worker_id
+        if not d.put_if_absent(page, worker_id):
+            continue  # another worker claimed this page (atomic dedup)
         crawled += 1
         q.put_many([p for p in links_of(page) if not d.contains(p)])
     return crawled

@@ -70,6 +70,10 @@ class Dict:
     def contains(self, k):
         return self._s.contains(k)
 
+    def put_if_absent(self, k, v) -> bool:
+        """Atomic insert-if-missing; True when this caller won the claim."""
+        return self._s.put_if_absent(k, v)
+
     def len(self):
         return self._s.len()
 

@@ -99,6 +99,16 @@ class DictStore:
         ).fetchone()
         return default if row is None else pickle.loads(row[0])
 
+    def put_if_absent(self, k, v) -> bool:
+        """Atomic claim: True iff the key was newly inserted (INSERT OR IGNORE)."""
+        conn = _DB.get()
+        with conn:
+            cur = conn.execute(
+                "INSERT OR IGNORE INTO kv VALUES (?,?,?,?)",
+                (self.ns, cloudpickle.dumps(k), cloudpickle.dumps(v), time.time()),
+            )
+        return cur.rowcount > 0
+
     def contains(self, k) -> bool:
         conn = _DB.get()
         row = conn.execute(
