@@ -6,7 +6,7 @@
 // v3 structure (one workgroup = 4 waves, MT m-tiles × 16 q-rows per wave):
 //   per 64-wide KV block:
 //     K staged row-major (vector 16B writes, XOR row swizzle — guide §6 G4),
-//     V staged into [kv/4][d/16][4][16] subtiles (still pure 16B writes),
+//     V staged into [kv/4][d/4][4][4]-halfword subtiles (pure 8B writes),
 //     QK^T: A=Q (registers), B=K (b128 reads hoisted across m-tiles),
 //     online softmax in MFMA C-layout (16-lane-group shuffle reductions),
 //     P staged through per-wave LDS,
@@ -18,9 +18,11 @@
 //
 // MFMA fragment layouts (cdna_hip_programming.md §3):
 //   A[l&15][(l>>4)*8+j], B[(l>>4)*8+j][l&15], C[(l>>4)*4+r][l&15].
-// ds_read_b64_tr_b16 semantics (guide §2, m156/m162): lane with halfword
-// address a receives halfwords a + j*16, j=0..3 — i.e. column (a mod 16) of
-// the [4][16] halfword tile at a&~15.
+// ds_read_b64_tr_b16 semantics (MEASURED on MI355X via tr16_probe, this
+// repo): lane with halfword address a receives halfwords a + 4*j, j=0..3 —
+// i.e. a column of the row-major [4][4]-halfword tile at a&~15.  V is
+// therefore staged in [kv/4][d/4] subtiles of [4kv][4d] halfwords so one
+// tr-read yields 4 consecutive kv rows at a fixed d.
 #include "common.h"
 
 #include <cstdio>
@@ -47,8 +49,8 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
   constexpr int QBLK = MT * 16 * NWAVES;
 
   __shared__ alignas(16) short Ks[KVBLK][KROW];
-  // V subtiles: [KVBLK/4][DT][4][16] halfwords
-  __shared__ alignas(16) short Vst[(KVBLK / 4) * DT * 64];
+  // V subtiles: [KVBLK/4][D/4][4][4] halfwords (tr-read tiles)
+  __shared__ alignas(16) short Vst[KVBLK * D];
   __shared__ alignas(16) short Ps[NWAVES][MT * 16][KVBLK + PPAD];
 
   const int tid = threadIdx.x;
@@ -115,10 +117,10 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
           vv8 = *(const bf16x8*)&V[kbase + (long long)kvp * D + c8 * 8];
         }
         *(bf16x8*)((char*)&Ks[row][0] + kv_swz(row, c8 * 16)) = kv8;
-        // V subtile slot: ((row/4)*DT + c8/2)*64 + (row%4)*16 + (c8%2)*8
-        int vh = (((row >> 2) * DT + (c8 >> 1)) << 6) + ((row & 3) << 4) +
-                 ((c8 & 1) << 3);
-        *(bf16x8*)&Vst[vh] = vv8;
+        // two 4x4 subtile rows: ((kv/4)*(D/4) + ds)*16 + (kv%4)*4
+        int vh = (((row >> 2) * (D / 4) + c8 * 2) << 4) + ((row & 3) << 2);
+        *(bf16x4_t*)&Vst[vh] = bf16x4_t{vv8[0], vv8[1], vv8[2], vv8[3]};
+        *(bf16x4_t*)&Vst[vh + 16] = bf16x4_t{vv8[4], vv8[5], vv8[6], vv8[7]};
       }
     }
     __syncthreads();
@@ -209,12 +211,14 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
         pfrag[mt] = *(const bf16x8*)&Ps[w][mt * 16 + lr][kc2 * 32 + lg * 8];
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
-        // lane halfword addr: subtile (kc2*8 + lg*2 + h, dt), column lr
-        int sub0 = (kc2 * 8 + lg * 2) * DT + dt;
+        // lane reads kv column at d = dt*16 + lr: subtile (ks, d/4), col d%4
+        int d = dt * 16 + lr;
+        int ks0 = kc2 * 8 + lg * 2;
+        int a0 = ((ks0 * (D / 4) + (d >> 2)) << 4) + (d & 3);
         bf16x4_t lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-            (lds_tr_ptr)&Vst[(sub0 << 6) + lr]);
+            (lds_tr_ptr)&Vst[a0]);
         bf16x4_t hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-            (lds_tr_ptr)&Vst[((sub0 + DT) << 6) + lr]);
+            (lds_tr_ptr)&Vst[a0 + ((D / 4) << 4)]);
         bf16x8 vfrag;
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
