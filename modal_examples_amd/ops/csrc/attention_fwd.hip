@@ -18,11 +18,12 @@
 //
 // MFMA fragment layouts (cdna_hip_programming.md §3):
 //   A[l&15][(l>>4)*8+j], B[(l>>4)*8+j][l&15], C[(l>>4)*4+r][l&15].
-// ds_read_b64_tr_b16 semantics (MEASURED on MI355X via tr16_probe, this
-// repo): lane with halfword address a receives halfwords a + 4*j, j=0..3 —
-// i.e. a column of the row-major [4][4]-halfword tile at a&~15.  V is
-// therefore staged in [kv/4][d/4] subtiles of [4kv][4d] halfwords so one
-// tr-read yields 4 consecutive kv rows at a fixed d.
+// ds_read_b64_tr_b16 semantics (MEASURED on MI355X, scripts/probe_tr16.cpp):
+// within each 16-lane group the instruction transposes a 4x4 grid of lanes:
+// lane 4a+b receives element j = halfword b of the 8-byte read issued by
+// lane 4j+a.  So with V row-major, lane 4j+a addressing V[kv0+j][d0+4a]
+// delivers lane 4a+b the column V[kv0+j][d0+4a+b] — a free 4-row transpose
+// with NO special LDS layout (V shares K's swizzled row-major staging).
 #include "common.h"
 
 #include <cstdio>
@@ -49,8 +50,7 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
   constexpr int QBLK = MT * 16 * NWAVES;
 
   __shared__ alignas(16) short Ks[KVBLK][KROW];
-  // V subtiles: [KVBLK/4][D/4][4][4] halfwords (tr-read tiles)
-  __shared__ alignas(16) short Vst[KVBLK * D];
+  __shared__ alignas(16) short Vs[KVBLK][KROW];
   __shared__ alignas(16) short Ps[NWAVES][MT * 16][KVBLK + PPAD];
 
   const int tid = threadIdx.x;
@@ -116,11 +116,9 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
           kv8 = *(const bf16x8*)&K[kbase + (long long)kvp * D + c8 * 8];
           vv8 = *(const bf16x8*)&V[kbase + (long long)kvp * D + c8 * 8];
         }
-        *(bf16x8*)((char*)&Ks[row][0] + kv_swz(row, c8 * 16)) = kv8;
-        // two 4x4 subtile rows: ((kv/4)*(D/4) + ds)*16 + (kv%4)*4
-        int vh = (((row >> 2) * (D / 4) + c8 * 2) << 4) + ((row & 3) << 2);
-        *(bf16x4_t*)&Vst[vh] = bf16x4_t{vv8[0], vv8[1], vv8[2], vv8[3]};
-        *(bf16x4_t*)&Vst[vh + 16] = bf16x4_t{vv8[4], vv8[5], vv8[6], vv8[7]};
+        int boff = kv_swz(row, c8 * 16);
+        *(bf16x8*)((char*)&Ks[row][0] + boff) = kv8;
+        *(bf16x8*)((char*)&Vs[row][0] + boff) = vv8;
       }
     }
     __syncthreads();
@@ -211,14 +209,16 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
         pfrag[mt] = *(const bf16x8*)&Ps[w][mt * 16 + lr][kc2 * 32 + lg * 8];
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
-        // lane reads kv column at d = dt*16 + lr: subtile (ks, d/4), col d%4
-        int d = dt * 16 + lr;
-        int ks0 = kc2 * 8 + lg * 2;
-        int a0 = ((ks0 * (D / 4) + (d >> 2)) << 4) + (d & 3);
-        bf16x4_t lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-            (lds_tr_ptr)&Vst[a0]);
-        bf16x4_t hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-            (lds_tr_ptr)&Vst[a0 + ((D / 4) << 4)]);
+        // this lane (4a+b = lr) issues the read for source role (j=lr>>2,
+        // a=lr&3): V[kv0 + (lr>>2)][dt*16 + 4*(lr&3) ..+3]; the 4x4 lane
+        // transpose hands back V[kv0+j][dt*16+lr] for j=0..3
+        int row0 = kc2 * 32 + lg * 8 + (lr >> 2);
+        int binrow = dt * 32 + (lr & 3) * 8;  // byte offset in the V row
+        bf16x4_t lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds_tr_ptr)(
+            (char*)&Vs[0][0] + row0 * (KROW * 2) + kv_swz(row0, binrow)));
+        bf16x4_t hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds_tr_ptr)(
+            (char*)&Vs[0][0] + (row0 + 4) * (KROW * 2) +
+            kv_swz(row0 + 4, binrow)));
         bf16x8 vfrag;
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
