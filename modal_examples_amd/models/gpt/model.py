@@ -47,9 +47,8 @@ class Block(nn.Module):
     def forward(self, x):
         B, S, C = x.shape
         qkv = self.qkv(self.ln1(x)).view(B, S, 3, self.h, self.d)
-        q, k, v = (qkv[:, :, i].transpose(1, 2).contiguous() for i in range(3))
-        o = OF.attention(q, k, v, causal=True)
-        x = x + self.proj(o.transpose(1, 2).reshape(B, S, C))
+        o = OF.attention_qkv(qkv[:, :, 0], qkv[:, :, 1], qkv[:, :, 2], causal=True)
+        x = x + self.proj(o)
         h = torch.nn.functional.gelu(self.fc(self.ln2(x)), approximate="tanh")
         return x + self.fc_proj(h)
 

@@ -150,8 +150,9 @@ class LlamaEngine:
         offs = pos % BLOCK
 
         def kv_writer(li, k, v):
-            self.cache_k[li][blks, :, offs] = k[0].permute(1, 0, 2)
-            self.cache_v[li][blks, :, offs] = v[0].permute(1, 0, 2)
+            # k/v arrive as [B=1, S, nkv, hd] (BSHD views)
+            self.cache_k[li][blks, :, offs] = k[0]
+            self.cache_v[li][blks, :, offs] = v[0]
 
         logits = self.model.prefill(toks, kv_writer)
         r.pos = L
@@ -202,8 +203,9 @@ class LlamaEngine:
         offs = (pos % BLOCK).long()
 
         def kv_append(li, k, v):
-            self.cache_k[li][blks, :, offs] = k[:, :, 0]
-            self.cache_v[li][blks, :, offs] = v[:, :, 0]
+            # k/v arrive as [B, 1, nkv, hd]
+            self.cache_k[li][blks, :, offs] = k[:, 0]
+            self.cache_v[li][blks, :, offs] = v[:, 0]
 
         def kv_attend(li, q):
             return OF.paged_decode(q, self.cache_k[li], self.cache_v[li],

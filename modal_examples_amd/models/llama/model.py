@@ -60,16 +60,15 @@ class LlamaBlock(nn.Module):
         self.ffn_dim = cfg.ffn_dim
 
     def project_qkv(self, x):
-        """x [B,S,d] → q [B,nq,S,hd], k/v [B,nkv,S,hd]."""
+        """x [B,S,d] → BSHD views of the fused projection (zero copies):
+        q [B,S,nq,hd], k/v [B,S,nkv,hd]."""
         B, S, _ = x.shape
         qkv = self.qkv(x)
         q, k, v = qkv.split(
             [self.nq * self.hd, self.nkv * self.hd, self.nkv * self.hd], dim=-1
         )
-        q = q.view(B, S, self.nq, self.hd).transpose(1, 2).contiguous()
-        k = k.view(B, S, self.nkv, self.hd).transpose(1, 2).contiguous()
-        v = v.view(B, S, self.nkv, self.hd).transpose(1, 2).contiguous()
-        return q, k, v
+        return (q.view(B, S, self.nq, self.hd), k.view(B, S, self.nkv, self.hd),
+                v.view(B, S, self.nkv, self.hd))
 
     def ffn(self, x):
         gu = self.gate_up(x)
@@ -102,16 +101,15 @@ class LlamaModel(nn.Module):
         kv_writer(layer_idx, k, v): callback storing [B,nkv,S,hd] into cache."""
         x = self.embed(tokens)
         rc, rs = self._rope_tables(x.device)
+        gpu = x.is_cuda
         for li, blk in enumerate(self.blocks):
             h = blk.attn_norm(x)
-            q, k, v = blk.project_qkv(h)
-            q = OF.rope(q, rc, rs)
-            k = OF.rope(k, rc, rs)
+            q, k, v = blk.project_qkv(h)  # BSHD views
+            q = OF.rope(q.transpose(1, 2), rc, rs, inplace=gpu).transpose(1, 2)
+            k = OF.rope(k.transpose(1, 2), rc, rs, inplace=gpu).transpose(1, 2)
             if kv_writer is not None:
-                kv_writer(li, k, v)
-            o = OF.attention(q, k, v, causal=True)
-            B, _, S, _ = o.shape
-            x = x + blk.o_proj(o.transpose(1, 2).reshape(B, S, -1))
+                kv_writer(li, k, v)  # [B,S,nkv,hd]
+            x = x + blk.o_proj(OF.attention_qkv(q, k, v, causal=True))
             x = x + blk.ffn(blk.ffn_norm(x))
         x = self.norm(x[:, -1:])
         return self.lm_head(x)[:, 0].float()
@@ -123,13 +121,16 @@ class LlamaModel(nn.Module):
         kv_attend(li, q): paged attention of q [B,nq,hd] vs the cache."""
         x = self.embed(tokens).unsqueeze(1)  # [B,1,d]
         rc, rs = self._rope_tables(x.device)
+        gpu = x.is_cuda
         for li, blk in enumerate(self.blocks):
             h = blk.attn_norm(x)
-            q, k, v = blk.project_qkv(h)  # [B,h,1,hd]
-            q = OF.rope(q, rc, rs, positions=positions)
-            k = OF.rope(k, rc, rs, positions=positions)
-            kv_append(li, k, v)
-            o = kv_attend(li, q[:, :, 0])  # [B,nq,hd]
+            q, k, v = blk.project_qkv(h)  # BSHD views [B,1,h,hd]
+            q = OF.rope(q.transpose(1, 2), rc, rs, positions=positions,
+                        inplace=gpu).transpose(1, 2)
+            k = OF.rope(k.transpose(1, 2), rc, rs, positions=positions,
+                        inplace=gpu).transpose(1, 2)
+            kv_append(li, k, v)  # [B,1,nkv,hd]
+            o = kv_attend(li, q[:, 0].contiguous() if not q[:, 0].is_contiguous() else q[:, 0])
             x = x + blk.o_proj(o.reshape(o.shape[0], 1, -1))
             x = x + blk.ffn(blk.ffn_norm(x))
         x = self.norm(x[:, -1:])

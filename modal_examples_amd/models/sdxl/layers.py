@@ -69,9 +69,9 @@ class SelfAttention(nn.Module):
     def forward(self, x):
         B, S, C = x.shape
         qkv = self.qkv(x).view(B, S, 3, self.heads, self.head_dim)
-        q, k, v = (qkv[:, :, i].transpose(1, 2).contiguous() for i in range(3))
-        o = OF.attention(q, k, v, causal=False)
-        return self.out(o.transpose(1, 2).reshape(B, S, C))
+        # transpose-free: the kernel reads the fused-projection slices directly
+        o = OF.attention_qkv(qkv[:, :, 0], qkv[:, :, 1], qkv[:, :, 2])
+        return self.out(o)
 
 
 class CrossAttention(nn.Module):
@@ -88,12 +88,10 @@ class CrossAttention(nn.Module):
     def forward(self, x, ctx):
         B, S, C = x.shape
         Sk = ctx.shape[1]
-        q = self.q(x).view(B, S, self.heads, self.head_dim).transpose(1, 2).contiguous()
+        q = self.q(x).view(B, S, self.heads, self.head_dim)
         kv = self.kv(ctx).view(B, Sk, 2, self.heads, self.head_dim)
-        k = kv[:, :, 0].transpose(1, 2).contiguous()
-        v = kv[:, :, 1].transpose(1, 2).contiguous()
-        o = OF.attention(q, k, v, causal=False)
-        return self.out(o.transpose(1, 2).reshape(B, S, C))
+        o = OF.attention_qkv(q, kv[:, :, 0], kv[:, :, 1])
+        return self.out(o)
 
 
 class GEGLUFeedForward(nn.Module):

@@ -62,20 +62,18 @@ class MHA(nn.Module):
         self.out = nn.Linear(state, state)
 
     def qkv(self, x, ctx=None):
+        """BSHD views: q [B,S,h,d]; k/v [B,Sk,h,d] (no copies)."""
         B, S, C = x.shape
         src = ctx if ctx is not None else x
         Sk = src.shape[1]
-        q = self.q(x).view(B, S, self.h, self.d).transpose(1, 2).contiguous()
+        q = self.q(x).view(B, S, self.h, self.d)
         kv = self.kv(src).view(B, Sk, 2, self.h, self.d)
-        k = kv[:, :, 0].transpose(1, 2).contiguous()
-        v = kv[:, :, 1].transpose(1, 2).contiguous()
-        return q, k, v
+        return q, kv[:, :, 0], kv[:, :, 1]
 
     def forward(self, x, ctx=None, causal=False):
         B, S, C = x.shape
         q, k, v = self.qkv(x, ctx)
-        o = OF.attention(q, k, v, causal=causal)
-        return self.out(o.transpose(1, 2).reshape(B, S, C))
+        return self.out(OF.attention_qkv(q, k, v, causal=causal))
 
 
 class MLP(nn.Module):
@@ -157,14 +155,15 @@ class WhisperModel(nn.Module):
         for li, layer in enumerate(self.dec_layers):
             c = caches[li]
             h = layer.ln1(x)
-            q, k, v = layer.self_attn.qkv(h)
-            c["k"][:, :, :S] = k
-            c["v"][:, :, :S] = v
-            o = OF.attention(q, k, v, causal=True)
-            x = x + layer.self_attn.out(o.transpose(1, 2).reshape(B, S, -1))
+            q, k, v = layer.self_attn.qkv(h)  # BSHD
+            c["k"][:, :, :S] = k.transpose(1, 2)
+            c["v"][:, :, :S] = v.transpose(1, 2)
+            x = x + layer.self_attn.out(OF.attention_qkv(q, k, v, causal=True))
             if "ck" not in c:
-                _, ck, cv = layer.cross_attn.qkv(h, audio)
-                c["ck"], c["cv"] = ck.contiguous(), cv.contiguous()
+                _, ck, cv = layer.cross_attn.qkv(h, audio)  # BSHD
+                # decode kernel wants [B, H, T, D] contiguous caches
+                c["ck"] = ck.transpose(1, 2).contiguous()
+                c["cv"] = cv.transpose(1, 2).contiguous()
             x = x + layer.cross_attn(layer.ln2(x), ctx=audio)
             x = x + layer.mlp(layer.ln3(x))
         x = self.dec_ln(x[:, -1:])
@@ -181,10 +180,10 @@ class WhisperModel(nn.Module):
         for li, layer in enumerate(self.dec_layers):
             c = caches[li]
             h = layer.ln1(x)
-            q, k, v = layer.self_attn.qkv(h)
-            c["k"][:, :, pos:pos + 1] = k
-            c["v"][:, :, pos:pos + 1] = v
-            o = OF.paged_decode(q[:, :, 0], c["k"], c["v"], None, lens_self)
+            q, k, v = layer.self_attn.qkv(h)  # BSHD [B,1,h,d]
+            c["k"][:, :, pos:pos + 1] = k.transpose(1, 2)
+            c["v"][:, :, pos:pos + 1] = v.transpose(1, 2)
+            o = OF.paged_decode(q[:, 0].contiguous(), c["k"], c["v"], None, lens_self)
             x = x + layer.self_attn.out(o.reshape(B, 1, -1))
             h2 = layer.ln2(x)
             q2 = layer.cross_attn.q(h2).view(B, 1, -1)

@@ -39,11 +39,16 @@ DEV_INLINE int kv_swz(int row, int byte_off) {
   return byte_off ^ (((row >> 3) & 3) << 4);
 }
 
+struct FAStrides {
+  // element strides (d is always contiguous); one set per tensor
+  long long qb, qh, qs, kb, kh, ks, vb, vh, vs, ob, oh, os;
+};
+
 template <int D, int MT, bool CAUSAL>
 __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
     const short* __restrict__ Q, const short* __restrict__ K,
     const short* __restrict__ V, short* __restrict__ O,
-    int B, int Hq, int Hkv, int Sq, int Sk, float scale) {
+    int B, int Hq, int Hkv, int Sq, int Sk, float scale, FAStrides st) {
   constexpr int DCH = D / 32;     // QK^T k-chunks
   constexpr int DT = D / 16;      // d tiles
   constexpr int KROW = D + PPAD;  // K row pitch (halfwords)
@@ -64,8 +69,10 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
   const int b = blockIdx.z;
   const int hkv = h / (Hq / Hkv);
 
-  const long long qbase = (((long long)b * Hq + h) * Sq) * D;
-  const long long kbase = (((long long)b * Hkv + hkv) * Sk) * D;
+  const long long qoff = b * st.qb + h * st.qh;
+  const long long koff = b * st.kb + hkv * st.kh;
+  const long long voff = b * st.vb + hkv * st.vh;
+  const long long ooff = b * st.ob + h * st.oh;
   const int q0 = qblk * QBLK + w * MT * 16;
   const int causal_off = Sk - Sq;
 
@@ -76,7 +83,7 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
 #pragma unroll
     for (int kc = 0; kc < DCH; ++kc) {
       qf[mt][kc] = (qr < Sq)
-          ? *(const bf16x8*)&Q[qbase + (long long)qr * D + kc * 32 + lg * 8]
+          ? *(const bf16x8*)&Q[qoff + qr * st.qs + kc * 32 + lg * 8]
           : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
     }
   }
@@ -113,8 +120,8 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
         bf16x8 kv8 = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
         bf16x8 vv8 = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
         if (kvp < Sk) {
-          kv8 = *(const bf16x8*)&K[kbase + (long long)kvp * D + c8 * 8];
-          vv8 = *(const bf16x8*)&V[kbase + (long long)kvp * D + c8 * 8];
+          kv8 = *(const bf16x8*)&K[koff + kvp * st.ks + c8 * 8];
+          vv8 = *(const bf16x8*)&V[voff + kvp * st.vs + c8 * 8];
         }
         int boff = kv_swz(row, c8 * 16);
         *(bf16x8*)((char*)&Ks[row][0] + boff) = kv8;
@@ -252,29 +259,35 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
     for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
       for (int r = 0; r < 4; ++r)
-        O[qbase + (long long)row * D + dt * 16 + lg * 4 + r] =
+        O[ooff + row * st.os + dt * 16 + lg * 4 + r] =
             f2bf(acc[dt][mt][r] * inv);
   }
 }
 
-extern "C" void fa_fwd_bf16(const void* q, const void* k, const void* v, void* o,
-                            int B, int Hq, int Hkv, int Sq, int Sk, int D,
-                            float scale, int causal, hipStream_t stream) {
+extern "C" void fa_fwd_strided_bf16(
+    const void* q, const void* k, const void* v, void* o, int B, int Hq,
+    int Hkv, int Sq, int Sk, int D, float scale, int causal,
+    const long long* strides /*[12]: qb qh qs kb kh ks vb vh vs ob oh os*/,
+    hipStream_t stream) {
   const short* Qp = (const short*)q;
   const short* Kp = (const short*)k;
   const short* Vp = (const short*)v;
   short* Op = (short*)o;
+  FAStrides st;
+  st.qb = strides[0]; st.qh = strides[1]; st.qs = strides[2];
+  st.kb = strides[3]; st.kh = strides[4]; st.ks = strides[5];
+  st.vb = strides[6]; st.vh = strides[7]; st.vs = strides[8];
+  st.ob = strides[9]; st.oh = strides[10]; st.os = strides[11];
 #define LAUNCH(DD, MM, CC)                                                    \
   do {                                                                        \
     dim3 grid((Sq + (MM * 16 * NWAVES) - 1) / (MM * 16 * NWAVES), Hq, B);     \
     hipLaunchKernelGGL((fa_fwd_kernel<DD, MM, CC>), grid,                     \
                        dim3(NWAVES * WAVE), 0, stream, Qp, Kp, Vp, Op, B, Hq, \
-                       Hkv, Sq, Sk, scale);                                   \
+                       Hkv, Sq, Sk, scale, st);                               \
   } while (0)
   if (D == 64) {
     if (causal) LAUNCH(64, 2, true); else LAUNCH(64, 2, false);
   } else if (D == 128) {
-    // MT=1 (16 q-rows/wave) keeps D=128 at >=2 waves/SIMD occupancy
     static int mt128 = -1;
     if (mt128 < 0) {
       const char* e = getenv("MODAL_AMD_FA_MT128");
@@ -286,10 +299,25 @@ extern "C" void fa_fwd_bf16(const void* q, const void* k, const void* v, void* o
       if (causal) LAUNCH(128, 1, true); else LAUNCH(128, 1, false);
     }
   } else {
-    fprintf(stderr, "fa_fwd_bf16: unsupported head_dim %d (need 64 or 128)\n", D);
+    fprintf(stderr, "fa_fwd: unsupported head_dim %d (need 64 or 128)\n", D);
     abort();
   }
 #undef LAUNCH
+}
+
+extern "C" void fa_fwd_bf16(const void* q, const void* k, const void* v,
+                            void* o, int B, int Hq, int Hkv, int Sq, int Sk,
+                            int D, float scale, int causal,
+                            hipStream_t stream) {
+  // contiguous [B,H,S,D] convenience wrapper
+  long long st[12] = {
+      (long long)Hq * Sq * D,  (long long)Sq * D,  D,
+      (long long)Hkv * Sk * D, (long long)Sk * D,  D,
+      (long long)Hkv * Sk * D, (long long)Sk * D,  D,
+      (long long)Hq * Sq * D,  (long long)Sq * D,  D,
+  };
+  fa_fwd_strided_bf16(q, k, v, o, B, Hq, Hkv, Sq, Sk, D, scale, causal, st,
+                      stream);
 }
 
 // ---- semantics probe for ds_read_b64_tr_b16 (debug aid; see tests) ----

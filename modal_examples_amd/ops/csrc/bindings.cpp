@@ -16,11 +16,15 @@
 extern "C" {
 void fa_fwd_bf16(const void*, const void*, const void*, void*, int, int, int,
                  int, int, int, float, int, hipStream_t);
+void fa_fwd_strided_bf16(const void*, const void*, const void*, void*, int,
+                         int, int, int, int, int, float, int,
+                         const long long*, hipStream_t);
 void paged_decode_bf16(const void*, const void*, const void*, const int*,
                        const int*, void*, int, int, int, int, int, int, float,
                        hipStream_t);
-void groupnorm_silu_bf16(const void*, void*, const float*, const float*, int,
-                         int, long long, int, float, int, hipStream_t);
+void groupnorm_silu_bf16(const void*, void*, float*, const float*,
+                         const float*, int, int, long long, int, float, int,
+                         hipStream_t);
 void layernorm_bf16(const void*, void*, const float*, const float*, long long,
                     int, float, hipStream_t);
 void rmsnorm_bf16(const void*, void*, const float*, long long, int, float,
@@ -30,8 +34,8 @@ void cfg_euler_bf16(const void*, const void*, const void*, void*, float, float,
 void silu_mul_bf16(const void*, const void*, void*, long long, hipStream_t);
 void geglu_bf16(const void*, const void*, void*, long long, hipStream_t);
 void add_bf16(const void*, const void*, void*, long long, hipStream_t);
-void rope_bf16(void*, const float*, const float*, long long, int, int,
-               int, const int*, hipStream_t);
+void rope_bf16(void*, const float*, const float*, long long, int, int, int,
+               long long, long long, long long, const int*, hipStream_t);
 void adamw_step(void*, const void*, float*, float*, long long, float, float,
                 float, float, float, int, int, hipStream_t);
 void gumbel_sample(const float*, unsigned long long*, int*, int, int,
@@ -100,6 +104,30 @@ torch::Tensor paged_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
   return o;
 }
 
+// Stride-aware attention: q/k/v logically [B,H,S,D] with ANY strides
+// (d contiguous); output is written [B,S,H,D]-contiguous so callers in
+// sequence-major models (UNet/Llama/Whisper blocks) skip every transpose
+// copy around the kernel.
+torch::Tensor attention_bshd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                             bool causal, double scale) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1,
+              "head_dim must be contiguous");
+  int B = q.size(0), Hq = q.size(1), Sq = q.size(2), D = q.size(3);
+  int Hkv = k.size(1), Sk = k.size(2);
+  auto o = torch::empty({B, Sq, Hq, D}, q.options());
+  long long st[12] = {
+      q.stride(0), q.stride(1), q.stride(2),
+      k.stride(0), k.stride(1), k.stride(2),
+      v.stride(0), v.stride(1), v.stride(2),
+      (long long)Sq * Hq * D, D, (long long)Hq * D,
+  };
+  fa_fwd_strided_bf16(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                      B, Hq, Hkv, Sq, Sk, D, (float)scale, causal ? 1 : 0, st,
+                      cur_stream());
+  return o;
+}
+
 // ---------------------------------------------------------------- norms
 
 torch::Tensor groupnorm_silu(torch::Tensor x, torch::Tensor gamma,
@@ -110,9 +138,12 @@ torch::Tensor groupnorm_silu(torch::Tensor x, torch::Tensor gamma,
   int N = x.size(0), C = x.size(1);
   long long HW = (long long)x.size(2) * x.size(3);
   auto y = torch::empty_like(x);
-  groupnorm_silu_bf16(x.data_ptr(), y.data_ptr(), gamma.data_ptr<float>(),
-                      beta.data_ptr<float>(), N, C, HW, (int)groups, (float)eps,
-                      do_silu ? 1 : 0, cur_stream());
+  auto ws = torch::zeros({N * groups * 2},
+                         x.options().dtype(torch::kFloat32));
+  groupnorm_silu_bf16(x.data_ptr(), y.data_ptr(), ws.data_ptr<float>(),
+                      gamma.data_ptr<float>(), beta.data_ptr<float>(), N, C,
+                      HW, (int)groups, (float)eps, do_silu ? 1 : 0,
+                      cur_stream());
   return y;
 }
 
@@ -176,15 +207,16 @@ torch::Tensor add_residual(torch::Tensor a, torch::Tensor b) {
 
 void rope_(torch::Tensor qk, torch::Tensor cosv, torch::Tensor sinv,
            c10::optional<torch::Tensor> positions) {
-  check_bf16(qk, "qk");
-  TORCH_CHECK(qk.dim() == 4, "qk must be [B,H,S,D]");
+  TORCH_CHECK(qk.is_cuda() && qk.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(qk.dim() == 4 && qk.stride(3) == 1,
+              "qk must be [B,H,S,D] with contiguous head_dim");
   long long B = qk.size(0);
   int H = qk.size(1), S = qk.size(2), D = qk.size(3);
   if (positions.has_value())
     TORCH_CHECK(positions->numel() == B * S || positions->numel() == S,
                 "positions must be [B*S] (or [S] with B==1)");
   rope_bf16(qk.data_ptr(), cosv.data_ptr<float>(), sinv.data_ptr<float>(), B,
-            H, S, D,
+            H, S, D, qk.stride(0), qk.stride(1), qk.stride(2),
             positions.has_value() ? positions->data_ptr<int>() : nullptr,
             cur_stream());
 }
@@ -309,6 +341,7 @@ void snap_free(int64_t id) {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention", &attention, "flash attention fwd bf16 (K1/K5/K7)");
+  m.def("attention_bshd", &attention_bshd, "stride-aware attention, BSHD out");
   m.def("paged_decode", &paged_decode, "paged KV decode attention (K6)");
   m.def("groupnorm_silu", &groupnorm_silu);
   m.def("layernorm", &layernorm);

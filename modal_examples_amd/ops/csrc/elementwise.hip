@@ -169,6 +169,7 @@ extern "C" void add_bf16(const void* a, const void* b, void* y, long long n,
 __global__ __launch_bounds__(EW_BLOCK) void rope_kernel(
     short* __restrict__ Qk, const float* __restrict__ Cos,
     const float* __restrict__ Sin, long long B, int H, int S, int D,
+    long long bst, long long hst, long long sst,
     const int* __restrict__ positions /*nullable [B*S], per-seq offsets*/) {
   // one wave handles one (b, h, s) row; lanes cover D/2 rotation pairs
   long long rows = B * H * S;
@@ -176,9 +177,10 @@ __global__ __launch_bounds__(EW_BLOCK) void rope_kernel(
   for (long long row = blockIdx.x * (EW_BLOCK / WAVE) + threadIdx.x / WAVE;
        row < rows; row += (long long)gridDim.x * (EW_BLOCK / WAVE)) {
     int s = (int)(row % S);
+    int h = (int)((row / S) % H);
     long long b = row / ((long long)H * S);
     int pos = positions != nullptr ? positions[b * S + s] : s;
-    short* x = Qk + row * D;
+    short* x = Qk + b * bst + h * hst + s * sst;
     for (int i = l * 2; i < D / 2; i += WAVE * 2) {
       float c0 = Cos[(long long)pos * (D / 2) + i];
       float s0 = Sin[(long long)pos * (D / 2) + i];
@@ -195,13 +197,15 @@ __global__ __launch_bounds__(EW_BLOCK) void rope_kernel(
 }
 
 extern "C" void rope_bf16(void* qk, const float* cosv, const float* sinv,
-                          long long B, int H, int S, int D,
-                          const int* positions, hipStream_t stream) {
+                          long long B, int H, int S, int D, long long bst,
+                          long long hst, long long sst, const int* positions,
+                          hipStream_t stream) {
   long long rows = B * (long long)H * S;
   int grid = (int)((rows + 3) / 4);
   if (grid > 4096) grid = 4096;
   hipLaunchKernelGGL(rope_kernel, dim3(grid), dim3(EW_BLOCK), 0, stream,
-                     (short*)qk, cosv, sinv, B, H, S, D, positions);
+                     (short*)qk, cosv, sinv, B, H, S, D, bst, hst, sst,
+                     positions);
 }
 
 // ---------------------------------------------------------------- graph-capturable

@@ -15,23 +15,23 @@
 // the group's slab is C/G * HW contiguous elements — two-pass (sum/sqsum, then
 // normalize+affine+optional SiLU).
 
-__global__ __launch_bounds__(256) void groupnorm_silu_kernel(
-    const short* __restrict__ X, short* __restrict__ Y,
-    const float* __restrict__ gamma, const float* __restrict__ beta,
-    int N, int C, long long HW, int G, float eps, int do_silu) {
-  const int n = blockIdx.x / G;
-  const int g = blockIdx.x % G;
+__global__ __launch_bounds__(256) void gn_partial_kernel(
+    const short* __restrict__ X, float* __restrict__ WS, int N, int C,
+    long long HW, int G, int split) {
+  const int blk = blockIdx.x;
+  const int ng = blk / split;
+  const int sp = blk % split;
+  const int n = ng / G, g = ng % G;
   const int cpg = C / G;
-  const long long slab = (long long)cpg * HW;
   const short* x = X + ((long long)n * C + (long long)g * cpg) * HW;
-  short* y = Y + ((long long)n * C + (long long)g * cpg) * HW;
-
-  // channel-major iteration: no per-element division, gamma/beta hoisted
-  const long long hw_full = (HW / 8) * 8;
+  long long per = (((HW + split - 1) / split) + 7) & ~7LL;  // 16B-aligned lo
+  long long lo = sp * per, hi = min(HW, lo + per);
+  if (lo >= HW) return;
   float s = 0.f, ss = 0.f;
   for (int c = 0; c < cpg; ++c) {
     const short* xc = x + (long long)c * HW;
-    for (long long i = (long long)threadIdx.x * 8; i + 8 <= HW; i += 256 * 8) {
+    long long i = lo + (long long)threadIdx.x * 8;
+    for (; i + 8 <= hi; i += 256 * 8) {
       bf16x8 v = *(const bf16x8*)&xc[i];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -40,34 +40,50 @@ __global__ __launch_bounds__(256) void groupnorm_silu_kernel(
         ss += f * f;
       }
     }
-    for (long long i = hw_full + threadIdx.x; i < HW; i += 256) {
-      float f = bf2f(xc[i]);
-      s += f;
-      ss += f * f;
+    // ragged tail of this slice
+    long long tail = hi - ((hi - lo) % 8);
+    for (long long t = tail + threadIdx.x; t < hi; t += 256) {
+      if (t >= lo) {
+        float f = bf2f(xc[t]);
+        s += f;
+        ss += f * f;
+      }
     }
   }
-  // block reduce
-  __shared__ float red[2][4];
   s = wave_sum(s);
   ss = wave_sum(ss);
-  int w = threadIdx.x / WAVE, l = threadIdx.x % WAVE;
-  if (l == 0) {
-    red[0][w] = s;
-    red[1][w] = ss;
+  if ((threadIdx.x & (WAVE - 1)) == 0) {
+    atomicAdd(&WS[ng * 2], s);
+    atomicAdd(&WS[ng * 2 + 1], ss);
   }
-  __syncthreads();
-  s = red[0][0] + red[0][1] + red[0][2] + red[0][3];
-  ss = red[1][0] + red[1][1] + red[1][2] + red[1][3];
-  float mean = s / (float)slab;
-  float var = ss / (float)slab - mean * mean;
-  float rstd = rsqrtf(var + eps);
+}
 
+__global__ __launch_bounds__(256) void gn_norm_kernel(
+    const short* __restrict__ X, short* __restrict__ Y,
+    const float* __restrict__ WS, const float* __restrict__ gamma,
+    const float* __restrict__ beta, int N, int C, long long HW, int G,
+    float eps, int do_silu, int split) {
+  const int blk = blockIdx.x;
+  const int ng = blk / split;
+  const int sp = blk % split;
+  const int n = ng / G, g = ng % G;
+  const int cpg = C / G;
+  const long long slab = (long long)cpg * HW;
+  const short* x = X + ((long long)n * C + (long long)g * cpg) * HW;
+  short* y = Y + ((long long)n * C + (long long)g * cpg) * HW;
+  float mean = WS[ng * 2] / (float)slab;
+  float var = WS[ng * 2 + 1] / (float)slab - mean * mean;
+  float rstd = rsqrtf(var + eps);
+  long long per = (((HW + split - 1) / split) + 7) & ~7LL;
+  long long lo = sp * per, hi = min(HW, lo + per);
+  if (lo >= HW) return;
   for (int c = 0; c < cpg; ++c) {
     const short* xc = x + (long long)c * HW;
     short* yc = y + (long long)c * HW;
     float gam = gamma[g * cpg + c] * rstd;
     float bet = beta[g * cpg + c] - mean * gam;
-    for (long long i = (long long)threadIdx.x * 8; i + 8 <= HW; i += 256 * 8) {
+    long long i = lo + (long long)threadIdx.x * 8;
+    for (; i + 8 <= hi; i += 256 * 8) {
       bf16x8 v = *(const bf16x8*)&xc[i];
       bf16x8 o;
 #pragma unroll
@@ -78,22 +94,31 @@ __global__ __launch_bounds__(256) void groupnorm_silu_kernel(
       }
       *(bf16x8*)&yc[i] = o;
     }
-    for (long long i = hw_full + threadIdx.x; i < HW; i += 256) {
-      float f = bf2f(xc[i]) * gam + bet;
-      if (do_silu) f = f / (1.f + __expf(-f));
-      yc[i] = f2bf(f);
+    long long tail = hi - ((hi - lo) % 8);
+    for (long long t = tail + threadIdx.x; t < hi; t += 256) {
+      if (t >= lo) {
+        float f = bf2f(xc[t]) * gam + bet;
+        if (do_silu) f = f / (1.f + __expf(-f));
+        yc[t] = f2bf(f);
+      }
     }
   }
 }
 
-extern "C" void groupnorm_silu_bf16(const void* x, void* y, const float* gamma,
-                                    const float* beta, int N, int C,
-                                    long long HW, int G, float eps, int do_silu,
+extern "C" void groupnorm_silu_bf16(const void* x, void* y, float* ws,
+                                    const float* gamma, const float* beta,
+                                    int N, int C, long long HW, int G,
+                                    float eps, int do_silu,
                                     hipStream_t stream) {
-  dim3 grid(N * G);
-  hipLaunchKernelGGL(groupnorm_silu_kernel, grid, dim3(256), 0, stream,
-                     (const short*)x, (short*)y, gamma, beta, N, C, HW, G, eps,
-                     do_silu);
+  // fill the chip: split each (n, group) slab so total blocks >= ~1024
+  int split = 1;
+  while (N * G * split < 1024 && (long long)split * 2048 < HW) split *= 2;
+  dim3 grid(N * G * split);
+  hipLaunchKernelGGL(gn_partial_kernel, grid, dim3(256), 0, stream,
+                     (const short*)x, ws, N, C, HW, G, split);
+  hipLaunchKernelGGL(gn_norm_kernel, grid, dim3(256), 0, stream,
+                     (const short*)x, (short*)y, ws, gamma, beta, N, C, HW, G,
+                     eps, do_silu, split);
 }
 
 // ---------------------------------------------------------------- LayerNorm

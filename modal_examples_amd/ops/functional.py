@@ -32,6 +32,27 @@ def attention(q, k, v, causal: bool = False, scale: Optional[float] = None):
     return ext.attention(q.contiguous(), k.contiguous(), v.contiguous(), causal, scale)
 
 
+def attention_qkv(q, k, v, causal: bool = False, scale: Optional[float] = None):
+    """Transpose-free attention for sequence-major models.
+
+    q: [B, S, Hq, D] (any strides, d contiguous — e.g. a slice of a fused QKV
+    projection); k/v: [B, Sk, Hkv, D].  Returns [B, S, Hq*D] contiguous,
+    ready for the output projection.  On GPU the kernel reads the strided
+    layouts directly and writes BSHD — no .contiguous() copies at all.
+    """
+    B, S, Hq, D = q.shape
+    Sk, Hkv = k.shape[1], k.shape[2]
+    scale = scale if scale is not None else 1.0 / math.sqrt(D)
+    ext = _ext_for(q)
+    if ext is None:
+        o = ref.attention_ref(q.transpose(1, 2), k.transpose(1, 2),
+                              v.transpose(1, 2), causal, scale)
+        return o.transpose(1, 2).reshape(B, S, Hq * D)
+    o = ext.attention_bshd(q.transpose(1, 2), k.transpose(1, 2),
+                           v.transpose(1, 2), causal, scale)
+    return o.view(B, S, Hq * D)
+
+
 def paged_decode(q, k_cache, v_cache, block_table, seq_lens, block_size: int = 0,
                  scale: Optional[float] = None):
     """Decode attention over a (paged) KV cache (K6). q [B,Hq,D]."""
@@ -112,9 +133,10 @@ def rope(x, cos, sin, positions=None, inplace: bool = False):
             outs = [ref.rope_ref(x[b:b + 1], cos, sin, pos[b]) for b in range(B)]
             return torch.cat(outs, 0)
         return ref.rope_ref(x, cos, sin, positions.long() if positions is not None else None)
-    y = x.contiguous() if not (inplace and x.is_contiguous()) else x
-    if y.data_ptr() == x.data_ptr() and not inplace:
-        y = x.clone()
+    # kernel is stride-aware (d contiguous required); inplace works on views
+    # (e.g. q/k slices of a fused QKV projection — no copies at all)
+    assert x.stride(-1) == 1, "rope needs contiguous head_dim"
+    y = x if inplace else x.clone()
     pos = positions
     if pos is not None:
         pos = pos.int()
