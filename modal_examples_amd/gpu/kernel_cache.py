@@ -26,7 +26,8 @@ def _user_db_dir() -> Path:
 
 def restore(cache_dir: Path = DEFAULT_CACHE) -> int:
     """Copy a cached find-db into MIOpen's user db location. Returns files
-    restored (0 = no cache yet)."""
+    restored (0 = no cache yet).  Multi-rank safe: writes go to a temp file
+    then os.replace (atomic), so 8 concurrent ranks cannot corrupt the db."""
     cache_dir = Path(cache_dir)
     if not cache_dir.is_dir():
         return 0
@@ -37,7 +38,9 @@ def restore(cache_dir: Path = DEFAULT_CACHE) -> int:
         if f.is_file():
             target = dst / f.name
             if not target.exists() or target.stat().st_size != f.stat().st_size:
-                shutil.copy2(f, target)
+                tmp = dst / f"{f.name}.tmp{os.getpid()}"
+                shutil.copy2(f, tmp)
+                os.replace(tmp, target)
             n += 1
     return n
 
