@@ -109,26 +109,45 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
     if (lim < nkb) nkb = lim;
   }
 
-  for (int kb = 0; kb < nkb; ++kb) {
-    __syncthreads();
-    {
-      constexpr int CPR = D / 8;  // 16B chunks per row
-      constexpr int NCH = KVBLK * CPR;
-      for (int ci = tid; ci < NCH; ci += NWAVES * WAVE) {
-        int row = ci / CPR, c8 = ci % CPR;
-        int kvp = kb * KVBLK + row;
-        bf16x8 kv8 = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-        bf16x8 vv8 = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-        if (kvp < Sk) {
-          kv8 = *(const bf16x8*)&K[koff + kvp * st.ks + c8 * 8];
-          vv8 = *(const bf16x8*)&V[voff + kvp * st.vs + c8 * 8];
-        }
-        int boff = kv_swz(row, c8 * 16);
-        *(bf16x8*)((char*)&Ks[row][0] + boff) = kv8;
-        *(bf16x8*)((char*)&Vs[row][0] + boff) = vv8;
+  // async-staged pipeline (guide §6 G15): block kb+1's global loads are
+  // issued into registers while block kb computes; the LDS write happens
+  // after the compute barrier — HBM latency hides under QK/softmax/PV.
+  constexpr int CPR = D / 8;                      // 16B chunks per row
+  constexpr int CPT = (KVBLK * CPR) / (NWAVES * WAVE);  // chunks per thread
+  bf16x8 kreg[CPT], vreg[CPT];
+
+  auto load_chunks = [&](int kb) {
+#pragma unroll
+    for (int i = 0; i < CPT; ++i) {
+      int ci = i * NWAVES * WAVE + tid;
+      int row = ci / CPR, c8 = ci % CPR;
+      int kvp = kb * KVBLK + row;
+      if (kvp < Sk) {
+        kreg[i] = *(const bf16x8*)&K[koff + kvp * st.ks + c8 * 8];
+        vreg[i] = *(const bf16x8*)&V[voff + kvp * st.vs + c8 * 8];
+      } else {
+        kreg[i] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+        vreg[i] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
       }
     }
-    __syncthreads();
+  };
+  auto write_chunks = [&]() {
+#pragma unroll
+    for (int i = 0; i < CPT; ++i) {
+      int ci = i * NWAVES * WAVE + tid;
+      int row = ci / CPR, c8 = ci % CPR;
+      int boff = kv_swz(row, c8 * 16);
+      *(bf16x8*)((char*)&Ks[row][0] + boff) = kreg[i];
+      *(bf16x8*)((char*)&Vs[row][0] + boff) = vreg[i];
+    }
+  };
+
+  load_chunks(0);
+  for (int kb = 0; kb < nkb; ++kb) {
+    __syncthreads();  // previous block's LDS reads complete
+    write_chunks();
+    __syncthreads();  // tile staged
+    if (kb + 1 < nkb) load_chunks(kb + 1);  // overlap with compute below
 
     // ---- S = scale * Q K^T ----
     f32x4 s[MT][4];
