@@ -45,8 +45,10 @@ def main():
         import torch.distributed as dist
 
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        dist.init_process_group("nccl", rank=rank, world_size=world)
-        torch.cuda.set_device(local_rank)
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        dist.init_process_group(backend, rank=rank, world_size=world)
+        if torch.cuda.is_available():
+            torch.cuda.set_device(local_rank)
     device = f"cuda:{local_rank}" if torch.cuda.is_available() else "cpu"
 
     from modal_examples_amd.models.sdxl.pipeline import SDXLPipeline
@@ -83,7 +85,7 @@ def main():
     if dist_on:
         import torch.distributed as dist
 
-        t = torch.tensor([elapsed], device=device if "cuda" in device else None)
+        t = torch.tensor([elapsed], device=device if "cuda" in device else "cpu")
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
         dist.barrier()

@@ -444,6 +444,15 @@ class Function:
     def with_options(self, **overrides) -> "Function":
         return Function(self.app, self.raw, self.opts.merged(**overrides))
 
+    @staticmethod
+    def from_name(app_name: str, name: str) -> "Function":
+        """Look up a function on a deployed/registered app
+        (the modal.Function.from_name pattern, torch_profiling.py)."""
+        app = App._registry.get(app_name)
+        if app is None or name not in app.functions:
+            raise NotFoundError(f"function {app_name}/{name} not found")
+        return app.functions[name]
+
     def get_web_url(self):
         from .web.ingress import web_url_for
 
@@ -937,6 +946,13 @@ class App:
     def registry_lookup_cls(app_name, cls_name):
         app = App._registry.get(app_name)
         return app.classes.get(cls_name) if app else None
+
+    def include(self, other: "App"):
+        """Merge another app's functions/classes (modal app composition)."""
+        self.functions.update(other.functions)
+        self.classes.update(other.classes)
+        self.web_endpoints.update(other.web_endpoints)
+        return self
 
     @staticmethod
     def lookup(name: str, create_if_missing: bool = False) -> "App":

@@ -157,3 +157,22 @@ def test_aio_map():
         return out
 
     assert asyncio.run(go()) == [0, 1, 4, 9, 16]
+
+
+def test_function_from_name():
+    assert modal.Function.from_name("test-fn", "square").remote(3) == 9
+    import pytest as _pytest
+
+    with _pytest.raises(modal.NotFoundError):
+        modal.Function.from_name("test-fn", "nope")
+
+
+def test_app_include():
+    other = modal.App("test-fn-other")
+
+    @other.function()
+    def triple(x):
+        return 3 * x
+
+    merged = modal.App("test-fn-merged").include(other)
+    assert merged.functions["triple"].remote(4) == 12
