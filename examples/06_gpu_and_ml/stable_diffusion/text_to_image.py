@@ -78,6 +78,32 @@ def png_bytes(img_hwc_uint8) -> bytes:
             + chunk(b"IEND", b""))
 
 
+@app.function()
+@modal.asgi_app(label="ui")
+def ui():
+    """Minimal browser frontend (the Alpine.js-UI pattern, served inline)."""
+    from fastapi import FastAPI
+    from fastapi.responses import HTMLResponse
+
+    web = FastAPI()
+
+    @web.get("/")
+    def index():
+        return HTMLResponse("""<!doctype html>
+<title>MI355X txt2img</title>
+<body style='font-family:sans-serif;max-width:40em;margin:2em auto'>
+<h2>SDXL on MI355X</h2>
+<form action='/generate' method='get'>
+  <input name='prompt' size='40' value='a watercolor city'/>
+  <button>Generate</button>
+</form>
+<p>POSTs hit the <code>generate</code> endpoint; images land on the
+<code>txt2img-outputs</code> volume.</p>
+</body>""")
+
+    return web
+
+
 @app.local_entrypoint()
 def main(prompt: str = "a bicycle on the moon", batch: int = 1,
          steps: int = 4, seed: int = 42):
