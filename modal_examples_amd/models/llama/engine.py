@@ -133,7 +133,42 @@ class LlamaEngine:
     # ------------------------------------------------ prefill
 
     @torch.no_grad()
-    def _prefill(self, r: Request) -> bool:
+    def _prefill_group(self, group: List[Request]) -> None:
+        """Batched prefill of same-length requests: ONE padded forward instead
+        of per-request calls (the prefill half of continuous batching)."""
+        n = len(group)
+        L = len(group[0].prompt)
+        toks = torch.tensor([r.prompt for r in group], device=self.device)
+        pos = torch.arange(L, device=self.device)
+        blk_rows = torch.stack([
+            torch.tensor(r.blocks, device=self.device)[pos // BLOCK]
+            for r in group])  # [n, L]
+        blks = blk_rows.reshape(-1)
+        offs = (pos % BLOCK).repeat(n)
+
+        def kv_writer(li, k, v):
+            # k/v [n, L, nkv, hd] → flatten rows into the paged cache
+            self.cache_k[li][blks, :, offs] = k.reshape(n * L, k.shape[2], k.shape[3])
+            self.cache_v[li][blks, :, offs] = v.reshape(n * L, v.shape[2], v.shape[3])
+
+        logits = self.model.prefill(toks, kv_writer)
+        temps = torch.tensor([r.temperature for r in group])
+        first = self._sample_rows(logits, temps).cpu()
+        for i, r in enumerate(group):
+            r.pos = L
+            self._append_token(r, int(first[i]))
+            r.t_first_token = time.monotonic()
+            slot = r.slot
+            self.toks_d[slot] = int(r.out_tokens[-1])
+            self.pos_d[slot] = r.pos
+            self.lens_d[slot] = r.pos + 1
+            bt = torch.tensor(r.blocks, device=self.device, dtype=torch.int32)
+            self.bt_d[slot, : len(r.blocks)] = bt
+            self.active_d[slot] = 1
+            self.temps_d[slot] = r.temperature
+
+    def _admit(self, r: Request) -> bool:
+        """Reserve blocks + a slot (no compute)."""
         L = len(r.prompt)
         nblk = (L + BLOCK) // BLOCK + 1
         blocks = self._alloc_blocks(nblk)
@@ -143,30 +178,6 @@ class LlamaEngine:
             self.free_blocks.extend(blocks)
             return False
         r.blocks = blocks
-        toks = torch.tensor([r.prompt], device=self.device)
-        pos = torch.arange(L, device=self.device)
-        blk_t = torch.tensor(r.blocks, device=self.device)
-        blks = blk_t[pos // BLOCK]
-        offs = pos % BLOCK
-
-        def kv_writer(li, k, v):
-            # k/v arrive as [B=1, S, nkv, hd] (BSHD views)
-            self.cache_k[li][blks, :, offs] = k[0]
-            self.cache_v[li][blks, :, offs] = v[0]
-
-        logits = self.model.prefill(toks, kv_writer)
-        r.pos = L
-        tok = self._sample_rows(logits, torch.tensor([r.temperature]))
-        self._append_token(r, int(tok[0]))
-        r.t_first_token = time.monotonic()
-        # stage slot state (once per request, not per step)
-        i = r.slot
-        self.toks_d[i] = int(r.out_tokens[-1])
-        self.pos_d[i] = r.pos
-        self.lens_d[i] = r.pos + 1
-        self.bt_d[i, : len(r.blocks)] = blk_t.int()
-        self.active_d[i] = 1
-        self.temps_d[i] = r.temperature
         return True
 
     def _append_token(self, r: Request, tok: int):
@@ -262,15 +273,25 @@ class LlamaEngine:
     def step(self) -> List[Request]:
         """One engine iteration: admit + prefill waiters, one decode step for
         the running batch. Returns requests that finished this step."""
-        while self.waiting and len(self.running) < self.max_batch:
+        # admit as many waiters as fit, then prefill them in same-length
+        # batches (padding-free grouping; synthetic/serving loads are bucketed)
+        admitted: List[Request] = []
+        while self.waiting and len(self.running) + len(admitted) < self.max_batch:
             r = self.waiting[0]
-            if not self._prefill(r):
+            if not self._admit(r):
                 break  # no KV blocks / slots free — keep waiting
             self.waiting.pop(0)
-            if r.done:
-                self._retire(r)
-            else:
-                self.running.append(r)
+            admitted.append(r)
+        by_len: Dict[int, List[Request]] = {}
+        for r in admitted:
+            by_len.setdefault(len(r.prompt), []).append(r)
+        for group in by_len.values():
+            self._prefill_group(group)
+            for r in group:
+                if r.done:
+                    self._retire(r)
+                else:
+                    self.running.append(r)
 
         done_now = []
         if self.running:
