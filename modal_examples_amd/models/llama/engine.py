@@ -313,6 +313,12 @@ class LlamaEngine:
         self._release(r)
         self.finished[r.req_id] = r
 
+    def warmup(self):
+        """Capture the decode graph ahead of serving (cold-start work that
+        belongs with init, not with the first request)."""
+        if self.use_graph:
+            self._ensure_graph()
+
     def run_until_done(self, max_steps: int = 100000):
         steps = 0
         while (self.waiting or self.running) and steps < max_steps:

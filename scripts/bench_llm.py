@@ -40,6 +40,9 @@ def main():
     eng = LlamaEngine(cfg, device=device, dtype=dtype,
                       use_graph=not args.no_graph and device == "cuda",
                       max_batch=args.requests)
+    eng.warmup()  # decode-graph capture is cold-start work
+    if device == "cuda":
+        torch.cuda.synchronize()
     init_s = time.perf_counter() - t0
 
     prompt = [(i % 1000) + 10 for i in range(args.prompt_len)]
