@@ -35,13 +35,8 @@
 typedef __attribute__((ext_vector_type(4))) short bf16x4_t;
 typedef __attribute__((address_space(3))) bf16x4_t* lds_tr_ptr;
 
-// Bank swizzle for K/V/P tiles stored at EXACT power-of-two pitch (row
-// base bank-neutral): XOR the 16B-slot index by row bits 0-2 (spreads the
-// 16 fragment lanes) AND row bits 3-5 (separates the four 16-lane groups,
-// incl. the tr-read rows which differ only in bits >=3).  PMC-motivated:
-// padded-pitch layout measured 2.5 LDS bank conflicts per MFMA.
 DEV_INLINE int kv_swz(int row, int byte_off) {
-  return byte_off ^ ((((row & 7) ^ ((row >> 3) & 7)) << 4));
+  return byte_off ^ (((row >> 3) & 3) << 4);
 }
 
 struct FAStrides {
@@ -56,12 +51,12 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
     int B, int Hq, int Hkv, int Sq, int Sk, float scale, FAStrides st) {
   constexpr int DCH = D / 32;     // QK^T k-chunks
   constexpr int DT = D / 16;      // d tiles
-  constexpr int KROW = D;  // exact pitch: bank-neutral row base + XOR swizzle
+  constexpr int KROW = D + PPAD;  // K row pitch (halfwords)
   constexpr int QBLK = MT * 16 * NWAVES;
 
   __shared__ alignas(16) short Ks[KVBLK][KROW];
   __shared__ alignas(16) short Vs[KVBLK][KROW];
-  __shared__ alignas(16) short Ps[NWAVES][MT * 16][KVBLK];
+  __shared__ alignas(16) short Ps[NWAVES][MT * 16][KVBLK + PPAD];
 
   const int tid = threadIdx.x;
   const int w = tid / WAVE;
@@ -211,11 +206,8 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt)
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          int prow = mt * 16 + lg * 4 + r;
-          *(short*)((char*)&Ps[w][prow][0] +
-                    kv_swz(prow, (nt * 16 + lr) * 2)) = f2bf(s[mt][nt][r]);
-        }
+        for (int r = 0; r < 4; ++r)
+          Ps[w][mt * 16 + lg * 4 + r][nt * 16 + lr] = f2bf(s[mt][nt][r]);
     }
 
     // ---- rescale O^T (factor for column q = lr via 4-shuffle broadcast) ----
@@ -239,11 +231,8 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
     for (int kc2 = 0; kc2 < 2; ++kc2) {
       bf16x8 pfrag[MT];
 #pragma unroll
-      for (int mt = 0; mt < MT; ++mt) {
-        int prow = mt * 16 + lr;
-        pfrag[mt] = *(const bf16x8*)(
-            (char*)&Ps[w][prow][0] + kv_swz(prow, (kc2 * 32 + lg * 8) * 2));
-      }
+      for (int mt = 0; mt < MT; ++mt)
+        pfrag[mt] = *(const bf16x8*)&Ps[w][mt * 16 + lr][kc2 * 32 + lg * 8];
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
         // this lane (4a+b = lr) issues the read for source role (j=lr>>2,
