@@ -17,7 +17,16 @@ from . import reference as ref
 from ._build import get_ext
 
 
-def _ext_for(t: torch.Tensor):
+def _ext_for(t: torch.Tensor, *grad_tensors):
+    """Dispatch: hand kernel on GPU for inference; differentiable torch
+    composition when autograd needs to flow (training forward) — the custom
+    kernels are forward-only, so grad-mode falls back to reference math on
+    the SAME device (hipBLASLt/eager ROCm kernels), keeping LoRA/backward
+    correct end-to-end."""
+    if torch.is_grad_enabled() and any(
+            isinstance(g, torch.Tensor) and g.requires_grad
+            for g in (t, *grad_tensors)):
+        return None
     if t.is_cuda:
         return get_ext(required=True)
     return None
@@ -26,7 +35,7 @@ def _ext_for(t: torch.Tensor):
 def attention(q, k, v, causal: bool = False, scale: Optional[float] = None):
     """Flash attention fwd (K1/K5/K7). q [B,Hq,Sq,D], k/v [B,Hkv,Sk,D] bf16."""
     scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
-    ext = _ext_for(q)
+    ext = _ext_for(q, k, v)
     if ext is None:
         return ref.attention_ref(q, k, v, causal, scale)
     return ext.attention(q.contiguous(), k.contiguous(), v.contiguous(), causal, scale)
@@ -43,7 +52,7 @@ def attention_qkv(q, k, v, causal: bool = False, scale: Optional[float] = None):
     B, S, Hq, D = q.shape
     Sk, Hkv = k.shape[1], k.shape[2]
     scale = scale if scale is not None else 1.0 / math.sqrt(D)
-    ext = _ext_for(q)
+    ext = _ext_for(q, k, v)
     if ext is None:
         o = ref.attention_ref(q.transpose(1, 2), k.transpose(1, 2),
                               v.transpose(1, 2), causal, scale)
@@ -67,7 +76,7 @@ def paged_decode(q, k_cache, v_cache, block_table, seq_lens, block_size: int = 0
 
 def groupnorm_silu(x, gamma, beta, groups: int = 32, eps: float = 1e-5,
                    do_silu: bool = True):
-    ext = _ext_for(x)
+    ext = _ext_for(x, gamma, beta)
     if ext is None:
         return ref.groupnorm_silu_ref(x, gamma, beta, groups, eps, do_silu)
     return ext.groupnorm_silu(x.contiguous(), gamma.float().contiguous(),
@@ -75,7 +84,7 @@ def groupnorm_silu(x, gamma, beta, groups: int = 32, eps: float = 1e-5,
 
 
 def layernorm(x, gamma, beta, eps: float = 1e-5):
-    ext = _ext_for(x)
+    ext = _ext_for(x, gamma, beta)
     if ext is None:
         return ref.layernorm_ref(x, gamma, beta, eps)
     return ext.layernorm(x.contiguous(), gamma.float().contiguous(),
@@ -83,7 +92,7 @@ def layernorm(x, gamma, beta, eps: float = 1e-5):
 
 
 def rmsnorm(x, gamma, eps: float = 1e-6):
-    ext = _ext_for(x)
+    ext = _ext_for(x, gamma)
     if ext is None:
         return ref.rmsnorm_ref(x, gamma, eps)
     return ext.rmsnorm(x.contiguous(), gamma.float().contiguous(), eps)
@@ -100,21 +109,21 @@ def cfg_euler(x_t, eps_c, eps_u, guidance: float, dsigma: float):
 
 
 def silu_mul(a, b):
-    ext = _ext_for(a)
+    ext = _ext_for(a, b)
     if ext is None:
         return ref.silu_mul_ref(a, b)
     return ext.silu_mul(a.contiguous(), b.contiguous())
 
 
 def geglu(a, b):
-    ext = _ext_for(a)
+    ext = _ext_for(a, b)
     if ext is None:
         return ref.geglu_ref(a, b)
     return ext.geglu(a.contiguous(), b.contiguous())
 
 
 def add_residual(a, b):
-    ext = _ext_for(a)
+    ext = _ext_for(a, b)
     if ext is None:
         return (a.float() + b.float()).to(a.dtype)
     return ext.add_residual(a.contiguous(), b.contiguous())
