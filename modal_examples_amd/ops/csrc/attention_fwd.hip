@@ -6,7 +6,7 @@
 // v3 structure (one workgroup = 4 waves, MT m-tiles × 16 q-rows per wave):
 //   per 64-wide KV block:
 //     K staged row-major (vector 16B writes, XOR row swizzle — guide §6 G4),
-//     V staged into [kv/4][d/4][4][4]-halfword subtiles (pure 8B writes),
+//     V shares K's swizzled row-major staging (no transpose pass),
 //     QK^T: A=Q (registers), B=K (b128 reads hoisted across m-tiles),
 //     online softmax in MFMA C-layout (16-lane-group shuffle reductions),
 //     P staged through per-wave LDS,
