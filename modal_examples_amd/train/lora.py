@@ -25,10 +25,13 @@ class LoRALinear(nn.Module):
             p.requires_grad_(False)
         self.rank = rank
         self.scale = alpha / rank
+        dev = base.weight.device
         self.lora_a = nn.Parameter(
-            torch.randn(rank, base.in_features, dtype=torch.float32) * 0.01)
+            torch.randn(rank, base.in_features, dtype=torch.float32,
+                        device=dev) * 0.01)
         self.lora_b = nn.Parameter(
-            torch.zeros(base.out_features, rank, dtype=torch.float32))
+            torch.zeros(base.out_features, rank, dtype=torch.float32,
+                        device=dev))
 
     def forward(self, x):
         y = self.base(x)
