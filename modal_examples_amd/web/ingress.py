@@ -114,8 +114,10 @@ def build_ingress_app(app):
             import inspect as _inspect
 
             sig = _inspect.signature(f.raw)
+            target = f  # closed over — FastAPI deep-copies handler DEFAULTS,
+            # and Function objects hold locks
 
-            async def handler(request: Request, _f=f, _sig=sig):
+            async def handler(request: Request):
                 kwargs = dict(request.query_params)
                 if request.method in ("POST", "PUT"):
                     try:
@@ -126,13 +128,13 @@ def build_ingress_app(app):
                         pass
                 # coerce types using the signature annotations
                 coerced = {}
-                for name, p in _sig.parameters.items():
+                for name, p in sig.parameters.items():
                     if name in kwargs:
                         v = kwargs[name]
                         if p.annotation in (int, float, bool) and isinstance(v, str):
                             v = p.annotation(v) if p.annotation is not bool else v.lower() in ("1", "true", "yes")
                         coerced[name] = v
-                result = await asyncio.to_thread(f.remote, **coerced)
+                result = await asyncio.to_thread(target.remote, **coerced)
                 if hasattr(result, "__class__") and result.__class__.__name__ == "Response":
                     return result
                 if isinstance(result, (bytes, bytearray)):

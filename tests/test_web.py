@@ -1,0 +1,113 @@
+"""Web layer: ingress routes for fastapi/asgi/wsgi endpoints, class-based
+endpoints, and the OpenAI app (hermetic, via httpx ASGI transport)."""
+import asyncio
+
+import httpx
+import pytest
+
+import modal_examples_amd as modal
+from modal_examples_amd.web.ingress import build_ingress_app
+
+app = modal.App("test-web")
+
+
+@app.function()
+@modal.fastapi_endpoint(method="GET", label="double")
+def double(x: int = 1):
+    return {"doubled": x * 2}
+
+
+@app.function()
+@modal.fastapi_endpoint(method="POST", label="concat")
+def concat(a: str = "x", b: str = "y"):
+    return {"joined": a + b}
+
+
+@app.function()
+@modal.asgi_app(label="sub")
+def sub_app():
+    from fastapi import FastAPI
+
+    w = FastAPI()
+
+    @w.get("/ping")
+    def ping():
+        return {"pong": True}
+
+    return w
+
+
+@app.function()
+@modal.wsgi_app(label="wsgi")
+def wsgi_fn():
+    def application(environ, start_response):
+        start_response("200 OK", [("Content-Type", "text/plain")])
+        return [b"wsgi-ok"]
+
+    return application
+
+
+@app.cls()
+class Greeter:
+    @modal.enter()
+    def setup(self):
+        self.msg = "hi"
+
+    @modal.fastapi_endpoint(method="GET", label="greet")
+    def greet(self, name: str = "w"):
+        return {"m": f"{self.msg}-{name}"}
+
+
+def _client():
+    root = build_ingress_app(app)
+    return httpx.AsyncClient(transport=httpx.ASGITransport(app=root),
+                             base_url="http://t")
+
+
+def test_fastapi_get_with_query():
+    async def go():
+        async with _client() as c:
+            r = await c.get("/double", params={"x": 21})
+            assert r.json() == {"doubled": 42}
+
+    asyncio.run(go())
+
+
+def test_fastapi_post_with_body():
+    async def go():
+        async with _client() as c:
+            r = await c.post("/concat", json={"a": "mi", "b": "355x"})
+            assert r.json() == {"joined": "mi355x"}
+
+    asyncio.run(go())
+
+
+def test_asgi_mount():
+    async def go():
+        async with _client() as c:
+            r = await c.get("/sub/ping")
+            assert r.json() == {"pong": True}
+
+    asyncio.run(go())
+
+
+def test_wsgi_mount():
+    async def go():
+        async with _client() as c:
+            r = await c.get("/wsgi/anything")
+            assert r.text == "wsgi-ok"
+
+    asyncio.run(go())
+
+
+def test_cls_endpoint_runs_enter():
+    async def go():
+        async with _client() as c:
+            r = await c.get("/greet", params={"name": "q"})
+            assert r.json() == {"m": "hi-q"}
+
+    asyncio.run(go())
+
+
+def test_web_url_label():
+    assert "double" in double.get_web_url()
