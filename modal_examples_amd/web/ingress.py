@@ -143,6 +143,9 @@ def build_ingress_app(app):
                     return Response(content=bytes(result))
                 return JSONResponse(result) if not hasattr(result, "status_code") else result
 
+            # module uses `from __future__ import annotations`: give FastAPI a
+            # REAL class, not the string "Request"
+            handler.__annotations__ = {"request": Request}
             root.add_api_route(f"/{label}", handler, methods=[method])
         elif kind == "asgi":
             sub = f.raw()  # factory runs in-process
@@ -212,6 +215,7 @@ def _add_cls_route(root, c, mname, flags):
                 return Response(content=bytes(result))
             return JSONResponse(result)
 
+        handler.__annotations__ = {"request": Request}
         root.add_api_route(f"/{label}", handler, methods=[method])
     elif kind == "asgi":
         inst = obj._local_instance()
