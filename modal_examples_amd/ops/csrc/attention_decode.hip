@@ -9,7 +9,10 @@
 // all G = Hq/Hkv query heads of its group against them, so KV bytes are read
 // once per group instead of once per query head.
 //
-//   grid = (Hkv, B), block = 256 (4 waves)
+//   grid = (Hkv, B, SPLITS), block = 256 (4 waves); SPLITS > 1 engages
+//   flash-decoding: each split covers a kv range and writes an (m, l, acc)
+//   partial to a workspace, merged by a second kernel — small-batch decode
+//   otherwise uses only B*Hkv workgroups of a 256-CU chip
 //   lane split: D/8 lanes cover one KV row (16B each) → ROWS=64/(D/8) rows
 //   per wave step; each row-group keeps an online-softmax partial (m, l, acc)
 //   per query head; partials merge through LDS at the end (flash-decoding
@@ -34,15 +37,22 @@ __global__ __launch_bounds__(DEC_THREADS) void paged_decode_kernel(
     const int* __restrict__ block_table,  // [B, max_blocks] or nullptr
     const int* __restrict__ seq_lens,     // [B]
     short* __restrict__ O,             // [B, Hq, D]
-    int B, int Hq, int Hkv, int block_size, int max_blocks, float scale) {
+    float* __restrict__ WS,            // [B, Hq, splits, D+2] or nullptr
+    int B, int Hq, int Hkv, int block_size, int max_blocks, int splits,
+    float scale) {
   constexpr int LPR = D / 8;          // lanes per KV row (16B chunks)
   constexpr int ROWS = WAVE / LPR;    // KV rows per wave step
   constexpr int NPART = DEC_WAVES * ROWS;  // softmax partials to merge
 
   const int hkv = blockIdx.x;
   const int b = blockIdx.y;
+  const int split = blockIdx.z;
   const int G = Hq / Hkv;
-  const int S = seq_lens[b];
+  const int S_total = seq_lens[b];
+  // per-sequence balanced split range
+  const int chunk = (S_total + splits - 1) / splits;
+  const int kv_lo = split * chunk;
+  const int S = min(S_total, kv_lo + chunk);
 
   const int tid = threadIdx.x;
   const int w = tid / WAVE;
@@ -72,8 +82,8 @@ __global__ __launch_bounds__(DEC_THREADS) void paged_decode_kernel(
     for (int j = 0; j < 8; ++j) accv[g][j] = 0.f;
   }
 
-  // ---- sweep the KV rows of this (b, hkv) ----
-  for (int kv = part; kv < S; kv += NPART) {
+  // ---- sweep this split's KV rows of (b, hkv) ----
+  for (int kv = kv_lo + part; kv < S; kv += NPART) {
     long long row_off;
     if (block_table != nullptr) {
       int blk = block_table[(long long)b * max_blocks + kv / block_size];
@@ -135,34 +145,77 @@ __global__ __launch_bounds__(DEC_THREADS) void paged_decode_kernel(
       num += f * sacc[g][p_][d];
       den += f * sl[g][p_];
     }
-    O[(((long long)b * Hq) + hkv * G + g) * D + d] =
-        f2bf(den > 0.f ? num / den : 0.f);
+    int h = hkv * G + g;
+    if (WS == nullptr) {
+      O[(((long long)b * Hq) + h) * D + d] =
+          f2bf(den > 0.f ? num / den : 0.f);
+    } else {
+      float* w = WS + ((((long long)b * Hq) + h) * splits + split) * (D + 2);
+      w[d] = num;  // unnormalized, at this split's m_star
+      if (d == 0) {
+        w[D] = m_star;
+        w[D + 1] = den;
+      }
+    }
+  }
+}
+
+// merge the split partials: O[b,h,:] = sum_s e^{m_s - m*} num_s / sum den_s
+__global__ __launch_bounds__(256) void decode_merge_kernel(
+    const float* __restrict__ WS, short* __restrict__ O, int B, int Hq, int D,
+    int splits) {
+  int bh = blockIdx.x;
+  const float* base = WS + (long long)bh * splits * (D + 2);
+  __shared__ float m_star_s;
+  if (threadIdx.x == 0) {
+    float m = -1e30f;
+    for (int s_ = 0; s_ < splits; ++s_) m = fmaxf(m, base[s_ * (D + 2) + D]);
+    m_star_s = m;
+  }
+  __syncthreads();
+  float m_star = m_star_s;
+  for (int d = threadIdx.x; d < D; d += 256) {
+    float num = 0.f, den_acc = 0.f;
+    for (int s_ = 0; s_ < splits; ++s_) {
+      const float* w = base + s_ * (D + 2);
+      float f = __expf(w[D] - m_star);
+      num += f * w[d];
+      den_acc += f * w[D + 1];
+    }
+    O[(long long)bh * D + d] = f2bf(den_acc > 0.f ? num / den_acc : 0.f);
   }
 }
 
 extern "C" void paged_decode_bf16(const void* q, const void* kc, const void* vc,
                                   const int* block_table, const int* seq_lens,
-                                  void* o, int B, int Hq, int Hkv, int D,
-                                  int block_size, int max_blocks, float scale,
+                                  void* o, float* ws, int B, int Hq, int Hkv,
+                                  int D, int block_size, int max_blocks,
+                                  int splits, float scale,
                                   hipStream_t stream) {
-  dim3 grid(Hkv, B);
-  dim3 block(DEC_THREADS);
   if (Hq / Hkv > MAX_G) {
     fprintf(stderr, "paged_decode_bf16: GQA group %d > %d\n", Hq / Hkv, MAX_G);
     abort();
   }
+  if (splits < 1) splits = 1;
+  dim3 grid(Hkv, B, splits);
+  dim3 block(DEC_THREADS);
+  float* ws_arg = splits > 1 ? ws : nullptr;
+#define DLAUNCH(DD)                                                          \
+  hipLaunchKernelGGL((paged_decode_kernel<DD>), grid, block, 0, stream,      \
+                     (const short*)q, (const short*)kc, (const short*)vc,    \
+                     block_table, seq_lens, (short*)o, ws_arg, B, Hq, Hkv,   \
+                     block_size, max_blocks, splits, scale)
   if (D == 64) {
-    hipLaunchKernelGGL((paged_decode_kernel<64>), grid, block, 0, stream,
-                       (const short*)q, (const short*)kc, (const short*)vc,
-                       block_table, seq_lens, (short*)o, B, Hq, Hkv, block_size,
-                       max_blocks, scale);
+    DLAUNCH(64);
   } else if (D == 128) {
-    hipLaunchKernelGGL((paged_decode_kernel<128>), grid, block, 0, stream,
-                       (const short*)q, (const short*)kc, (const short*)vc,
-                       block_table, seq_lens, (short*)o, B, Hq, Hkv, block_size,
-                       max_blocks, scale);
+    DLAUNCH(128);
   } else {
     fprintf(stderr, "paged_decode_bf16: unsupported head_dim %d\n", D);
     abort();
+  }
+#undef DLAUNCH
+  if (splits > 1) {
+    hipLaunchKernelGGL(decode_merge_kernel, dim3(B * Hq), dim3(256), 0,
+                       stream, ws, (short*)o, B, Hq, D, splits);
   }
 }

@@ -20,8 +20,8 @@ void fa_fwd_strided_bf16(const void*, const void*, const void*, void*, int,
                          int, int, int, int, int, float, int,
                          const long long*, hipStream_t);
 void paged_decode_bf16(const void*, const void*, const void*, const int*,
-                       const int*, void*, int, int, int, int, int, int, float,
-                       hipStream_t);
+                       const int*, void*, float*, int, int, int, int, int,
+                       int, int, float, hipStream_t);
 void groupnorm_silu_bf16(const void*, void*, float*, const float*,
                          const float*, int, int, long long, int, float, int,
                          hipStream_t);
@@ -98,9 +98,25 @@ torch::Tensor paged_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
     block_size = kc.size(2);
   }
   auto o = torch::empty_like(q);
+  // flash-decoding split: fill the chip when B*Hkv is small; cap splits by
+  // the longest representable sequence so chunks stay >=256 kv rows
+  long long s_bound = block_table.has_value()
+      ? (long long)max_blocks * block_size : (long long)kc.size(2);
+  int splits = 1;
+  while (B * Hkv * splits < 512 && (long long)splits * 256 < s_bound &&
+         splits < 64)
+    splits *= 2;
+  torch::Tensor ws;
+  float* ws_ptr = nullptr;
+  if (splits > 1) {
+    ws = torch::empty({(long long)B * Hq * splits * (D + 2)},
+                      q.options().dtype(torch::kFloat32));
+    ws_ptr = ws.data_ptr<float>();
+  }
   paged_decode_bf16(q.data_ptr(), kc.data_ptr(), vc.data_ptr(), bt_ptr,
-                    seq_lens.data_ptr<int>(), o.data_ptr(), B, Hq, Hkv, D,
-                    (int)block_size, max_blocks, (float)scale, cur_stream());
+                    seq_lens.data_ptr<int>(), o.data_ptr(), ws_ptr, B, Hq,
+                    Hkv, D, (int)block_size, max_blocks, splits, (float)scale,
+                    cur_stream());
   return o;
 }
 
