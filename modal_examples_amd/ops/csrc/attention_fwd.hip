@@ -44,7 +44,7 @@ struct FAStrides {
   long long qb, qh, qs, kb, kh, ks, vb, vh, vs, ob, oh, os;
 };
 
-template <int D, int MT, bool CAUSAL>
+template <int D, int MT, bool CAUSAL, int ABL = 0>
 __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
     const short* __restrict__ Q, const short* __restrict__ K,
     const short* __restrict__ V, short* __restrict__ O,
@@ -144,10 +144,12 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
 
   load_chunks(0);
   for (int kb = 0; kb < nkb; ++kb) {
-    __syncthreads();  // previous block's LDS reads complete
-    write_chunks();
-    __syncthreads();  // tile staged
-    if (kb + 1 < nkb) load_chunks(kb + 1);  // overlap with compute below
+    if (ABL != 4) {
+      __syncthreads();  // previous block's LDS reads complete
+      write_chunks();
+      __syncthreads();  // tile staged
+      if (kb + 1 < nkb) load_chunks(kb + 1);  // overlap with compute below
+    }
 
     // ---- S = scale * Q K^T ----
     f32x4 s[MT][4];
@@ -155,6 +157,7 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
     for (int mt = 0; mt < MT; ++mt)
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) s[mt][nt] = f32x4{0.f, 0.f, 0.f, 0.f};
+    if (ABL != 3)
 #pragma unroll
     for (int nt = 0; nt < 4; ++nt) {
 #pragma unroll
@@ -173,6 +176,22 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
 
     // ---- mask + online softmax ----
     float fac[MT][4];
+    if (ABL == 1) {
+      // ablation: skip softmax, but keep s live and P written (raw values)
+#pragma unroll
+      for (int mt = 0; mt < MT; ++mt) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) fac[mt][r] = 1.0f;
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            asm volatile("" :: "v"(s[mt][nt][r]));  // keep QK^T live (rule 17)
+            int prow = mt * 16 + lg * 4 + r;
+            Ps[w][prow][nt * 16 + lr] = f2bf(s[mt][nt][r]);
+          }
+      }
+    } else
 #pragma unroll
     for (int mt = 0; mt < MT; ++mt) {
 #pragma unroll
@@ -211,6 +230,7 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
     }
 
     // ---- rescale O^T (factor for column q = lr via 4-shuffle broadcast) ----
+    if (ABL != 1 && ABL != 2)
 #pragma unroll
     for (int mt = 0; mt < MT; ++mt) {
       int src = ((lr >> 2) << 4) | (l & 15);
@@ -227,6 +247,7 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
     }
 
     // ---- O^T += V^T P^T (V^T via hardware transpose reads) ----
+    if (ABL != 2)
 #pragma unroll
     for (int kc2 = 0; kc2 < 2; ++kc2) {
       bf16x8 pfrag[MT];
@@ -305,7 +326,36 @@ extern "C" void fa_fwd_strided_bf16(
                        Hkv, Sq, Sk, scale, st);                               \
   } while (0)
   if (D == 64) {
-    if (causal) LAUNCH(64, 2, true); else LAUNCH(64, 2, false);
+    static int abl = -1;
+    if (abl < 0) {
+      const char* e = getenv("MODAL_AMD_FA_ABLATE");
+      abl = e ? atoi(e) : 0;
+    }
+    if (causal) {
+      LAUNCH(64, 2, true);
+    } else if (abl == 1) {
+      dim3 grid((Sq + 127) / 128, Hq, B);
+      hipLaunchKernelGGL((fa_fwd_kernel<64, 2, false, 1>), grid,
+                         dim3(NWAVES * WAVE), 0, stream, Qp, Kp, Vp, Op, B,
+                         Hq, Hkv, Sq, Sk, scale, st);
+    } else if (abl == 2) {
+      dim3 grid((Sq + 127) / 128, Hq, B);
+      hipLaunchKernelGGL((fa_fwd_kernel<64, 2, false, 2>), grid,
+                         dim3(NWAVES * WAVE), 0, stream, Qp, Kp, Vp, Op, B,
+                         Hq, Hkv, Sq, Sk, scale, st);
+    } else if (abl == 3) {
+      dim3 grid((Sq + 127) / 128, Hq, B);
+      hipLaunchKernelGGL((fa_fwd_kernel<64, 2, false, 3>), grid,
+                         dim3(NWAVES * WAVE), 0, stream, Qp, Kp, Vp, Op, B,
+                         Hq, Hkv, Sq, Sk, scale, st);
+    } else if (abl == 4) {
+      dim3 grid((Sq + 127) / 128, Hq, B);
+      hipLaunchKernelGGL((fa_fwd_kernel<64, 2, false, 4>), grid,
+                         dim3(NWAVES * WAVE), 0, stream, Qp, Kp, Vp, Op, B,
+                         Hq, Hkv, Sq, Sk, scale, st);
+    } else {
+      LAUNCH(64, 2, false);
+    }
   } else if (D == 128) {
     static int mt128 = -1;
     if (mt128 < 0) {
