@@ -56,7 +56,8 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
 
   __shared__ alignas(16) short Ks[KVBLK][KROW];
   __shared__ alignas(16) short Vs[KVBLK][KROW];
-  __shared__ alignas(16) short Ps[NWAVES][MT * 16][KVBLK + PPAD];
+  constexpr int PROW = MT * 16 + 8;  // P^T row pitch (halfwords)
+  __shared__ alignas(16) short Ps[NWAVES][KVBLK][PROW];
 
   const int tid = threadIdx.x;
   const int w = tid / WAVE;
@@ -183,25 +184,37 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
 #pragma unroll
         for (int r = 0; r < 4; ++r) fac[mt][r] = 1.0f;
 #pragma unroll
-        for (int nt = 0; nt < 4; ++nt)
+        for (int nt = 0; nt < 4; ++nt) {
 #pragma unroll
-          for (int r = 0; r < 4; ++r) {
+          for (int r = 0; r < 4; ++r)
             asm volatile("" :: "v"(s[mt][nt][r]));  // keep QK^T live (rule 17)
-            int prow = mt * 16 + lg * 4 + r;
-            Ps[w][prow][nt * 16 + lr] = f2bf(s[mt][nt][r]);
-          }
+          bf16x4_t pk = bf16x4_t{f2bf(s[mt][nt][0]), f2bf(s[mt][nt][1]),
+                                 f2bf(s[mt][nt][2]), f2bf(s[mt][nt][3])};
+          *(bf16x4_t*)&Ps[w][nt * 16 + lr][mt * 16 + lg * 4] = pk;
+        }
       }
     } else
 #pragma unroll
     for (int mt = 0; mt < MT; ++mt) {
+      // interior blocks (no Sk edge, no causal frontier) skip per-element
+      // masking — the common case for long sequences
+      bool full = (kb * KVBLK + KVBLK <= Sk) &&
+                  (!CAUSAL || kb * KVBLK + KVBLK - 1 <= q0 + mt * 16 + causal_off);
+      if (full) {
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
-        int col = kb * KVBLK + nt * 16 + lr;
+        for (int nt = 0; nt < 4; ++nt)
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          int row = q0 + mt * 16 + lg * 4 + r;
-          bool dead = (col >= Sk) || (CAUSAL && col > row + causal_off);
-          s[mt][nt][r] = dead ? -1e30f : s[mt][nt][r] * scale;
+          for (int r = 0; r < 4; ++r) s[mt][nt][r] *= scale;
+      } else {
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt) {
+          int col = kb * KVBLK + nt * 16 + lr;
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            int row = q0 + mt * 16 + lg * 4 + r;
+            bool dead = (col >= Sk) || (CAUSAL && col > row + causal_off);
+            s[mt][nt][r] = dead ? -1e30f : s[mt][nt][r] * scale;
+          }
         }
       }
 #pragma unroll
@@ -223,27 +236,38 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
         fac[mt][r] = rs;
       }
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt)
-#pragma unroll
-        for (int r = 0; r < 4; ++r)
-          Ps[w][mt * 16 + lg * 4 + r][nt * 16 + lr] = f2bf(s[mt][nt][r]);
+      for (int nt = 0; nt < 4; ++nt) {
+        bf16x4_t pk = bf16x4_t{f2bf(s[mt][nt][0]), f2bf(s[mt][nt][1]),
+                               f2bf(s[mt][nt][2]), f2bf(s[mt][nt][3])};
+        *(bf16x4_t*)&Ps[w][nt * 16 + lr][mt * 16 + lg * 4] = pk;
+      }
     }
 
-    // ---- rescale O^T (factor for column q = lr via 4-shuffle broadcast) ----
-    if (ABL != 1 && ABL != 2)
+    // ---- rescale O^T (skipped when no row max moved: then every factor
+    // is exactly 1 — and on the first block acc is still zero) ----
+    if (ABL != 1 && ABL != 2) {
+      bool changed = false;
+      if (kb > 0)
 #pragma unroll
-    for (int mt = 0; mt < MT; ++mt) {
-      int src = ((lr >> 2) << 4) | (l & 15);
-      float f0 = __shfl(fac[mt][0], src, WAVE);
-      float f1 = __shfl(fac[mt][1], src, WAVE);
-      float f2 = __shfl(fac[mt][2], src, WAVE);
-      float f3 = __shfl(fac[mt][3], src, WAVE);
-      int rsel = lr & 3;
-      float ft = rsel == 0 ? f0 : rsel == 1 ? f1 : rsel == 2 ? f2 : f3;
+        for (int mt = 0; mt < MT; ++mt)
 #pragma unroll
-      for (int dt = 0; dt < DT; ++dt)
+          for (int r = 0; r < 4; ++r) changed |= fac[mt][r] < 0.999999f;
+      if (kb > 0 && __ballot(changed) != 0ull) {
 #pragma unroll
-        for (int r = 0; r < 4; ++r) acc[dt][mt][r] *= ft;
+        for (int mt = 0; mt < MT; ++mt) {
+          int src = ((lr >> 2) << 4) | (l & 15);
+          float f0 = __shfl(fac[mt][0], src, WAVE);
+          float f1 = __shfl(fac[mt][1], src, WAVE);
+          float f2 = __shfl(fac[mt][2], src, WAVE);
+          float f3 = __shfl(fac[mt][3], src, WAVE);
+          int rsel = lr & 3;
+          float ft = rsel == 0 ? f0 : rsel == 1 ? f1 : rsel == 2 ? f2 : f3;
+#pragma unroll
+          for (int dt = 0; dt < DT; ++dt)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) acc[dt][mt][r] *= ft;
+        }
+      }
     }
 
     // ---- O^T += V^T P^T (V^T via hardware transpose reads) ----
