@@ -270,6 +270,11 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
       }
     }
 
+    // P^T writes above are consumed by tr16 reads below WITHOUT a wave
+    // barrier (same-wave LDS is in-order in HW) — stop the COMPILER from
+    // hoisting the reads across the stores:
+    asm volatile("" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
     // ---- O^T += V^T P^T (V^T via hardware transpose reads) ----
     if (ABL != 2)
 #pragma unroll
