@@ -95,6 +95,20 @@ def test_paged_decode_contiguous(D):
 
 
 @requires_gpu
+def test_paged_decode_split_s():
+    """Long contiguous cache at tiny batch engages flash-decoding splits."""
+    torch.manual_seed(55)
+    B, Hq, Hkv, S, D = 1, 8, 2, 4096, 128
+    q = torch.randn(B, Hq, D, device="cuda", dtype=torch.bfloat16)
+    kc = torch.randn(B, Hkv, S, D, device="cuda", dtype=torch.bfloat16)
+    vc = torch.randn(B, Hkv, S, D, device="cuda", dtype=torch.bfloat16)
+    lens = torch.tensor([3777], device="cuda", dtype=torch.int32)  # ragged
+    out = F.paged_decode(q, kc, vc, None, lens)
+    exp = ref.paged_decode_ref(q, kc, vc, None, lens.cpu(), S)
+    _close(out, exp)
+
+
+@requires_gpu
 def test_paged_decode_block_table():
     torch.manual_seed(6)
     B, Hq, Hkv, D, BS = 2, 32, 8, 128, 16
