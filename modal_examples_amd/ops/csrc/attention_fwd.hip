@@ -281,8 +281,21 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
     for (int kc2 = 0; kc2 < 2; ++kc2) {
       bf16x8 pfrag[MT];
 #pragma unroll
-      for (int mt = 0; mt < MT; ++mt)
-        pfrag[mt] = *(const bf16x8*)&Ps[w][mt * 16 + lr][kc2 * 32 + lg * 8];
+      for (int mt = 0; mt < MT; ++mt) {
+        // tr16: source lane 4j+a reads P^T[kv0+j][mt*16+4a..+3]; the 4x4
+        // lane transpose delivers lane 4a+b the column P^T[kv0+j][mt*16+lr]
+        int prow0 = kc2 * 32 + lg * 8 + (lr >> 2);
+        const short* pbase = &Ps[w][0][0];
+        bf16x4_t plo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_tr_ptr)&pbase[prow0 * PROW + mt * 16 + (lr & 3) * 4]);
+        bf16x4_t phi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_tr_ptr)&pbase[(prow0 + 4) * PROW + mt * 16 + (lr & 3) * 4]);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          pfrag[mt][j] = plo[j];
+          pfrag[mt][j + 4] = phi[j];
+        }
+      }
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
         // this lane (4a+b = lr) issues the read for source role (j=lr>>2,
