@@ -368,12 +368,16 @@ extern "C" void fa_fwd_strided_bf16(
                        Hkv, Sq, Sk, scale, st);                               \
   } while (0)
   if (D == 64) {
-    static int abl = -1;
+    static int abl = -1, mt64 = -1;
     if (abl < 0) {
       const char* e = getenv("MODAL_AMD_FA_ABLATE");
       abl = e ? atoi(e) : 0;
+      const char* m = getenv("MODAL_AMD_FA_MT64");
+      mt64 = m ? atoi(m) : 2;
     }
-    if (causal) {
+    if (mt64 == 1) {
+      if (causal) LAUNCH(64, 1, true); else LAUNCH(64, 1, false);
+    } else if (causal) {
       LAUNCH(64, 2, true);
     } else if (abl == 1) {
       dim3 grid((Sq + 127) / 128, Hq, B);
