@@ -141,6 +141,75 @@ def cmd_shell(target: str, extra_args):
     code.interact(banner=banner, local=ns)
 
 
+def cmd_volume(args):
+    """volume ls|get|put|rm <name> [paths] — the `modal volume` surface."""
+    from .resources.volume import Volume
+
+    sub, name, *rest = args
+    vol = Volume.from_name(name, create_if_missing=True)
+    if sub == "ls":
+        for f in sorted(vol.listdir(rest[0] if rest else "/")):
+            print(f)
+    elif sub == "get":
+        remote, local = rest[0], (rest[1] if len(rest) > 1 else Path(rest[0]).name)
+        Path(local).write_bytes(vol.read_file(remote))
+        print(f"wrote {local}")
+    elif sub == "put":
+        local, remote = rest[0], (rest[1] if len(rest) > 1 else Path(rest[0]).name)
+        dst = vol.path / remote.lstrip("/")
+        dst.parent.mkdir(parents=True, exist_ok=True)
+        dst.write_bytes(Path(local).read_bytes())
+        vol.commit()
+        print(f"put {remote}")
+    elif sub == "rm":
+        vol.remove_file(rest[0], recursive="-r" in rest)
+        print(f"removed {rest[0]}")
+    else:
+        raise SystemExit(f"unknown volume subcommand {sub}")
+
+
+def cmd_dict(args):
+    from .resources.dict_queue import Dict
+
+    sub, name, *rest = args
+    d = Dict.from_name(name)
+    if sub == "get":
+        print(d.get(rest[0]))
+    elif sub == "set":
+        d[rest[0]] = rest[1]
+        print("ok")
+    elif sub == "items":
+        for k, v in d.items():
+            print(f"{k}\t{v}")
+    elif sub == "clear":
+        d.clear()
+
+
+def cmd_queue(args):
+    from .resources.dict_queue import Queue
+
+    sub, name, *rest = args
+    q = Queue.from_name(name)
+    if sub == "len":
+        print(q.len(rest[0] if rest else None))
+    elif sub == "put":
+        q.put(rest[0])
+    elif sub == "get":
+        print(q.get(block=False))
+    elif sub == "clear":
+        q.clear(all=True)
+
+
+def cmd_secret(args):
+    from .resources.secret import Secret
+
+    sub, name, *rest = args
+    if sub == "create":
+        env = dict(kv.split("=", 1) for kv in rest)
+        Secret.create(name, env)
+        print(f"created secret {name} with {len(env)} keys")
+
+
 def main(argv=None):
     argv = list(sys.argv[1:] if argv is None else argv)
     if not argv or argv[0] in ("-h", "--help"):
@@ -155,6 +224,16 @@ def main(argv=None):
         cmd_deploy(rest[0], rest[1:])
     elif cmd == "shell":
         cmd_shell(rest[0] if rest else None, rest[1:])
+    elif cmd == "volume":
+        cmd_volume(rest)
+    elif cmd == "dict":
+        cmd_dict(rest)
+    elif cmd == "queue":
+        cmd_queue(rest)
+    elif cmd == "secret":
+        cmd_secret(rest)
     else:
-        raise SystemExit(f"unknown command {cmd!r}; use run/serve/deploy/shell")
+        raise SystemExit(
+            f"unknown command {cmd!r}; use run/serve/deploy/shell/"
+            "volume/dict/queue/secret")
     return 0

@@ -51,13 +51,14 @@ class LlamaEngine:
     def __init__(self, cfg: Optional[LlamaConfig] = None, device: str = "cuda",
                  dtype=torch.bfloat16, max_batch: int = 64,
                  kv_blocks: Optional[int] = None, use_graph: bool = True,
-                 eos_id: int = 2, seed: int = 0):
+                 eos_id: int = 2, seed: int = 0, top_p: float = 1.0):
         self.cfg = cfg or LlamaConfig.llama3_8b()
         self.device = torch.device(device)
         self.dtype = dtype
         self.max_batch = max_batch
         self.use_graph = use_graph and self.device.type == "cuda"
         self.eos_id = eos_id
+        self.top_p = top_p
         torch.manual_seed(seed)
         with torch.device(self.device):
             self.model = LlamaModel(self.cfg).to(self.device, dtype)
@@ -153,7 +154,7 @@ class LlamaEngine:
 
         logits = self.model.prefill(toks, kv_writer)
         temps = torch.tensor([r.temperature for r in group])
-        first = self._sample_rows(logits, temps).cpu()
+        first = self._sample_rows(logits, temps, self.top_p).cpu()
         for i, r in enumerate(group):
             r.pos = L
             self._append_token(r, int(first[i]))
@@ -192,13 +193,22 @@ class LlamaEngine:
 
     # ------------------------------------------------ sampling
 
-    def _sample_rows(self, logits: torch.Tensor, temps: torch.Tensor) -> torch.Tensor:
+    def _sample_rows(self, logits: torch.Tensor, temps: torch.Tensor,
+                     top_p: float = 1.0) -> torch.Tensor:
         temps = temps.to(logits.device)
         self._step_count += 1
         greedy = logits.argmax(-1).int()
         if (temps <= 0).all():
             return greedy
         scaled = logits / temps.clamp_min(1e-5).unsqueeze(-1)
+        if top_p < 1.0:
+            # nucleus filter: mask tokens outside the top-p probability mass
+            sorted_logits, idx = scaled.sort(-1, descending=True)
+            probs = torch.softmax(sorted_logits.float(), -1)
+            cum = probs.cumsum(-1)
+            drop_sorted = cum - probs > top_p  # keep first token crossing p
+            drop = torch.zeros_like(drop_sorted).scatter(-1, idx, drop_sorted)
+            scaled = scaled.masked_fill(drop, float("-inf"))
         sampled = OF.sample(scaled, 1.0, seed=0x5EED + self._step_count)
         return torch.where(temps <= 0, greedy, sampled.to(greedy.device))
 
@@ -248,7 +258,7 @@ class LlamaEngine:
         else:
             lim = max(r.slot for r in self.running) + 1
             logits = self._run_decode(lim)
-        new_toks = self._sample_rows(logits, self.temps_d[:lim])
+        new_toks = self._sample_rows(logits, self.temps_d[:lim], self.top_p)
         # advance device state without host staging
         self.toks_d[:lim].copy_(new_toks.long())
         self.pos_d.add_(self.active_d)

@@ -98,6 +98,19 @@ def web_url_for(fn) -> str:
     return f"http://127.0.0.1:{port}/{label}"
 
 
+def _proxy_auth_ok(request) -> bool:
+    """Proxy-auth check (basic_web.py:178-180 pattern): token id/secret via
+    Modal-Key/Modal-Secret headers, verified against the local keystore env."""
+    import os
+
+    want_key = os.environ.get("MODAL_AMD_PROXY_TOKEN_ID", "")
+    want_secret = os.environ.get("MODAL_AMD_PROXY_TOKEN_SECRET", "")
+    if not want_key:
+        return False  # locked endpoints stay locked until a token is set
+    return (request.headers.get("Modal-Key") == want_key and
+            request.headers.get("Modal-Secret", "") == want_secret)
+
+
 def build_ingress_app(app):
     """Build one FastAPI app exposing every web endpoint of ``app``."""
     from fastapi import FastAPI, Request
@@ -117,7 +130,12 @@ def build_ingress_app(app):
             target = f  # closed over — FastAPI deep-copies handler DEFAULTS,
             # and Function objects hold locks
 
+            needs_auth = bool(flags.get("requires_proxy_auth"))
+
             async def handler(request: Request):
+                if needs_auth and not _proxy_auth_ok(request):
+                    return JSONResponse({"detail": "proxy auth required"},
+                                        status_code=401)
                 kwargs = dict(request.query_params)
                 if request.method in ("POST", "PUT"):
                     try:

@@ -161,3 +161,19 @@ def test_openai_streaming():
         asyncio.run(go())
     finally:
         srv.shutdown()
+
+
+def test_top_p_sampling():
+    """Nucleus filter keeps only the top-p mass (low-prob tokens never drawn)."""
+    import torch
+
+    eng = make_engine()
+    eng.top_p = 0.5
+    logits = torch.full((1, 100), -10.0)
+    logits[0, 7] = 5.0
+    logits[0, 9] = 4.9
+    temps = torch.tensor([1.0])
+    for s in range(20):
+        eng._step_count = s * 7
+        tok = int(eng._sample_rows(logits, temps, eng.top_p)[0])
+        assert tok in (7, 9), tok
