@@ -15,7 +15,25 @@ import torch.nn as nn
 from ...ops import functional as OF
 
 
-class LayerNormK(nn.Module):
+class _F32ParamCache:
+    """Norm params live in whatever dtype the module was cast to; the kernels
+    want f32.  In eval mode the f32 copies are cached (profiling showed
+    thousands of tiny bf16→f32 casts per SDXL step); training recomputes so
+    autograd sees the live parameters."""
+
+    def _f32(self, name: str):
+        p = getattr(self, name)
+        if self.training or p.requires_grad and torch.is_grad_enabled():
+            return p.float()
+        cache = self.__dict__.setdefault("_f32_cache", {})
+        ent = cache.get(name)
+        if ent is None or ent[1] is not p or ent[2] != p._version:
+            ent = (p.detach().float(), p, p._version)
+            cache[name] = ent
+        return ent[0]
+
+
+class LayerNormK(nn.Module, _F32ParamCache):
     """LayerNorm over last dim via the gfx950 kernel (f32 affine params)."""
 
     def __init__(self, dim: int, eps: float = 1e-5):
@@ -25,20 +43,20 @@ class LayerNormK(nn.Module):
         self.eps = eps
 
     def forward(self, x):
-        return OF.layernorm(x, self.weight.float(), self.bias.float(), self.eps)
+        return OF.layernorm(x, self._f32("weight"), self._f32("bias"), self.eps)
 
 
-class RMSNormK(nn.Module):
+class RMSNormK(nn.Module, _F32ParamCache):
     def __init__(self, dim: int, eps: float = 1e-6):
         super().__init__()
         self.weight = nn.Parameter(torch.ones(dim))
         self.eps = eps
 
     def forward(self, x):
-        return OF.rmsnorm(x, self.weight.float(), self.eps)
+        return OF.rmsnorm(x, self._f32("weight"), self.eps)
 
 
-class GroupNormSiLU(nn.Module):
+class GroupNormSiLU(nn.Module, _F32ParamCache):
     """Fused GroupNorm(+SiLU) NCHW via the gfx950 kernel."""
 
     def __init__(self, channels: int, groups: int = 32, eps: float = 1e-5,
@@ -51,7 +69,7 @@ class GroupNormSiLU(nn.Module):
         self.silu = silu
 
     def forward(self, x):
-        return OF.groupnorm_silu(x, self.weight.float(), self.bias.float(),
+        return OF.groupnorm_silu(x, self._f32("weight"), self._f32("bias"),
                                  self.groups, self.eps, self.silu)
 
 
