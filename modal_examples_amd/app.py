@@ -373,6 +373,12 @@ class Function:
                 gpu_count = min(gpu_count, rt.device_pool.n)
                 raw = self.raw
                 env = {}
+                if self.opts.image is not None:
+                    env.update(getattr(self.opts.image, "build_env", {}) or {})
+                    try:
+                        self.opts.image.build()  # run_function layers, once
+                    except Exception:
+                        pass
                 for s in self.opts.secrets or []:
                     env.update(getattr(s, "env", {}))
                 vols = {p: getattr(v, "name", str(v)) for p, v in (self.opts.volumes or {}).items()}
@@ -416,6 +422,7 @@ class Function:
                     retries=policy,
                     single_use_containers=self.opts.single_use_containers,
                     max_inputs_per_worker=max_inputs,
+                    target_inputs_per_worker=int(self._flags.get("target_inputs", 0) or 0),
                     device_pool=rt.device_pool,
                 )
                 self._pool.on_spawned_result = _persist_spawned
@@ -618,6 +625,12 @@ class _Obj:
                 rt = _Runtime.get()
                 gpu_count = min(parse_gpu(cls.opts.gpu), rt.device_pool.n)
                 env = {}
+                if cls.opts.image is not None:
+                    env.update(getattr(cls.opts.image, "build_env", {}) or {})
+                    try:
+                        cls.opts.image.build()
+                    except Exception:
+                        pass
                 for s in cls.opts.secrets or []:
                     env.update(getattr(s, "env", {}))
                 target_blob = ipc.dumps(cls.user_cls)
@@ -627,6 +640,9 @@ class _Obj:
                 name = f"{cls.name}({','.join(f'{k}={v}' for k, v in sorted(params.items()))})" if params else cls.name
                 mem_snap = bool(cls.opts.enable_memory_snapshot)
                 gpu_snap = bool((cls.opts.experimental_options or {}).get("enable_gpu_snapshot"))
+
+                vols = {p_: getattr(v, "name", str(v))
+                        for p_, v in (cls.opts.volumes or {}).items()}
 
                 def make_spec(devices):
                     return ipc.ServiceSpec(
@@ -638,6 +654,7 @@ class _Obj:
                         max_inputs=max_inputs,
                         gpu_devices=devices,
                         env=dict(env),
+                        volumes=vols,
                         enable_memory_snapshot=mem_snap,
                         enable_gpu_snapshot=gpu_snap,
                     )

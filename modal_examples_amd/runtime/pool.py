@@ -153,6 +153,7 @@ class Pool:
         retries: Optional[RetryPolicy] = None,
         single_use_containers: bool = False,
         max_inputs_per_worker: int = 1,
+        target_inputs_per_worker: int = 0,
         device_pool=None,
     ):
         self.name = name
@@ -166,6 +167,7 @@ class Pool:
         self.retries = retries or RetryPolicy(max_retries=0)
         self.single_use = single_use_containers
         self.max_inputs_per_worker = max(1, max_inputs_per_worker)
+        self.target_inputs_per_worker = target_inputs_per_worker or self.max_inputs_per_worker
         self.device_pool = device_pool
 
         self.result_q = _mp.Queue()
@@ -358,6 +360,11 @@ class Pool:
                 if started is None:
                     return  # all busy / can't grow: leave pending
                 continue  # newly started worker may not be ready; loop picks ready ones
+            # target_inputs autoscale signal (@modal.concurrent): prefer
+            # growing the pool over loading a worker past its target
+            if (len(w.inflight) >= self.target_inputs_per_worker
+                    and len(self.workers) < self.max_containers):
+                self._maybe_scale_up()
             call = self.pending.pop(0)
             self._dispatch(call, w)
         # buffer containers: keep `buffer` idle warm workers beyond demand

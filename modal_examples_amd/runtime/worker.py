@@ -23,6 +23,28 @@ from concurrent.futures import ThreadPoolExecutor
 from . import ipc
 
 
+def _mount_volumes(volumes: dict) -> None:
+    """Symlink mount paths to the shared volume directories (the worker runs
+    as root on this node — mirrors containers mounting at /cache etc.)."""
+    from .. import config
+
+    for mount, name in (volumes or {}).items():
+        if name.startswith("bucket:"):
+            target = config.state_dir() / "buckets" / name.split(":", 1)[1]
+        else:
+            target = config.state_dir() / "volumes" / name
+        target.mkdir(parents=True, exist_ok=True)
+        try:
+            if os.path.islink(mount) or os.path.exists(mount):
+                continue
+            parent = os.path.dirname(mount.rstrip("/"))
+            if parent and not os.path.exists(parent):
+                os.makedirs(parent, exist_ok=True)
+            os.symlink(target, mount)
+        except OSError:
+            pass  # unmountable path: functions can still use volume.path
+
+
 def _apply_env(spec_env: dict, gpu_devices: tuple) -> None:
     if gpu_devices:
         vis = ",".join(str(d) for d in gpu_devices)
@@ -74,6 +96,7 @@ def _resolve_target(spec: "ipc.ServiceSpec"):
 def worker_main(worker_id: int, spec_blob: bytes, task_q, result_q) -> None:
     spec: ipc.ServiceSpec = ipc.loads(spec_blob)
     _apply_env(spec.env, spec.gpu_devices)
+    _mount_volumes(spec.volumes)
     os.environ["MODAL_TASK_ID"] = f"ta-local-{spec.name}-{worker_id}"
 
     def post(kind, call_id=None, payload=None, text=""):

@@ -155,3 +155,35 @@ def test_gpu_string_parsing():
     assert parse_gpu("MI355X:4") == 4
     assert parse_gpu("H100!") == 1
     assert parse_gpu(["H100:2", "A100"]) == 2
+
+
+def test_volume_mount_path_in_worker():
+    """volumes={'/mnt/x': vol} symlinks the mount path inside the worker."""
+    vol = modal.Volume.from_name("mount-test-vol", create_if_missing=True)
+    (vol.path / "probe.txt").write_text("mounted!")
+    app = modal.App("test-mount-app")
+
+    @app.function(volumes={"/tmp/mxa_mount_test": vol})
+    def read_mounted():
+        with open("/tmp/mxa_mount_test/probe.txt") as f:
+            return f.read()
+
+    assert read_mounted.remote() == "mounted!"
+    modal.Volume.delete("mount-test-vol")
+    import os
+
+    if os.path.islink("/tmp/mxa_mount_test"):
+        os.unlink("/tmp/mxa_mount_test")
+
+
+def test_image_env_reaches_worker():
+    img = modal.Image.debian_slim().env({"IMG_LAYER_VAR": "layered"})
+    app = modal.App("test-imgenv-app")
+
+    @app.function(image=img)
+    def read_env():
+        import os
+
+        return os.environ.get("IMG_LAYER_VAR")
+
+    assert read_env.remote() == "layered"
