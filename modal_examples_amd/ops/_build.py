@@ -21,6 +21,7 @@ ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
 HIP_SOURCES = [
     "attention_fwd.hip",
+    "attention_fwd32.hip",
     "attention_decode.hip",
     "norms.hip",
     "elementwise.hip",

@@ -346,11 +346,27 @@ __global__ __launch_bounds__(NWAVES * WAVE) void fa_fwd_kernel(
   }
 }
 
+extern "C" void fa32_fwd_strided_bf16(
+    const void*, const void*, const void*, void*, int, int, int, int, int,
+    int, float, int, const long long*, hipStream_t);
+
 extern "C" void fa_fwd_strided_bf16(
     const void* q, const void* k, const void* v, void* o, int B, int Hq,
     int Hkv, int Sq, int Sk, int D, float scale, int causal,
     const long long* strides /*[12]: qb qh qs kb kh ks vb vh vs ob oh os*/,
     hipStream_t stream) {
+  // v7 (32x32 MFMA, lane-local softmax) is the default path; set
+  // MODAL_AMD_FA_V7=0 to fall back to the 16x16 structure.
+  static int v7 = -1;
+  if (v7 < 0) {
+    const char* e = getenv("MODAL_AMD_FA_V7");
+    v7 = e ? atoi(e) : 1;
+  }
+  if (v7 && (D == 64 || D == 128)) {
+    fa32_fwd_strided_bf16(q, k, v, o, B, Hq, Hkv, Sq, Sk, D, scale, causal,
+                          strides, stream);
+    return;
+  }
   const short* Qp = (const short*)q;
   const short* Kp = (const short*)k;
   const short* Vp = (const short*)v;

@@ -1,0 +1,262 @@
+// Flash-attention forward v7 — 32x32x16 MFMA with LANE-LOCAL online softmax.
+//
+// The v3-v6 kernels (attention_fwd.hip) are softmax-LATENCY-bound: ablation
+// measured the 16-lane-group shuffle/exp chains at ~56% of kernel time.  This
+// structure eliminates cross-lane reductions almost entirely (the m214-style
+// swapped-operand recipe, re-derived for gfx950):
+//
+//   QK^T is computed TRANSPOSED with mfma_f32_32x32x16_bf16:
+//     S^T = K · Q^T  →  each lane holds 16 of the 32 kv-scores for ONE query
+//     (C layout col = q = lane&31); row-softmax becomes 15 in-register ops +
+//     ONE shfl_xor(32) to combine with the partner lane.  Rescale factors are
+//     lane-local (no broadcast).
+//   P converts to bf16 B-fragments IN REGISTERS: v_cvt_pk_bf16_f32 pairs +
+//     permlane32_swap half-exchanges (semantics measured: result pair is
+//     (A_lo||B_lo , A_hi||B_hi)) — P never touches LDS.
+//   PV: O^T = V^T · P^T, V^T fragments via ds_read_b64_tr_b16 hardware
+//     transpose reads (4x4 lane-grid exchange, measured in probe_tr16.cpp).
+//
+// Fragment layouts (verified on-device, scripts/probe_mfma32.cpp):
+//   A[m][k]: lane holds A[l&31][(l>>5)*8+j];  B[k][n]: B[(l>>5)*8+j][l&31]
+//   C[m][n]: lane holds C[(reg&3)+8*(reg>>2)+4*(l>>5)][l&31], reg 0..15.
+#include "common.h"
+
+#include <cstdio>
+
+#define FA32_NWAVES 4
+#define FA32_KVBLK 32
+#define FA32_PPAD 8
+
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+typedef __attribute__((ext_vector_type(4))) short bf16x4_v7;
+typedef __attribute__((address_space(3))) bf16x4_v7* lds_tr_ptr7;
+typedef __attribute__((ext_vector_type(2))) int i32x2;
+
+DEV_INLINE int kv_swz7(int row, int byte_off) {
+  return byte_off ^ (((row >> 3) & 3) << 4);
+}
+
+DEV_INLINE unsigned cvt_pk_bf16(float lo, float hi) {
+  unsigned r;
+  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+
+struct FA32Strides {
+  long long qb, qh, qs, kb, kh, ks, vb, vh, vs, ob, oh, os;
+};
+
+template <int D, bool CAUSAL>
+__global__ __launch_bounds__(FA32_NWAVES * WAVE) void fa32_kernel(
+    const short* __restrict__ Q, const short* __restrict__ K,
+    const short* __restrict__ V, short* __restrict__ O,
+    int B, int Hq, int Hkv, int Sq, int Sk, float scale, FA32Strides st) {
+  constexpr int KROW = D + FA32_PPAD;
+  constexpr int DCH = D / 16;  // QK^T k-chunks (K-dim 16)
+  constexpr int DT = D / 32;   // PV d-tiles (M-dim 32)
+  constexpr int QBLK = 32 * FA32_NWAVES;
+
+  __shared__ alignas(16) short Ks[FA32_KVBLK][KROW];
+  __shared__ alignas(16) short Vs[FA32_KVBLK][KROW];
+
+  const int tid = threadIdx.x;
+  const int w = tid / WAVE;
+  const int l = tid % WAVE;
+  const int l31 = l & 31;
+  const int hi5 = l >> 5;  // 0 for lanes 0-31, 1 for 32-63
+
+  const int qblk = blockIdx.x;
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int hkv = h / (Hq / Hkv);
+
+  const long long qoff = b * st.qb + h * st.qh;
+  const long long koff = b * st.kb + hkv * st.kh;
+  const long long voff = b * st.vb + hkv * st.vh;
+  const long long ooff = b * st.ob + h * st.oh;
+  const int q0 = qblk * QBLK + w * 32;
+  const int my_q = q0 + l31;  // this lane's query row
+  const int causal_off = Sk - Sq;
+
+  // Q^T B-fragments: qreg[c][j] = Q[my_q][c*16 + hi5*8 + j]
+  bf16x8 qreg[DCH];
+#pragma unroll
+  for (int c = 0; c < DCH; ++c) {
+    qreg[c] = (my_q < Sq)
+        ? *(const bf16x8*)&Q[qoff + (long long)my_q * st.qs + c * 16 + hi5 * 8]
+        : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+  }
+
+  f32x16 acc[DT];
+#pragma unroll
+  for (int dt = 0; dt < DT; ++dt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) acc[dt][r] = 0.f;
+  float m_run = -1e30f, l_run = 0.f;
+
+  int nkb = (Sk + FA32_KVBLK - 1) / FA32_KVBLK;
+  if (CAUSAL) {
+    int max_kv = qblk * QBLK + QBLK - 1 + causal_off;
+    int lim = (max_kv + FA32_KVBLK) / FA32_KVBLK;
+    if (lim < nkb) nkb = lim;
+  }
+
+  for (int kb = 0; kb < nkb; ++kb) {
+    __syncthreads();
+    {
+      constexpr int CPR = D / 8;
+      constexpr int NCH = FA32_KVBLK * CPR;
+      for (int ci = tid; ci < NCH; ci += FA32_NWAVES * WAVE) {
+        int row = ci / CPR, c8 = ci % CPR;
+        int kvp = kb * FA32_KVBLK + row;
+        bf16x8 kv8 = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+        bf16x8 vv8 = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+        if (kvp < Sk) {
+          kv8 = *(const bf16x8*)&K[koff + kvp * st.ks + c8 * 8];
+          vv8 = *(const bf16x8*)&V[voff + kvp * st.vs + c8 * 8];
+        }
+        int boff = kv_swz7(row, c8 * 16);
+        *(bf16x8*)((char*)&Ks[row][0] + boff) = kv8;
+        *(bf16x8*)((char*)&Vs[row][0] + boff) = vv8;
+      }
+    }
+    __syncthreads();
+
+    // ---- S^T = K · Q^T ----
+    f32x16 s;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) s[r] = 0.f;
+#pragma unroll
+    for (int c = 0; c < DCH; ++c) {
+      // A = K[kv][d]: lane holds K[l31][c*16 + hi5*8 + j]
+      bf16x8 kfrag = *(const bf16x8*)(
+          (char*)&Ks[l31][0] + kv_swz7(l31, (c * 16 + hi5 * 8) * 2));
+      __builtin_amdgcn_s_setprio(1);
+      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qreg[c], s, 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+
+    // ---- mask + lane-local online softmax (this lane owns query my_q) ----
+    bool full = (kb * FA32_KVBLK + FA32_KVBLK <= Sk) &&
+                (!CAUSAL || kb * FA32_KVBLK + FA32_KVBLK - 1 <= my_q + causal_off);
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int kvp = kb * FA32_KVBLK + (r & 3) + 8 * (r >> 2) + 4 * hi5;
+      bool dead = !full && ((kvp >= Sk) || (CAUSAL && kvp > my_q + causal_off));
+      s[r] = dead ? -1e30f : s[r] * scale;
+    }
+    float smax = s[0];
+#pragma unroll
+    for (int r = 1; r < 16; ++r) smax = fmaxf(smax, s[r]);
+    smax = fmaxf(smax, __shfl_xor(smax, 32, WAVE));  // partner combine
+    float m_new = fmaxf(m_run, smax);
+    float rs = __expf(m_run - m_new);
+    float psum = 0.f;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      s[r] = __expf(s[r] - m_new);
+      psum += s[r];
+    }
+    psum += __shfl_xor(psum, 32, WAVE);
+    l_run = (kb == 0) ? psum : l_run * rs + psum;
+    m_run = m_new;
+    if (kb > 0 && __ballot(rs < 0.999999f) != 0ull) {
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) acc[dt][r] *= rs;
+    }
+
+    // ---- P → bf16 B-fragments in registers (cvt_pk + permlane32_swap) ----
+    // own regs pack kv pairs; half-exchange composes the 16-kv chunks:
+    //   chunk c frag words = [A'(pk01,pk45), A'(pk23,pk67), B'(same), B'(same)]
+    bf16x8 pfrag[2];
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int base = c * 8;
+      int pk01 = (int)cvt_pk_bf16(s[base + 0], s[base + 1]);
+      int pk23 = (int)cvt_pk_bf16(s[base + 2], s[base + 3]);
+      int pk45 = (int)cvt_pk_bf16(s[base + 4], s[base + 5]);
+      int pk67 = (int)cvt_pk_bf16(s[base + 6], s[base + 7]);
+      i32x2 x = __builtin_amdgcn_permlane32_swap(pk01, pk45, false, false);
+      i32x2 y = __builtin_amdgcn_permlane32_swap(pk23, pk67, false, false);
+      union {
+        int w[4];
+        bf16x8 v;
+      } u;
+      u.w[0] = x[0];
+      u.w[1] = y[0];
+      u.w[2] = x[1];
+      u.w[3] = y[1];
+      pfrag[c] = u.v;
+    }
+
+    // ---- O^T += V^T · P^T (V^T via tr16 hardware transpose reads) ----
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        // lane (within its 16-group: 4a+b) issues the source read for
+        // V[c*16 + hi5*8 + (l&15)>>2 .. +4][dt*32 + ((l>>4)&1)*16 + 4a]
+        int row0 = c * 16 + hi5 * 8 + ((l & 15) >> 2);
+        int dbase = dt * 32 + ((l >> 4) & 1) * 16 + ((l & 3) * 4);
+        bf16x4_v7 lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds_tr_ptr7)(
+            (char*)&Vs[0][0] + row0 * (KROW * 2) + kv_swz7(row0, dbase * 2)));
+        bf16x4_v7 hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds_tr_ptr7)(
+            (char*)&Vs[0][0] + (row0 + 4) * (KROW * 2) +
+            kv_swz7(row0 + 4, dbase * 2)));
+        bf16x8 vfrag;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          vfrag[j] = lo[j];
+          vfrag[j + 4] = hi[j];
+        }
+        __builtin_amdgcn_s_setprio(1);
+        acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfrag, pfrag[c],
+                                                          acc[dt], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+      }
+    }
+  }
+
+  // ---- epilogue: O[q][d] = O^T[d][q] / l ----
+  if (my_q < Sq) {
+    float inv = l_run > 0.f ? 1.f / l_run : 0.f;
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int d = dt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi5;
+        O[ooff + (long long)my_q * st.os + d] = f2bf(acc[dt][r] * inv);
+      }
+  }
+}
+
+extern "C" void fa32_fwd_strided_bf16(
+    const void* q, const void* k, const void* v, void* o, int B, int Hq,
+    int Hkv, int Sq, int Sk, int D, float scale, int causal,
+    const long long* strides, hipStream_t stream) {
+  const short* Qp = (const short*)q;
+  const short* Kp = (const short*)k;
+  const short* Vp = (const short*)v;
+  short* Op = (short*)o;
+  FA32Strides st;
+  st.qb = strides[0]; st.qh = strides[1]; st.qs = strides[2];
+  st.kb = strides[3]; st.kh = strides[4]; st.ks = strides[5];
+  st.vb = strides[6]; st.vh = strides[7]; st.vs = strides[8];
+  st.ob = strides[9]; st.oh = strides[10]; st.os = strides[11];
+  dim3 grid((Sq + 127) / 128, Hq, B);
+  dim3 block(FA32_NWAVES * WAVE);
+#define L32(DD, CC)                                                           \
+  hipLaunchKernelGGL((fa32_kernel<DD, CC>), grid, block, 0, stream, Qp, Kp,   \
+                     Vp, Op, B, Hq, Hkv, Sq, Sk, scale, st)
+  if (D == 64) {
+    if (causal) L32(64, true); else L32(64, false);
+  } else if (D == 128) {
+    if (causal) L32(128, true); else L32(128, false);
+  } else {
+    fprintf(stderr, "fa32: unsupported head_dim %d\n", D);
+    abort();
+  }
+#undef L32
+}
