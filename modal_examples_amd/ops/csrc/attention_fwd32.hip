@@ -101,26 +101,42 @@ __global__ __launch_bounds__(FA32_NWAVES * WAVE) void fa32_kernel(
     if (lim < nkb) nkb = lim;
   }
 
-  for (int kb = 0; kb < nkb; ++kb) {
-    __syncthreads();
-    {
-      constexpr int CPR = D / 8;
-      constexpr int NCH = FA32_KVBLK * CPR;
-      for (int ci = tid; ci < NCH; ci += FA32_NWAVES * WAVE) {
-        int row = ci / CPR, c8 = ci % CPR;
-        int kvp = kb * FA32_KVBLK + row;
-        bf16x8 kv8 = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-        bf16x8 vv8 = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-        if (kvp < Sk) {
-          kv8 = *(const bf16x8*)&K[koff + kvp * st.ks + c8 * 8];
-          vv8 = *(const bf16x8*)&V[voff + kvp * st.vs + c8 * 8];
-        }
-        int boff = kv_swz7(row, c8 * 16);
-        *(bf16x8*)((char*)&Ks[row][0] + boff) = kv8;
-        *(bf16x8*)((char*)&Vs[row][0] + boff) = vv8;
+  // async register staging: next block loads issue before this block computes
+  constexpr int CPR = D / 8;
+  constexpr int CPT = (FA32_KVBLK * CPR) / (FA32_NWAVES * WAVE);
+  bf16x8 kreg[CPT], vreg[CPT];
+  auto load_chunks = [&](int kb) {
+#pragma unroll
+    for (int i = 0; i < CPT; ++i) {
+      int ci = i * FA32_NWAVES * WAVE + tid;
+      int row = ci / CPR, c8 = ci % CPR;
+      int kvp = kb * FA32_KVBLK + row;
+      if (kvp < Sk) {
+        kreg[i] = *(const bf16x8*)&K[koff + kvp * st.ks + c8 * 8];
+        vreg[i] = *(const bf16x8*)&V[voff + kvp * st.vs + c8 * 8];
+      } else {
+        kreg[i] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+        vreg[i] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
       }
     }
+  };
+  auto write_chunks = [&]() {
+#pragma unroll
+    for (int i = 0; i < CPT; ++i) {
+      int ci = i * FA32_NWAVES * WAVE + tid;
+      int row = ci / CPR, c8 = ci % CPR;
+      int boff = kv_swz7(row, c8 * 16);
+      *(bf16x8*)((char*)&Ks[row][0] + boff) = kreg[i];
+      *(bf16x8*)((char*)&Vs[row][0] + boff) = vreg[i];
+    }
+  };
+
+  load_chunks(0);
+  for (int kb = 0; kb < nkb; ++kb) {
     __syncthreads();
+    write_chunks();
+    __syncthreads();
+    if (kb + 1 < nkb) load_chunks(kb + 1);
 
     // ---- S^T = K · Q^T ----
     f32x16 s;
