@@ -78,13 +78,21 @@ __global__ __launch_bounds__(FA32_NWAVES * WAVE) void fa32_kernel(
   const int my_q = q0 + l31;  // this lane's query row
   const int causal_off = Sk - Sq;
 
-  // Q^T B-fragments: qreg[c][j] = Q[my_q][c*16 + hi5*8 + j]
-  bf16x8 qreg[DCH];
+  // Q^T B-fragments: qreg[c][j] = Q[my_q][c*16 + hi5*8 + j].
+  // D=128 would spend 32 VGPR on Q (dropping occupancy to 2 waves/SIMD);
+  // since Q is re-read once per KV block and stays L2-resident, the large-D
+  // path reloads fragments from global inside the QK loop instead.
+  constexpr bool QREG = (D <= 64);
+  bf16x8 qreg[QREG ? DCH : 1];
+  const long long qrow_off = qoff + (long long)my_q * st.qs;
+  const bool q_ok = my_q < Sq;
+  if (QREG) {
 #pragma unroll
-  for (int c = 0; c < DCH; ++c) {
-    qreg[c] = (my_q < Sq)
-        ? *(const bf16x8*)&Q[qoff + (long long)my_q * st.qs + c * 16 + hi5 * 8]
-        : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    for (int c = 0; c < (QREG ? DCH : 1); ++c) {
+      qreg[c] = q_ok
+          ? *(const bf16x8*)&Q[qrow_off + c * 16 + hi5 * 8]
+          : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
   }
 
   f32x16 acc[DT];
@@ -147,8 +155,15 @@ __global__ __launch_bounds__(FA32_NWAVES * WAVE) void fa32_kernel(
       // A = K[kv][d]: lane holds K[l31][c*16 + hi5*8 + j]
       bf16x8 kfrag = *(const bf16x8*)(
           (char*)&Ks[l31][0] + kv_swz7(l31, (c * 16 + hi5 * 8) * 2));
+      bf16x8 qf;
+      if (QREG) {
+        qf = qreg[c % (QREG ? DCH : 1)];
+      } else {
+        qf = q_ok ? *(const bf16x8*)&Q[qrow_off + c * 16 + hi5 * 8]
+                  : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      }
       __builtin_amdgcn_s_setprio(1);
-      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qreg[c], s, 0, 0, 0);
+      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qf, s, 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
     }
 
