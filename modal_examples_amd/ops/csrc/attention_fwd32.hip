@@ -82,7 +82,9 @@ __global__ __launch_bounds__(FA32_NWAVES * WAVE) void fa32_kernel(
   // D=128 would spend 32 VGPR on Q (dropping occupancy to 2 waves/SIMD);
   // since Q is re-read once per KV block and stays L2-resident, the large-D
   // path reloads fragments from global inside the QK loop instead.
-  constexpr bool QREG = (D <= 64);
+  // measured: reloading Q from L2 regressed D=128 22-30% (occupancy stayed
+  // at 2 waves; the extra global traffic was pure cost) — registers for all D
+  constexpr bool QREG = true;
   bf16x8 qreg[QREG ? DCH : 1];
   const long long qrow_off = qoff + (long long)my_q * st.qs;
   const bool q_ok = my_q < Sq;
