@@ -18,6 +18,27 @@ app = modal.App("example-lora-finetune")
 weights = modal.Volume.from_name("lora-weights", create_if_missing=True)
 
 
+@app.function(gpu="mi355x:8", timeout=3600)
+def train_distributed(max_steps: int = 3) -> str:
+    """DP over every visible GPU: the reference's `accelerate launch`
+    subprocess pattern (double process boundary) becomes torchrun — one rank
+    per GPU, bucketed RCCL all-reduce inside the trainer."""
+    import subprocess
+    import sys
+
+    import torch
+
+    n = max(1, torch.cuda.device_count())
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           f"--nproc-per-node={n}", "--master-addr", "127.0.0.1",
+           "--master-port", "29561", "scripts/bench_train.py",
+           "--steps", str(max_steps), "--warmup", "0"]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=3000)
+    print(r.stdout[-800:])
+    assert r.returncode == 0, r.stderr[-800:]
+    return f"trained on {n} rank(s)"
+
+
 @app.function(gpu="mi355x", timeout=3600)
 def train(max_steps: int = 3) -> str:
     import torch
