@@ -56,10 +56,8 @@ __global__ __launch_bounds__(FA32_NWAVES * WAVE) void fa32_kernel(
   constexpr int DT = D / 32;   // PV d-tiles (M-dim 32)
   constexpr int QBLK = 32 * FA32_NWAVES;
 
-  // double-buffered tiles: stage block kb+1 while computing kb — ONE
-  // barrier per block (write lands in the idle buffer)
-  __shared__ alignas(16) short Ks[2][FA32_KVBLK][KROW];
-  __shared__ alignas(16) short Vs[2][FA32_KVBLK][KROW];
+  __shared__ alignas(16) short Ks[FA32_KVBLK][KROW];
+  __shared__ alignas(16) short Vs[FA32_KVBLK][KROW];
 
   const int tid = threadIdx.x;
   const int w = tid / WAVE;
@@ -132,22 +130,22 @@ __global__ __launch_bounds__(FA32_NWAVES * WAVE) void fa32_kernel(
       }
     }
   };
-  auto write_chunks = [&](int buf) {
+  auto write_chunks = [&]() {
 #pragma unroll
     for (int i = 0; i < CPT; ++i) {
       int ci = i * FA32_NWAVES * WAVE + tid;
       int row = ci / CPR, c8 = ci % CPR;
       int boff = kv_swz7(row, c8 * 16);
-      *(bf16x8*)((char*)&Ks[buf][row][0] + boff) = kreg[i];
-      *(bf16x8*)((char*)&Vs[buf][row][0] + boff) = vreg[i];
+      *(bf16x8*)((char*)&Ks[row][0] + boff) = kreg[i];
+      *(bf16x8*)((char*)&Vs[row][0] + boff) = vreg[i];
     }
   };
 
   load_chunks(0);
-  write_chunks(0);
-  __syncthreads();
   for (int kb = 0; kb < nkb; ++kb) {
-    const int buf = kb & 1;
+    __syncthreads();
+    write_chunks();
+    __syncthreads();
     if (kb + 1 < nkb) load_chunks(kb + 1);
 
     // ---- S^T = K · Q^T ----
@@ -158,7 +156,7 @@ __global__ __launch_bounds__(FA32_NWAVES * WAVE) void fa32_kernel(
     for (int c = 0; c < DCH; ++c) {
       // A = K[kv][d]: lane holds K[l31][c*16 + hi5*8 + j]
       bf16x8 kfrag = *(const bf16x8*)(
-          (char*)&Ks[buf][l31][0] + kv_swz7(l31, (c * 16 + hi5 * 8) * 2));
+          (char*)&Ks[l31][0] + kv_swz7(l31, (c * 16 + hi5 * 8) * 2));
       bf16x8 qf;
       if (QREG) {
         qf = qreg[c % (QREG ? DCH : 1)];
@@ -236,10 +234,9 @@ __global__ __launch_bounds__(FA32_NWAVES * WAVE) void fa32_kernel(
         int row0 = c * 16 + hi5 * 8 + ((l & 15) >> 2);
         int dbase = dt * 32 + ((l >> 4) & 1) * 16 + ((l & 3) * 4);
         bf16x4_v7 lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds_tr_ptr7)(
-            (char*)&Vs[buf][0][0] + row0 * (KROW * 2) +
-            kv_swz7(row0, dbase * 2)));
+            (char*)&Vs[0][0] + row0 * (KROW * 2) + kv_swz7(row0, dbase * 2)));
         bf16x4_v7 hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds_tr_ptr7)(
-            (char*)&Vs[buf][0][0] + (row0 + 4) * (KROW * 2) +
+            (char*)&Vs[0][0] + (row0 + 4) * (KROW * 2) +
             kv_swz7(row0 + 4, dbase * 2)));
         bf16x8 vfrag;
 #pragma unroll
@@ -253,9 +250,6 @@ __global__ __launch_bounds__(FA32_NWAVES * WAVE) void fa32_kernel(
         __builtin_amdgcn_s_setprio(0);
       }
     }
-
-    if (kb + 1 < nkb) write_chunks(buf ^ 1);
-    __syncthreads();  // next tile staged AND this tile's readers done
   }
 
   // ---- epilogue: O[q][d] = O^T[d][q] / l ----
