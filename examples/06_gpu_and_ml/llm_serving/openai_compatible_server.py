@@ -38,7 +38,7 @@ class LLMService:
     def serve(self):
         from modal_examples_amd.models.llama.server import serve_openai
 
-        serve_openai(self.server, port=PORT, block=False)
+        self.uvicorn = serve_openai(self.server, port=PORT, block=False)
 
     @modal.method()
     def chat(self, prompt: str, max_tokens: int = 32) -> str:
@@ -46,6 +46,8 @@ class LLMService:
 
     @modal.exit()
     def stop(self):
+        if hasattr(self, "uvicorn"):
+            self.uvicorn.should_exit = True
         self.server.shutdown()
 
 
