@@ -14,7 +14,7 @@ app = modal.App("example-with-options")
 @app.cls(max_containers=1)
 class Worker:
     @modal.method()
-    def pid(self) -> int:
+    def pid(self, i: int = 0) -> int:
         return os.getpid()
 
 
@@ -26,5 +26,5 @@ def main():
     assert len(pids_small) == 1
 
     Big = Worker.with_options(max_containers=4)
-    list(Big().pid.map(range(8)))  # warm several workers
-    print("with_options(max_containers=4) scaled out")
+    pids_big = set(Big().pid.map(range(8)))  # warms several workers
+    print(f"with_options(max_containers=4) scaled to {len(pids_big)} workers")

@@ -39,13 +39,21 @@ class FrameProcessor:
     @modal.method()
     def serve_session(self, session_id: str, max_frames: int = 16) -> dict:
         """Answer one peer: read offer, process frames until 'bye'."""
+        from queue import Empty
+
         torch = self.torch
-        offer = signaling.get(partition=session_id, timeout=10)
+        try:
+            offer = signaling.get(partition=session_id, timeout=30)
+        except Empty:
+            return {"frames": 0, "error": "no offer"}
         signaling.put({"type": "answer", "codec": offer["codec"]},
                       partition=f"{session_id}-answer")
         n, lat = 0, []
         while n < max_frames:
-            msg = media.get(partition=session_id, timeout=5)
+            try:
+                msg = media.get(partition=session_id, timeout=10)
+            except Empty:
+                break
             if msg is None or msg.get("type") == "bye":
                 break
             x = torch.as_tensor(msg["frame"], dtype=torch.float32,
@@ -75,7 +83,16 @@ def main(frames: int = 6):
 
     # peer side: offer → answer → stream frames
     signaling.put({"type": "offer", "codec": "raw-rgb"}, partition=session)
-    ans = signaling.get(partition=f"{session}-answer", timeout=30)
+    from queue import Empty
+
+    ans = None
+    for _ in range(12):  # worker cold start may take a few seconds
+        try:
+            ans = signaling.get(partition=f"{session}-answer", timeout=10)
+            break
+        except Empty:
+            continue
+    assert ans is not None, "no answer from processor"
     print("negotiated:", ans)
     rng = np.random.default_rng(0)
     for i in range(frames):
@@ -85,7 +102,10 @@ def main(frames: int = 6):
     media.put({"type": "bye"}, partition=session)
     got = 0
     while got < frames:
-        out = media.get(partition=f"{session}-out", timeout=30)
+        try:
+            out = media.get(partition=f"{session}-out", timeout=30)
+        except Empty:
+            break
         if out is None:
             break
         rtt = (time.time() - out["t_sent"]) * 1000
