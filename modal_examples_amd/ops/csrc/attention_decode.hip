@@ -83,16 +83,30 @@ __global__ __launch_bounds__(DEC_THREADS) void paged_decode_kernel(
   }
 
   // ---- sweep this split's KV rows of (b, hkv) ----
-  for (int kv = kv_lo + part; kv < S; kv += NPART) {
-    long long row_off;
+  // software-pipelined: the NEXT row's K/V loads issue before computing the
+  // current row, hiding the ~300-cycle HBM latency under the dot products
+  auto row_addr = [&](int kv) -> long long {
     if (block_table != nullptr) {
       int blk = block_table[(long long)b * max_blocks + kv / block_size];
-      row_off = (((long long)blk * Hkv + hkv) * block_size + kv % block_size) * D;
-    } else {
-      row_off = (((long long)b * Hkv + hkv) * (long long)block_size + kv) * D;
+      return (((long long)blk * Hkv + hkv) * block_size + kv % block_size) * D;
     }
-    bf16x8 k8 = *(const bf16x8*)&Kc[row_off + slot * 8];
-    bf16x8 v8 = *(const bf16x8*)&Vc[row_off + slot * 8];
+    return (((long long)b * Hkv + hkv) * (long long)block_size + kv) * D;
+  };
+  bf16x8 k8 = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+  bf16x8 v8 = k8;
+  int kv0 = kv_lo + part;
+  if (kv0 < S) {
+    long long off0 = row_addr(kv0);
+    k8 = *(const bf16x8*)&Kc[off0 + slot * 8];
+    v8 = *(const bf16x8*)&Vc[off0 + slot * 8];
+  }
+  for (int kv = kv0; kv < S; kv += NPART) {
+    bf16x8 k8n, v8n;
+    if (kv + NPART < S) {
+      long long offn = row_addr(kv + NPART);
+      k8n = *(const bf16x8*)&Kc[offn + slot * 8];
+      v8n = *(const bf16x8*)&Vc[offn + slot * 8];
+    }
     float kfl[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) kfl[j] = bf2f(k8[j]);
@@ -115,6 +129,8 @@ __global__ __launch_bounds__(DEC_THREADS) void paged_decode_kernel(
 #pragma unroll
       for (int j = 0; j < 8; ++j) accv[g][j] = accv[g][j] * rs + p * bf2f(v8[j]);
     }
+    k8 = k8n;
+    v8 = v8n;
   }
 
   // ---- merge partials through LDS ----
