@@ -92,27 +92,21 @@ __global__ __launch_bounds__(DEC_THREADS) void paged_decode_kernel(
     }
     return (((long long)b * Hkv + hkv) * (long long)block_size + kv) * D;
   };
-  bf16x8 kpipe[2], vpipe[2];
-  kpipe[0] = kpipe[1] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-  vpipe[0] = vpipe[1] = kpipe[0];
+  bf16x8 k8 = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+  bf16x8 v8 = k8;
   int kv0 = kv_lo + part;
-#pragma unroll
-  for (int pf = 0; pf < 2; ++pf) {
-    if (kv0 + pf * NPART < S) {
-      long long off0 = row_addr(kv0 + pf * NPART);
-      kpipe[pf] = *(const bf16x8*)&Kc[off0 + slot * 8];
-      vpipe[pf] = *(const bf16x8*)&Vc[off0 + slot * 8];
-    }
+  if (kv0 < S) {
+    long long off0 = row_addr(kv0);
+    k8 = *(const bf16x8*)&Kc[off0 + slot * 8];
+    v8 = *(const bf16x8*)&Vc[off0 + slot * 8];
   }
-  int ph = 0;
   for (int kv = kv0; kv < S; kv += NPART) {
-    bf16x8 k8 = kpipe[ph], v8 = vpipe[ph];
-    if (kv + 2 * NPART < S) {
-      long long offn = row_addr(kv + 2 * NPART);
-      kpipe[ph] = *(const bf16x8*)&Kc[offn + slot * 8];
-      vpipe[ph] = *(const bf16x8*)&Vc[offn + slot * 8];
+    bf16x8 k8n, v8n;
+    if (kv + NPART < S) {
+      long long offn = row_addr(kv + NPART);
+      k8n = *(const bf16x8*)&Kc[offn + slot * 8];
+      v8n = *(const bf16x8*)&Vc[offn + slot * 8];
     }
-    ph ^= 1;
     float kfl[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) kfl[j] = bf2f(k8[j]);
@@ -135,6 +129,8 @@ __global__ __launch_bounds__(DEC_THREADS) void paged_decode_kernel(
 #pragma unroll
       for (int j = 0; j < 8; ++j) accv[g][j] = accv[g][j] * rs + p * bf2f(v8[j]);
     }
+    k8 = k8n;
+    v8 = v8n;
   }
 
   // ---- merge partials through LDS ----
