@@ -92,20 +92,28 @@ __global__ __launch_bounds__(DEC_THREADS) void paged_decode_kernel(
     }
     return (((long long)b * Hkv + hkv) * (long long)block_size + kv) * D;
   };
-  bf16x8 k8 = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-  bf16x8 v8 = k8;
+  // depth-2 pipeline with NAMED registers (a runtime-indexed ring spills to
+  // scratch — measured 3.5x slower; guide rule #20)
+  bf16x8 kA = bf16x8{0, 0, 0, 0, 0, 0, 0, 0}, vA = kA, kB = kA, vB = kA;
   int kv0 = kv_lo + part;
   if (kv0 < S) {
-    long long off0 = row_addr(kv0);
-    k8 = *(const bf16x8*)&Kc[off0 + slot * 8];
-    v8 = *(const bf16x8*)&Vc[off0 + slot * 8];
+    long long o0 = row_addr(kv0);
+    kA = *(const bf16x8*)&Kc[o0 + slot * 8];
+    vA = *(const bf16x8*)&Vc[o0 + slot * 8];
+  }
+  if (kv0 + NPART < S) {
+    long long o1 = row_addr(kv0 + NPART);
+    kB = *(const bf16x8*)&Kc[o1 + slot * 8];
+    vB = *(const bf16x8*)&Vc[o1 + slot * 8];
   }
   for (int kv = kv0; kv < S; kv += NPART) {
-    bf16x8 k8n, v8n;
-    if (kv + NPART < S) {
-      long long offn = row_addr(kv + NPART);
-      k8n = *(const bf16x8*)&Kc[offn + slot * 8];
-      v8n = *(const bf16x8*)&Vc[offn + slot * 8];
+    bf16x8 k8 = kA, v8 = vA;
+    kA = kB;
+    vA = vB;
+    if (kv + 2 * NPART < S) {
+      long long offn = row_addr(kv + 2 * NPART);
+      kB = *(const bf16x8*)&Kc[offn + slot * 8];
+      vB = *(const bf16x8*)&Vc[offn + slot * 8];
     }
     float kfl[8];
 #pragma unroll
