@@ -52,6 +52,13 @@ class Volume:
 
         return ctx()
 
+    def read_only(self) -> "Volume":
+        """A read-only view for mounting (reference restricted-volume usage,
+        08_advanced).  Write APIs raise ``InvalidError``; workers additionally
+        enforce it at the filesystem level by bind-mounting the volume
+        read-only inside a private mount namespace (runtime/worker.py)."""
+        return _ReadOnlyVolume(self)
+
     def commit(self):
         """Flush writes so other workers observe them (fsync the tree)."""
         d = os.open(self.path, os.O_RDONLY)
@@ -122,6 +129,40 @@ class Volume:
         return _Batch()
 
 
+class _ReadOnlyVolume(Volume):
+    """Read-only view over a Volume.  ``name`` carries an ``ro:`` prefix that
+    the worker mount logic interprets as "bind-mount read-only"."""
+
+    def __init__(self, base: Volume):  # noqa: super-init-not-called (view)
+        self._base = base
+        self.version = base.version
+
+    @property
+    def name(self) -> str:  # type: ignore[override]
+        return f"ro:{self._base.name}"
+
+    @property
+    def path(self) -> Path:
+        return self._base.path
+
+    def read_only(self) -> "Volume":
+        return self
+
+    def _refuse(self):
+        from ..exception import InvalidError
+
+        raise InvalidError(f"volume {self._base.name!r} is mounted read-only")
+
+    def commit(self):
+        self._refuse()
+
+    def remove_file(self, path: str, recursive: bool = False):
+        self._refuse()
+
+    def batch_upload(self):
+        self._refuse()
+
+
 class CloudBucketMount:
     """S3/GCS bucket mount shim (10_integrations/s3_bucket_mount.py:66): no
     network locally, so this maps to a named local directory that tests can
@@ -137,4 +178,5 @@ class CloudBucketMount:
 
     @property
     def name(self) -> str:
-        return f"bucket:{self.bucket_name}"
+        base = f"bucket:{self.bucket_name}"
+        return f"ro:{base}" if self.read_only else base

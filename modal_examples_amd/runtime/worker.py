@@ -23,18 +23,58 @@ from concurrent.futures import ThreadPoolExecutor
 from . import ipc
 
 
+def _bind_read_only(pairs: list) -> bool:
+    """Enforce read-only mounts at the filesystem level: unshare this worker's
+    mount namespace (mounts stay private to the process) and bind-mount each
+    volume directory read-only at its mount path.  Returns False if the kernel
+    refuses (no CAP_SYS_ADMIN) — callers fall back to a plain symlink with
+    API-level enforcement only.  Must run before any threads start:
+    unshare(CLONE_NEWNS) fails EINVAL in a multithreaded process."""
+    import ctypes
+
+    try:
+        libc = ctypes.CDLL("libc.so.6", use_errno=True)
+        CLONE_NEWNS = 0x00020000
+        MS_RDONLY, MS_BIND, MS_REC, MS_REMOUNT, MS_PRIVATE = 1, 0x1000, 0x4000, 32, 0x40000
+        if libc.unshare(CLONE_NEWNS) != 0:
+            return False
+        if libc.mount(b"none", b"/", None, MS_REC | MS_PRIVATE, None) != 0:
+            return False
+        for mount, target in pairs:
+            os.makedirs(mount, exist_ok=True)
+            m, t = str(mount).encode(), str(target).encode()
+            if libc.mount(t, m, None, MS_BIND, None) != 0:
+                return False
+            if libc.mount(b"none", m, None, MS_BIND | MS_REMOUNT | MS_RDONLY, None) != 0:
+                return False
+        return True
+    except OSError:
+        return False
+
+
 def _mount_volumes(volumes: dict) -> None:
     """Symlink mount paths to the shared volume directories (the worker runs
-    as root on this node — mirrors containers mounting at /cache etc.)."""
+    as root on this node — mirrors containers mounting at /cache etc.).
+    Names carrying an ``ro:`` prefix (Volume.read_only()) get a read-only
+    bind mount in a private mount namespace instead of a symlink."""
     from .. import config
 
+    ro_pairs = []
     for mount, name in (volumes or {}).items():
+        ro = name.startswith("ro:")
+        if ro:
+            name = name[3:]
         if name.startswith("bucket:"):
             target = config.state_dir() / "buckets" / name.split(":", 1)[1]
         else:
             target = config.state_dir() / "volumes" / name
         target.mkdir(parents=True, exist_ok=True)
         try:
+            if ro:
+                if os.path.islink(mount):
+                    continue  # an earlier symlink owns this path
+                ro_pairs.append((mount, target))  # bind over dir is fine
+                continue
             if os.path.islink(mount) or os.path.exists(mount):
                 continue
             parent = os.path.dirname(mount.rstrip("/"))
@@ -43,6 +83,13 @@ def _mount_volumes(volumes: dict) -> None:
             os.symlink(target, mount)
         except OSError:
             pass  # unmountable path: functions can still use volume.path
+    if ro_pairs and not _bind_read_only(ro_pairs):
+        for mount, target in ro_pairs:  # fallback: API-level enforcement only
+            try:
+                if not os.path.exists(mount):
+                    os.symlink(target, mount)
+            except OSError:
+                pass
 
 
 def _apply_env(spec_env: dict, gpu_devices: tuple) -> None:

@@ -187,3 +187,58 @@ def test_image_env_reaches_worker():
         return os.environ.get("IMG_LAYER_VAR")
 
     assert read_env.remote() == "layered"
+
+
+def test_volume_read_only_api():
+    """Volume.read_only(): write APIs raise, reads still work."""
+    vol = modal.Volume.from_name("ro-api-vol", create_if_missing=True)
+    (vol.path / "d.txt").write_text("ro")
+    ro = vol.read_only()
+    assert ro.read_file("d.txt") == b"ro"
+    assert "d.txt" in ro.listdir("/")
+    from modal_examples_amd.exception import InvalidError
+
+    with pytest.raises(InvalidError):
+        ro.commit()
+    with pytest.raises(InvalidError):
+        ro.remove_file("d.txt")
+    with pytest.raises(InvalidError):
+        ro.batch_upload()
+    assert ro.read_only() is ro
+    modal.Volume.delete("ro-api-vol")
+
+
+def test_volume_read_only_mount_enforced_in_worker():
+    """A read-only mount is a ro bind mount in the worker's private mount
+    namespace: raw file writes fail with EROFS, reads work, and the parent
+    namespace (this process) never sees the mount."""
+    vol = modal.Volume.from_name("ro-mount-vol", create_if_missing=True)
+    (vol.path / "probe.txt").write_text("ro-mounted")
+    app = modal.App("test-ro-mount-app")
+    mnt = "/tmp/mxa_ro_mount_test"
+
+    @app.function(volumes={mnt: vol.read_only()})
+    def probe():
+        import errno
+
+        with open(f"{mnt}/probe.txt") as f:
+            content = f.read()
+        try:
+            open(f"{mnt}/new.txt", "w")
+            write = "allowed"
+        except OSError as e:
+            write = "EROFS" if e.errno == errno.EROFS else f"errno={e.errno}"
+        return content, write
+
+    content, write = probe.remote()
+    assert content == "ro-mounted"
+    assert write == "EROFS"
+    # volume itself untouched and still writable from here
+    assert not (vol.path / "new.txt").exists()
+    (vol.path / "after.txt").write_text("ok")
+    modal.Volume.delete("ro-mount-vol")
+    import os
+    import shutil
+
+    if os.path.isdir(mnt) and not os.path.islink(mnt):
+        shutil.rmtree(mnt, ignore_errors=True)
