@@ -84,8 +84,12 @@ class Image:
 
     def run_function(self, fn: Callable, gpu=None, volumes=None, secrets=None, **kw) -> "Image":
         """Build-time function execution (llm-serving/sglang_snapshot.py:145-149).
-        Locally: executed once, lazily, on first worker use of this image."""
-        return self._with("run_function", fn)
+        Runs once, lazily, on first use of this image — in a WORKER process
+        with the requested gpu/volumes/secrets (reference semantics: build
+        steps execute in a container with those resources, e.g. weight
+        downloads onto a volume), not in the client process."""
+        return self._with("run_function",
+                          (fn, {"gpu": gpu, "volumes": volumes, "secrets": secrets}))
 
     def pip_install_from_requirements(self, path, **kw) -> "Image":
         return self._with("pip_req", str(path))
@@ -117,10 +121,24 @@ class Image:
         return dict(self._env)
 
     def build(self):
-        """Execute run_function layers (once per process)."""
-        for fn in self._build_fns:
+        """Execute run_function layers (once per process), each in a one-shot
+        worker carrying the layer's gpu/volumes/secrets request."""
+        for entry in self._build_fns:
+            fn, opts = entry if isinstance(entry, tuple) else (entry, {})
             key = f"_built_{id(fn)}"
-            if not getattr(self, key, False):
-                fn()
-                setattr(self, key, True)
+            if getattr(self, key, False):
+                continue
+            self._run_build_fn(fn, opts or {})
+            setattr(self, key, True)
         return self
+
+    def _run_build_fn(self, fn: Callable, opts: dict) -> None:
+        from ..app import App  # lazy: resources must not import app at module load
+
+        app = App(f"image-build-{self.content_hash()}")
+        builder = app.function(
+            gpu=opts.get("gpu"),
+            volumes=opts.get("volumes") or {},
+            secrets=opts.get("secrets") or [],
+        )(fn)
+        builder.remote()

@@ -242,3 +242,34 @@ def test_volume_read_only_mount_enforced_in_worker():
 
     if os.path.isdir(mnt) and not os.path.islink(mnt):
         shutil.rmtree(mnt, ignore_errors=True)
+
+
+def test_image_run_function_runs_in_worker():
+    """Image.run_function build steps execute in a worker process (with the
+    requested volumes), not in the client (reference: build-time weight
+    downloads run inside a container)."""
+    import os
+
+    vol = modal.Volume.from_name("img-build-vol", create_if_missing=True)
+
+    def download_weights():
+        v = modal.Volume.from_name("img-build-vol")
+        (v.path / "weights.bin").write_bytes(b"W" * 16)
+        (v.path / "pid.txt").write_text(str(os.getpid()))
+        v.commit()
+
+    img = modal.Image.debian_slim().run_function(
+        download_weights, volumes={"/tmp/mxa_imgbuild_vol": vol})
+    app = modal.App("test-img-build", image=img)
+
+    @app.function()
+    def use_weights():
+        v = modal.Volume.from_name("img-build-vol")
+        return v.read_file("weights.bin")
+
+    assert use_weights.remote() == b"W" * 16
+    # the build step ran in a different process than this client
+    assert (vol.path / "pid.txt").read_text() != str(os.getpid())
+    modal.Volume.delete("img-build-vol")
+    if os.path.islink("/tmp/mxa_imgbuild_vol"):
+        os.unlink("/tmp/mxa_imgbuild_vol")
