@@ -312,6 +312,7 @@ class FunctionOptions:
     buffer_containers: int = 0
     scaledown_window: Optional[float] = None
     single_use_containers: bool = False
+    sticky: bool = False  # web sessions pin to one container (Modal Server flag)
     schedule: Any = None
     enable_memory_snapshot: bool = False
     experimental_options: dict = field(default_factory=dict)
@@ -471,11 +472,12 @@ class Function:
 
     # -- verbs --
 
-    def _submit(self, args, kwargs, is_gen=False, spawned=False) -> Call:
+    def _submit(self, args, kwargs, is_gen=False, spawned=False, sticky_key=None) -> Call:
         pool = self._get_pool()
         if self.is_batched:
             return self._batcher.enqueue(args, kwargs)
-        return pool.submit("", args, kwargs, is_gen=is_gen, spawned=spawned)
+        return pool.submit("", args, kwargs, is_gen=is_gen, spawned=spawned,
+                           sticky_key=sticky_key)
 
     def _remote(self, *args, **kwargs):
         if self.is_generator:
@@ -741,7 +743,7 @@ class _BoundMethod:
         self.for_each = _Verb(self._for_each)
         self.remote_gen = _Verb(self._remote_gen, _agen_from_sync(self._remote_gen))
 
-    def _submit(self, args, kwargs, is_gen=False, spawned=False) -> Call:
+    def _submit(self, args, kwargs, is_gen=False, spawned=False, sticky_key=None) -> Call:
         pool = self.obj._get_pool()
         if self.is_batched:
             if self._batcher is None:
@@ -749,7 +751,8 @@ class _BoundMethod:
                                                int(self.flags["max_batch_size"]),
                                                float(self.flags["wait_ms"]))
             return self._batcher.enqueue(args, kwargs)
-        return pool.submit(self.name, args, kwargs, is_gen=is_gen, spawned=spawned)
+        return pool.submit(self.name, args, kwargs, is_gen=is_gen, spawned=spawned,
+                           sticky_key=sticky_key)
 
     def _remote(self, *args, **kwargs):
         if self.is_generator:

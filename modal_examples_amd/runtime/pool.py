@@ -67,9 +67,11 @@ class Call:
         "spawned",
         "done",
         "t_submit",
+        "sticky_key",
     )
 
-    def __init__(self, method_name, args_blob, is_gen, timeout, retries: RetryPolicy, spawned=False):
+    def __init__(self, method_name, args_blob, is_gen, timeout, retries: RetryPolicy, spawned=False,
+                 sticky_key=None):
         self.id = "fc-" + uuid.uuid4().hex[:16]
         self.method_name = method_name
         self.args_blob = args_blob
@@ -86,6 +88,7 @@ class Call:
         self.spawned = spawned
         self.done = False
         self.t_submit = time.monotonic()
+        self.sticky_key = sticky_key
 
     def wait(self, timeout: Optional[float] = None):
         if not self.event.wait(timeout):
@@ -174,6 +177,7 @@ class Pool:
         self.workers: Dict[int, WorkerHandle] = {}
         self.calls: Dict[str, Call] = {}
         self.pending: List[Call] = []
+        self.sticky: Dict[str, int] = {}  # session key -> worker id
         self.lock = threading.RLock()
         self.closed = False
         self._threads_started = False
@@ -183,7 +187,7 @@ class Pool:
     # ---------------- public API ----------------
 
     def submit(self, method_name, args, kwargs, is_gen=False, spawned=False,
-               timeout: Optional[float] = None) -> Call:
+               timeout: Optional[float] = None, sticky_key: Optional[str] = None) -> Call:
         call = Call(
             method_name,
             ipc.dumps((args, kwargs)),
@@ -191,6 +195,7 @@ class Pool:
             timeout if timeout is not None else self.timeout,
             self.retries,
             spawned,
+            sticky_key=sticky_key,
         )
         with self.lock:
             self._ensure_threads()
@@ -351,10 +356,29 @@ class Pool:
             if time.monotonic() > deadline and not self.workers:
                 raise ExecutionError(f"pool {self.name}: no worker available")
 
+    def _sticky_worker(self, key: str):
+        """(worker, wait): the worker bound to ``key``, or a fresh binding.
+        wait=True means the bound worker exists but is at capacity — the call
+        must queue for THAT worker (sticky semantics: same container serves
+        the whole session)."""
+        wid = self.sticky.get(key)
+        w = self.workers.get(wid) if wid is not None else None
+        if w is not None and w.ready and not w.shutting_down:
+            if len(w.inflight) < self.max_inputs_per_worker:
+                return w, False
+            return None, True
+        return self._pick_worker(), False  # no binding or stale: rebind below
+
     def _pump(self):
         """Dispatch pending calls onto ready workers; scale up when starved."""
         while self.pending:
-            w = self._pick_worker()
+            head = self.pending[0]
+            if head.sticky_key is not None:
+                w, wait = self._sticky_worker(head.sticky_key)
+                if wait:
+                    return  # re-pumped when the bound worker resolves a call
+            else:
+                w = self._pick_worker()
             if w is None:
                 started = self._maybe_scale_up()
                 if started is None:
@@ -366,6 +390,8 @@ class Pool:
                     and len(self.workers) < self.max_containers):
                 self._maybe_scale_up()
             call = self.pending.pop(0)
+            if call.sticky_key is not None:
+                self.sticky[call.sticky_key] = w.id
             self._dispatch(call, w)
         # buffer containers: keep `buffer` idle warm workers beyond demand
         if self.buffer_containers:

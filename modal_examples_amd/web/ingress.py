@@ -131,6 +131,7 @@ def build_ingress_app(app):
             # and Function objects hold locks
 
             needs_auth = bool(flags.get("requires_proxy_auth"))
+            is_sticky = bool(getattr(f.opts, "sticky", False))
 
             async def handler(request: Request):
                 if needs_auth and not _proxy_auth_ok(request):
@@ -152,7 +153,16 @@ def build_ingress_app(app):
                         if p.annotation in (int, float, bool) and isinstance(v, str):
                             v = p.annotation(v) if p.annotation is not bool else v.lower() in ("1", "true", "yes")
                         coerced[name] = v
-                result = await asyncio.to_thread(target.remote, **coerced)
+                if is_sticky:
+                    # Modal Server sticky routing: all requests of one client
+                    # session land on one container.  Session = Modal-Session
+                    # header, else the client address.
+                    key = request.headers.get("Modal-Session") or (
+                        request.client.host if request.client else "anon")
+                    result = await asyncio.to_thread(
+                        lambda: target._submit((), coerced, sticky_key=key).wait())
+                else:
+                    result = await asyncio.to_thread(target.remote, **coerced)
                 if hasattr(result, "__class__") and result.__class__.__name__ == "Response":
                     return result
                 if isinstance(result, (bytes, bytearray)):

@@ -176,3 +176,22 @@ def test_app_include():
 
     merged = modal.App("test-fn-merged").include(other)
     assert merged.functions["triple"].remote(4) == 12
+
+
+def test_sticky_key_routes_to_same_worker():
+    """Calls carrying the same sticky_key are all served by one worker even
+    when the pool is allowed to scale; the binding survives across calls."""
+    import os as _os
+
+    app2 = modal.App("test-sticky-pool")
+
+    @app2.function(max_containers=4)
+    def worker_pid(i: int = 0) -> int:
+        return _os.getpid()
+
+    same = {worker_pid._submit((i,), {}, sticky_key="sess-A").wait()
+            for i in range(5)}
+    assert len(same) == 1
+    a = worker_pid._submit((0,), {}, sticky_key="k1").wait()
+    b = worker_pid._submit((1,), {}, sticky_key="k1").wait()
+    assert a == b

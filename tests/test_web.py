@@ -111,3 +111,32 @@ def test_cls_endpoint_runs_enter():
 
 def test_web_url_label():
     assert "double" in double.get_web_url()
+
+
+app_sticky = modal.App("test-web-sticky")
+
+
+@app_sticky.function(sticky=True, max_containers=4)
+@modal.fastapi_endpoint(method="GET", label="whoami")
+def whoami():
+    import os
+
+    return {"pid": os.getpid()}
+
+
+def test_sticky_session_pins_worker():
+    """sticky=True: requests sharing a Modal-Session header are served by one
+    container (Modal Server sticky routing)."""
+    root = build_ingress_app(app_sticky)
+
+    async def go():
+        async with httpx.AsyncClient(transport=httpx.ASGITransport(app=root),
+                                     base_url="http://t") as c:
+            pids = set()
+            for _ in range(4):
+                r = await c.get("/whoami", headers={"Modal-Session": "s-1"})
+                assert r.status_code == 200
+                pids.add(r.json()["pid"])
+            assert len(pids) == 1
+
+    asyncio.run(go())
