@@ -133,8 +133,41 @@ def cmd_deploy(target: str, extra_args):
 
 
 def cmd_shell(target: str, extra_args):
+    """``shell file.py`` → local REPL over the module; ``shell
+    file.py::FuncOrCls`` → attach to a RUNNING container of that service
+    (warming one if needed) and exec lines inside the worker process, where
+    the live models/caches are loaded."""
     import code
 
+    if target and "::" in target:
+        path, name = target.split("::", 1)
+        mod = load_module(path)
+        app = find_app(mod)
+        fn = app.functions.get(name)
+        if fn is None and name in app.classes:
+            c = app.classes[name]()
+            fn = next(m for m in (getattr(c, n, None) for n in dir(c))
+                      if hasattr(m, "_submit"))
+        if fn is None:
+            raise SystemExit(f"no function or class {name!r} in {path}")
+        pool = fn._get_pool() if hasattr(fn, "_get_pool") else fn.obj._get_pool()
+        print(f"attached to a worker of {name} — `obj` is the live instance, "
+              "`methods` the service methods; blank line or Ctrl-D exits.")
+        while True:
+            try:
+                line = input(f"({name}) >>> ")
+            except EOFError:
+                break
+            if not line.strip():
+                break
+            try:
+                out = pool.submit("__debug_exec__", (line,), {},
+                                  sticky_key="__shell__").wait()
+                if out:
+                    print(out, end="" if out.endswith("\n") else "\n")
+            except BaseException as e:  # noqa: BLE001
+                print(f"{type(e).__name__}: {e}")
+        return
     mod = load_module(target) if target else None
     banner = "modal_examples_amd shell"
     ns = dict(vars(mod)) if mod else {}

@@ -158,9 +158,32 @@ def worker_main(worker_id: int, spec_blob: bytes, task_q, result_q) -> None:
 
     import inspect
 
+    # `modal shell` into THIS running container: exec code in a persistent
+    # namespace that can see the live service (loaded models, caches, ...).
+    debug_ns = {"methods": methods, "spec": spec, "os": os}
+    debug_ns["obj"] = getattr(  # the live class instance, for @app.cls services
+        next((m for m in methods.values() if m is not None), None), "__self__", None)
+
+    def _debug_exec(src: str):
+        import io
+        from contextlib import redirect_stdout
+
+        buf = io.StringIO()
+        with redirect_stdout(buf):
+            try:
+                value = eval(compile(src, "<shell>", "eval"), debug_ns)
+                if value is not None:
+                    print(repr(value))
+            except SyntaxError:
+                exec(compile(src, "<shell>", "exec"), debug_ns)
+        return buf.getvalue()
+
     def run_one(call_id: str, method_name: str, args_blob: bytes) -> None:
         try:
             args, kwargs = ipc.loads(args_blob)
+            if method_name == "__debug_exec__":
+                post(ipc.RESULT, call_id, ipc.dumps(_debug_exec(args[0])))
+                return
             fn = methods.get(method_name)
             if fn is None:
                 raise RuntimeError(f"no method {method_name!r} on service {spec.name}")

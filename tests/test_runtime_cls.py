@@ -141,3 +141,31 @@ def test_concurrent_overlaps():
 def test_with_options():
     M2 = Model.with_options(max_containers=2)
     assert M2().run.remote("w") == ">>loaded:w"
+
+
+def test_debug_exec_shell_into_running_worker():
+    """`modal shell app.py::Cls` path: __debug_exec__ runs code inside the
+    live worker with access to the service instance (`obj`)."""
+    app2 = modal.App("test-shell-attach")
+
+    @app2.cls()
+    class Svc:
+        @modal.enter()
+        def boot(self):
+            self.loaded = "model-v1"
+
+        @modal.method()
+        def f(self) -> int:
+            return 1
+
+    svc = Svc()
+    assert svc.f.remote() == 1  # worker is now running with state loaded
+    pool = svc.f.obj._get_pool()
+    out = pool.submit("__debug_exec__", ("obj.loaded",), {},
+                      sticky_key="__shell__").wait()
+    assert out.strip() == "'model-v1'"
+    # statements work too, and mutate the LIVE instance
+    pool.submit("__debug_exec__", ("obj.loaded = 'patched'",), {},
+                sticky_key="__shell__").wait()
+    assert pool.submit("__debug_exec__", ("print(obj.loaded)",), {},
+                       sticky_key="__shell__").wait().strip() == "patched"
