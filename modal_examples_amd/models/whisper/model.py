@@ -146,6 +146,28 @@ class WhisperModel(nn.Module):
 
     # ------------------------------------------------ decoder
 
+    def forward_train(self, mel: torch.Tensor, tokens: torch.Tensor) -> torch.Tensor:
+        """Teacher-forced training forward for ASR fine-tuning: grad-enabled,
+        full-sequence causal decoder over encoded audio.  Returns logits
+        [B, S, vocab] (fp32).  Attention dispatches to differentiable refs
+        when grads are required (ops/functional grad-aware dispatch)."""
+        x = torch.nn.functional.gelu(self.conv1(mel))
+        x = torch.nn.functional.gelu(self.conv2(x))
+        x = x.permute(0, 2, 1)
+        x = x + self.pos_audio[: x.shape[1]].to(x.dtype)
+        for layer in self.enc_layers:
+            x = layer(x)
+        audio = self.enc_ln(x)
+
+        B, S = tokens.shape
+        y = self.tok_embed(tokens) + self.pos_embed[:S].to(audio.dtype)
+        for layer in self.dec_layers:
+            y = y + layer.self_attn(layer.ln1(y), causal=True)
+            y = y + layer.cross_attn(layer.ln2(y), ctx=audio)
+            y = y + layer.mlp(layer.ln3(y))
+        y = self.dec_ln(y)
+        return (y @ self.tok_embed.weight.T.to(y.dtype)).float()
+
     @torch.no_grad()
     def decode_prefill(self, tokens: torch.Tensor, audio: torch.Tensor, caches):
         """tokens [B,S]; audio [B,T,state]; caches: per-layer dict with

@@ -82,3 +82,28 @@ def test_pipeline_batch_matches_single():
     solo2 = pipe.transcribe([a2], max_tokens=5)[0]
     assert both[0] == solo1
     assert both[1] == solo2
+
+
+def test_forward_train_grads_flow_and_match_decode_shapes():
+    """forward_train: teacher-forced logits [B,S,V], grads reach LoRA-style
+    trainable params."""
+    import torch
+
+    from modal_examples_amd.models.whisper.model import WhisperConfig, WhisperModel
+
+    cfg = WhisperConfig.small_test()
+    torch.manual_seed(0)
+    m = WhisperModel(cfg)
+    mel = torch.randn(2, cfg.n_mels, 2 * cfg.n_audio_ctx)
+    toks = torch.randint(0, cfg.vocab_size, (2, 8))
+    logits = m.forward_train(mel, toks)
+    assert logits.shape == (2, 8, cfg.vocab_size)
+    loss = logits.float().logsumexp(-1).mean()
+    loss.backward()
+    assert m.tok_embed.weight.grad is not None
+    # causality: token t logits must not depend on token t+1
+    with torch.no_grad():
+        toks2 = toks.clone()
+        toks2[:, -1] = (toks2[:, -1] + 1) % cfg.vocab_size
+        l2 = m.forward_train(mel, toks2)
+        assert torch.allclose(logits[:, :-1], l2[:, :-1], atol=1e-4)
