@@ -64,6 +64,17 @@ class GPT(nn.Module):
         self.ln_f = LayerNormK(cfg.n_embd)
         self.head = nn.Linear(cfg.n_embd, cfg.vocab_size, bias=False)
         self.head.weight = self.tok.weight  # weight tying
+        # GPT-2-style init: torch's default N(0,1) embeddings give ±15-unit
+        # logits at init, which makes sampling deterministic (argmax) — bad
+        # for anything exploring from the prior (e.g. RL rollouts).
+        self.apply(self._init_weights)
+
+    @staticmethod
+    def _init_weights(m: nn.Module):
+        if isinstance(m, (nn.Linear, nn.Embedding)):
+            nn.init.normal_(m.weight, mean=0.0, std=0.02)
+            if isinstance(m, nn.Linear) and m.bias is not None:
+                nn.init.zeros_(m.bias)
 
     def forward(self, idx: torch.Tensor, targets: torch.Tensor = None):
         B, S = idx.shape
