@@ -24,7 +24,9 @@ import json, os, torch, torch.distributed as dist
 from modal_examples_amd.parallel.tp import TPGroup, shard_linear
 
 rank, world = int(os.environ["RANK"]), int(os.environ["WORLD_SIZE"])
-backend = "nccl" if torch.cuda.is_available() else "gloo"
+# RCCL needs one GPU per rank; fall back to gloo on a smaller box
+backend = ("nccl" if torch.cuda.is_available()
+           and torch.cuda.device_count() >= world else "gloo")
 dist.init_process_group(backend, rank=rank, world_size=world)
 device = f"cuda:{rank}" if backend == "nccl" else "cpu"
 if backend == "nccl":
