@@ -37,7 +37,10 @@ def _ext_for(t: torch.Tensor, *grad_tensors, any_dtype: bool = False):
         if not any_dtype and t.dtype is not torch.bfloat16:
             get_ext(required=True)  # still verify the .so is present
             return None
-        return get_ext(required=True)
+        ext = get_ext(required=True)
+        from ..gpu.guard import SyncProxy, debug_sync_enabled
+
+        return SyncProxy(ext) if debug_sync_enabled() else ext
     return None
 
 

@@ -252,3 +252,33 @@ def test_snapshot_roundtrip():
     for k in tensors:
         assert torch.equal(tensors[k], originals[k]), k
     snap.close()
+
+
+@pytest.mark.gpu
+def test_guard_band_around_inplace_rope():
+    """Guard-band canaries survive the in-place RoPE kernel (no OOB writes),
+    and the sync-debug proxy path executes the same kernel correctly."""
+    import os
+
+    import torch
+
+    from modal_examples_amd.gpu.guard import GuardBand
+    from modal_examples_amd.ops import functional as OF
+
+    B, H, S, D = 2, 4, 64, 128
+    g = GuardBand((B, H, S, D), dtype=torch.bfloat16, device="cuda")
+    torch.manual_seed(0)
+    g.tensor.copy_(torch.randn(B, H, S, D, device="cuda").bfloat16())
+    cos, sin = OF.rope_tables(S, D, device="cuda")
+    ref = OF.rope(g.tensor.clone(), cos, sin)
+    OF.rope(g.tensor, cos, sin, inplace=True)
+    torch.cuda.synchronize()
+    g.check()
+    assert torch.equal(g.tensor, ref)
+
+    os.environ["MODAL_AMD_DEBUG_SYNC"] = "1"
+    try:
+        y2 = OF.rope(ref.clone(), cos, sin)
+        assert torch.equal(y2, OF.rope(ref.clone(), cos, sin))
+    finally:
+        os.environ.pop("MODAL_AMD_DEBUG_SYNC", None)
