@@ -234,6 +234,16 @@ class LlamaEngine:
 
         return self.model.decode_step(toks, pos, kv_append, kv_attend)
 
+    def close(self):
+        """Release the captured graph while the HIP runtime is still alive.
+        A CUDAGraph destroyed during interpreter teardown (after the runtime)
+        aborts the process with 'terminate called without an active
+        exception' — call this from worker @exit hooks."""
+        self._graph = None
+        self.logits_d = None
+        if self.device.type == "cuda":
+            torch.cuda.synchronize()
+
     def _ensure_graph(self):
         if self._graph is not None:
             return

@@ -93,6 +93,13 @@ class LLMServer:
     def shutdown(self):
         self._stop = True
         self._wake.set()
+        if self._t.is_alive():
+            self._t.join(timeout=5)
+        # drop the captured decode graph while the HIP runtime is healthy
+        # (a graph destructor at interpreter teardown aborts the worker)
+        close = getattr(self.engine, "close", None)
+        if close is not None:
+            close()
 
 
 def create_openai_app(server: LLMServer):
