@@ -2,6 +2,8 @@
 # cmd: ["python", "-m", "modal_examples_amd", "run", "examples/06_gpu_and_ml/audio_generation/musicgen_tokens.py"]
 # ---
 # # Music generation: autoregressive audio-token LM
+# (reference: 06_gpu_and_ml/text-to-audio/generate_music.py — MusicGen, an
+# AR transformer over EnCodec audio tokens)
 #
 # The musicgen shape — an autoregressive transformer over discrete audio
 # tokens — end to end and hermetic: the "codec" is 8-bit mu-law at 4 kHz

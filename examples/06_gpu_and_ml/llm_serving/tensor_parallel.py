@@ -3,7 +3,8 @@
 # ---
 # # Tensor-parallel big-model serving
 #
-# The big-model TP shape (reference: Llama-70B+ split across 4-8 GPUs):
+# The big-model TP shape (reference: --tensor-parallel-size at
+# llm-serving/vllm_inference.py:180, sglang --tp, TRT-LLM tensor_parallel_size):
 # Megatron-style column→row sharding from `parallel/tp.py`, one all-reduce
 # per transformer block over RCCL/xGMI.  On MI355X (288 GB HBM3E) TP only
 # matters for the 70B-405B class — each rank below holds 1/world of the

@@ -3,7 +3,8 @@
 # ---
 # # Fine-tune Whisper with LoRA
 #
-# The ASR fine-tuning shape (reference: whisper fine-tune on a GPU fleet):
+# The ASR fine-tuning shape (reference:
+# 06_gpu_and_ml/openai_whisper/fine_tune_asr.py + finetuning/):
 # LoRA adapters on the decoder projections, teacher-forced cross-entropy on
 # synthetic (mel, transcript) pairs, fused-AdamW optimizer, adapter saved to
 # a Volume and reloaded for a before/after comparison.  Runs on CPU with the

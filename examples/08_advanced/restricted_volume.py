@@ -7,6 +7,7 @@
 # `vol.read_only()` — write attempts inside the worker fail with EROFS (the
 # runtime bind-mounts the volume read-only in the worker's private mount
 # namespace), so a buggy consumer can't corrupt shared artifacts.
+# Reference shape: 08_advanced/restricted_volumes.py.
 
 import modal_examples_amd as modal
 

@@ -2,6 +2,7 @@
 # cmd: ["python", "-m", "modal_examples_amd", "run", "examples/09_job_queues/doc_ocr_webapp.py"]
 # ---
 # # Document-parser webapp: JS frontend + job queue backend
+# (reference: 09_job_queues/doc_ocr_webapp.py + doc_ocr_frontend/app.jsx)
 #
 # The doc-OCR webapp shape: a static JS single-page app (doc_ocr_frontend/)
 # submits documents to a REST API that spawns parse jobs and polls results.

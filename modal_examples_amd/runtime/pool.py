@@ -360,7 +360,7 @@ class Pool:
         """(worker, wait): the worker bound to ``key``, or a fresh binding.
         wait=True means the bound worker exists but is at capacity — the call
         must queue for THAT worker (sticky semantics: same container serves
-        the whole session)."""
+        the whole session; reference: 07_web/server_sticky.py:8-18)."""
         wid = self.sticky.get(key)
         w = self.workers.get(wid) if wid is not None else None
         if w is not None and w.ready and not w.shutting_down:
