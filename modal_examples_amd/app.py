@@ -20,6 +20,7 @@ from __future__ import annotations
 import asyncio
 import inspect
 import itertools
+import os
 import threading
 import time
 from dataclasses import dataclass, field
@@ -105,6 +106,42 @@ class _Runtime:
         self.device_pool = DevicePool()
         self.pools: List[Pool] = []
         self.device_pool.register_reclaim_hook(self._reclaim_idle)
+        self._stop_dispatch = False
+        self._dispatcher = threading.Thread(target=self._named_dispatch_loop,
+                                            daemon=True)
+        self._dispatcher.start()
+
+    def _named_dispatch_loop(self):
+        """Execute by-name invocations enqueued by workers
+        (_NamedFunctionStub.spawn): look the function up in this process's
+        app registry, run it on its pool, persist the durable result."""
+        q = store.QueueStore(_NAMED_SPAWN_QUEUE)
+        while not self._stop_dispatch:
+            try:
+                reqs = q.get_many(1, block=True, timeout=0.3)
+            except Exception:
+                time.sleep(0.3)
+                continue
+            if not reqs:
+                continue
+            req = reqs[0]
+            app = App._registry.get(req.get("app"))
+            fn = app.functions.get(req.get("fn")) if app else None
+            if fn is None:
+                q.put_many([req])  # another client may own this app
+                time.sleep(0.2)
+                continue
+
+            def run(req=req, fn=fn):
+                import traceback as _tb
+
+                try:
+                    val = fn.remote(*req["args"], **req["kwargs"])
+                    store.put_result(req["call_id"], True, val)
+                except BaseException as e:  # noqa: BLE001
+                    store.put_result(req["call_id"], False, e, _tb.format_exc())
+
+            threading.Thread(target=run, daemon=True).start()
 
     @classmethod
     def get(cls) -> "_Runtime":
@@ -121,6 +158,7 @@ class _Runtime:
         self.pools.append(pool)
 
     def shutdown(self):
+        self._stop_dispatch = True
         for p in self.pools:
             try:
                 p.shutdown()
@@ -255,6 +293,35 @@ class FunctionCall:
 def gather(*calls: FunctionCall):
     """modal.functions.gather analog (08_advanced/parallel_execution.py:41)."""
     return FunctionCall.gather(*calls)
+
+
+_NAMED_SPAWN_QUEUE = "__named_spawns__"
+
+
+class _NamedFunctionStub:
+    """Store-backed handle for `Function.from_name` inside a worker process.
+
+    The worker can't reach the client's pools directly, so spawn/remote
+    enqueue the invocation on a shared queue; the client-side dispatcher
+    (started with the runtime) executes it on the real Function and writes
+    the durable result, which FunctionCall.get polls.  This is the local
+    analog of calling a DEPLOYED function by name from anywhere."""
+
+    def __init__(self, app_name: str, name: str):
+        self.app_name = app_name
+        self.name = name
+
+    def spawn(self, *args, **kwargs) -> FunctionCall:
+        import uuid
+
+        call_id = "fc-named-" + uuid.uuid4().hex[:16]
+        store.QueueStore(_NAMED_SPAWN_QUEUE).put_many([
+            {"call_id": call_id, "app": self.app_name, "fn": self.name,
+             "args": args, "kwargs": kwargs}])
+        return FunctionCall(object_id=call_id)
+
+    def remote(self, *args, **kwargs):
+        return self.spawn(*args, **kwargs).get()
 
 
 # ---------------------------------------------------------------- verbs
@@ -455,9 +522,17 @@ class Function:
     @staticmethod
     def from_name(app_name: str, name: str) -> "Function":
         """Look up a function on a deployed/registered app
-        (the modal.Function.from_name pattern, torch_profiling.py)."""
+        (the modal.Function.from_name pattern, torch_profiling.py).
+
+        Inside a WORKER process the client's app registry is not available;
+        the lookup returns a store-backed stub whose spawn/remote enqueue the
+        invocation for the client-side dispatcher — this is what lets
+        pipeline stages hand off by name from within workers
+        (09_job_queues/pipeline_orchestration.py)."""
         app = App._registry.get(app_name)
         if app is None or name not in app.functions:
+            if os.environ.get("MODAL_TASK_ID"):  # running inside a worker
+                return _NamedFunctionStub(app_name, name)
             raise NotFoundError(f"function {app_name}/{name} not found")
         return app.functions[name]
 

@@ -195,3 +195,21 @@ def test_sticky_key_routes_to_same_worker():
     a = worker_pid._submit((0,), {}, sticky_key="k1").wait()
     b = worker_pid._submit((1,), {}, sticky_key="k1").wait()
     assert a == b
+
+
+def test_function_from_name_handoff_from_worker():
+    """A worker resolves a peer function by name and hands off to it: the
+    store-backed stub enqueues, the client dispatcher executes, the durable
+    result comes back to the worker (pipeline_orchestration pattern)."""
+    app3 = modal.App("test-handoff-app")
+
+    @app3.function()
+    def stage_two(x: int) -> int:
+        return x * 10
+
+    @app3.function()
+    def stage_one(x: int) -> int:
+        fc = modal.Function.from_name("test-handoff-app", "stage_two").spawn(x + 1)
+        return fc.get(timeout=30)
+
+    assert stage_one.remote(4) == 50
