@@ -117,6 +117,11 @@ class _Runtime:
         app registry, run it on its pool, persist the durable result."""
         q = store.QueueStore(_NAMED_SPAWN_QUEUE)
         while not self._stop_dispatch:
+            if not App._registry:
+                # nothing registered here (e.g. a worker process) — leave the
+                # queue to the client process that owns the apps
+                time.sleep(0.5)
+                continue
             try:
                 reqs = q.get_many(1, block=True, timeout=0.3)
             except Exception:
