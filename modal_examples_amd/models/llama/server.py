@@ -109,6 +109,24 @@ def create_openai_app(server: LLMServer):
 
     app = FastAPI(title="modal_examples_amd LLM server")
 
+    # built-in chat UI (the reference's llm-frontend JS app role): static ES
+    # modules streaming /v1/chat/completions SSE deltas into the page
+    from pathlib import Path
+
+    frontend = Path(__file__).parent / "chat_frontend"
+
+    @app.get("/")
+    async def index():
+        from fastapi.responses import FileResponse
+
+        return FileResponse(frontend / "index.html")
+
+    @app.get("/chat.js")
+    async def chat_js():
+        from fastapi.responses import FileResponse
+
+        return FileResponse(frontend / "chat.js", media_type="text/javascript")
+
     @app.get("/health")
     async def health():
         return {"status": "ok"}

@@ -177,3 +177,30 @@ def test_top_p_sampling():
         eng._step_count = s * 7
         tok = int(eng._sample_rows(logits, temps, eng.top_p)[0])
         assert tok in (7, 9), tok
+
+
+def test_chat_frontend_served():
+    """The OpenAI server ships its own JS chat UI at / (llm-frontend role)."""
+    import asyncio
+
+    import httpx
+
+    eng = make_engine()
+    srv = LLMServer(eng, "test-model")
+    try:
+        api = create_openai_app(srv)
+
+        async def go():
+            transport = httpx.ASGITransport(app=api)
+            async with httpx.AsyncClient(transport=transport,
+                                         base_url="http://llm") as c:
+                page = await c.get("/")
+                assert page.status_code == 200
+                assert "LLM chat" in page.text
+                js = await c.get("/chat.js")
+                assert js.status_code == 200
+                assert "v1/chat/completions" in js.text
+
+        asyncio.run(go())
+    finally:
+        srv.shutdown()
