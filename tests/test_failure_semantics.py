@@ -92,3 +92,24 @@ def test_worker_crash_surfaces_error():
 def test_single_use_containers_fresh_pid():
     pids = {single_use.remote() for _ in range(3)}
     assert len(pids) == 3, "single_use_containers must not reuse workers"
+
+
+@app.function(retries=modal.Retries(max_retries=2, initial_delay=0.1,
+                                    backoff_coefficient=1.0))
+def preemptible(key):
+    """First attempt dies by SIGKILL (preemption analog); retry must land on
+    a FRESH worker and succeed (kill-worker fault injection)."""
+    import signal
+
+    d = modal.Dict.from_name(_attempt_dict_name)
+    if d.put_if_absent(key, os.getpid()):
+        os.kill(os.getpid(), signal.SIGKILL)
+    return os.getpid()
+
+
+def test_worker_sigkill_midcall_retries_on_fresh_worker():
+    d = modal.Dict.from_name(_attempt_dict_name, create_if_missing=True)
+    key = f"preempt-{time.time()}"
+    pid = preemptible.remote(key)
+    assert isinstance(pid, int)
+    assert pid != d.get(key), "retry must run in a new worker process"
