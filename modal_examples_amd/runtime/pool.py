@@ -370,7 +370,12 @@ class Pool:
         return self._pick_worker(), False  # no binding or stale: rebind below
 
     def _pump(self):
-        """Dispatch pending calls onto ready workers; scale up when starved."""
+        """Dispatch pending calls onto ready workers; scale up when starved.
+
+        Dispatch is FIFO.  A sticky call whose bound worker is at capacity
+        parks the queue until that worker frees a slot (head-of-line by
+        design: session ordering must hold, and a sticky pool is normally
+        dedicated to its sessions)."""
         while self.pending:
             head = self.pending[0]
             if head.sticky_key is not None:
