@@ -140,5 +140,7 @@ class Image:
             gpu=opts.get("gpu"),
             volumes=opts.get("volumes") or {},
             secrets=opts.get("secrets") or [],
+            single_use_containers=True,  # build steps are one-shot: don't
+            scaledown_window=1.0,        # leave warm workers behind
         )(fn)
         builder.remote()
