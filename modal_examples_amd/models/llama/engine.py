@@ -83,6 +83,7 @@ class LlamaEngine:
         self.finished: Dict[int, Request] = {}
         self._next_id = 1
         self._step_count = 0
+        self.preemptions = 0  # KV-pressure evictions (served, not errors)
         self._graph = None
         self._slots: List[Optional[Request]] = [None] * max_batch
         # persistent device-side slot state (double as the graph's static bufs)
@@ -138,8 +139,8 @@ class LlamaEngine:
         """Batched prefill of same-length requests: ONE padded forward instead
         of per-request calls (the prefill half of continuous batching)."""
         n = len(group)
-        L = len(group[0].prompt)
-        toks = torch.tensor([r.prompt for r in group], device=self.device)
+        L = len(self._feed(group[0]))
+        toks = torch.tensor([self._feed(r) for r in group], device=self.device)
         pos = torch.arange(L, device=self.device)
         blk_rows = torch.stack([
             torch.tensor(r.blocks, device=self.device)[pos // BLOCK]
@@ -158,7 +159,8 @@ class LlamaEngine:
         for i, r in enumerate(group):
             r.pos = L
             self._append_token(r, int(first[i]))
-            r.t_first_token = time.monotonic()
+            if r.t_first_token is None:
+                r.t_first_token = time.monotonic()
             slot = r.slot
             self.toks_d[slot] = int(r.out_tokens[-1])
             self.pos_d[slot] = r.pos
@@ -168,9 +170,15 @@ class LlamaEngine:
             self.active_d[slot] = 1
             self.temps_d[slot] = r.temperature
 
+    @staticmethod
+    def _feed(r: Request) -> List[int]:
+        """Tokens to (re)prefill: the prompt plus anything already generated —
+        after a preemption the request recomputes its full context."""
+        return r.prompt + r.out_tokens
+
     def _admit(self, r: Request) -> bool:
         """Reserve blocks + a slot (no compute)."""
-        L = len(r.prompt)
+        L = len(self._feed(r))
         nblk = (L + BLOCK) // BLOCK + 1
         blocks = self._alloc_blocks(nblk)
         if blocks is None:
@@ -299,12 +307,20 @@ class LlamaEngine:
         while self.waiting and len(self.running) + len(admitted) < self.max_batch:
             r = self.waiting[0]
             if not self._admit(r):
+                need = (len(self._feed(r)) + BLOCK) // BLOCK + 1
+                if not self.running and not admitted and need > self.num_blocks - 1:
+                    # can NEVER fit even in an empty cache: fail it instead of
+                    # livelocking the scheduler
+                    self.waiting.pop(0)
+                    r.done = True
+                    self._retire(r)
+                    continue
                 break  # no KV blocks / slots free — keep waiting
             self.waiting.pop(0)
             admitted.append(r)
         by_len: Dict[int, List[Request]] = {}
         for r in admitted:
-            by_len.setdefault(len(r.prompt), []).append(r)
+            by_len.setdefault(len(self._feed(r)), []).append(r)
         for group in by_len.values():
             self._prefill_group(group)
             for r in group:
@@ -317,7 +333,15 @@ class LlamaEngine:
         if self.running:
             for r in self.running:
                 if not self._ensure_blocks(r):
-                    r.done = True  # out of memory: finish it
+                    # KV cache exhausted: preempt (vLLM-style) — release this
+                    # request's blocks+slot and requeue it for recompute once
+                    # capacity frees up; emitted tokens are kept
+                    self._release(r)
+                    self.preemptions += 1
+                    self.waiting.insert(0, r)
+            self.running = [r for r in self.running if r.slot >= 0]
+            if not self.running:
+                return done_now
             self._decode_batch()
             still = []
             for r in self.running:

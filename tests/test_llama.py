@@ -17,8 +17,9 @@ from modal_examples_amd.models.llama.server import (
 
 def make_engine(**kw):
     torch.manual_seed(0)
+    kw.setdefault("kv_blocks", 128)
     return LlamaEngine(LlamaConfig.small(), device="cpu", dtype=torch.float32,
-                       use_graph=False, kv_blocks=128, **kw)
+                       use_graph=False, **kw)
 
 
 def test_decode_matches_full_prefill():
@@ -204,3 +205,30 @@ def test_chat_frontend_served():
         asyncio.run(go())
     finally:
         srv.shutdown()
+
+
+def test_kv_exhaustion_preempts_and_recomputes():
+    """Under KV pressure the engine preempts a running request (vLLM-style:
+    release blocks, requeue, recompute prompt+generated) instead of
+    truncating it — every request still completes at full length."""
+    from modal_examples_amd.models.llama.engine import BLOCK, LlamaEngine
+
+    eng = make_engine(kv_blocks=14, max_batch=4)  # ~13 usable blocks
+    want = 40  # forces several block growths per request
+    ids = [eng.add_request(list(range(3, 3 + BLOCK * 2)), max_new_tokens=want,
+                           temperature=0.0) for _ in range(3)]
+    eng.run_until_done(max_steps=5000)
+    assert eng.preemptions > 0, "test must actually hit KV pressure"
+    for rid in ids:
+        r = eng.finished[rid]
+        assert len(r.out_tokens) == want or r.out_tokens[-1] == eng.eos_id, (
+            rid, len(r.out_tokens))
+
+
+def test_impossible_request_fails_not_livelocks():
+    from modal_examples_amd.models.llama.engine import BLOCK
+
+    eng = make_engine(kv_blocks=6, max_batch=2)
+    rid = eng.add_request(list(range(3, 3 + BLOCK * 8)), max_new_tokens=8)
+    eng.run_until_done(max_steps=50)
+    assert rid in eng.finished  # retired (failed) rather than spinning
