@@ -536,7 +536,11 @@ class Function:
         (09_job_queues/pipeline_orchestration.py)."""
         app = App._registry.get(app_name)
         if app is None or name not in app.functions:
-            if os.environ.get("MODAL_TASK_ID"):  # running inside a worker
+            # not in this process — route via the store-backed stub when the
+            # lookup can plausibly be served elsewhere: inside a worker (the
+            # client owns the app) or when a deployed process recorded it
+            if os.environ.get("MODAL_TASK_ID") or \
+                    store.DictStore("__deployments__").contains(app_name):
                 return _NamedFunctionStub(app_name, name)
             raise NotFoundError(f"function {app_name}/{name} not found")
         return app.functions[name]
@@ -1040,6 +1044,11 @@ class App:
         from .runtime.cron import start_schedules
 
         start_schedules(self)
+        # record the deployment and run the named-spawn dispatcher, so OTHER
+        # processes can invoke this app's functions via Function.from_name
+        # while this process lives (the deployed-app invocation pattern)
+        store.DictStore("__deployments__").put(self.name, time.time())
+        _Runtime.get()
         return self
 
     @staticmethod

@@ -213,3 +213,45 @@ def test_function_from_name_handoff_from_worker():
         return fc.get(timeout=30)
 
     assert stage_one.remote(4) == 50
+
+
+def test_deployed_app_invoked_from_other_process(tmp_path):
+    """app.deploy() in one process lets ANOTHER process call its functions by
+    name: from_name returns the store-backed stub and the deployed process's
+    dispatcher serves the call (deployed-app invocation pattern)."""
+    import subprocess
+    import sys
+    import textwrap
+    import time as _time
+    from pathlib import Path
+
+    mod = textwrap.dedent("""
+        import time
+        import modal_examples_amd as modal
+        app = modal.App("test-deployed-demo")
+
+        @app.function()
+        def tripled(x: int) -> int:
+            return x * 3
+
+        if __name__ == "__main__":
+            app.deploy()
+            print("DEPLOYED", flush=True)
+            time.sleep(60)
+    """)
+    path = tmp_path / "deployed_mod.py"
+    path.write_text(mod)
+    import os as _os
+
+    repo = str(Path(__file__).resolve().parent.parent)
+    env = dict(_os.environ)
+    env["PYTHONPATH"] = repo + _os.pathsep + env.get("PYTHONPATH", "")
+    server = subprocess.Popen([sys.executable, str(path)], cwd=repo, env=env,
+                              stdout=subprocess.PIPE, text=True)
+    try:
+        assert "DEPLOYED" in server.stdout.readline()
+        f = modal.Function.from_name("test-deployed-demo", "tripled")
+        assert f.spawn(14).get(timeout=45) == 42
+    finally:
+        server.terminate()
+        server.wait(timeout=10)
