@@ -116,7 +116,16 @@ class _Runtime:
         (_NamedFunctionStub.spawn): look the function up in this process's
         app registry, run it on its pool, persist the durable result."""
         q = store.QueueStore(_NAMED_SPAWN_QUEUE)
+        deployments = store.DictStore("__deployments__")
+        last_beat = 0.0
         while not self._stop_dispatch:
+            now = time.monotonic()
+            if App._registry and now - last_beat > 5.0:
+                # heartbeat: from_name in other processes only trusts fresh
+                # records, so a dead deploy process stops attracting calls
+                for app_name in list(App._registry):
+                    deployments.put(app_name, time.time())
+                last_beat = now
             if not App._registry:
                 # nothing registered here (e.g. a worker process) — leave the
                 # queue to the client process that owns the apps
@@ -539,10 +548,12 @@ class Function:
             # not in this process — route via the store-backed stub when the
             # lookup can plausibly be served elsewhere: inside a worker (the
             # client owns the app) or when a deployed process recorded it
-            if os.environ.get("MODAL_TASK_ID") or \
-                    store.DictStore("__deployments__").contains(app_name):
+            beat = store.DictStore("__deployments__").get(app_name)
+            fresh = beat is not None and time.time() - beat < 30.0
+            if os.environ.get("MODAL_TASK_ID") or fresh:
                 return _NamedFunctionStub(app_name, name)
-            raise NotFoundError(f"function {app_name}/{name} not found")
+            raise NotFoundError(f"function {app_name}/{name} not found"
+                                + (" (deployment heartbeat stale)" if beat else ""))
         return app.functions[name]
 
     def get_web_url(self):
