@@ -544,17 +544,19 @@ class Function:
         pipeline stages hand off by name from within workers
         (09_job_queues/pipeline_orchestration.py)."""
         app = App._registry.get(app_name)
-        if app is None or name not in app.functions:
-            # not in this process — route via the store-backed stub when the
-            # lookup can plausibly be served elsewhere: inside a worker (the
-            # client owns the app) or when a deployed process recorded it
-            beat = store.DictStore("__deployments__").get(app_name)
-            fresh = beat is not None and time.time() - beat < 30.0
-            if os.environ.get("MODAL_TASK_ID") or fresh:
-                return _NamedFunctionStub(app_name, name)
-            raise NotFoundError(f"function {app_name}/{name} not found"
-                                + (" (deployment heartbeat stale)" if beat else ""))
-        return app.functions[name]
+        if app is not None:
+            if name not in app.functions:  # app is local: missing fn is final
+                raise NotFoundError(f"function {app_name}/{name} not found")
+            return app.functions[name]
+        # app not in this process — route via the store-backed stub when the
+        # lookup can plausibly be served elsewhere: inside a worker (the
+        # client owns the app) or when a live deployment heartbeats it
+        beat = store.DictStore("__deployments__").get(app_name)
+        fresh = beat is not None and time.time() - beat < 30.0
+        if os.environ.get("MODAL_TASK_ID") or fresh:
+            return _NamedFunctionStub(app_name, name)
+        raise NotFoundError(f"function {app_name}/{name} not found"
+                            + (" (deployment heartbeat stale)" if beat else ""))
 
     def get_web_url(self):
         from .web.ingress import web_url_for
