@@ -124,7 +124,8 @@ class _Runtime:
                 # heartbeat: from_name in other processes only trusts fresh
                 # records, so a dead deploy process stops attracting calls
                 for app_name in list(App._registry):
-                    deployments.put(app_name, time.time())
+                    deployments.put(app_name, {"t": time.time(),
+                                               "pid": os.getpid()})
                 last_beat = now
             if not App._registry:
                 # nothing registered here (e.g. a worker process) — leave the
@@ -552,7 +553,8 @@ class Function:
         # lookup can plausibly be served elsewhere: inside a worker (the
         # client owns the app) or when a live deployment heartbeats it
         beat = store.DictStore("__deployments__").get(app_name)
-        fresh = beat is not None and time.time() - beat < 30.0
+        ts = beat.get("t") if isinstance(beat, dict) else beat
+        fresh = ts is not None and time.time() - ts < 30.0
         if os.environ.get("MODAL_TASK_ID") or fresh:
             return _NamedFunctionStub(app_name, name)
         raise NotFoundError(f"function {app_name}/{name} not found"
@@ -1060,7 +1062,8 @@ class App:
         # record the deployment and run the named-spawn dispatcher, so OTHER
         # processes can invoke this app's functions via Function.from_name
         # while this process lives (the deployed-app invocation pattern)
-        store.DictStore("__deployments__").put(self.name, time.time())
+        store.DictStore("__deployments__").put(
+            self.name, {"t": time.time(), "pid": os.getpid()})
         _Runtime.get()
         return self
 

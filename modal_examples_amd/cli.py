@@ -8,12 +8,16 @@ booleans, flux.py:170-176).  Usage::
     python -m modal_examples_amd run file.py::entry --arg 3
     python -m modal_examples_amd serve examples/07_web/basic_web.py
     python -m modal_examples_amd deploy examples/05_scheduling/schedule_simple.py
+    python -m modal_examples_amd app list            # deployed apps + heartbeats
+    python -m modal_examples_amd shell file.py::Svc  # attach into a live worker
 """
 from __future__ import annotations
 
 import argparse
 import importlib.util
 import inspect
+import os
+import signal
 import sys
 import time
 from pathlib import Path
@@ -243,6 +247,42 @@ def cmd_secret(args):
         print(f"created secret {name} with {len(env)} keys")
 
 
+def cmd_app(args):
+    """app list|stop <name> — deployed-app management (the `modal app`
+    surface): list shows heartbeat state, stop SIGTERMs the deploy process
+    and clears the record."""
+    from .runtime.store import DictStore
+
+    d = DictStore("__deployments__")
+    sub = args[0] if args else "list"
+    if sub == "list":
+        now = time.time()
+        items = sorted(d.items())
+        if not items:
+            print("no deployments")
+            return
+        for name, beat in items:
+            ts = beat.get("t") if isinstance(beat, dict) else beat
+            pid = beat.get("pid") if isinstance(beat, dict) else None
+            age = now - (ts or 0)
+            state = "live" if age < 30 else f"stale {int(age)}s"
+            print(f"{name:36s} {state:14s} pid={pid or '?'}")
+    elif sub == "stop":
+        name = args[1]
+        beat = d.get(name)
+        pid = beat.get("pid") if isinstance(beat, dict) else None
+        if pid:
+            try:
+                os.kill(int(pid), signal.SIGTERM)
+                print(f"sent SIGTERM to {name} (pid {pid})")
+            except (ProcessLookupError, PermissionError):
+                print(f"{name}: deploy process already gone")
+        d.delete(name)
+        print(f"cleared deployment record for {name}")
+    else:
+        raise SystemExit("usage: app list | app stop <name>")
+
+
 def main(argv=None):
     argv = list(sys.argv[1:] if argv is None else argv)
     if not argv or argv[0] in ("-h", "--help"):
@@ -265,8 +305,10 @@ def main(argv=None):
         cmd_queue(rest)
     elif cmd == "secret":
         cmd_secret(rest)
+    elif cmd == "app":
+        cmd_app(rest)
     else:
         raise SystemExit(
-            f"unknown command {cmd!r}; use run/serve/deploy/shell/"
+            f"unknown command {cmd!r}; use run/serve/deploy/shell/app/"
             "volume/dict/queue/secret")
     return 0
