@@ -45,16 +45,20 @@ def test_bench_single_process_contract():
     _check_contract(_parse_json_line(r.stdout), 1)
 
 
-def test_bench_torchrun_world2_contract():
+import pytest
+
+
+@pytest.mark.parametrize("n", [2, 4])
+def test_bench_torchrun_contract(n):
     """The driver's N>1 launcher: torch.distributed.run, one rank per GPU."""
     port = str(_free_port())
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", port, "bench.py", "--gpus", "2", "--steps", "1",
-         "--warmup", "1", "--small", "--latent", "32", "--batch", "1"],
+         "--nproc-per-node", str(n), "--master-addr", "127.0.0.1",
+         "--master-port", port, "bench.py", "--gpus", str(n), "--steps", "1",
+         "--warmup", "1", "--small", "--latent", "16", "--batch", "1"],
         capture_output=True, text=True, cwd=REPO, timeout=420)
     assert r.returncode == 0, r.stderr[-2000:]
     d = _parse_json_line(r.stdout)
-    _check_contract(d, 2)
-    assert d["config"]["global_batch"] == 2  # whole-job aggregate, not per-rank
+    _check_contract(d, n)
+    assert d["config"]["global_batch"] == n  # whole-job aggregate, not per-rank
