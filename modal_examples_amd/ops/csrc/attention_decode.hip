@@ -137,8 +137,6 @@ __global__ __launch_bounds__(DEC_THREADS) void paged_decode_kernel(
 #pragma unroll
       for (int j = 0; j < 8; ++j) accv[g][j] = accv[g][j] * rs + p * bf2f(v8[j]);
     }
-    k8 = k8n;
-    v8 = v8n;
   }
 
   // ---- merge partials through LDS ----
