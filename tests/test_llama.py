@@ -232,3 +232,33 @@ def test_impossible_request_fails_not_livelocks():
     rid = eng.add_request(list(range(3, 3 + BLOCK * 8)), max_new_tokens=8)
     eng.run_until_done(max_steps=50)
     assert rid in eng.finished  # retired (failed) rather than spinning
+
+
+def test_engine_randomized_stress_all_requests_complete():
+    """Randomized scheduler stress: mixed prompt/gen lengths under tight KV
+    pressure (forces admission queuing, block growth, preemption, recompute).
+    Invariants: every request finishes, generates exactly max_new tokens (or
+    EOS), and the block allocator ends balanced (no leaks)."""
+    import random
+
+    from modal_examples_amd.models.llama.engine import BLOCK
+
+    rng = random.Random(7)
+    eng = make_engine(kv_blocks=24, max_batch=3)
+    total_blocks = len(eng.free_blocks)
+    want = {}
+    for i in range(12):
+        plen = rng.randint(1, 2 * BLOCK)
+        gen = rng.randint(1, 24)
+        rid = eng.add_request([3 + (j % 250) for j in range(plen)],
+                              max_new_tokens=gen, temperature=0.0)
+        want[rid] = gen
+    eng.run_until_done(max_steps=20000)
+    assert not eng.waiting and not eng.running
+    for rid, gen in want.items():
+        r = eng.finished[rid]
+        assert r.done
+        assert len(r.out_tokens) == gen or r.out_tokens[-1] == eng.eos_id, (
+            rid, len(r.out_tokens), gen)
+    assert len(eng.free_blocks) == total_blocks, "block leak"
+    assert all(s is None for s in eng._slots), "slot leak"
