@@ -79,3 +79,21 @@ def test_example_runs_end_to_end(stem):
         capture_output=True, text=True, cwd=REPO, timeout=300,
     )
     assert r.returncode == 0, f"{stem} failed:\n{r.stdout[-2000:]}\n{r.stderr[-2000:]}"
+
+
+def test_diff_matrix_tool():
+    """tools/generate_diff_matrix.py: diff-driven CI fan-out (reference
+    internal/generate_diff_matrix.py role) emits valid JSON and flags
+    runtime-wide changes."""
+    import json
+
+    r = subprocess.run(
+        [sys.executable, str(REPO / "tools" / "generate_diff_matrix.py"),
+         "HEAD~3"],
+        capture_output=True, text=True, cwd=REPO, timeout=60)
+    assert r.returncode == 0, r.stderr[-1000:]
+    out = json.loads(r.stdout)
+    assert set(out) == {"runtime_changed", "examples"}
+    assert isinstance(out["examples"], list)
+    if out["runtime_changed"]:  # runtime edits fan out to every example
+        assert len(out["examples"]) == len(EXAMPLES)
