@@ -3,8 +3,10 @@
 # ---
 # # Load testing the OpenAI-compatible server
 #
-# The locust-on-Modal analog: N concurrent client workers hammer the chat
-# endpoint, latency percentiles land in a CSV on a Volume.
+# The locust analog: N concurrent user sessions (client threads, like locust
+# users) hammer ONE serving pool — `@modal.concurrent(max_inputs=32)` lets a
+# single replica interleave them, continuous batching does the rest.
+# Latency percentiles land in a CSV on a Volume.
 
 import time
 
@@ -38,23 +40,26 @@ class Target:
         return self.server.generate(prompt, max_tokens=max_tokens)
 
 
-@app.function()
-def user_session(user_id: int, n_requests: int) -> list:
-    t = Target()
+def user_session(target, user_id: int, n_requests: int) -> list:
     lat = []
     for i in range(n_requests):
         t0 = time.perf_counter()
-        t.chat.remote(f"user {user_id} message {i} about wavefronts", 16)
+        target.chat.remote(f"user {user_id} message {i} about wavefronts", 16)
         lat.append(time.perf_counter() - t0)
     return lat
 
 
 @app.local_entrypoint()
 def main(users: int = 8, requests_per_user: int = 3):
+    import concurrent.futures
+
+    target = Target()
+    target.chat.remote("warmup", 4)  # model load outside the timed window
     t0 = time.perf_counter()
-    all_lat = sorted(
-        x for lats in user_session.map(range(users), [requests_per_user] * users)
-        for x in lats)
+    with concurrent.futures.ThreadPoolExecutor(users) as pool:
+        futs = [pool.submit(user_session, target, u, requests_per_user)
+                for u in range(users)]
+        all_lat = sorted(x for f in futs for x in f.result())
     wall = time.perf_counter() - t0
     n = len(all_lat)
     stats = {

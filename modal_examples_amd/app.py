@@ -432,7 +432,9 @@ class Function:
         self.is_batched = bool(flags.get("batched"))
         self.max_inputs = int(flags.get("max_inputs", 1)) if flags.get("concurrent") else 1
         self._flags = flags
-        # verbs
+        self._install_verbs()
+
+    def _install_verbs(self):
         self.remote = _Verb(self._remote)
         self.local = _Verb(self._local)
         self.spawn = _Verb(self._spawn)
@@ -440,6 +442,24 @@ class Function:
         self.starmap = _Verb(self._starmap, _agen_from_sync(self._starmap))
         self.for_each = _Verb(self._for_each)
         self.remote_gen = _Verb(self._remote_gen, _agen_from_sync(self._remote_gen))
+
+    def __getstate__(self):
+        """Functions travel inside worker payloads whenever user code
+        references one from another function (pipelines, load tests).  Drop
+        the per-process runtime state (lock, pool, batcher, bound verbs);
+        the unpickled copy rebuilds them lazily in its own process."""
+        d = dict(self.__dict__)
+        for k in ("_lock", "_pool", "_batcher", "remote", "local", "spawn",
+                  "map", "starmap", "for_each", "remote_gen"):
+            d.pop(k, None)
+        return d
+
+    def __setstate__(self, d):
+        self.__dict__.update(d)
+        self._pool = None
+        self._batcher = None
+        self._lock = threading.Lock()
+        self._install_verbs()
 
     def __call__(self, *args, **kwargs):
         return self.raw(*args, **kwargs)
@@ -718,6 +738,12 @@ class _Obj:
         object.__setattr__(self, "_methods", {})
         object.__setattr__(self, "_local_obj", None)
         object.__setattr__(self, "_lock", threading.Lock())
+
+    def __getstate__(self):
+        return {"_cls": self._cls, "_params": self._params}
+
+    def __setstate__(self, d):
+        _Obj.__init__(self, d["_cls"], d["_params"])
 
     def _get_pool(self) -> Pool:
         with self._lock:

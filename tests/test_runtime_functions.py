@@ -255,3 +255,28 @@ def test_deployed_app_invoked_from_other_process(tmp_path):
     finally:
         server.terminate()
         server.wait(timeout=10)
+
+
+def test_function_and_cls_objects_are_picklable():
+    """User code that references a Function/Cls from inside another function
+    must serialize: runtime state (locks, pools, verbs) is dropped and
+    rebuilt on unpickle (caught by the load_test example)."""
+    import cloudpickle
+
+    app4 = modal.App("test-pickle-fn")
+
+    @app4.function()
+    def helper(x: int) -> int:
+        return x + 1
+
+    @app4.cls()
+    class Svc:
+        @modal.method()
+        def m(self) -> int:
+            return 7
+
+    blob = cloudpickle.dumps({"fn": helper, "cls": Svc, "obj": Svc()})
+    back = cloudpickle.loads(blob)
+    # the revived Function still works (rebuilds its pool in this process)
+    assert back["fn"].remote(41) == 42
+    assert back["obj"].m.remote() == 7
