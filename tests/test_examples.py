@@ -68,7 +68,9 @@ def test_manifest_json():
 FAST_RUN = ["hello_world", "generators", "basic_grid_search",
             "dynamic_batching", "parallel_execution", "secret_env",
             "dicts_and_queues", "volume_ingest", "safe_code_execution",
-            "pushgateway_metrics"]
+            "pushgateway_metrics", "restricted_volume", "code_interpreter",
+            "flask_app", "mcp_server", "tensor_parallel",
+            "pipeline_orchestration", "doc_ocr_webapp"]
 
 
 @pytest.mark.parametrize("stem", FAST_RUN)
