@@ -70,12 +70,14 @@ def _mount_volumes(volumes: dict) -> None:
             target = config.state_dir() / "volumes" / name
         target.mkdir(parents=True, exist_ok=True)
         try:
+            if os.path.islink(mount):
+                if os.path.realpath(mount) == os.path.realpath(target):
+                    continue
+                os.unlink(mount)  # stale link into another state dir
             if ro:
-                if os.path.islink(mount):
-                    continue  # an earlier symlink owns this path
                 ro_pairs.append((mount, target))  # bind over dir is fine
                 continue
-            if os.path.islink(mount) or os.path.exists(mount):
+            if os.path.exists(mount):
                 continue
             parent = os.path.dirname(mount.rstrip("/"))
             if parent and not os.path.exists(parent):
