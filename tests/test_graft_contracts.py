@@ -1,0 +1,28 @@
+"""Driver-facing contracts: __graft_entry__ exposes build()/smoke(), bench.py
+parses its contract flags and defaults to N=1."""
+import subprocess
+import sys
+from pathlib import Path
+
+REPO = Path(__file__).resolve().parent.parent
+
+
+def test_graft_entry_surface():
+    sys.path.insert(0, str(REPO))
+    import __graft_entry__ as g
+
+    assert callable(g.build) and callable(g.smoke)
+    import inspect
+
+    assert not any(p.default is inspect.Parameter.empty
+                   for p in inspect.signature(g.build).parameters.values())
+    assert not any(p.default is inspect.Parameter.empty
+                   for p in inspect.signature(g.smoke).parameters.values())
+
+
+def test_bench_help_and_defaults():
+    r = subprocess.run([sys.executable, "bench.py", "--help"],
+                       capture_output=True, text=True, cwd=REPO, timeout=120)
+    assert r.returncode == 0
+    for flag in ("--gpus", "--steps", "--warmup"):
+        assert flag in r.stdout
