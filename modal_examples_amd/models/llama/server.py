@@ -84,11 +84,25 @@ class LLMServer:
             raise TimeoutError(f"request {rid} did not finish")
         return r
 
+    @staticmethod
+    def apply_stop(text: str, stop) -> tuple:
+        """OpenAI `stop` semantics: truncate BEFORE the first occurrence of
+        any stop sequence; returns (text, finish_reason)."""
+        if not stop:
+            return text, "stop"
+        seqs = [stop] if isinstance(stop, str) else list(stop)
+        cut = min((text.find(q) for q in seqs if q and text.find(q) >= 0),
+                  default=-1)
+        if cut < 0:
+            return text, "stop"
+        return text[:cut], "stop"
+
     def generate(self, prompt: str, max_tokens: int = 64,
-                 temperature: float = 0.0) -> str:
+                 temperature: float = 0.0, stop=None) -> str:
         rid = self.submit(prompt, max_tokens, temperature)
         r = self.wait(rid)
-        return self.tok.decode(r.out_tokens)
+        text, _ = self.apply_stop(self.tok.decode(r.out_tokens), stop)
+        return text
 
     def shutdown(self):
         self._stop = True
@@ -137,13 +151,14 @@ def create_openai_app(server: LLMServer):
                 "data": [{"id": server.model_name, "object": "model"}]}
 
     async def _run(prompt: str, max_tokens: int, temperature: float,
-                   stream: bool, chat: bool):
+                   stream: bool, chat: bool, stop=None):
         created = int(time.time())
         rid_str = f"cmpl-{uuid.uuid4().hex[:12]}"
         if not stream:
             loop = asyncio.get_running_loop()
             text = await loop.run_in_executor(
-                None, server.generate, prompt, max_tokens, temperature)
+                None, lambda: server.generate(prompt, max_tokens, temperature,
+                                              stop=stop))
             usage = {"prompt_tokens": len(server.tok.encode(prompt)),
                      "completion_tokens": len(text.split()),
                      "total_tokens": len(server.tok.encode(prompt)) + len(text.split())}
@@ -172,6 +187,7 @@ def create_openai_app(server: LLMServer):
 
         async def gen():
             sent = 0
+            acc = ""
             while True:
                 try:
                     tok_id = await asyncio.wait_for(q.get(), timeout=120)
@@ -179,6 +195,23 @@ def create_openai_app(server: LLMServer):
                     break
                 sent += 1
                 piece = f"t{tok_id} "
+                if stop:
+                    probe, reason = LLMServer.apply_stop(acc + piece, stop)
+                    if reason == "stop" and len(probe) < len(acc + piece):
+                        tail = probe[len(acc):]
+                        if tail:
+                            payload = {"id": rid_str,
+                                       "object": "chat.completion.chunk" if chat
+                                       else "text_completion",
+                                       "created": created,
+                                       "model": server.model_name,
+                                       "choices": [{"index": 0,
+                                                    **({"delta": {"content": tail}}
+                                                       if chat else {"text": tail}),
+                                                    "finish_reason": None}]}
+                            yield f"data: {json.dumps(payload)}\n\n"
+                        break
+                    acc += piece
                 if chat:
                     payload = {"id": rid_str, "object": "chat.completion.chunk",
                                "created": created, "model": server.model_name,
@@ -200,7 +233,8 @@ def create_openai_app(server: LLMServer):
     async def completions(body: dict):
         return await _run(body.get("prompt", ""), int(body.get("max_tokens", 64)),
                           float(body.get("temperature", 0.0)),
-                          bool(body.get("stream", False)), chat=False)
+                          bool(body.get("stream", False)), chat=False,
+                          stop=body.get("stop"))
 
     @app.post("/v1/chat/completions")
     async def chat_completions(body: dict):
@@ -208,7 +242,8 @@ def create_openai_app(server: LLMServer):
         prompt = "\n".join(m.get("content", "") for m in msgs)
         return await _run(prompt, int(body.get("max_tokens", 64)),
                           float(body.get("temperature", 0.0)),
-                          bool(body.get("stream", False)), chat=True)
+                          bool(body.get("stream", False)), chat=True,
+                          stop=body.get("stop"))
 
     return app
 

@@ -262,3 +262,45 @@ def test_engine_randomized_stress_all_requests_complete():
             rid, len(r.out_tokens), gen)
     assert len(eng.free_blocks) == total_blocks, "block leak"
     assert all(s is None for s in eng._slots), "slot leak"
+
+
+def test_stop_sequences_truncate():
+    """OpenAI `stop` param: generation text is cut before the first stop
+    sequence, in both non-streaming and SSE paths."""
+    import asyncio
+
+    import httpx
+
+    eng = make_engine()
+    srv = LLMServer(eng, "test-model")
+    try:
+        full = srv.generate("stop test prompt", max_tokens=8)
+        words = full.split()
+        assert len(words) >= 3
+        stop_word = words[2]
+        cut = srv.generate("stop test prompt", max_tokens=8, stop=stop_word)
+        assert cut == " ".join(words[:2]) + " "
+        assert stop_word not in cut
+
+        api = create_openai_app(srv)
+
+        async def go():
+            transport = httpx.ASGITransport(app=api)
+            async with httpx.AsyncClient(transport=transport,
+                                         base_url="http://t") as c:
+                r = await c.post("/v1/completions",
+                                 json={"prompt": "stop test prompt",
+                                       "max_tokens": 8, "stop": [stop_word]})
+                text = r.json()["choices"][0]["text"]
+                assert stop_word not in text
+                # streaming: chunks stop before the stop sequence
+                r2 = await c.post("/v1/completions",
+                                  json={"prompt": "stop test prompt",
+                                        "max_tokens": 8, "stream": True,
+                                        "stop": stop_word})
+                body = r2.text
+                assert stop_word not in body.replace("[DONE]", "")
+
+        asyncio.run(go())
+    finally:
+        srv.shutdown()
