@@ -57,6 +57,16 @@ from .web.ingress import (  # noqa: F401
 
 __version__ = "0.1.0"
 
+
+def is_local() -> bool:
+    """True in the client process, False inside a worker "container"
+    (reference idiom: guard client-only imports/config, e.g.
+    llm-serving/openai_compatible/load_test.py:7)."""
+    import os
+
+    return not os.environ.get("MODAL_TASK_ID")
+
+
 # `modal.functions.gather` style access
 from . import app as _app_mod
 

@@ -280,3 +280,15 @@ def test_function_and_cls_objects_are_picklable():
     # the revived Function still works (rebuilds its pool in this process)
     assert back["fn"].remote(41) == 42
     assert back["obj"].m.remote() == 7
+
+
+def test_is_local_true_in_client_false_in_worker():
+    app5 = modal.App("test-is-local")
+
+    @app5.function()
+    def where() -> bool:
+        return modal.is_local()
+
+    assert modal.is_local() is True
+    assert where.remote() is False
+    assert where.local() is True  # .local runs in the client
