@@ -67,6 +67,14 @@ def is_local() -> bool:
     return not os.environ.get("MODAL_TASK_ID")
 
 
+def current_function_call_id():
+    """Id of the currently-executing invocation (inside a worker; reference:
+    09_job_queues/pipeline_orchestration.py:111).  None in the client."""
+    from .runtime.worker import current_call_id
+
+    return current_call_id()
+
+
 # `modal.functions.gather` style access
 from . import app as _app_mod
 

@@ -22,6 +22,14 @@ from concurrent.futures import ThreadPoolExecutor
 
 from . import ipc
 
+_call_ctx = threading.local()
+
+
+def current_call_id():
+    """The FunctionCall id of the invocation running on THIS thread (None
+    outside a call) — backs modal.current_function_call_id()."""
+    return getattr(_call_ctx, "call_id", None)
+
 
 def _bind_read_only(pairs: list) -> bool:
     """Enforce read-only mounts at the filesystem level: unshare this worker's
@@ -185,6 +193,7 @@ def worker_main(worker_id: int, spec_blob: bytes, task_q, result_q) -> None:
         return buf.getvalue()
 
     def run_one(call_id: str, method_name: str, args_blob: bytes) -> None:
+        _call_ctx.call_id = call_id
         try:
             args, kwargs = ipc.loads(args_blob)
             if method_name == "__debug_exec__":
@@ -207,6 +216,8 @@ def worker_main(worker_id: int, spec_blob: bytes, task_q, result_q) -> None:
             except Exception:
                 blob = ipc.dumps(RuntimeError(repr(e)))
             post(ipc.ERROR, call_id, blob, traceback.format_exc())
+        finally:
+            _call_ctx.call_id = None
 
     def run_batch(call_ids, method_name: str, args_blobs) -> None:
         """@modal.batched: collect single-input calls into one list-shaped call.

@@ -292,3 +292,15 @@ def test_is_local_true_in_client_false_in_worker():
     assert modal.is_local() is True
     assert where.remote() is False
     assert where.local() is True  # .local runs in the client
+
+
+def test_current_function_call_id_inside_worker():
+    app6 = modal.App("test-call-id")
+
+    @app6.function()
+    def my_id() -> str:
+        return modal.current_function_call_id()
+
+    assert modal.current_function_call_id() is None
+    call = my_id.spawn()
+    assert call.get(timeout=30) == call.object_id
