@@ -1015,6 +1015,10 @@ class App:
 
     def function(self, _fn=None, **opts):
         def deco(fn):
+            # legacy spelling of @modal.concurrent (misc/gpt_oss_sglang.py:56)
+            aci = opts.pop("allow_concurrent_inputs", None)
+            if aci:
+                _set_flag(fn, concurrent=True, max_inputs=int(aci))
             o = self._make_opts(opts)
             f = Function(self, fn, o)
             self.functions[f.name] = f

@@ -304,3 +304,22 @@ def test_current_function_call_id_inside_worker():
     assert modal.current_function_call_id() is None
     call = my_id.spawn()
     assert call.get(timeout=30) == call.object_id
+
+
+def test_allow_concurrent_inputs_legacy_kwarg():
+    """@app.function(allow_concurrent_inputs=N): legacy spelling of
+    @modal.concurrent — inputs overlap inside one worker."""
+    import time as _time
+
+    app7 = modal.App("test-legacy-concurrent")
+
+    @app7.function(allow_concurrent_inputs=4)
+    def slow(i: int) -> int:
+        _time.sleep(0.3)
+        return i
+
+    t0 = _time.monotonic()
+    out = list(slow.map(range(4)))
+    dt = _time.monotonic() - t0
+    assert sorted(out) == [0, 1, 2, 3]
+    assert dt < 1.0, f"inputs did not overlap: {dt:.2f}s"
