@@ -1,0 +1,64 @@
+"""CLI surface (`python -m modal_examples_amd …`): volume/dict/queue/app
+subcommands against an isolated state dir, matching the `modal` CLI verbs."""
+import os
+import subprocess
+import sys
+from pathlib import Path
+
+import pytest
+
+REPO = Path(__file__).resolve().parent.parent
+
+
+@pytest.fixture()
+def cli_env(tmp_path):
+    env = dict(os.environ)
+    env["MODAL_AMD_STATE_DIR"] = str(tmp_path / "state")
+    env["MODAL_AMD_NUM_GPUS"] = "0"
+    return env
+
+
+def run_cli(env, *args, **kw):
+    return subprocess.run([sys.executable, "-m", "modal_examples_amd", *args],
+                          capture_output=True, text=True, cwd=REPO, env=env,
+                          timeout=kw.pop("timeout", 120))
+
+
+def test_volume_put_ls_get_rm(cli_env, tmp_path):
+    src = tmp_path / "weights.bin"
+    src.write_bytes(b"W" * 32)
+    assert run_cli(cli_env, "volume", "put", "cli-vol", str(src),
+                   "models/w.bin").returncode == 0
+    ls = run_cli(cli_env, "volume", "ls", "cli-vol", "models")
+    assert "w.bin" in ls.stdout
+    dst = tmp_path / "back.bin"
+    assert run_cli(cli_env, "volume", "get", "cli-vol", "models/w.bin",
+                   str(dst)).returncode == 0
+    assert dst.read_bytes() == b"W" * 32
+    assert run_cli(cli_env, "volume", "rm", "cli-vol",
+                   "models/w.bin").returncode == 0
+    assert "w.bin" not in run_cli(cli_env, "volume", "ls", "cli-vol",
+                                  "models").stdout
+
+
+def test_dict_and_queue_subcommands(cli_env):
+    assert run_cli(cli_env, "dict", "set", "cli-d", "k1", "v1").returncode == 0
+    assert run_cli(cli_env, "dict", "get", "cli-d", "k1").stdout.strip() == "v1"
+    assert "k1\tv1" in run_cli(cli_env, "dict", "items", "cli-d").stdout
+    assert run_cli(cli_env, "queue", "put", "cli-q", "job-1").returncode == 0
+    assert run_cli(cli_env, "queue", "len", "cli-q").stdout.strip() == "1"
+    assert run_cli(cli_env, "queue", "get", "cli-q").stdout.strip() == "job-1"
+
+
+def test_app_list_empty_then_run(cli_env):
+    out = run_cli(cli_env, "app", "list").stdout
+    assert "no deployments" in out
+    r = run_cli(cli_env, "run",
+                "examples/01_getting_started/hello_world.py", timeout=180)
+    assert r.returncode == 0, r.stderr[-1000:]
+
+
+def test_unknown_command_errors(cli_env):
+    r = run_cli(cli_env, "frobnicate")
+    assert r.returncode != 0
+    assert "unknown command" in (r.stderr + r.stdout)
