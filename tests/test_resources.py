@@ -273,3 +273,36 @@ def test_image_run_function_runs_in_worker():
     modal.Volume.delete("img-build-vol")
     if os.path.islink("/tmp/mxa_imgbuild_vol"):
         os.unlink("/tmp/mxa_imgbuild_vol")
+
+
+def test_sandbox_from_id_cross_process():
+    """A sandbox created in one process can be attached from ANOTHER via the
+    store-backed registry: exec in its workdir, poll, terminate."""
+    import subprocess
+    import sys
+    import textwrap
+    from pathlib import Path
+
+    sb = modal.Sandbox.create("sleep", "30")
+    (Path(sb.workdir) / "marker.txt").write_text("shared")
+    probe = textwrap.dedent(f"""
+        import modal_examples_amd as modal
+        h = modal.Sandbox.from_id({sb.object_id!r})
+        assert h.poll() is None          # still running
+        p = h.exec("cat", "marker.txt")  # runs in the sandbox workdir
+        assert p.wait() == 0
+        print("READ:" + p.stdout.read().strip())
+        h.terminate()
+    """)
+    repo = str(Path(__file__).resolve().parent.parent)
+    r = subprocess.run([sys.executable, "-c", probe], capture_output=True,
+                       text=True, cwd=repo, timeout=60)
+    assert r.returncode == 0, r.stderr[-1500:]
+    assert "READ:shared" in r.stdout
+    import time as _time
+
+    for _ in range(50):  # the other process killed the main proc
+        if sb.poll() is not None:
+            break
+        _time.sleep(0.1)
+    assert sb.poll() is not None
