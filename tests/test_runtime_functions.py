@@ -323,3 +323,21 @@ def test_allow_concurrent_inputs_legacy_kwarg():
     dt = _time.monotonic() - t0
     assert sorted(out) == [0, 1, 2, 3]
     assert dt < 1.0, f"inputs did not overlap: {dt:.2f}s"
+
+
+def test_named_handoff_parallel_burst():
+    """Dispatcher under load: many workers hand off by name simultaneously;
+    every durable result resolves."""
+    app8 = modal.App("test-handoff-burst")
+
+    @app8.function()
+    def double(x: int) -> int:
+        return x * 2
+
+    @app8.function()
+    def relay(x: int) -> int:
+        fc = modal.Function.from_name("test-handoff-burst", "double").spawn(x)
+        return fc.get(timeout=60)
+
+    out = sorted(relay.map(range(10)))
+    assert out == [2 * x for x in range(10)]
