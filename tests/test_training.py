@@ -172,3 +172,21 @@ def test_trainer_dp2_params_stay_synced():
             p.join(timeout=180)
             assert p.exitcode == 0
         assert results[0] == pytest.approx(results[1], rel=1e-6)
+
+
+def test_grad_reducer_single_process_noop_and_bucket_edges():
+    """GradReducer with world=1 (no dist): grads pass through unchanged, and
+    params larger than the bucket budget still get their own bucket."""
+    lin = torch.nn.Linear(2048, 2048)  # 16 MB fp32 weight > 0.001 MB budget
+    from modal_examples_amd.parallel.ddp import GradReducer
+
+    red = GradReducer(list(lin.parameters()), bucket_mb=0.001)
+    x = torch.randn(4, 2048)
+    lin(x).sum().backward()
+    before = lin.weight.grad.clone()
+    red.finish()
+    assert torch.allclose(lin.weight.grad, before)  # world=1: no averaging
+    red.remove()
+    # hooks removed: a second backward accumulates normally
+    lin(x).sum().backward()
+    assert torch.allclose(lin.weight.grad, 2 * before, atol=1e-5)
