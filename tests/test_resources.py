@@ -306,3 +306,26 @@ def test_sandbox_from_id_cross_process():
             break
         _time.sleep(0.1)
     assert sb.poll() is not None
+
+
+def test_period_schedule_fires_live():
+    """A deployed Period(seconds=...) schedule actually invokes the function
+    repeatedly (live end of runtime/cron.py, beyond the matching logic)."""
+    import time
+
+    app_s = modal.App("test-live-schedule")
+    counter = modal.Dict.from_name("sched-count", create_if_missing=True)
+    counter.clear()
+    counter["n"] = 0
+
+    @app_s.function(schedule=modal.Period(seconds=0.3))
+    def tick():
+        d = modal.Dict.from_name("sched-count")
+        d["n"] = d.get("n", 0) + 1
+
+    app_s.deploy()
+    deadline = time.time() + 6
+    while counter.get("n", 0) < 2 and time.time() < deadline:
+        time.sleep(0.1)
+    assert counter.get("n", 0) >= 2, "schedule did not fire repeatedly"
+    modal.Dict.delete("sched-count")
