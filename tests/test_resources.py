@@ -328,4 +328,7 @@ def test_period_schedule_fires_live():
     while counter.get("n", 0) < 2 and time.time() < deadline:
         time.sleep(0.1)
     assert counter.get("n", 0) >= 2, "schedule did not fire repeatedly"
+    from modal_examples_amd.runtime.cron import stop_schedules
+
+    stop_schedules("test-live-schedule")
     modal.Dict.delete("sched-count")
