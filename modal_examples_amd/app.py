@@ -174,6 +174,9 @@ class _Runtime:
 
     def shutdown(self):
         self._stop_dispatch = True
+        from .runtime.cron import stop_schedules
+
+        stop_schedules()
         for p in self.pools:
             try:
                 p.shutdown()
