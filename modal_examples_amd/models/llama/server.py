@@ -145,6 +145,22 @@ def create_openai_app(server: LLMServer):
     async def health():
         return {"status": "ok"}
 
+    @app.get("/metrics")
+    async def metrics():
+        """Prometheus exposition (vLLM serves /metrics; scrape or push)."""
+        from fastapi.responses import PlainTextResponse
+
+        from ...observability import metrics as M
+
+        eng = server.engine
+        M.observe("llm_running_requests", float(len(eng.running)))
+        M.observe("llm_waiting_requests", float(len(eng.waiting)))
+        M.observe("llm_free_kv_blocks", float(len(eng.free_blocks)))
+        M.inc("llm_preemptions_total", 0)  # ensure series exists
+        if eng.preemptions:
+            M.observe("llm_preemptions", float(eng.preemptions))
+        return PlainTextResponse(M.render_prometheus())
+
     @app.get("/v1/models")
     async def models():
         return {"object": "list",

@@ -304,3 +304,27 @@ def test_stop_sequences_truncate():
         asyncio.run(go())
     finally:
         srv.shutdown()
+
+
+def test_metrics_endpoint_prometheus_format():
+    import asyncio
+
+    import httpx
+
+    eng = make_engine()
+    srv = LLMServer(eng, "test-model")
+    try:
+        api = create_openai_app(srv)
+
+        async def go():
+            transport = httpx.ASGITransport(app=api)
+            async with httpx.AsyncClient(transport=transport,
+                                         base_url="http://t") as c:
+                r = await c.get("/metrics")
+                assert r.status_code == 200
+                assert "llm_free_kv_blocks" in r.text
+                assert "llm_running_requests" in r.text
+
+        asyncio.run(go())
+    finally:
+        srv.shutdown()
