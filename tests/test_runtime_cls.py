@@ -169,3 +169,30 @@ def test_debug_exec_shell_into_running_worker():
                 sticky_key="__shell__").wait()
     assert pool.submit("__debug_exec__", ("print(obj.loaded)",), {},
                        sticky_key="__shell__").wait().strip() == "patched"
+
+
+def test_exit_hook_runs_on_scaledown():
+    """@modal.exit fires when the worker is reaped (container shutdown)."""
+    appx = modal.App("test-exit-hook")
+    d = modal.Dict.from_name("exit-hook-proof", create_if_missing=True)
+    d.clear()
+
+    @appx.cls()
+    class Svc:
+        @modal.method()
+        def ping(self) -> int:
+            return 1
+
+        @modal.exit()
+        def bye(self):
+            modal.Dict.from_name("exit-hook-proof")["ran"] = True
+
+    s = Svc()
+    assert s.ping.remote() == 1
+    pool = s.ping.obj._get_pool()
+    pool.reap_idle(force=True)
+    deadline = time.monotonic() + 10
+    while not d.get("ran") and time.monotonic() < deadline:
+        time.sleep(0.05)
+    assert d.get("ran") is True, "exit hook did not run at worker shutdown"
+    modal.Dict.delete("exit-hook-proof")
