@@ -62,3 +62,38 @@ def test_unknown_command_errors(cli_env):
     r = run_cli(cli_env, "frobnicate")
     assert r.returncode != 0
     assert "unknown command" in (r.stderr + r.stdout)
+
+
+def test_serve_binds_and_answers(cli_env):
+    """`serve` boots the uvicorn ingress, answers HTTP, and exits after
+    --timeout (the MODAL_SERVE_TIMEOUT self-termination contract)."""
+    import socket
+    import threading
+    import time
+    import urllib.request
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "modal_examples_amd", "serve",
+         "examples/07_web/basic_web.py", "--port", str(port),
+         "--timeout", "8"],
+        cwd=REPO, env=cli_env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        text=True)
+    try:
+        body, deadline = None, time.time() + 15
+        while time.time() < deadline:
+            try:
+                with urllib.request.urlopen(
+                        f"http://127.0.0.1:{port}/docs", timeout=2) as r:
+                    body = r.read().decode()
+                break
+            except Exception:
+                time.sleep(0.3)
+        assert body and "swagger" in body.lower()
+        assert proc.wait(timeout=20) == 0  # self-terminates after --timeout
+    finally:
+        if proc.poll() is None:
+            proc.terminate()
