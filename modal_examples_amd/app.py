@@ -143,6 +143,13 @@ class _Runtime:
             app = App._registry.get(req.get("app"))
             fn = app.functions.get(req.get("fn")) if app else None
             if fn is None:
+                req["hops"] = req.get("hops", 0) + 1
+                if req["hops"] > 150:  # ~30 s of requeue: no owner alive
+                    store.put_result(
+                        req["call_id"], False,
+                        NotFoundError(
+                            f"no process serves {req.get('app')}/{req.get('fn')}"))
+                    continue
                 q.put_many([req])  # another client may own this app
                 time.sleep(0.2)
                 continue
