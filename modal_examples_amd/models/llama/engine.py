@@ -101,6 +101,12 @@ class LlamaEngine:
 
     def add_request(self, prompt: List[int], max_new_tokens: int = 64,
                     temperature: float = 0.0, stream_cb=None) -> int:
+        limit = self.cfg.max_seq
+        if len(prompt) >= limit:
+            prompt = prompt[-(limit - 1):]  # keep the most recent context
+        if len(prompt) + max_new_tokens > limit:
+            # a request past max_seq would overflow the per-slot block table
+            max_new_tokens = max(1, limit - len(prompt))
         r = Request(self._next_id, list(prompt), max_new_tokens, temperature,
                     stream_cb=stream_cb)
         self._next_id += 1

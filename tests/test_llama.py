@@ -328,3 +328,15 @@ def test_metrics_endpoint_prometheus_format():
         asyncio.run(go())
     finally:
         srv.shutdown()
+
+
+def test_requests_clamped_to_max_seq():
+    """Prompts/generation beyond cfg.max_seq are clamped instead of
+    overflowing the per-slot block table."""
+    eng = make_engine()
+    limit = eng.cfg.max_seq
+    rid = eng.add_request(list(range(3, 3 + limit + 50)), max_new_tokens=64)
+    eng.run_until_done(max_steps=5000)
+    r = eng.finished[rid]
+    assert len(r.prompt) + len(r.out_tokens) <= limit
+    assert r.done
