@@ -185,7 +185,8 @@ class LlamaEngine:
     def _admit(self, r: Request) -> bool:
         """Reserve blocks + a slot (no compute)."""
         L = len(self._feed(r))
-        nblk = (L + BLOCK) // BLOCK + 1
+        # +1 spare block of headroom, but never beyond the per-slot table
+        nblk = min((L + BLOCK) // BLOCK + 1, self.max_blocks_per_seq)
         blocks = self._alloc_blocks(nblk)
         if blocks is None:
             return False
