@@ -79,11 +79,11 @@ def test_serve_binds_and_answers(cli_env):
     proc = subprocess.Popen(
         [sys.executable, "-m", "modal_examples_amd", "serve",
          "examples/07_web/basic_web.py", "--port", str(port),
-         "--timeout", "8"],
+         "--timeout", "20"],
         cwd=REPO, env=cli_env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
         text=True)
     try:
-        body, deadline = None, time.time() + 15
+        body, deadline = None, time.time() + 30
         while time.time() < deadline:
             try:
                 with urllib.request.urlopen(
@@ -93,7 +93,7 @@ def test_serve_binds_and_answers(cli_env):
             except Exception:
                 time.sleep(0.3)
         assert body and "swagger" in body.lower()
-        assert proc.wait(timeout=20) == 0  # self-terminates after --timeout
+        assert proc.wait(timeout=40) == 0  # self-terminates after --timeout
     finally:
         if proc.poll() is None:
             proc.terminate()
