@@ -191,7 +191,7 @@ def test_exit_hook_runs_on_scaledown():
     assert s.ping.remote() == 1
     pool = s.ping.obj._get_pool()
     pool.reap_idle(force=True)
-    deadline = time.monotonic() + 10
+    deadline = time.monotonic() + 25
     while not d.get("ran") and time.monotonic() < deadline:
         time.sleep(0.05)
     assert d.get("ran") is True, "exit hook did not run at worker shutdown"

@@ -114,14 +114,14 @@ def test_function_call_from_id():
     assert fc.get() == 25
     time.sleep(0.2)  # allow durable store write
     fc2 = modal.FunctionCall.from_id(fc.object_id)
-    assert fc2.get(timeout=5) == 25
+    assert fc2.get(timeout=20) == 25
 
 
 def test_get_timeout_zero_raises():
     fc = slow_echo.spawn("hi")
     with pytest.raises(TimeoutError):
         fc.get(timeout=0)
-    assert fc.get(timeout=10) == "hi"
+    assert fc.get(timeout=30) == "hi"
 
 
 def test_remote_gen():
