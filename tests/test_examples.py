@@ -70,7 +70,7 @@ FAST_RUN = ["hello_world", "generators", "basic_grid_search",
             "dicts_and_queues", "volume_ingest", "safe_code_execution",
             "pushgateway_metrics", "restricted_volume", "code_interpreter",
             "flask_app", "mcp_server", "tensor_parallel",
-            "pipeline_orchestration", "doc_ocr_webapp"]
+            "pipeline_orchestration", "doc_ocr_webapp", "feed_alerts"]
 
 
 @pytest.mark.parametrize("stem", FAST_RUN)
