@@ -140,3 +140,37 @@ def test_sticky_session_pins_worker():
             assert len(pids) == 1
 
     asyncio.run(go())
+
+
+def test_proxy_auth_enforced(monkeypatch):
+    """requires_proxy_auth: 401 without Modal-Key/Modal-Secret headers, 200
+    with the keystore credentials; locked when no token is configured."""
+    app_auth = modal.App("test-web-auth")
+
+    @app_auth.function()
+    @modal.fastapi_endpoint(method="GET", label="secret-square",
+                            requires_proxy_auth=True)
+    def ssq(x: int = 2):
+        return {"sq": x * x}
+
+    root = build_ingress_app(app_auth)
+
+    async def go():
+        async with httpx.AsyncClient(transport=httpx.ASGITransport(app=root),
+                                     base_url="http://t") as c:
+            # no token configured: endpoint stays locked
+            monkeypatch.delenv("MODAL_AMD_PROXY_TOKEN_ID", raising=False)
+            r = await c.get("/secret-square", params={"x": 3})
+            assert r.status_code == 401
+            monkeypatch.setenv("MODAL_AMD_PROXY_TOKEN_ID", "wk-1")
+            monkeypatch.setenv("MODAL_AMD_PROXY_TOKEN_SECRET", "ws-2")
+            r = await c.get("/secret-square", params={"x": 3})
+            assert r.status_code == 401  # still no headers
+            r = await c.get("/secret-square", params={"x": 3},
+                            headers={"Modal-Key": "wk-1", "Modal-Secret": "ws-2"})
+            assert r.status_code == 200 and r.json() == {"sq": 9}
+            r = await c.get("/secret-square",
+                            headers={"Modal-Key": "wk-1", "Modal-Secret": "bad"})
+            assert r.status_code == 401
+
+    asyncio.run(go())
