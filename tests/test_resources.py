@@ -332,3 +332,35 @@ def test_period_schedule_fires_live():
 
     stop_schedules("test-live-schedule")
     modal.Dict.delete("sched-count")
+
+
+def test_bucket_mount_read_only():
+    """CloudBucketMount(read_only=True) gets the same filesystem-level
+    enforcement as Volume.read_only() (ro bind mount in the worker)."""
+    bucket = modal.CloudBucketMount("models-bucket", read_only=True)
+    # pre-populate the local bucket dir
+    (bucket.path / "weights.txt").write_text("frozen")
+    app_b = modal.App("test-bucket-ro")
+    mnt = "/tmp/mxa_bucket_ro_test"
+
+    @app_b.function(volumes={mnt: bucket})
+    def probe():
+        import errno
+
+        with open(f"{mnt}/weights.txt") as f:
+            content = f.read()
+        try:
+            open(f"{mnt}/new.txt", "w")
+            blocked = False
+        except OSError as e:
+            blocked = e.errno == errno.EROFS
+        return content, blocked
+
+    content, blocked = probe.remote()
+    assert content == "frozen"
+    assert blocked
+    import os
+    import shutil
+
+    if os.path.isdir(mnt) and not os.path.islink(mnt):
+        shutil.rmtree(mnt, ignore_errors=True)
