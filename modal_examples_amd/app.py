@@ -303,7 +303,20 @@ class FunctionCall:
         return await asyncio.to_thread(self.get, timeout)
 
     def cancel(self):
-        pass  # local calls are not cancellable mid-flight yet
+        """Cancel a spawned call: queued inputs are dropped, an executing
+        input's container is terminated; a later ``get()`` raises
+        ``FunctionCancelledError``.  (08_advanced/poll_delayed_result.py)"""
+        from .exception import FunctionCancelledError
+
+        if self._call is not None and getattr(self._call, "pool", None) is not None:
+            self._call.pool.cancel(self._call.id)
+            return
+        # cross-process handle: record the cancellation in the durable store
+        # unless the call already finished (finished results win)
+        found, *_ = store.get_result(self.object_id)
+        if not found:
+            store.put_result(self.object_id, False,
+                             FunctionCancelledError(f"{self.object_id} cancelled"))
 
     @staticmethod
     def from_id(object_id: str) -> "FunctionCall":

@@ -35,6 +35,11 @@ class ExecutionError(Error):
     """Worker died or the runtime failed internally while executing a call."""
 
 
+class FunctionCancelledError(Error):
+    """The call was cancelled via ``FunctionCall.cancel()``
+    (08_advanced/poll_delayed_result.py relies on cancellation semantics)."""
+
+
 class DeserializationError(Error):
     """Result or argument could not be (un)pickled."""
 

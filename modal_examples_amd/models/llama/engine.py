@@ -42,6 +42,7 @@ class Request:
     pos: int = 0  # tokens stored in cache
     slot: int = -1
     done: bool = False
+    error: Optional[str] = None
     t_arrive: float = field(default_factory=time.monotonic)
     t_first_token: Optional[float] = None
     stream_cb: Optional[object] = None
@@ -59,6 +60,7 @@ class LlamaEngine:
         self.use_graph = use_graph and self.device.type == "cuda"
         self.eos_id = eos_id
         self.top_p = top_p
+        self.seed = seed
         torch.manual_seed(seed)
         with torch.device(self.device):
             self.model = LlamaModel(self.cfg).to(self.device, dtype)
@@ -224,7 +226,8 @@ class LlamaEngine:
             drop_sorted = cum - probs > top_p  # keep first token crossing p
             drop = torch.zeros_like(drop_sorted).scatter(-1, idx, drop_sorted)
             scaled = scaled.masked_fill(drop, float("-inf"))
-        sampled = OF.sample(scaled, 1.0, seed=0x5EED + self._step_count)
+        sampled = OF.sample(scaled, 1.0,
+                            seed=self.seed ^ (0x5EED + self._step_count))
         return torch.where(temps <= 0, greedy, sampled.to(greedy.device))
 
     # ------------------------------------------------ decode
@@ -320,6 +323,9 @@ class LlamaEngine:
                     # livelocking the scheduler
                     self.waiting.pop(0)
                     r.done = True
+                    r.error = (f"prompt of {len(self._feed(r))} tokens cannot "
+                               f"fit in the KV cache ({self.num_blocks - 1} "
+                               f"blocks of {BLOCK})")
                     self._retire(r)
                     continue
                 break  # no KV blocks / slots free — keep waiting

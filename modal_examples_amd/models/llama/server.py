@@ -82,6 +82,8 @@ class LLMServer:
         r = self.engine.finished.get(rid)
         if r is None:
             raise TimeoutError(f"request {rid} did not finish")
+        if getattr(r, "error", None):
+            raise ValueError(r.error)
         return r
 
     @staticmethod

@@ -33,7 +33,7 @@ class Volume:
 
     @staticmethod
     def from_name(name: str, create_if_missing: bool = False, version: int = 2) -> "Volume":
-        return Volume(name, _create=create_if_missing or True, version=version)
+        return Volume(name, _create=create_if_missing, version=version)
 
     @staticmethod
     def ephemeral():

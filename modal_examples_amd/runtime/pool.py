@@ -68,6 +68,7 @@ class Call:
         "done",
         "t_submit",
         "sticky_key",
+        "pool",
     )
 
     def __init__(self, method_name, args_blob, is_gen, timeout, retries: RetryPolicy, spawned=False,
@@ -89,6 +90,7 @@ class Call:
         self.done = False
         self.t_submit = time.monotonic()
         self.sticky_key = sticky_key
+        self.pool = None  # set at submit; lets FunctionCall.cancel reach us
 
     def wait(self, timeout: Optional[float] = None):
         if not self.event.wait(timeout):
@@ -197,6 +199,7 @@ class Pool:
             spawned,
             sticky_key=sticky_key,
         )
+        call.pool = self
         with self.lock:
             self._ensure_threads()
             self.calls[call.id] = call
@@ -230,6 +233,31 @@ class Pool:
                 (ipc.T_BATCH, [c.id for c in calls], "", [c.args_blob for c in calls])
             )
         return calls
+
+    def cancel(self, call_id: str) -> bool:
+        """Cancel a submitted call: drop it from the queue, or — if already
+        executing — terminate its worker (mid-flight work cannot be interrupted
+        in-process; the container is torn down, matching the platform's input
+        cancellation).  Returns True if a live call was cancelled."""
+        from ..exception import FunctionCancelledError
+
+        with self.lock:
+            call = self.calls.get(call_id)
+            if call is None or call.done:
+                return False
+            if call in self.pending:
+                self.pending.remove(call)
+                self._resolve(call, exc=FunctionCancelledError(
+                    f"{self.name} call {call_id} cancelled"))
+                return True
+            w = self.workers.get(call.worker_id)
+            if w is not None:
+                w.inflight.discard(call.id)
+                self._stop_worker(w, graceful=False)
+                self._release_worker(w)
+            self._resolve(call, exc=FunctionCancelledError(
+                f"{self.name} call {call_id} cancelled"))
+            return True
 
     def warm(self, n: Optional[int] = None):
         """Pre-start ``min_containers`` (or n) workers."""
@@ -385,6 +413,13 @@ class Pool:
             else:
                 w = self._pick_worker()
             if w is None:
+                # demand-bound scale-up: workers already starting (spawned,
+                # not yet READY) will absorb pending inputs — don't burst to
+                # max_containers on a cold fan-out
+                starting = sum(1 for h in self.workers.values()
+                               if not h.ready and not h.shutting_down)
+                if starting * self.max_inputs_per_worker >= len(self.pending):
+                    return  # capacity is on the way; re-pumped on READY
                 started = self._maybe_scale_up()
                 if started is None:
                     return  # all busy / can't grow: leave pending

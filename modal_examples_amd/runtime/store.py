@@ -21,6 +21,26 @@ from .. import config
 from ..exception import NotFoundError
 
 
+import contextlib
+
+
+@contextlib.contextmanager
+def _write_txn(conn: sqlite3.Connection):
+    """BEGIN IMMEDIATE so a read-then-delete pair is one atomic write
+    transaction.  Python's sqlite3 only opens the implicit transaction at the
+    first DML statement — a bare ``with conn:`` around SELECT+DELETE lets two
+    processes both read the same rows before either deletes (duplicate job
+    delivery in the cross-process dispatcher)."""
+    conn.execute("BEGIN IMMEDIATE")
+    try:
+        yield conn
+    except BaseException:
+        conn.rollback()
+        raise
+    else:
+        conn.commit()
+
+
 class _DB:
     """One sqlite connection per (process, path), WAL mode for cross-process use."""
 
@@ -119,7 +139,7 @@ class DictStore:
     def pop(self, k):
         conn = _DB.get()
         kb = cloudpickle.dumps(k)
-        with conn:
+        with _write_txn(conn):
             row = conn.execute(
                 "SELECT v FROM kv WHERE ns=? AND k=?", (self.ns, kb)
             ).fetchone()
@@ -175,7 +195,7 @@ class QueueStore:
         part = partition or ""
         while True:
             conn = _DB.get()
-            with conn:
+            with _write_txn(conn):
                 rows = conn.execute(
                     "SELECT rowid, v FROM fifo WHERE ns=? AND partition=?"
                     " ORDER BY rowid LIMIT ?",

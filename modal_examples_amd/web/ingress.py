@@ -60,7 +60,8 @@ def web_server(port: int, startup_timeout: float = 60.0, label: Optional[str] = 
                custom_domains=None, requires_proxy_auth: bool = False):
     def deco(fn):
         return _set_flag(fn, web=True, web_kind="server", port=port,
-                         startup_timeout=startup_timeout, label=label)
+                         startup_timeout=startup_timeout, label=label,
+                         requires_proxy_auth=requires_proxy_auth)
     return deco
 
 
@@ -109,6 +110,41 @@ def _proxy_auth_ok(request) -> bool:
         return False  # locked endpoints stay locked until a token is set
     return (request.headers.get("Modal-Key") == want_key and
             request.headers.get("Modal-Secret", "") == want_secret)
+
+
+class _ProxyAuthASGI:
+    """ASGI middleware enforcing proxy auth on mounted asgi/wsgi sub-apps —
+    the locked-endpoint contract of basic_web.py:178-180 applies to every
+    web kind, not just function-level fastapi routes."""
+
+    def __init__(self, inner):
+        self.inner = inner
+
+    async def __call__(self, scope, receive, send):
+        if scope["type"] in ("http", "websocket"):
+            headers = {k.decode("latin1").lower(): v.decode("latin1")
+                       for k, v in scope.get("headers", [])}
+            if not _proxy_auth_headers_ok(headers):
+                if scope["type"] == "websocket":
+                    await send({"type": "websocket.close", "code": 4401})
+                    return
+                await send({"type": "http.response.start", "status": 401,
+                            "headers": [(b"content-type", b"application/json")]})
+                await send({"type": "http.response.body",
+                            "body": b'{"detail": "proxy auth required"}'})
+                return
+        await self.inner(scope, receive, send)
+
+
+def _proxy_auth_headers_ok(headers: dict) -> bool:
+    import os
+
+    want_key = os.environ.get("MODAL_AMD_PROXY_TOKEN_ID", "")
+    want_secret = os.environ.get("MODAL_AMD_PROXY_TOKEN_SECRET", "")
+    if not want_key:
+        return False
+    return (headers.get("modal-key") == want_key and
+            headers.get("modal-secret", "") == want_secret)
 
 
 def build_ingress_app(app):
@@ -177,18 +213,27 @@ def build_ingress_app(app):
             root.add_api_route(f"/{label}", handler, methods=[method])
         elif kind == "asgi":
             sub = f.raw()  # factory runs in-process
+            if flags.get("requires_proxy_auth"):
+                sub = _ProxyAuthASGI(sub)
             root.mount(f"/{label}" if label else "", sub)
         elif kind == "wsgi":
             from starlette.middleware.wsgi import WSGIMiddleware
 
-            sub = f.raw()
-            root.mount(f"/{label}" if label else "", WSGIMiddleware(sub))
+            sub = WSGIMiddleware(f.raw())
+            if flags.get("requires_proxy_auth"):
+                sub = _ProxyAuthASGI(sub)
+            root.mount(f"/{label}" if label else "", sub)
         elif kind == "server":
             port = flags.get("port")
+            needs_auth = bool(flags.get("requires_proxy_auth"))
             f.spawn()  # starts the server inside a worker
 
             @root.get(f"/{label}/_url")
-            async def url_handler(_port=port):
+            async def url_handler(request: Request, _port=port,
+                                  _auth=needs_auth):
+                if _auth and not _proxy_auth_ok(request):
+                    return JSONResponse({"detail": "proxy auth required"},
+                                        status_code=401)
                 return {"url": f"http://127.0.0.1:{_port}"}
 
     for f in app.web_endpoints.values():
@@ -213,11 +258,16 @@ def _add_cls_route(root, c, mname, flags):
     label = flags.get("label") or mname
     obj = c()
 
+    needs_auth = bool(flags.get("requires_proxy_auth"))
+
     if kind == "fastapi":
         sig = _inspect.signature(getattr(c.user_cls, mname))
         method = flags.get("http_method", "GET")
 
         async def handler(request: Request):
+            if needs_auth and not _proxy_auth_ok(request):
+                return JSONResponse({"detail": "proxy auth required"},
+                                    status_code=401)
             kwargs = dict(request.query_params)
             if request.method in ("POST", "PUT"):
                 try:
@@ -248,6 +298,8 @@ def _add_cls_route(root, c, mname, flags):
     elif kind == "asgi":
         inst = obj._local_instance()
         sub = getattr(inst, mname)()
+        if needs_auth:
+            sub = _ProxyAuthASGI(sub)
         root.mount(f"/{label}" if label else "", sub)
 
 

@@ -364,3 +364,12 @@ def test_bucket_mount_read_only():
 
     if os.path.isdir(mnt) and not os.path.islink(mnt):
         shutil.rmtree(mnt, ignore_errors=True)
+
+
+def test_volume_from_name_missing_raises():
+    """from_name(create_if_missing=False) on a nonexistent volume raises
+    NotFoundError instead of silently creating it (ADVICE r1 finding)."""
+    import pytest as _pytest
+
+    with _pytest.raises(modal.NotFoundError):
+        modal.Volume.from_name("never-created-vol-xyz")

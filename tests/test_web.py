@@ -174,3 +174,60 @@ def test_proxy_auth_enforced(monkeypatch):
             assert r.status_code == 401
 
     asyncio.run(go())
+
+
+def test_proxy_auth_enforced_all_kinds(monkeypatch):
+    """requires_proxy_auth applies to asgi/wsgi mounts, web_server helper
+    routes, and class-based routes — not only function-level fastapi routes
+    (ADVICE r1 high finding)."""
+    app_auth2 = modal.App("test-web-auth-kinds")
+
+    @app_auth2.function()
+    @modal.asgi_app(label="locked-sub", requires_proxy_auth=True)
+    def locked_sub():
+        from fastapi import FastAPI
+
+        w = FastAPI()
+
+        @w.get("/ping")
+        def ping():
+            return {"pong": True}
+
+        return w
+
+    @app_auth2.function()
+    @modal.wsgi_app(label="locked-wsgi", requires_proxy_auth=True)
+    def locked_wsgi():
+        def application(environ, start_response):
+            start_response("200 OK", [("Content-Type", "text/plain")])
+            return [b"wsgi-ok"]
+
+        return application
+
+    @app_auth2.cls()
+    class LockedGreeter:
+        @modal.fastapi_endpoint(method="GET", label="locked-greet",
+                                requires_proxy_auth=True)
+        def greet(self, name: str = "w"):
+            return {"m": name}
+
+    root = build_ingress_app(app_auth2)
+    ok_headers = {"Modal-Key": "wk-1", "Modal-Secret": "ws-2"}
+
+    async def go():
+        async with httpx.AsyncClient(transport=httpx.ASGITransport(app=root),
+                                     base_url="http://t") as c:
+            monkeypatch.setenv("MODAL_AMD_PROXY_TOKEN_ID", "wk-1")
+            monkeypatch.setenv("MODAL_AMD_PROXY_TOKEN_SECRET", "ws-2")
+            for path in ("/locked-sub/ping", "/locked-wsgi/x", "/locked-greet"):
+                r = await c.get(path)
+                assert r.status_code == 401, path
+            r = await c.get("/locked-sub/ping", headers=ok_headers)
+            assert r.status_code == 200 and r.json() == {"pong": True}
+            r = await c.get("/locked-wsgi/x", headers=ok_headers)
+            assert r.status_code == 200 and r.text == "wsgi-ok"
+            r = await c.get("/locked-greet", params={"name": "q"},
+                            headers=ok_headers)
+            assert r.status_code == 200 and r.json() == {"m": "q"}
+
+    asyncio.run(go())
