@@ -73,6 +73,65 @@ class GroupNormSiLU(nn.Module, _F32ParamCache):
                                  self.groups, self.eps, self.silu)
 
 
+class Conv3x3(nn.Conv2d):
+    """3x3 stride-1 pad-1 conv on the hand NCHW implicit-GEMM kernel (K3).
+
+    Drop-in for nn.Conv2d(c_in, c_out, 3, padding=1): same parameters and
+    state_dict.  Inference on GPU bf16 runs the MFMA kernel (with an optional
+    fused residual add); training / CPU / odd dtypes use F.conv2d so autograd
+    and hermetic tests stay correct.  The [9,Kpad,C16] weight repack is cached
+    against the parameter version."""
+
+    def __init__(self, c_in: int, c_out: int):
+        super().__init__(c_in, c_out, 3, padding=1)
+
+    def _packed(self):
+        w = self.weight
+        ent = self.__dict__.get("_wr_cache")
+        if ent is None or ent[1] is not w or ent[2] != w._version:
+            wr = OF.repack_conv3x3_weight(w.detach())
+            b = (self.bias.detach().float().contiguous() if self.bias is not None
+                 else torch.zeros(w.shape[0], device=w.device))
+            ent = ((wr, b), w, w._version)
+            self.__dict__["_wr_cache"] = ent
+        return ent[0]
+
+    def _use_kernel(self, x) -> bool:
+        if not x.is_cuda or x.dtype is not torch.bfloat16 or self.training:
+            return False
+        if torch.is_grad_enabled() and (x.requires_grad or
+                                        self.weight.requires_grad):
+            return False
+        return True
+
+    def forward(self, x, residual=None):
+        if self._use_kernel(x):
+            wr, b = self._packed()
+            return OF.conv3x3(x, wr, b, self.out_channels, residual=residual,
+                              raw_weight=self.weight)
+        y = nn.functional.conv2d(x, self.weight, self.bias, padding=1)
+        return y if residual is None else y + residual
+
+
+class Conv1x1(nn.Conv2d):
+    """1x1 conv as a plain hipBLASLt GEMM over [C, N*H*W] — keeps MIOpen (and
+    its internal NCHW<->NHWC transposes) off the skip-connection path."""
+
+    def __init__(self, c_in: int, c_out: int):
+        super().__init__(c_in, c_out, 1)
+
+    def forward(self, x):
+        if not x.is_cuda or self.training or (
+                torch.is_grad_enabled() and x.requires_grad):
+            return nn.functional.conv2d(x, self.weight, self.bias)
+        N, C, H, W = x.shape
+        w = self.weight.reshape(self.out_channels, C)
+        y = (w @ x.reshape(N, C, H * W)).reshape(N, self.out_channels, H, W)
+        if self.bias is not None:
+            y = y + self.bias.reshape(1, -1, 1, 1)
+        return y
+
+
 class SelfAttention(nn.Module):
     """Fused-QKV self-attention over [B, S, C] → gfx950 flash kernel."""
 

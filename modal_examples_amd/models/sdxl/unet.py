@@ -20,6 +20,8 @@ import torch
 import torch.nn as nn
 
 from .layers import (
+    Conv1x1,
+    Conv3x3,
     GroupNormSiLU,
     TransformerBlock,
     timestep_embedding,
@@ -57,17 +59,17 @@ class ResnetBlock(nn.Module):
     def __init__(self, c_in: int, c_out: int, temb_dim: int):
         super().__init__()
         self.norm1 = GroupNormSiLU(c_in)
-        self.conv1 = nn.Conv2d(c_in, c_out, 3, padding=1)
+        self.conv1 = Conv3x3(c_in, c_out)
         self.temb_proj = nn.Linear(temb_dim, c_out)
         self.norm2 = GroupNormSiLU(c_out)
-        self.conv2 = nn.Conv2d(c_out, c_out, 3, padding=1)
-        self.skip = nn.Conv2d(c_in, c_out, 1) if c_in != c_out else nn.Identity()
+        self.conv2 = Conv3x3(c_out, c_out)
+        self.skip = Conv1x1(c_in, c_out) if c_in != c_out else nn.Identity()
 
     def forward(self, x, temb):
         h = self.conv1(self.norm1(x))
         h = h + self.temb_proj(torch.nn.functional.silu(temb))[:, :, None, None]
-        h = self.conv2(self.norm2(h))
-        return self.skip(x) + h
+        # conv2 fuses the residual add (K3 epilogue)
+        return self.conv2(self.norm2(h), residual=self.skip(x))
 
 
 class SpatialTransformer(nn.Module):
@@ -105,7 +107,7 @@ class Downsample(nn.Module):
 class Upsample(nn.Module):
     def __init__(self, c):
         super().__init__()
-        self.conv = nn.Conv2d(c, c, 3, padding=1)
+        self.conv = Conv3x3(c, c)
 
     def forward(self, x):
         return self.conv(torch.nn.functional.interpolate(x, scale_factor=2.0, mode="nearest"))
@@ -119,7 +121,7 @@ class UNetXL(nn.Module):
         ch = cfg.channels
         ted = cfg.time_embed_dim
 
-        self.conv_in = nn.Conv2d(cfg.in_channels, ch[0], 3, padding=1)
+        self.conv_in = Conv3x3(cfg.in_channels, ch[0])
         self.time_mlp = nn.Sequential(
             nn.Linear(ch[0], ted), nn.SiLU(), nn.Linear(ted, ted)
         )
@@ -182,7 +184,7 @@ class UNetXL(nn.Module):
                 self.upsamplers.append(nn.Identity())
 
         self.norm_out = GroupNormSiLU(ch[0])
-        self.conv_out = nn.Conv2d(ch[0], cfg.out_channels, 3, padding=1)
+        self.conv_out = Conv3x3(ch[0], cfg.out_channels)
 
     def forward(self, x, t, ctx, pooled_add):
         """x [B,4,H,W] bf16, t [B] f32 timesteps, ctx [B,77,ctx_dim] bf16,

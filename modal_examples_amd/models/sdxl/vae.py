@@ -14,21 +14,22 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
-from .layers import GroupNormSiLU
+from .layers import Conv1x1, Conv3x3, GroupNormSiLU
 
 
 class VAEResnet(nn.Module):
     def __init__(self, c_in, c_out):
         super().__init__()
         self.norm1 = GroupNormSiLU(c_in)
-        self.conv1 = nn.Conv2d(c_in, c_out, 3, padding=1)
+        self.conv1 = Conv3x3(c_in, c_out)
         self.norm2 = GroupNormSiLU(c_out)
-        self.conv2 = nn.Conv2d(c_out, c_out, 3, padding=1)
-        self.skip = nn.Conv2d(c_in, c_out, 1) if c_in != c_out else nn.Identity()
+        self.conv2 = Conv3x3(c_out, c_out)
+        self.skip = Conv1x1(c_in, c_out) if c_in != c_out else nn.Identity()
 
     def forward(self, x):
-        h = self.conv2(self.norm2(self.conv1(self.norm1(x))))
-        return self.skip(x) + h
+        h = self.conv1(self.norm1(x))
+        # conv2 fuses the skip add (K3 epilogue)
+        return self.conv2(self.norm2(h), residual=self.skip(x))
 
 
 class VAEMidAttention(nn.Module):
@@ -63,7 +64,7 @@ class VAEDecoder(nn.Module):
         super().__init__()
         self.scaling_factor = scaling_factor
         c0 = channels[0]
-        self.conv_in = nn.Conv2d(latent_channels, c0, 3, padding=1)
+        self.conv_in = Conv3x3(latent_channels, c0)
         self.mid_res1 = VAEResnet(c0, c0)
         self.mid_attn = VAEMidAttention(c0)
         self.mid_res2 = VAEResnet(c0, c0)
@@ -74,10 +75,10 @@ class VAEDecoder(nn.Module):
             for _ in range(resnets_per_level):
                 blocks.append(VAEResnet(c_prev, c))
                 c_prev = c
-            up = nn.Conv2d(c, c, 3, padding=1) if li < len(channels) - 1 else None
+            up = Conv3x3(c, c) if li < len(channels) - 1 else None
             self.levels.append(nn.ModuleList([blocks, nn.ModuleList([up] if up else [])]))
         self.norm_out = GroupNormSiLU(channels[-1])
-        self.conv_out = nn.Conv2d(channels[-1], out_channels, 3, padding=1)
+        self.conv_out = Conv3x3(channels[-1], out_channels)
 
     def forward(self, z):
         h = self.conv_in(z / self.scaling_factor)

@@ -27,6 +27,7 @@ HIP_SOURCES = [
     "elementwise.hip",
     "adamw.hip",
     "sampling.hip",
+    "conv3x3.hip",
 ]
 
 

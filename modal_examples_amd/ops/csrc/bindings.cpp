@@ -48,6 +48,8 @@ void cfg_euler_dev_bf16(const void*, const void*, const void*, void*,
                         const float*, const long long*, float, long long,
                         hipStream_t);
 void advance_step(long long*, hipStream_t);
+void conv3x3_bf16(const void*, const void*, const void*, const void*, void*,
+                  int, int, int, int, int, int, int, hipStream_t);
 }
 
 namespace {
@@ -248,6 +250,32 @@ void adamw_(torch::Tensor p, torch::Tensor g, torch::Tensor m, torch::Tensor v,
              cur_stream());
 }
 
+// ---------------------------------------------------------------- conv (K3)
+
+torch::Tensor conv3x3(torch::Tensor x, torch::Tensor wr, torch::Tensor bias,
+                      c10::optional<torch::Tensor> residual, int64_t K) {
+  check_bf16(x, "x");
+  check_bf16(wr, "wr");
+  TORCH_CHECK(x.dim() == 4, "x must be NCHW");
+  TORCH_CHECK(wr.dim() == 3 && wr.size(0) == 9,
+              "wr must be [9, Kpad, C16] (repacked 3x3 weights)");
+  TORCH_CHECK(bias.scalar_type() == torch::kFloat && bias.is_contiguous(),
+              "bias must be fp32 contiguous");
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  int Kpad = wr.size(1), C16 = wr.size(2);
+  TORCH_CHECK(K <= Kpad && C <= C16, "weight pack smaller than conv");
+  auto out = torch::empty({N, K, H, W}, x.options());
+  const void* resp = nullptr;
+  if (residual.has_value()) {
+    check_bf16(*residual, "residual");
+    TORCH_CHECK(residual->sizes() == out.sizes(), "residual shape mismatch");
+    resp = residual->data_ptr();
+  }
+  conv3x3_bf16(x.data_ptr(), wr.data_ptr(), bias.data_ptr(), resp,
+               out.data_ptr(), N, C, H, W, (int)K, C16, Kpad, cur_stream());
+  return out;
+}
+
 torch::Tensor sample_gumbel(torch::Tensor logits, double temperature,
                             int64_t seed) {
   TORCH_CHECK(logits.is_cuda() && logits.scalar_type() == torch::kFloat32,
@@ -369,6 +397,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_", &rope_, "in-place RoPE with host cos/sin tables");
   m.def("adamw_", &adamw_, "fused AdamW step (K9)");
   m.def("sample_gumbel", &sample_gumbel, "fused sampling (K8)");
+  m.def("conv3x3", &conv3x3, "NCHW implicit-GEMM 3x3 conv, fused bias+res (K3)");
   m.def("softmax_fwd", &softmax_fwd);
   m.def("tr16_probe", [](torch::Tensor out) {
     tr16_probe((short*)out.data_ptr(), cur_stream());

@@ -1,0 +1,200 @@
+// K3: hand-written NCHW implicit-GEMM 3x3 conv (stride 1, pad 1) on MFMA.
+//
+// Replaces MIOpen/CK for the SDXL VAE/UNet conv pyramid (reference trigger:
+// 06_gpu_and_ml/stable_diffusion/text_to_image.py:114 VAE decode path,
+// flux.py:259-261).  MIOpen's fastest solvers insert NCHW<->NHWC transpose
+// pairs (~19 ms of the 390 ms SDXL step); this kernel consumes NCHW directly
+// and does its one unavoidable transpose (x-contiguous -> c-contiguous) in
+// LDS during staging.
+//
+// Formulation: 9 shifted GEMMs.  out[k][p] = sum_tap sum_c W[tap][k][c] *
+// in[c][p+tap].  M = 64 out-channels, N = 256 pixels (8 rows x 32 cols),
+// K-dim = C in chunks of 16 (mfma_f32_32x32x16_bf16).
+//
+//  - Input patch staged per 16-channel chunk as LDS [10 rows][34 cols][16 c]
+//    (c innermost): the MFMA B-fragment read (8 consecutive c at the lane's
+//    pixel) is one ds_read_b128 at 32 B col-stride -> 64 lanes cover a
+//    contiguous span, every bank hit evenly (conflict-free; guide §6 G4
+//    applies to same-column strides, not contiguous spans).
+//  - Weights repacked on host to [9][Kpad][C16] (c contiguous) so the MFMA
+//    A-fragment is one bf16x8 global read; per-k-tile weight working set
+//    (64 x C x 9 x 2B <= 590 KB) stays L2-resident across the XCD-chunked
+//    run of pixel tiles (bijective chunk swizzle, guide §5 T1/m204).
+//  - Wave w owns output rows {2w, 2w+1} and BOTH 32-k fragments: 36 MFMA per
+//    18 ds_read_b128 per chunk (2:1, the m97 GEMM ratio).
+//  - Epilogue fuses bias add and an optional residual add (VAEResnet skip).
+//
+// The kernel is stride-1/pad-1 only; stride-2 downsample convs (3 per UNet
+// fwd) stay on the library path.
+#include "common.h"
+
+#include <cstdio>
+
+#define CV_NW 4      // waves per block
+#define CV_ROWS 8    // output rows per block
+#define CV_COLS 32   // output cols per block
+#define CV_BK 64     // output channels per block
+#define CV_CC 16     // input-channel chunk (MFMA K)
+
+typedef __attribute__((ext_vector_type(16))) float f32x16c;
+
+// LDS patch: [CV_ROWS+2][CV_COLS+2][CV_CC] bf16, c innermost.
+#define PATCH_R (CV_ROWS + 2)
+#define PATCH_C (CV_COLS + 2)
+#define PATCH_ELEMS (PATCH_R * PATCH_C * CV_CC)
+#define STAGE_SLOTS (PATCH_R * PATCH_C * 2)             // (row, col, c-oct)
+#define SLOTS_PER_T ((STAGE_SLOTS + CV_NW * WAVE - 1) / (CV_NW * WAVE))
+
+__global__ __launch_bounds__(CV_NW * WAVE) void conv3x3_kernel(
+    const short* __restrict__ in, const short* __restrict__ wr,
+    const float* __restrict__ bias, const short* __restrict__ res,
+    short* __restrict__ out, int C, int H, int W, int K, int C16, int Kpad,
+    int npix_x, int npix, int nk) {
+  __shared__ alignas(16) short patch[PATCH_ELEMS];
+
+  const int tid = threadIdx.x;
+  const int w = tid / WAVE;
+  const int l = tid % WAVE;
+  const int l31 = l & 31;
+  const int hi5 = l >> 5;
+  const int n = blockIdx.y;
+
+  // bijective XCD-chunk swizzle (m204): each XCD runs a contiguous range of
+  // block ids = consecutive pixel tiles of ONE k-tile -> weights L2-hit.
+  int bid = blockIdx.x;
+  {
+    int nwg = npix * nk;
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = bid & 7, pos = bid >> 3;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const int ktile = bid / npix;
+  const int pix = bid - ktile * npix;
+  const int y0 = (pix / npix_x) * CV_ROWS;
+  const int x0 = (pix - (pix / npix_x) * npix_x) * CV_COLS;
+  const int k0 = ktile * CV_BK;
+
+  const long long in_n = (long long)n * C * H * W;
+  const int nc = C16 / CV_CC;
+
+  // ---- staging: slot s = (row, col, c-oct); thread gathers 8 strided
+  // channel values (coalesced across lanes: consecutive threads read
+  // consecutive x) and writes ONE bf16x8 to the c-contiguous LDS slot.
+  bf16x8 sreg[SLOTS_PER_T];
+  auto stage_load = [&](int cc) {
+    const int c0 = cc * CV_CC;
+#pragma unroll
+    for (int i = 0; i < SLOTS_PER_T; ++i) {
+      int s = i * CV_NW * WAVE + tid;
+      bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (s < STAGE_SLOTS) {
+        int col = s % PATCH_C;
+        int u = s / PATCH_C;
+        int oct = u & 1;
+        int row = u >> 1;
+        int y = y0 + row - 1;
+        int x = x0 + col - 1;
+        if (y >= 0 && y < H && x >= 0 && x < W) {
+          const short* src = in + in_n + (long long)(c0 + oct * 8) * H * W +
+                             (long long)y * W + x;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            int c = c0 + oct * 8 + j;
+            v[j] = (c < C) ? src[(long long)j * H * W] : (short)0;
+          }
+        }
+      }
+      sreg[i] = v;
+    }
+  };
+  auto stage_write = [&]() {
+#pragma unroll
+    for (int i = 0; i < SLOTS_PER_T; ++i) {
+      int s = i * CV_NW * WAVE + tid;
+      if (s < STAGE_SLOTS) {
+        int col = s % PATCH_C;
+        int u = s / PATCH_C;
+        int oct = u & 1;
+        int row = u >> 1;
+        *(bf16x8*)&patch[(row * PATCH_C + col) * CV_CC + oct * 8] = sreg[i];
+      }
+    }
+  };
+
+  // accumulators: [row-pair rr][k-fragment mf]
+  f32x16c acc[2][2];
+#pragma unroll
+  for (int a = 0; a < 2; ++a)
+#pragma unroll
+    for (int b = 0; b < 2; ++b)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc[a][b][r] = 0.f;
+
+  stage_load(0);
+  for (int cc = 0; cc < nc; ++cc) {
+    __syncthreads();  // previous compute done; LDS free
+    stage_write();
+    __syncthreads();  // patch ready
+    if (cc + 1 < nc) stage_load(cc + 1);
+
+    const int cfrag = cc * CV_CC + hi5 * 8;
+#pragma unroll
+    for (int tap = 0; tap < 9; ++tap) {
+      const int dy = tap / 3, dx = tap - 3 * (tap / 3);
+      // A-fragments for both 32-k halves (L2-resident after first pixel tile)
+      const long long wbase = ((long long)tap * Kpad + k0) * C16 + cfrag;
+      bf16x8 a0 = *(const bf16x8*)&wr[wbase + (long long)l31 * C16];
+      bf16x8 a1 = *(const bf16x8*)&wr[wbase + (long long)(l31 + 32) * C16];
+#pragma unroll
+      for (int rr = 0; rr < 2; ++rr) {
+        const int row = w * 2 + rr;
+        bf16x8 b = *(const bf16x8*)&patch[((row + dy) * PATCH_C + l31 + dx) *
+                                              CV_CC + hi5 * 8];
+        __builtin_amdgcn_s_setprio(1);
+        acc[rr][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b, acc[rr][0], 0, 0, 0);
+        acc[rr][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b, acc[rr][1], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+      }
+    }
+  }
+
+  // ---- epilogue: bias (+ residual) add, masked NCHW store.
+  const int x = x0 + l31;
+  if (x < W) {
+#pragma unroll
+    for (int rr = 0; rr < 2; ++rr) {
+      const int y = y0 + w * 2 + rr;
+      if (y >= H) continue;
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int k = k0 + mf * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi5;
+          if (k < K) {
+            const long long o = ((long long)n * K + k) * H * W +
+                                (long long)y * W + x;
+            float v = acc[rr][mf][r] + bias[k];
+            if (res != nullptr) v += bf2f(res[o]);
+            out[o] = f2bf(v);
+          }
+        }
+      }
+    }
+  }
+}
+
+extern "C" void conv3x3_bf16(const void* in, const void* wrepack,
+                             const void* bias, const void* residual, void* out,
+                             int N, int C, int H, int W, int K, int C16,
+                             int Kpad, hipStream_t stream) {
+  const int npix_x = (W + CV_COLS - 1) / CV_COLS;
+  const int npix_y = (H + CV_ROWS - 1) / CV_ROWS;
+  const int npix = npix_x * npix_y;
+  const int nk = (K + CV_BK - 1) / CV_BK;
+  dim3 grid(npix * nk, N);
+  dim3 block(CV_NW * WAVE);
+  hipLaunchKernelGGL(conv3x3_kernel, grid, block, 0, stream,
+                     (const short*)in, (const short*)wrepack,
+                     (const float*)bias, (const short*)residual, (short*)out,
+                     C, H, W, K, C16, Kpad, npix_x, npix, nk);
+}

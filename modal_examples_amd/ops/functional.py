@@ -175,6 +175,37 @@ def rope_tables(max_seq: int, dim: int, base: float = 10000.0, device="cpu"):
     return (freqs.cos().float().to(device), freqs.sin().float().to(device))
 
 
+def repack_conv3x3_weight(weight: torch.Tensor) -> torch.Tensor:
+    """[K,C,3,3] conv weight → the kernel's [9, Kpad, C16] layout (c
+    contiguous per (tap, k) row so the MFMA A-fragment is one bf16x8 read;
+    K padded to the 64-channel block, C to the 16-channel chunk)."""
+    K, C = weight.shape[0], weight.shape[1]
+    kpad = (K + 63) // 64 * 64
+    c16 = (C + 15) // 16 * 16
+    wr = torch.zeros(9, kpad, c16, dtype=torch.bfloat16, device=weight.device)
+    w = weight.permute(2, 3, 0, 1).reshape(9, K, C)  # [tap, k, c]
+    wr[:, :K, :C] = w.to(torch.bfloat16)
+    return wr.contiguous()
+
+
+def conv3x3(x, wr, bias, K: int, residual=None, raw_weight=None):
+    """3x3 stride-1 pad-1 conv (K3). x [N,C,H,W] bf16; wr from
+    repack_conv3x3_weight; bias fp32 [K]; optional fused residual add.
+    raw_weight [K,C,3,3] drives the CPU/autograd reference path."""
+    ext = _ext_for(x, raw_weight if raw_weight is not None else x)
+    if ext is None:
+        assert raw_weight is not None, "reference conv path needs raw_weight"
+        y = torch.nn.functional.conv2d(
+            x.float(), raw_weight.float(), bias.float(), padding=1)
+        if residual is not None:
+            y = y + residual.float()
+        return y.to(x.dtype)
+    out = ext.conv3x3(x.contiguous(), wr, bias,
+                      residual.contiguous() if residual is not None else None,
+                      K)
+    return out
+
+
 def adamw_step(p, g, m, v, lr, beta1=0.9, beta2=0.999, eps=1e-8, wd=0.01, step=1):
     """Fused AdamW (K9); in-place on p/m/v."""
     ext = _ext_for(p, any_dtype=True)

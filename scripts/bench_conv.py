@@ -58,6 +58,55 @@ def gemm_exp():
         print(f"  64x{k} @ {k}x{n}: {dt:6.3f} ms  {gb/dt*1000:6.0f} GB/s")
 
 
+
+
+def kernel_exp():
+    """Hand conv3x3 kernel (K3) vs MIOpen NCHW per hot shape + TF rate."""
+    import math
+
+    from modal_examples_amd.gpu import kernel_cache
+    from modal_examples_amd.ops import functional as F
+
+    kernel_cache.restore()
+    shapes = [  # (B, Cin, Cout, H, W)
+        (4, 128, 128, 1024, 1024),
+        (4, 256, 256, 512, 512),
+        (4, 512, 512, 256, 256),
+        (4, 512, 512, 128, 128),
+        (4, 512, 256, 512, 512),
+        (4, 320, 320, 128, 128),
+        (4, 640, 640, 64, 64),
+        (4, 1280, 1280, 32, 32),
+        (4, 4, 512, 128, 128),
+    ]
+    print(f"{'shape':26s} {'miopen':>8s} {'hand':>8s} {'speedup':>8s} {'TF':>6s}")
+    tot_mi = tot_hand = 0.0
+    for B, Ci, Co, H, W in shapes:
+        x = torch.randn(B, Ci, H, W, device="cuda", dtype=torch.bfloat16)
+        w = torch.randn(Co, Ci, 3, 3, device="cuda",
+                        dtype=torch.bfloat16) / math.sqrt(Ci * 9)
+        b = torch.randn(Co, device="cuda", dtype=torch.float32)
+        conv = torch.nn.Conv2d(Ci, Co, 3, padding=1).to("cuda", torch.bfloat16)
+        with torch.no_grad():
+            conv.weight.copy_(w)
+            conv.bias.copy_(b.to(torch.bfloat16))
+        t_mi = timeit(lambda: conv(x))
+        wr = F.repack_conv3x3_weight(w)
+        t_k = timeit(lambda: F.conv3x3(x, wr, b, Co, raw_weight=w))
+        flops = 2.0 * B * Ci * Co * 9 * H * W
+        tf = flops / (t_k * 1e-3) / 1e12
+        tot_mi += t_mi
+        tot_hand += t_k
+        print(f"{B}x{Ci}->{Co}@{H:<4}x{W:<6} {t_mi:8.3f} {t_k:8.3f} "
+              f"{t_mi / t_k:7.2f}x {tf:6.0f}")
+    print(f"{'TOTAL':26s} {tot_mi:8.3f} {tot_hand:8.3f} {tot_mi/tot_hand:7.2f}x")
+
+
 if __name__ == "__main__":
-    conv_exp()
-    gemm_exp()
+    import sys as _sys
+
+    if "--kernel" in _sys.argv:
+        kernel_exp()
+    else:
+        conv_exp()
+        gemm_exp()

@@ -282,3 +282,38 @@ def test_guard_band_around_inplace_rope():
         assert torch.equal(y2, OF.rope(ref.clone(), cos, sin))
     finally:
         os.environ.pop("MODAL_AMD_DEBUG_SYNC", None)
+
+
+@requires_gpu
+@pytest.mark.parametrize("shape", [
+    (2, 64, 64, 32, 32),      # generic
+    (1, 512, 512, 64, 64),    # VAE-class
+    (2, 4, 320, 16, 32),      # conv_in: tiny C, padded chunk
+    (1, 128, 3, 24, 32),      # conv_out: K < 32, masked stores
+    (1, 96, 64, 19, 45),      # ragged H/W: edge masking both dims
+])
+def test_conv3x3_vs_fp32(shape):
+    """K3 NCHW implicit-GEMM conv vs torch fp32 conv2d."""
+    N, C, K, H, W = shape
+    torch.manual_seed(0)
+    x = torch.randn(N, C, H, W, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(K, C, 3, 3, device="cuda", dtype=torch.bfloat16) / math.sqrt(C * 9)
+    b = torch.randn(K, device="cuda", dtype=torch.float32)
+    wr = F.repack_conv3x3_weight(w)
+    got = F.conv3x3(x, wr, b.contiguous(), K, raw_weight=w)
+    want = torch.nn.functional.conv2d(x.float(), w.float(), b, padding=1)
+    _close(got, want, atol=5e-2, rtol=5e-2)
+
+
+@requires_gpu
+def test_conv3x3_fused_residual():
+    N, C, K, H, W = 2, 64, 64, 16, 32
+    torch.manual_seed(1)
+    x = torch.randn(N, C, H, W, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(K, C, 3, 3, device="cuda", dtype=torch.bfloat16) / math.sqrt(C * 9)
+    b = torch.zeros(K, device="cuda", dtype=torch.float32)
+    r = torch.randn(N, K, H, W, device="cuda", dtype=torch.bfloat16)
+    wr = F.repack_conv3x3_weight(w)
+    got = F.conv3x3(x, wr, b, K, residual=r, raw_weight=w)
+    want = torch.nn.functional.conv2d(x.float(), w.float(), b, padding=1) + r.float()
+    _close(got, want, atol=5e-2, rtol=5e-2)

@@ -58,3 +58,37 @@ def test_pipeline_small_cpu_end_to_end():
     # determinism with same seed
     imgs2 = pipe.generate(["test prompt"], steps=2)
     assert torch.equal(imgs, imgs2)
+
+
+def test_conv3x3_module_cpu_fallback_matches_conv2d():
+    """Conv3x3 module on CPU == nn.Conv2d with the same params (the kernel
+    path is GPU-only; CPU uses F.conv2d)."""
+    import torch
+
+    from modal_examples_amd.models.sdxl.layers import Conv1x1, Conv3x3
+
+    torch.manual_seed(0)
+    m = Conv3x3(8, 16).eval()
+    x = torch.randn(2, 8, 12, 12)
+    want = torch.nn.functional.conv2d(x, m.weight, m.bias, padding=1)
+    assert torch.allclose(m(x), want, atol=1e-6)
+    r = torch.randn(2, 16, 12, 12)
+    assert torch.allclose(m(x, residual=r), want + r, atol=1e-6)
+    s = Conv1x1(8, 16).eval()
+    assert torch.allclose(
+        s(x), torch.nn.functional.conv2d(x, s.weight, s.bias), atol=1e-6)
+
+
+def test_repack_conv3x3_weight_layout():
+    import torch
+
+    from modal_examples_amd.ops import functional as OF
+
+    w = torch.arange(5 * 7 * 9, dtype=torch.float32).reshape(5, 7, 3, 3)
+    wr = OF.repack_conv3x3_weight(w)
+    assert wr.shape == (9, 64, 16)
+    # wr[tap, k, c] == w[k, c, tap//3, tap%3]
+    for tap in (0, 4, 8):
+        assert torch.equal(wr[tap, :5, :7],
+                           w[:, :, tap // 3, tap % 3].to(torch.bfloat16))
+    assert (wr[:, 5:, :] == 0).all() and (wr[:, :, 7:] == 0).all()
