@@ -110,7 +110,8 @@ class Upsample(nn.Module):
         self.conv = Conv3x3(c, c)
 
     def forward(self, x):
-        return self.conv(torch.nn.functional.interpolate(x, scale_factor=2.0, mode="nearest"))
+        # nearest-2x fused into the conv's staging read (K3 UP variant)
+        return self.conv(x, upsample=True)
 
 
 class UNetXL(nn.Module):

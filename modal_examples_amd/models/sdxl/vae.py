@@ -87,8 +87,7 @@ class VAEDecoder(nn.Module):
             for blk in blocks:
                 h = blk(h)
             if len(ups):
-                h = torch.nn.functional.interpolate(h, scale_factor=2.0, mode="nearest")
-                h = ups[0](h)
+                h = ups[0](h, upsample=True)  # fused nearest-2x + conv
         return self.conv_out(self.norm_out(h))
 
 

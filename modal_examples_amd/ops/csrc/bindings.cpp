@@ -49,7 +49,7 @@ void cfg_euler_dev_bf16(const void*, const void*, const void*, void*,
                         hipStream_t);
 void advance_step(long long*, hipStream_t);
 void conv3x3_bf16(const void*, const void*, const void*, const void*, void*,
-                  int, int, int, int, int, int, int, hipStream_t);
+                  int, int, int, int, int, int, int, int, hipStream_t);
 }
 
 namespace {
@@ -253,7 +253,8 @@ void adamw_(torch::Tensor p, torch::Tensor g, torch::Tensor m, torch::Tensor v,
 // ---------------------------------------------------------------- conv (K3)
 
 torch::Tensor conv3x3(torch::Tensor x, torch::Tensor wr, torch::Tensor bias,
-                      c10::optional<torch::Tensor> residual, int64_t K) {
+                      c10::optional<torch::Tensor> residual, int64_t K,
+                      bool upsample) {
   check_bf16(x, "x");
   check_bf16(wr, "wr");
   TORCH_CHECK(x.dim() == 4, "x must be NCHW");
@@ -262,6 +263,7 @@ torch::Tensor conv3x3(torch::Tensor x, torch::Tensor wr, torch::Tensor bias,
   TORCH_CHECK(bias.scalar_type() == torch::kFloat && bias.is_contiguous(),
               "bias must be fp32 contiguous");
   int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  if (upsample) { H *= 2; W *= 2; }
   int Kpad = wr.size(1), C16 = wr.size(2);
   TORCH_CHECK(K <= Kpad && C <= C16, "weight pack smaller than conv");
   auto out = torch::empty({N, K, H, W}, x.options());
@@ -272,7 +274,8 @@ torch::Tensor conv3x3(torch::Tensor x, torch::Tensor wr, torch::Tensor bias,
     resp = residual->data_ptr();
   }
   conv3x3_bf16(x.data_ptr(), wr.data_ptr(), bias.data_ptr(), resp,
-               out.data_ptr(), N, C, H, W, (int)K, C16, Kpad, cur_stream());
+               out.data_ptr(), N, C, H, W, (int)K, C16, Kpad,
+               upsample ? 1 : 0, cur_stream());
   return out;
 }
 

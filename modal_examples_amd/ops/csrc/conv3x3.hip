@@ -45,12 +45,25 @@ typedef __attribute__((ext_vector_type(16))) float f32x16c;
 #define STAGE_SLOTS (PATCH_R * PATCH_C * 2)             // (row, col, c-oct)
 #define SLOTS_PER_T ((STAGE_SLOTS + CV_NW * WAVE - 1) / (CV_NW * WAVE))
 
+// weight tile in LDS: [9 taps][CV_BK k][CV_CC c], staged once per c-chunk and
+// shared by all 4 waves (per-wave global A-reads were 4x-redundant L2
+// traffic — the v1 bottleneck).
+#define WLDS_ELEMS (9 * CV_BK * CV_CC)
+#define WSLOTS (WLDS_ELEMS / 8)  // bf16x8 units
+#define WSLOTS_PER_T ((WSLOTS + CV_NW * WAVE - 1) / (CV_NW * WAVE))
+
+// UP: fused nearest-2x upsample — the conv reads the half-res source
+// directly (out pixel (y,x) <- src[y/2][x/2]), eliminating the materialized
+// F.interpolate pass before every Upsample conv.  H/W are OUTPUT dims;
+// Hs/Ws the source's.
+template <bool UP>
 __global__ __launch_bounds__(CV_NW * WAVE) void conv3x3_kernel(
     const short* __restrict__ in, const short* __restrict__ wr,
     const float* __restrict__ bias, const short* __restrict__ res,
-    short* __restrict__ out, int C, int H, int W, int K, int C16, int Kpad,
-    int npix_x, int npix, int nk) {
+    short* __restrict__ out, int C, int H, int W, int Hs, int Ws, int K,
+    int C16, int Kpad, int npix_x, int npix, int nk) {
   __shared__ alignas(16) short patch[PATCH_ELEMS];
+  __shared__ alignas(16) short wlds[WLDS_ELEMS];
 
   const int tid = threadIdx.x;
   const int w = tid / WAVE;
@@ -74,7 +87,7 @@ __global__ __launch_bounds__(CV_NW * WAVE) void conv3x3_kernel(
   const int x0 = (pix - (pix / npix_x) * npix_x) * CV_COLS;
   const int k0 = ktile * CV_BK;
 
-  const long long in_n = (long long)n * C * H * W;
+  const long long in_n = (long long)n * C * Hs * Ws;
   const int nc = C16 / CV_CC;
 
   // ---- staging: slot s = (row, col, c-oct); thread gathers 8 strided
@@ -95,12 +108,15 @@ __global__ __launch_bounds__(CV_NW * WAVE) void conv3x3_kernel(
         int y = y0 + row - 1;
         int x = x0 + col - 1;
         if (y >= 0 && y < H && x >= 0 && x < W) {
-          const short* src = in + in_n + (long long)(c0 + oct * 8) * H * W +
-                             (long long)y * W + x;
+          const int sy = UP ? (y >> 1) : y;
+          const int sx = UP ? (x >> 1) : x;
+          const long long hw = (long long)Hs * Ws;
+          const short* src = in + in_n + (long long)(c0 + oct * 8) * hw +
+                             (long long)sy * Ws + sx;
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
             int c = c0 + oct * 8 + j;
-            v[j] = (c < C) ? src[(long long)j * H * W] : (short)0;
+            v[j] = (c < C) ? src[(long long)j * hw] : (short)0;
           }
         }
       }
@@ -121,6 +137,32 @@ __global__ __launch_bounds__(CV_NW * WAVE) void conv3x3_kernel(
     }
   };
 
+  // weight staging: slot = (tap, k-row, c-oct); one bf16x8 each.
+  bf16x8 wreg[WSLOTS_PER_T];
+  auto wstage_load = [&](int cc) {
+    const int c0 = cc * CV_CC;
+#pragma unroll
+    for (int i = 0; i < WSLOTS_PER_T; ++i) {
+      int s = i * CV_NW * WAVE + tid;
+      if (s < WSLOTS) {
+        int oct = s & 1;
+        int kr = (s >> 1) & (CV_BK - 1);
+        int tap = (s >> 1) >> 6;  // 64 k-rows per tap
+        wreg[i] = *(const bf16x8*)&wr[((long long)tap * Kpad + k0 + kr) * C16 +
+                                      c0 + oct * 8];
+      }
+    }
+  };
+  auto wstage_write = [&]() {
+#pragma unroll
+    for (int i = 0; i < WSLOTS_PER_T; ++i) {
+      int s = i * CV_NW * WAVE + tid;
+      if (s < WSLOTS) {
+        *(bf16x8*)&wlds[s * 8] = wreg[i];
+      }
+    }
+  };
+
   // accumulators: [row-pair rr][k-fragment mf]
   f32x16c acc[2][2];
 #pragma unroll
@@ -131,20 +173,22 @@ __global__ __launch_bounds__(CV_NW * WAVE) void conv3x3_kernel(
       for (int r = 0; r < 16; ++r) acc[a][b][r] = 0.f;
 
   stage_load(0);
+  wstage_load(0);
   for (int cc = 0; cc < nc; ++cc) {
     __syncthreads();  // previous compute done; LDS free
     stage_write();
-    __syncthreads();  // patch ready
-    if (cc + 1 < nc) stage_load(cc + 1);
+    wstage_write();
+    __syncthreads();  // patch + weights ready
+    if (cc + 1 < nc) {
+      stage_load(cc + 1);
+      wstage_load(cc + 1);
+    }
 
-    const int cfrag = cc * CV_CC + hi5 * 8;
 #pragma unroll
     for (int tap = 0; tap < 9; ++tap) {
       const int dy = tap / 3, dx = tap - 3 * (tap / 3);
-      // A-fragments for both 32-k halves (L2-resident after first pixel tile)
-      const long long wbase = ((long long)tap * Kpad + k0) * C16 + cfrag;
-      bf16x8 a0 = *(const bf16x8*)&wr[wbase + (long long)l31 * C16];
-      bf16x8 a1 = *(const bf16x8*)&wr[wbase + (long long)(l31 + 32) * C16];
+      bf16x8 a0 = *(const bf16x8*)&wlds[(tap * CV_BK + l31) * CV_CC + hi5 * 8];
+      bf16x8 a1 = *(const bf16x8*)&wlds[(tap * CV_BK + 32 + l31) * CV_CC + hi5 * 8];
 #pragma unroll
       for (int rr = 0; rr < 2; ++rr) {
         const int row = w * 2 + rr;
@@ -186,15 +230,24 @@ __global__ __launch_bounds__(CV_NW * WAVE) void conv3x3_kernel(
 extern "C" void conv3x3_bf16(const void* in, const void* wrepack,
                              const void* bias, const void* residual, void* out,
                              int N, int C, int H, int W, int K, int C16,
-                             int Kpad, hipStream_t stream) {
+                             int Kpad, int upsample, hipStream_t stream) {
+  // H/W are the OUTPUT dims; with upsample the source is H/2 x W/2.
+  const int Hs = upsample ? H / 2 : H;
+  const int Ws = upsample ? W / 2 : W;
   const int npix_x = (W + CV_COLS - 1) / CV_COLS;
   const int npix_y = (H + CV_ROWS - 1) / CV_ROWS;
   const int npix = npix_x * npix_y;
   const int nk = (K + CV_BK - 1) / CV_BK;
   dim3 grid(npix * nk, N);
   dim3 block(CV_NW * WAVE);
-  hipLaunchKernelGGL(conv3x3_kernel, grid, block, 0, stream,
-                     (const short*)in, (const short*)wrepack,
-                     (const float*)bias, (const short*)residual, (short*)out,
-                     C, H, W, K, C16, Kpad, npix_x, npix, nk);
+  if (upsample)
+    hipLaunchKernelGGL(conv3x3_kernel<true>, grid, block, 0, stream,
+                       (const short*)in, (const short*)wrepack,
+                       (const float*)bias, (const short*)residual, (short*)out,
+                       C, H, W, Hs, Ws, K, C16, Kpad, npix_x, npix, nk);
+  else
+    hipLaunchKernelGGL(conv3x3_kernel<false>, grid, block, 0, stream,
+                       (const short*)in, (const short*)wrepack,
+                       (const float*)bias, (const short*)residual, (short*)out,
+                       C, H, W, Hs, Ws, K, C16, Kpad, npix_x, npix, nk);
 }

@@ -188,21 +188,27 @@ def repack_conv3x3_weight(weight: torch.Tensor) -> torch.Tensor:
     return wr.contiguous()
 
 
-def conv3x3(x, wr, bias, K: int, residual=None, raw_weight=None):
+def conv3x3(x, wr, bias, K: int, residual=None, raw_weight=None,
+            upsample: bool = False):
     """3x3 stride-1 pad-1 conv (K3). x [N,C,H,W] bf16; wr from
-    repack_conv3x3_weight; bias fp32 [K]; optional fused residual add.
+    repack_conv3x3_weight; bias fp32 [K]; optional fused residual add;
+    upsample=True fuses a nearest-2x upsample of x into the conv read.
     raw_weight [K,C,3,3] drives the CPU/autograd reference path."""
     ext = _ext_for(x, raw_weight if raw_weight is not None else x)
     if ext is None:
         assert raw_weight is not None, "reference conv path needs raw_weight"
+        xin = x
+        if upsample:
+            xin = torch.nn.functional.interpolate(x, scale_factor=2.0,
+                                                  mode="nearest")
         y = torch.nn.functional.conv2d(
-            x.float(), raw_weight.float(), bias.float(), padding=1)
+            xin.float(), raw_weight.float(), bias.float(), padding=1)
         if residual is not None:
             y = y + residual.float()
         return y.to(x.dtype)
     out = ext.conv3x3(x.contiguous(), wr, bias,
                       residual.contiguous() if residual is not None else None,
-                      K)
+                      K, upsample)
     return out
 
 

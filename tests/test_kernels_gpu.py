@@ -317,3 +317,19 @@ def test_conv3x3_fused_residual():
     got = F.conv3x3(x, wr, b, K, residual=r, raw_weight=w)
     want = torch.nn.functional.conv2d(x.float(), w.float(), b, padding=1) + r.float()
     _close(got, want, atol=5e-2, rtol=5e-2)
+
+
+@requires_gpu
+def test_conv3x3_fused_upsample():
+    """UP variant == interpolate(nearest,2x) + conv2d."""
+    N, C, K, H, W = 2, 32, 48, 12, 16
+    torch.manual_seed(2)
+    x = torch.randn(N, C, H, W, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(K, C, 3, 3, device="cuda", dtype=torch.bfloat16) / math.sqrt(C * 9)
+    b = torch.randn(K, device="cuda", dtype=torch.float32)
+    wr = F.repack_conv3x3_weight(w)
+    got = F.conv3x3(x, wr, b, K, raw_weight=w, upsample=True)
+    up = torch.nn.functional.interpolate(x.float(), scale_factor=2.0, mode="nearest")
+    want = torch.nn.functional.conv2d(up, w.float(), b, padding=1)
+    assert got.shape == want.shape
+    _close(got, want, atol=5e-2, rtol=5e-2)
