@@ -3,66 +3,57 @@
 # ---
 # # Multi-GPU process groups (the clusters example, single-node analog)
 #
-# `@modal.experimental.clustered(size=n)` runs n simultaneous ranks;
-# `get_cluster_info()` provides rank + peer IPs (loopback on one node), and
-# torch.distributed rendezvous over 127.0.0.1 — RCCL over xGMI between GPUs,
-# gloo on CPU.  The payload: rank 0 broadcasts a tensor, every rank verifies.
-
-import os
-import subprocess
-import sys
+# `@modal.experimental.clustered(size=n)` makes `fn.remote()` launch n
+# simultaneous rank containers — the RUNTIME does the fan-out (reference
+# contract: 14_clusters/simple_torch_cluster.py:96-130): each rank worker gets
+# RANK/WORLD_SIZE/MASTER_ADDR/MASTER_PORT injected, `get_cluster_info()`
+# reports rank + peer IPs (loopback on one node), and the caller receives
+# rank 0's return value once every rank finished.  torch.distributed
+# rendezvous over 127.0.0.1 — RCCL over xGMI between pinned GPUs, gloo on CPU.
 
 import modal_examples_amd as modal
 
 app = modal.App("example-torch-cluster")
 
-WORLD = int(os.environ.get("CLUSTER_WORLD", "2"))
-
-WORKER = r"""
-import os, torch, torch.distributed as dist
-from modal_examples_amd.parallel.cluster import get_cluster_info
-
-info = get_cluster_info()
-rank, world = info.rank, len(info.container_ips)
-backend = "nccl" if torch.cuda.is_available() else "gloo"
-dist.init_process_group(backend, rank=rank, world_size=world,
-                        init_method=f"tcp://127.0.0.1:{os.environ['CLUSTER_PORT']}")
-dev = f"cuda:{rank}" if backend == "nccl" else "cpu"
-if backend == "nccl":
-    torch.cuda.set_device(rank)
-t = torch.full((4,), float(rank), device=dev)
-if rank == 0:
-    t.fill_(42.0)
-dist.broadcast(t, src=0)
-assert t.eq(42).all(), t
-dist.barrier()
-print(f"rank {rank}/{world} on {dev}: broadcast ok -> {t.tolist()}")
-dist.destroy_process_group()
-"""
+WORLD = 2
 
 
-@app.function(gpu=f"mi355x:{WORLD}", timeout=300)
-def run_cluster() -> int:
-    """Launches one process per GPU (torchrun-style) inside the allocation."""
-    from modal_examples_amd.parallel.cluster import free_port
+def _gpu_request():
+    try:
+        import torch
 
-    port = free_port()
-    procs = []
-    for rank in range(WORLD):
-        env = dict(os.environ)
-        env.update({
-            "MODAL_AMD_CLUSTER_RANK": str(rank),
-            "MODAL_AMD_CLUSTER_SIZE": str(WORLD),
-            "CLUSTER_PORT": str(port),
-            "MASTER_ADDR": "127.0.0.1",
-        })
-        procs.append(subprocess.Popen([sys.executable, "-c", WORKER], env=env))
-    rcs = [p.wait(timeout=240) for p in procs]
-    assert all(rc == 0 for rc in rcs), rcs
-    return WORLD
+        return f"mi355x:1" if torch.cuda.device_count() >= WORLD else None
+    except Exception:
+        return None
+
+
+@app.function(gpu=_gpu_request(), timeout=300)
+@modal.experimental.clustered(size=WORLD)
+def all_ranks_broadcast() -> str:
+    """Runs in EVERY rank container; rank 0's return value is the caller's."""
+    import torch
+    import torch.distributed as dist
+
+    from modal_examples_amd.parallel.cluster import get_cluster_info
+
+    info = get_cluster_info()
+    rank, world = info.rank, len(info.container_ips)
+    use_gpu = torch.cuda.is_available()
+    backend = "nccl" if use_gpu else "gloo"  # "nccl" IS RCCL on ROCm
+    dist.init_process_group(backend, init_method="env://")
+    dev = "cuda:0" if use_gpu else "cpu"  # each rank sees only its pinned GPU
+    t = torch.full((4,), float(rank), device=dev)
+    if rank == 0:
+        t.fill_(42.0)
+    dist.broadcast(t, src=0)
+    assert t.eq(42).all(), t
+    dist.barrier()
+    print(f"rank {rank}/{world} [{backend}] broadcast ok -> {t.tolist()}")
+    dist.destroy_process_group()
+    return f"rank0 of {world} [{backend}]"
 
 
 @app.local_entrypoint()
 def main():
-    n = run_cluster.remote()
-    print(f"cluster of {n} ranks completed the broadcast")
+    out = all_ranks_broadcast.remote()
+    print(f"cluster done: {out}")

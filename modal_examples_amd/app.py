@@ -439,6 +439,51 @@ class FunctionOptions:
         return out
 
 
+class ClusterCall:
+    """Aggregate of one Call per cluster rank: resolves to rank 0's result
+    once EVERY rank finished; any rank's failure propagates.  Quacks like a
+    Call for FunctionCall/_remote (id/wait/gen_q/pool)."""
+
+    def __init__(self, calls):
+        self.calls = calls
+        self.id = calls[0].id
+        self.gen_q = None
+        self.spawned = calls[0].spawned
+        self.pool = self  # FunctionCall.cancel reaches us via call.pool
+
+    def wait(self, timeout: Optional[float] = None):
+        deadline = None if timeout is None else time.monotonic() + timeout
+        first_exc = None
+        for c in self.calls:
+            left = None if deadline is None else max(0.0, deadline - time.monotonic())
+            try:
+                c.wait(left)
+            except BaseException as e:  # noqa: BLE001 — collect, drain the rest
+                if first_exc is None:
+                    first_exc = e
+        if first_exc is not None:
+            raise first_exc
+        return self.calls[0].value
+
+    def cancel(self, call_id=None) -> bool:
+        any_live = False
+        for c in self.calls:
+            if getattr(c, "pool", None) is not None and c.pool.cancel(c.id):
+                any_live = True
+        return any_live
+
+    @property
+    def exc(self):
+        for c in self.calls:
+            if c.exc is not None:
+                return c.exc
+        return None
+
+    @property
+    def value(self):
+        return self.calls[0].value
+
+
 class Function:
     """A decorated function bound to an autoscaling worker pool."""
 
@@ -454,6 +499,9 @@ class Function:
         self.is_generator = inspect.isgeneratorfunction(raw)
         self.is_batched = bool(flags.get("batched"))
         self.max_inputs = int(flags.get("max_inputs", 1)) if flags.get("concurrent") else 1
+        self.is_clustered = bool(flags.get("clustered"))
+        self.cluster_size = int(flags.get("cluster_size", 1) or 1)
+        self._cluster_pools = None
         self._flags = flags
         self._install_verbs()
 
@@ -472,8 +520,8 @@ class Function:
         the per-process runtime state (lock, pool, batcher, bound verbs);
         the unpickled copy rebuilds them lazily in its own process."""
         d = dict(self.__dict__)
-        for k in ("_lock", "_pool", "_batcher", "remote", "local", "spawn",
-                  "map", "starmap", "for_each", "remote_gen"):
+        for k in ("_lock", "_pool", "_batcher", "_cluster_pools", "remote",
+                  "local", "spawn", "map", "starmap", "for_each", "remote_gen"):
             d.pop(k, None)
         return d
 
@@ -481,6 +529,7 @@ class Function:
         self.__dict__.update(d)
         self._pool = None
         self._batcher = None
+        self._cluster_pools = None
         self._lock = threading.Lock()
         self._install_verbs()
 
@@ -515,18 +564,45 @@ class Function:
                 mem_snap = bool(self.opts.enable_memory_snapshot)
                 gpu_snap = bool((self.opts.experimental_options or {}).get("enable_gpu_snapshot"))
 
-                def make_spec(devices):
-                    return ipc.ServiceSpec(
-                        app_name=app_name,
-                        name=fn_name,
-                        target_blob=target_blob,
-                        max_inputs=max_inputs,
-                        gpu_devices=devices,
-                        env=dict(env),
-                        volumes=vols,
-                        enable_memory_snapshot=mem_snap,
-                        enable_gpu_snapshot=gpu_snap,
-                    )
+                def mk_make_spec(extra_env):
+                    merged = dict(env)
+                    merged.update(extra_env or {})
+
+                    def make_spec(devices):
+                        return ipc.ServiceSpec(
+                            app_name=app_name,
+                            name=fn_name,
+                            target_blob=target_blob,
+                            max_inputs=max_inputs,
+                            gpu_devices=devices,
+                            env=dict(merged),
+                            volumes=vols,
+                            enable_memory_snapshot=mem_snap,
+                            enable_gpu_snapshot=gpu_snap,
+                        )
+
+                    return make_spec
+
+                cluster_env = {}
+                if self.is_clustered:
+                    from .parallel.cluster import free_port
+
+                    port = str(free_port())
+                    cluster_env = {
+                        "MODAL_AMD_CLUSTER_SIZE": str(self.cluster_size),
+                        "MODAL_AMD_CLUSTER_PORT": port,
+                        "MASTER_ADDR": "127.0.0.1",
+                        "MASTER_PORT": port,
+                        "WORLD_SIZE": str(self.cluster_size),
+                    }
+
+                def rank_env(r):
+                    e = dict(cluster_env)
+                    e["MODAL_AMD_CLUSTER_RANK"] = str(r)
+                    e["RANK"] = str(r)
+                    return e
+
+                make_spec = mk_make_spec(rank_env(0) if self.is_clustered else {})
 
                 retries = self.opts.retries
                 if isinstance(retries, Retries):
@@ -541,7 +617,8 @@ class Function:
                     make_spec=make_spec,
                     gpu_count=gpu_count,
                     min_containers=self.opts.min_containers,
-                    max_containers=self.opts.max_containers or default_max,
+                    max_containers=(1 if self.is_clustered
+                                    else self.opts.max_containers or default_max),
                     buffer_containers=self.opts.buffer_containers,
                     scaledown_window=self.opts.scaledown_window,
                     timeout=self.opts.timeout,
@@ -553,6 +630,25 @@ class Function:
                 )
                 self._pool.on_spawned_result = _persist_spawned
                 rt.register(self._pool)
+                if self.is_clustered and self.cluster_size > 1:
+                    # ranks 1..n-1: one dedicated single-container pool each,
+                    # same spec but rank-specific rendezvous env
+                    # (14_clusters/simple_torch_cluster.py:96-130 contract)
+                    self._cluster_pools = []
+                    for r in range(1, self.cluster_size):
+                        p = Pool(
+                            name=f"{self.app.name}.{self.name}.rank{r}",
+                            make_spec=mk_make_spec(rank_env(r)),
+                            gpu_count=gpu_count,
+                            min_containers=0,
+                            max_containers=1,
+                            timeout=self.opts.timeout,
+                            retries=RetryPolicy(),
+                            max_inputs_per_worker=max_inputs,
+                            device_pool=rt.device_pool,
+                        )
+                        rt.register(p)
+                        self._cluster_pools.append(p)
                 if self.is_batched:
                     self._batcher = Batcher(
                         self._pool,
@@ -616,6 +712,14 @@ class Function:
 
     def _submit(self, args, kwargs, is_gen=False, spawned=False, sticky_key=None) -> Call:
         pool = self._get_pool()
+        if self.is_clustered and self.cluster_size > 1:
+            if is_gen:
+                raise InvalidError("clustered functions cannot be generators")
+            # launch ALL ranks simultaneously; the caller sees rank 0's result
+            calls = [pool.submit("", args, kwargs, spawned=spawned)]
+            for p in self._cluster_pools or []:
+                calls.append(p.submit("", args, kwargs, spawned=spawned))
+            return ClusterCall(calls)
         if self.is_batched:
             return self._batcher.enqueue(args, kwargs)
         return pool.submit("", args, kwargs, is_gen=is_gen, spawned=spawned,

@@ -34,9 +34,12 @@ def get_cluster_info() -> ClusterInfo:
 def clustered(size: int, rdma: bool = False):
     """Decorator marking a function to run as ``size`` simultaneous ranks.
 
-    ``fn.remote()`` then launches all ranks (one worker process each, one GPU
-    each by default) and returns rank 0's result.
-    """
+    ``fn.remote()`` launches all ranks — one dedicated worker container per
+    rank with RANK / WORLD_SIZE / MASTER_ADDR=127.0.0.1 / MASTER_PORT and the
+    MODAL_AMD_CLUSTER_* env injected — waits for every rank and returns
+    rank 0's result (any rank's exception propagates).  ``.spawn()`` returns
+    a FunctionCall over the whole gang.  Wiring lives in
+    ``app.Function._submit`` + ``app.ClusterCall``."""
 
     def deco(fn):
         from ..app import _set_flag

@@ -4,6 +4,8 @@ reproduce the full single-process result bit-for-bit-ish over gloo world 2
 import os
 import socket
 
+import pytest
+
 import torch
 
 
@@ -95,3 +97,69 @@ def test_tp_single_process_degenerates_to_local():
     assert torch.allclose(row(x), lin(x), atol=1e-6)
     mlp = TPMLP(8, 16)
     assert mlp(x).shape == (2, 8)
+
+
+def test_clustered_launches_all_ranks_gloo():
+    """@clustered(size=4): fn.remote() launches 4 rank workers with injected
+    rendezvous env; a gloo all_reduce proves they form one process group;
+    caller gets rank 0's result (14_clusters contract, now runtime-wired)."""
+    import modal_examples_amd as modal
+
+    app = modal.App("test-clustered-gloo")
+
+    @app.function(timeout=120)
+    @modal.experimental.clustered(size=4)
+    def rank_sum() -> float:
+        import os
+
+        import torch
+        import torch.distributed as dist
+
+        from modal_examples_amd.parallel.cluster import get_cluster_info
+
+        info = get_cluster_info()
+        assert len(info.container_ips) == 4
+        assert os.environ["MASTER_ADDR"] == "127.0.0.1"
+        dist.init_process_group("gloo", init_method="env://")
+        t = torch.tensor([float(info.rank)])
+        dist.all_reduce(t)
+        dist.barrier()
+        dist.destroy_process_group()
+        return float(t.item()) + 100.0 * info.rank  # rank-dependent result
+
+    out = rank_sum.remote()
+    assert out == 6.0  # 0+1+2+3, rank 0's value (no +100 offset)
+
+
+def test_clustered_rank_failure_propagates():
+    import modal_examples_amd as modal
+
+    app = modal.App("test-clustered-fail")
+
+    @app.function(timeout=60, retries=0)
+    @modal.experimental.clustered(size=2)
+    def fail_on_rank1() -> int:
+        from modal_examples_amd.parallel.cluster import get_cluster_info
+
+        if get_cluster_info().rank == 1:
+            raise RuntimeError("rank 1 exploded")
+        return 7
+
+    with pytest.raises(Exception, match="rank 1 exploded"):
+        fail_on_rank1.remote()
+
+
+def test_clustered_spawn_gather():
+    import modal_examples_amd as modal
+
+    app = modal.App("test-clustered-spawn")
+
+    @app.function(timeout=60)
+    @modal.experimental.clustered(size=2)
+    def whoami() -> int:
+        from modal_examples_amd.parallel.cluster import get_cluster_info
+
+        return get_cluster_info().rank
+
+    fc = whoami.spawn()
+    assert fc.get(timeout=60) == 0
