@@ -231,3 +231,73 @@ def test_proxy_auth_enforced_all_kinds(monkeypatch):
             assert r.status_code == 200 and r.json() == {"m": "q"}
 
     asyncio.run(go())
+
+
+def test_websocket_through_asgi_mount():
+    """WebSocket serving: a WS route on a mounted @modal.asgi_app works
+    through the ingress (streaming-audio family dependency,
+    streaming_kyutai_stt.py:334-390)."""
+    app_ws = modal.App("test-web-ws")
+
+    @app_ws.function()
+    @modal.asgi_app(label="wsapp")
+    def wsapp():
+        from fastapi import FastAPI, WebSocket
+
+        w = FastAPI()
+
+        @w.websocket("/echo")
+        async def echo(ws: WebSocket):
+            await ws.accept()
+            while True:
+                msg = await ws.receive_text()
+                if msg == "bye":
+                    break
+                await ws.send_text(msg.upper())
+            await ws.close()
+
+        return w
+
+    from starlette.testclient import TestClient
+
+    root = build_ingress_app(app_ws)
+    with TestClient(root) as c:
+        with c.websocket_connect("/wsapp/echo") as ws:
+            ws.send_text("hello")
+            assert ws.receive_text() == "HELLO"
+            ws.send_text("bye")
+
+
+def test_websocket_proxy_auth(monkeypatch):
+    """A locked asgi mount refuses the WS handshake without credentials."""
+    app_ws2 = modal.App("test-web-ws-auth")
+
+    @app_ws2.function()
+    @modal.asgi_app(label="lockedws", requires_proxy_auth=True)
+    def lockedws():
+        from fastapi import FastAPI, WebSocket
+
+        w = FastAPI()
+
+        @w.websocket("/s")
+        async def s(ws: WebSocket):
+            await ws.accept()
+            await ws.send_text("in")
+            await ws.close()
+
+        return w
+
+    from starlette.testclient import TestClient
+
+    monkeypatch.setenv("MODAL_AMD_PROXY_TOKEN_ID", "k")
+    monkeypatch.setenv("MODAL_AMD_PROXY_TOKEN_SECRET", "s")
+    root = build_ingress_app(app_ws2)
+    with TestClient(root) as c:
+        import pytest as _pytest
+
+        with _pytest.raises(Exception):
+            with c.websocket_connect("/lockedws/s"):
+                pass
+        with c.websocket_connect(
+                "/lockedws/s", headers={"Modal-Key": "k", "Modal-Secret": "s"}) as ws:
+            assert ws.receive_text() == "in"
