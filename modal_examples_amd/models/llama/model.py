@@ -96,9 +96,12 @@ class LlamaModel(nn.Module):
         return self._rope_cache[key]
 
     @torch.no_grad()
-    def prefill(self, tokens: torch.Tensor, kv_writer=None):
+    def prefill(self, tokens: torch.Tensor, kv_writer=None, last_pos=None):
         """tokens [B,S] → logits [B, vocab] (last position only).
-        kv_writer(layer_idx, k, v): callback storing [B,nkv,S,hd] into cache."""
+        kv_writer(layer_idx, k, v): callback storing [B,nkv,S,hd] into cache.
+        last_pos [B] int64: per-row index of the last REAL token — the ragged
+        path right-pads shorter prompts and samples at each row's own end
+        (vLLM-style mixed-length prefill; role of vllm_inference.py:158-209)."""
         x = self.embed(tokens)
         rc, rs = self._rope_tables(x.device)
         gpu = x.is_cuda
@@ -111,7 +114,11 @@ class LlamaModel(nn.Module):
                 kv_writer(li, k, v)  # [B,S,nkv,hd]
             x = x + blk.o_proj(OF.attention_qkv(q, k, v, causal=True))
             x = x + blk.ffn(blk.ffn_norm(x))
-        x = self.norm(x[:, -1:])
+        if last_pos is None:
+            x = self.norm(x[:, -1:])
+        else:
+            idx = last_pos.view(-1, 1, 1).expand(-1, 1, x.shape[-1])
+            x = self.norm(x.gather(1, idx))
         return self.lm_head(x)[:, 0].float()
 
     @torch.no_grad()
