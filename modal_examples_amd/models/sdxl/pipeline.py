@@ -20,6 +20,7 @@ from typing import List, Optional
 
 import torch
 
+from ...gpu.graphs import GraphLRU
 from ...ops import functional as OF
 from .text import encode_prompts, fourier_time_ids
 from .unet import UNetConfig, UNetXL
@@ -65,7 +66,7 @@ class SDXLPipeline:
             self.vae = vae_cls().to(self.device, dtype)
         self.unet.eval()
         self.vae.eval()
-        self._graphs = {}  # (batch, steps, cfg_on) -> graph state
+        self._graphs = GraphLRU(4)  # (batch, steps, cfg_on) -> graph state, LRU-bounded
         self.image_size = latent_size * 8
 
     # -------------------------------------------------- conditioning
@@ -104,8 +105,9 @@ class SDXLPipeline:
 
     def _get_graph(self, batch: int, steps: int, guidance: float):
         key = (batch, steps, guidance > 1.0)
-        if key in self._graphs:
-            return self._graphs[key]
+        st = self._graphs.get(key)
+        if st is not None:
+            return st
         from ...ops._build import get_ext
 
         ext = get_ext(required=True)
@@ -146,7 +148,7 @@ class SDXLPipeline:
         with torch.cuda.graph(g), torch.no_grad():
             one_step()
         st["graph"] = g
-        self._graphs[key] = st
+        self._graphs.put(key, st)
         return st
 
     # -------------------------------------------------- public API
