@@ -548,8 +548,12 @@ class Function:
                 gpu_count = min(gpu_count, rt.device_pool.n)
                 raw = self.raw
                 env = {}
+                python_exe = None
                 if self.opts.image is not None:
                     env.update(getattr(self.opts.image, "build_env", {}) or {})
+                    build_venv = getattr(self.opts.image, "build_venv", None)
+                    if build_venv is not None:
+                        python_exe = build_venv()  # loud on unsatisfiable layer
                     try:
                         self.opts.image.build()  # run_function layers, once
                     except Exception:
@@ -627,6 +631,7 @@ class Function:
                     max_inputs_per_worker=max_inputs,
                     target_inputs_per_worker=int(self._flags.get("target_inputs", 0) or 0),
                     device_pool=rt.device_pool,
+                    python_exe=python_exe,
                 )
                 self._pool.on_spawned_result = _persist_spawned
                 rt.register(self._pool)

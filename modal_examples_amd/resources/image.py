@@ -116,6 +116,64 @@ class Image:
         ).encode()
         return hashlib.sha256(blob).hexdigest()[:16]
 
+    # ---- environment materialization (venv isolation) ----
+
+    @property
+    def _pip_layers(self):
+        return [p for k, p in self.layers if k in ("pip", "pip_req")]
+
+    def build_venv(self) -> Optional[str]:
+        """Materialize the pip layers as a content-hashed venv the workers
+        exec from (reference: 02_building_containers/install_flash_attn.py:
+        17-24 — an image's installs define the interpreter environment, not
+        an importability assertion).
+
+        Offline semantics (this node has no network): the venv is created
+        with --system-site-packages and each pip layer installs with
+        --no-index, which succeeds for anything the local wheel environment
+        already satisfies and FAILS LOUDLY for a requirement it cannot meet —
+        the build error the reference would surface at image-build time.
+        Returns the venv's python path, or None if there are no pip layers.
+        """
+        if not self._pip_layers:
+            return None
+        import subprocess
+        import sys
+
+        from .. import config
+
+        root = config.state_dir() / "images" / self.content_hash()
+        venv = root / "venv"
+        py = venv / "bin" / "python"
+        manifest = root / "manifest.json"
+        if manifest.exists() and py.exists():
+            return str(py)
+        root.mkdir(parents=True, exist_ok=True)
+        import venv as venv_mod
+
+        # with_pip=False: this distro python ships no ensurepip wheels; the
+        # SYSTEM pip installs into the venv via --prefix instead (and, with
+        # --no-index, validates that each requirement is already satisfiable
+        # from the local wheel environment)
+        venv_mod.EnvBuilder(system_site_packages=True, with_pip=False,
+                            symlinks=True).create(str(venv))
+        installed = []
+        for pkgs in self._pip_layers:
+            args = list(pkgs) if isinstance(pkgs, list) else ["-r", str(pkgs)]
+            r = subprocess.run(
+                [sys.executable, "-m", "pip", "install", "--no-index",
+                 "--prefix", str(venv), *args],
+                capture_output=True, text=True)
+            if r.returncode != 0:
+                raise RuntimeError(
+                    f"image build failed: pip layer {args} cannot be satisfied "
+                    f"offline:\n{r.stderr[-2000:]}")
+            installed.append(args)
+        manifest.write_text(json.dumps({
+            "base": self.base, "python": sys.version, "pip": installed,
+            "hash": self.content_hash()}))
+        return str(py)
+
     @property
     def build_env(self) -> dict:
         return dict(self._env)

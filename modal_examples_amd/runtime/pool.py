@@ -17,6 +17,7 @@ from __future__ import annotations
 
 import itertools
 import multiprocessing as mp
+import sys
 import threading
 import time
 import uuid
@@ -33,6 +34,7 @@ from . import ipc
 from .worker import worker_main
 
 _mp = mp.get_context("spawn")
+_spawn_exe_lock = threading.Lock()
 
 
 @dataclass
@@ -160,6 +162,7 @@ class Pool:
         max_inputs_per_worker: int = 1,
         target_inputs_per_worker: int = 0,
         device_pool=None,
+        python_exe: Optional[str] = None,
     ):
         self.name = name
         self.make_spec = make_spec
@@ -174,6 +177,7 @@ class Pool:
         self.max_inputs_per_worker = max(1, max_inputs_per_worker)
         self.target_inputs_per_worker = target_inputs_per_worker or self.max_inputs_per_worker
         self.device_pool = device_pool
+        self.python_exe = python_exe  # image-venv interpreter for workers
 
         self.result_q = _mp.Queue()
         self.workers: Dict[int, WorkerHandle] = {}
@@ -331,7 +335,18 @@ class Pool:
             daemon=True,
             name=f"mxa-{self.name}-{wid}",
         )
-        proc.start()
+        if self.python_exe:
+            # image-built venv: the worker process execs the venv interpreter
+            # (spawn replicates sys.path, so the package resolves; venv
+            # site-packages take precedence over system for pip layers)
+            with _spawn_exe_lock:
+                _mp.set_executable(self.python_exe)
+                try:
+                    proc.start()
+                finally:
+                    _mp.set_executable(sys.executable)
+        else:
+            proc.start()
         w = WorkerHandle(wid, proc, task_q, devices)
         self.workers[wid] = w
         return w
