@@ -45,12 +45,15 @@ def train_variant(n_layer: int, n_embd: int, steps: int) -> dict:
         p.data = p.data.float() if device == "cpu" else p.data
     opt = FusedAdamW([p for p in model.parameters()], lr=3e-4)
     name = f"gpt_l{n_layer}_d{n_embd}"
+    from modal_examples_amd.observability.board import log_scalar
+
     for step in range(steps):
         x, y = make_batch(cfg, step, device)
         _, loss = model(x, y)
         loss.backward()
         opt.step()
         opt.zero_grad()
+        log_scalar(runs.path / "logs" / name, "train/loss", step, float(loss))
     vx, vy = make_batch(cfg, 10_000, device)
     with torch.no_grad():
         _, val_loss = model(vx, vy)
@@ -87,6 +90,19 @@ def resume_best(name: str, extra_steps: int) -> dict:
         opt.zero_grad()
     return {"name": name, "resumed_from": ck["steps"],
             "now_at": ck["steps"] + extra_steps}
+
+
+@app.function()
+@modal.wsgi_app(label="hp-board")
+def board():
+    """TensorBoard-on-Volume role: scalar dashboard over the sweep's logs,
+    volume reloaded before every request (hp_sweep_gpt.py:396-414 contract)."""
+    from modal_examples_amd.observability.board import (
+        VolumeReloadMiddleware,
+        make_board_wsgi,
+    )
+
+    return VolumeReloadMiddleware(make_board_wsgi(runs.path / "logs"), runs)
 
 
 @app.cls(gpu="mi355x")
@@ -126,3 +142,24 @@ def main(variants: int = 4, steps: int = 8):
     srv = GPTServer(variant=best["name"])
     completion = srv.complete.remote([1, 2, 3], 8)
     print("completion:", completion)
+
+    # the TensorBoard-role dashboard serves the sweep's loss curves
+    import asyncio
+
+    import httpx
+
+    from modal_examples_amd.web.ingress import build_ingress_app
+
+    async def check_board():
+        root = build_ingress_app(app)
+        async with httpx.AsyncClient(transport=httpx.ASGITransport(app=root),
+                                     base_url="http://t") as c:
+            r = await c.get("/hp-board/data")
+            data = r.json()
+            assert any("train/loss" in tags for tags in data.values()), data.keys()
+            html = (await c.get("/hp-board/")).text
+            assert "train/loss" in html and "<svg" in html
+            return len(data)
+
+    n_runs = asyncio.run(check_board())
+    print(f"board serves {n_runs} run(s) of loss curves")

@@ -34,6 +34,49 @@ def profile_workload() -> str:
     return path
 
 
+@app.function()
+@modal.wsgi_app(label="trace-board")
+def board():
+    """TensorBoard-role dashboard over the trace Volume, reloaded per request
+    (torch_profiling.py:294-316 contract): lists saved traces + any scalar
+    events alongside them."""
+    import json
+    from pathlib import Path
+
+    from modal_examples_amd.observability.board import (
+        VolumeReloadMiddleware,
+        make_board_wsgi,
+    )
+
+    scalars = make_board_wsgi(traces.path)
+
+    def wsgi(environ, start_response):
+        if environ.get("PATH_INFO", "/").rstrip("/").endswith("traces"):
+            files = sorted(str(f.relative_to(traces.path))
+                           for f in Path(traces.path).glob("**/*.json"))
+            start_response("200 OK", [("Content-Type", "application/json")])
+            return [json.dumps({"traces": files}).encode()]
+        return scalars(environ, start_response)
+
+    return VolumeReloadMiddleware(wsgi, traces)
+
+
 @app.local_entrypoint()
 def main():
     print("trace written to:", profile_workload.remote())
+
+    import asyncio
+
+    import httpx
+
+    from modal_examples_amd.web.ingress import build_ingress_app
+
+    async def check():
+        root = build_ingress_app(app)
+        async with httpx.AsyncClient(transport=httpx.ASGITransport(app=root),
+                                     base_url="http://t") as c:
+            r = (await c.get("/trace-board/traces")).json()
+            assert r["traces"], r
+            return r["traces"]
+
+    print("board lists traces:", asyncio.run(check()))
