@@ -439,6 +439,20 @@ class FunctionOptions:
         return out
 
 
+def _maybe_start_s3(vols: dict, env: dict) -> None:
+    """If any mount is an S3 bucket, start the local S3 endpoint and hand
+    its URL to workers (CloudBucketMount prefix-sync, s3_bucket_mount.py)."""
+    for v in vols.values():
+        name = str(v)
+        if name.startswith("ro:"):
+            name = name[3:]
+        if name.startswith("s3:"):
+            from .resources.s3local import start_s3_server
+
+            env["MODAL_AMD_S3_ENDPOINT"] = start_s3_server()
+            return
+
+
 class ClusterCall:
     """Aggregate of one Call per cluster rank: resolves to rank 0's result
     once EVERY rank finished; any rank's failure propagates.  Quacks like a
@@ -561,6 +575,7 @@ class Function:
                 for s in self.opts.secrets or []:
                     env.update(getattr(s, "env", {}))
                 vols = {p: getattr(v, "name", str(v)) for p, v in (self.opts.volumes or {}).items()}
+                _maybe_start_s3(vols, env)
                 fn_name = self.name
                 app_name = self.app.name
                 max_inputs = self.max_inputs
@@ -902,6 +917,7 @@ class _Obj:
 
                 vols = {p_: getattr(v, "name", str(v))
                         for p_, v in (cls.opts.volumes or {}).items()}
+                _maybe_start_s3(vols, env)
 
                 def make_spec(devices):
                     return ipc.ServiceSpec(

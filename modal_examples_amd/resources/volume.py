@@ -164,19 +164,25 @@ class _ReadOnlyVolume(Volume):
 
 
 class CloudBucketMount:
-    """S3/GCS bucket mount shim (10_integrations/s3_bucket_mount.py:66): no
-    network locally, so this maps to a named local directory that tests can
-    pre-populate."""
+    """S3-backed bucket mount (10_integrations/s3_bucket_mount.py:63-80).
+
+    Real bucket semantics on a no-network node: objects live behind a LOCAL
+    S3-compatible REST endpoint (resources/s3local.py); mounting a bucket in
+    a worker prefix-DOWNLOADS the objects into a private per-worker directory
+    over HTTP (not a shared-dir symlink) and, unless read_only, writes
+    changes back to the endpoint at worker shutdown."""
 
     def __init__(self, bucket_name: str, secret=None, read_only: bool = False,
                  key_prefix: Optional[str] = None, **kw):
         self.bucket_name = bucket_name
         self.read_only = read_only
-        self.key_prefix = key_prefix or ""
+        self.key_prefix = (key_prefix or "").strip("/")
+        # direct server-side path (client-side seeding/tests); workers see a
+        # synced copy, never this directory
         self.path = config.state_dir() / "buckets" / bucket_name
         self.path.mkdir(parents=True, exist_ok=True)
 
     @property
     def name(self) -> str:
-        base = f"bucket:{self.bucket_name}"
+        base = f"s3:{self.bucket_name}:{self.key_prefix}"
         return f"ro:{base}" if self.read_only else base

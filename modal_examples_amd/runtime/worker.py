@@ -14,6 +14,7 @@ container≈worker-process lifecycle).  A worker:
 from __future__ import annotations
 
 import os
+from pathlib import Path
 import queue as _queue
 import sys
 import threading
@@ -60,6 +61,70 @@ def _bind_read_only(pairs: list) -> bool:
         return False
 
 
+_bucket_syncs = []  # (bucket, prefix, local_dir) pending write-back
+
+
+def _mount_bucket(mount: str, name: str, ro: bool):
+    """s3:<bucket>:<prefix> mount: prefix-sync download into a private dir
+    via the local S3 endpoint; non-ro mounts write back at worker exit.
+    Returns the private dir for ro mounts (caller bind-mounts it read-only)
+    or None when the mount was placed here."""
+    import tempfile
+
+    from ..resources.s3local import S3Client
+
+    _, bucket, prefix = name.split(":", 2)
+    endpoint = os.environ.get("MODAL_AMD_S3_ENDPOINT", "")
+    local = Path(tempfile.mkdtemp(prefix=f"s3mount-{bucket}-"))
+    if endpoint:
+        try:
+            S3Client(endpoint).sync_down(bucket, prefix, local)
+        except Exception:
+            pass  # empty bucket / server racing: mount starts empty
+    else:
+        # no endpoint (direct in-process use): fall back to the server-side dir
+        from .. import config
+
+        src = config.state_dir() / "buckets" / bucket / prefix
+        if src.exists():
+            import shutil
+
+            shutil.copytree(src, local, dirs_exist_ok=True)
+    if ro:
+        return local  # caller adds (mount, local) to the RO bind-mount set
+    try:
+        if os.path.islink(mount):
+            os.unlink(mount)
+        if not os.path.exists(mount):
+            parent = os.path.dirname(str(mount).rstrip("/"))
+            if parent and not os.path.exists(parent):
+                os.makedirs(parent, exist_ok=True)
+            os.symlink(local, mount)
+    except OSError:
+        pass
+    _bucket_syncs.append((bucket, prefix, local))
+    return None
+
+
+def _writeback_buckets() -> None:
+    from ..resources.s3local import S3Client
+
+    endpoint = os.environ.get("MODAL_AMD_S3_ENDPOINT", "")
+    for bucket, prefix, local in _bucket_syncs:
+        try:
+            if endpoint:
+                S3Client(endpoint).sync_up(bucket, prefix, local)
+            else:
+                from .. import config
+
+                import shutil
+
+                dst = config.state_dir() / "buckets" / bucket / prefix
+                shutil.copytree(local, dst, dirs_exist_ok=True)
+        except Exception:
+            pass
+
+
 def _mount_volumes(volumes: dict) -> None:
     """Symlink mount paths to the shared volume directories (the worker runs
     as root on this node — mirrors containers mounting at /cache etc.).
@@ -72,6 +137,11 @@ def _mount_volumes(volumes: dict) -> None:
         ro = name.startswith("ro:")
         if ro:
             name = name[3:]
+        if name.startswith("s3:"):
+            ro_target = _mount_bucket(mount, name, ro)
+            if ro_target is not None:
+                ro_pairs.append((mount, ro_target))
+            continue
         if name.startswith("bucket:"):
             target = config.state_dir() / "buckets" / name.split(":", 1)[1]
         else:
@@ -264,6 +334,7 @@ def worker_main(worker_id: int, spec_blob: bytes, task_q, result_q) -> None:
             run_batch(call_ids, method_name, args_blobs)
 
     pool.shutdown(wait=True)
+    _writeback_buckets()
     for hook in exit_hooks:
         try:
             hook()

@@ -406,3 +406,56 @@ def test_image_unsatisfiable_pip_layer_fails_loudly():
 
     with pytest.raises(Exception, match="cannot be satisfied offline"):
         f.remote()
+
+
+def test_s3_endpoint_roundtrip():
+    """Local S3-compatible endpoint: PUT/GET/LIST/DELETE over real HTTP."""
+    from modal_examples_amd.resources.s3local import S3Client, start_s3_server
+
+    c = S3Client(start_s3_server())
+    c.put("t-bucket", "a/x.txt", b"hello")
+    c.put("t-bucket", "a/y.txt", b"world")
+    c.put("t-bucket", "b/z.txt", b"nope")
+    assert sorted(c.list("t-bucket", "a/")) == ["a/x.txt", "a/y.txt"]
+    assert c.get("t-bucket", "a/x.txt") == b"hello"
+    c.delete("t-bucket", "a/y.txt")
+    assert c.list("t-bucket", "a/") == ["a/x.txt"]
+
+
+def test_bucket_mount_prefix_sync_and_writeback(tmp_path):
+    """CloudBucketMount: worker sees a PRIVATE synced copy of the bucket
+    prefix (downloaded over the S3 endpoint, not a shared symlink) and its
+    writes land back in the bucket at worker exit."""
+    import time
+
+    from modal_examples_amd.resources.s3local import S3Client, start_s3_server
+
+    ep = start_s3_server()
+    c = S3Client(ep)
+    c.put("sync-bkt", "data/in.txt", b"seeded")
+    c.put("sync-bkt", "other/skip.txt", b"outside prefix")
+
+    app = modal.App("test-s3-mount")
+    mnt = modal.CloudBucketMount("sync-bkt", key_prefix="data")
+
+    @app.function(volumes={"/mnt/bkt": mnt}, scaledown_window=0.5)
+    def roundtrip() -> dict:
+        import os
+        from pathlib import Path
+
+        p = Path("/mnt/bkt")
+        seen = sorted(f.name for f in p.glob("**/*") if f.is_file())
+        (p / "out.txt").write_text("produced")
+        real = os.path.realpath(p)
+        return {"seen": seen, "private": "s3mount" in real}
+
+    out = roundtrip.remote()
+    assert out["seen"] == ["in.txt"], out
+    assert out["private"], out  # synced copy, not the server-side dir
+    # writeback happens at worker shutdown (scaledown reaps the idle worker)
+    deadline = time.monotonic() + 20
+    while time.monotonic() < deadline:
+        if "data/out.txt" in c.list("sync-bkt", "data/"):
+            break
+        time.sleep(0.3)
+    assert c.get("sync-bkt", "data/out.txt") == b"produced"
