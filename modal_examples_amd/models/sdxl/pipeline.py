@@ -41,7 +41,8 @@ def euler_sigmas(steps: int, num_train: int = 1000, beta_start: float = 0.00085,
 class SDXLPipeline:
     def __init__(self, cfg: Optional[UNetConfig] = None, device: str = "cuda",
                  dtype=torch.bfloat16, latent_size: int = 128,
-                 use_graph: bool = True, seed: int = 0):
+                 use_graph: bool = True, seed: int = 0,
+                 init_weights: bool = True):
         self.cfg = cfg or UNetConfig.sdxl()
         self.device = torch.device(device)
         self.dtype = dtype
@@ -60,10 +61,19 @@ class SDXLPipeline:
             torch.backends.cudnn.benchmark = (
                 _os.environ.get("MODAL_AMD_CONV_BENCHMARK", "0") == "1")
         torch.manual_seed(seed)
-        with torch.device(self.device):
-            self.unet = UNetXL(self.cfg).to(self.device, dtype)
-            vae_cls = VAEDecoder if self.cfg.channels[0] >= 320 else VAEDecoderSmall
-            self.vae = vae_cls().to(self.device, dtype)
+        vae_cls = VAEDecoder if self.cfg.channels[0] >= 320 else VAEDecoderSmall
+        if init_weights:
+            with torch.device(self.device):
+                self.unet = UNetXL(self.cfg).to(self.device, dtype)
+                self.vae = vae_cls().to(self.device, dtype)
+        else:
+            # cold-restore path: build on meta (no init compute), materialize
+            # EMPTY on device — load_state_dict fills it (bench_cold.py)
+            with torch.device("meta"):
+                self.unet = UNetXL(self.cfg)
+                self.vae = vae_cls()
+            self.unet = self.unet.to_empty(device=self.device).to(dtype)
+            self.vae = self.vae.to_empty(device=self.device).to(dtype)
         self.unet.eval()
         self.vae.eval()
         self._graphs = GraphLRU(4)  # (batch, steps, cfg_on) -> graph state, LRU-bounded

@@ -59,19 +59,23 @@ def child(model: str, shm: str) -> None:
         from modal_examples_amd.models.sdxl.pipeline import SDXLPipeline
 
         t0 = time.perf_counter()
-        sd_u = torch.load(f"{shm}/unet.pt", map_location="cpu")
-        sd_v = torch.load(f"{shm}/vae.pt", map_location="cpu")
-        pipe = SDXLPipeline(device=dev, dtype=torch.bfloat16)
-        pipe.unet.load_state_dict(sd_u)
-        pipe.vae.load_state_dict(sd_v)
+        # straight-to-device load; model built EMPTY on device (meta init)
+        sd_u = torch.load(f"{shm}/unet.pt", map_location=dev)
+        sd_v = torch.load(f"{shm}/vae.pt", map_location=dev)
+        pipe = SDXLPipeline(device=dev, dtype=torch.bfloat16,
+                            init_weights=False)
+        pipe.unet.load_state_dict(sd_u, assign=True)
+        pipe.vae.load_state_dict(sd_v, assign=True)
         if dev == "cuda":
             torch.cuda.synchronize()
         t_restore = time.perf_counter() - t0
         nbytes = sum(v.numel() * v.element_size() for v in sd_u.values())
         nbytes += sum(v.numel() * v.element_size() for v in sd_v.values())
+        t1 = time.perf_counter()
         pipe.generate(["cold start probe"] * 1, steps=4)
         if dev == "cuda":
             torch.cuda.synchronize()
+        t_first = time.perf_counter() - t1
     else:
         from modal_examples_amd.models.llama.engine import LlamaEngine
         from modal_examples_amd.models.llama.model import LlamaConfig
@@ -87,14 +91,17 @@ def child(model: str, shm: str) -> None:
         t_restore = time.perf_counter() - t0
         nbytes = sum(v.numel() * v.element_size() for v in sd.values())
         srv = LLMServer(eng, model_name="cold-probe")
+        t1 = time.perf_counter()
         srv.generate("cold start probe", max_tokens=4)
         srv.shutdown()
+        t_first = time.perf_counter() - t1
     total = time.perf_counter() - t_start
     print(json.dumps({
         "cold_start_s": round(total, 3),
         "import_s": round(t_import - t_start, 3),
         "restore_s": round(t_restore, 3),
         "restore_gb_s": round(nbytes / t_restore / 1e9, 2),
+        "first_call_s": round(t_first, 3),
         "weight_gb": round(nbytes / 1e9, 2)}), flush=True)
 
 
