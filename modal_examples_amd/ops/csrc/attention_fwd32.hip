@@ -22,6 +22,7 @@
 #include "common.h"
 
 #include <cstdio>
+#include <cstdlib>
 
 #define FA32_NWAVES 4
 #define FA32_KVBLK 32
@@ -46,7 +47,12 @@ struct FA32Strides {
   long long qb, qh, qs, kb, kh, ks, vb, vh, vs, ob, oh, os;
 };
 
-template <int D, bool CAUSAL>
+// KVB: kv rows per block iteration.  64 halves the barrier/softmax rounds
+// per token vs the r1 kernel's 32 (the m214-ladder "KVBLK=64" lever).
+// DEFER: defer-max rescale skipping (HK THR=8) — keep the old running max
+// while per-block growth <= 8, so P = exp(s - m_old) <= e^8 (f32 acc safe)
+// and the O-wide rescale is skipped entirely.
+template <int D, bool CAUSAL, int KVB, bool DEFER>
 __global__ __launch_bounds__(FA32_NWAVES * WAVE) void fa32_kernel(
     const short* __restrict__ Q, const short* __restrict__ K,
     const short* __restrict__ V, short* __restrict__ O,
@@ -55,9 +61,10 @@ __global__ __launch_bounds__(FA32_NWAVES * WAVE) void fa32_kernel(
   constexpr int DCH = D / 16;  // QK^T k-chunks (K-dim 16)
   constexpr int DT = D / 32;   // PV d-tiles (M-dim 32)
   constexpr int QBLK = 32 * FA32_NWAVES;
+  constexpr int KF = KVB / 32;  // 32-kv score fragments per iteration
 
-  __shared__ alignas(16) short Ks[FA32_KVBLK][KROW];
-  __shared__ alignas(16) short Vs[FA32_KVBLK][KROW];
+  __shared__ alignas(16) short Ks[KVB][KROW];
+  __shared__ alignas(16) short Vs[KVB][KROW];
 
   const int tid = threadIdx.x;
   const int w = tid / WAVE;
@@ -104,23 +111,23 @@ __global__ __launch_bounds__(FA32_NWAVES * WAVE) void fa32_kernel(
     for (int r = 0; r < 16; ++r) acc[dt][r] = 0.f;
   float m_run = -1e30f, l_run = 0.f;
 
-  int nkb = (Sk + FA32_KVBLK - 1) / FA32_KVBLK;
+  int nkb = (Sk + KVB - 1) / KVB;
   if (CAUSAL) {
     int max_kv = qblk * QBLK + QBLK - 1 + causal_off;
-    int lim = (max_kv + FA32_KVBLK) / FA32_KVBLK;
+    int lim = (max_kv + KVB) / KVB;
     if (lim < nkb) nkb = lim;
   }
 
   // async register staging: next block loads issue before this block computes
   constexpr int CPR = D / 8;
-  constexpr int CPT = (FA32_KVBLK * CPR) / (FA32_NWAVES * WAVE);
+  constexpr int CPT = (KVB * CPR) / (FA32_NWAVES * WAVE);
   bf16x8 kreg[CPT], vreg[CPT];
   auto load_chunks = [&](int kb) {
 #pragma unroll
     for (int i = 0; i < CPT; ++i) {
       int ci = i * FA32_NWAVES * WAVE + tid;
       int row = ci / CPR, c8 = ci % CPR;
-      int kvp = kb * FA32_KVBLK + row;
+      int kvp = kb * KVB + row;
       if (kvp < Sk) {
         kreg[i] = *(const bf16x8*)&K[koff + kvp * st.ks + c8 * 8];
         vreg[i] = *(const bf16x8*)&V[voff + kvp * st.vs + c8 * 8];
@@ -148,69 +155,85 @@ __global__ __launch_bounds__(FA32_NWAVES * WAVE) void fa32_kernel(
     __syncthreads();
     if (kb + 1 < nkb) load_chunks(kb + 1);
 
-    // ---- S^T = K · Q^T ----
-    f32x16 s;
+    // ---- S^T = K · Q^T (KF fragments of 32 kv each) ----
+    f32x16 s[KF];
 #pragma unroll
-    for (int r = 0; r < 16; ++r) s[r] = 0.f;
+    for (int f = 0; f < KF; ++f)
 #pragma unroll
-    for (int c = 0; c < DCH; ++c) {
-      // A = K[kv][d]: lane holds K[l31][c*16 + hi5*8 + j]
-      bf16x8 kfrag = *(const bf16x8*)(
-          (char*)&Ks[l31][0] + kv_swz7(l31, (c * 16 + hi5 * 8) * 2));
-      bf16x8 qf;
-      if (QREG) {
-        qf = qreg[c % (QREG ? DCH : 1)];
-      } else {
-        qf = q_ok ? *(const bf16x8*)&Q[qrow_off + c * 16 + hi5 * 8]
-                  : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      for (int r = 0; r < 16; ++r) s[f][r] = 0.f;
+#pragma unroll
+    for (int f = 0; f < KF; ++f) {
+#pragma unroll
+      for (int c = 0; c < DCH; ++c) {
+        // A = K[kv][d]: lane holds K[f*32 + l31][c*16 + hi5*8 + j]
+        int krow = f * 32 + l31;
+        bf16x8 kfrag = *(const bf16x8*)(
+            (char*)&Ks[krow][0] + kv_swz7(krow, (c * 16 + hi5 * 8) * 2));
+        bf16x8 qf;
+        if (QREG) {
+          qf = qreg[c % (QREG ? DCH : 1)];
+        } else {
+          qf = q_ok ? *(const bf16x8*)&Q[qrow_off + c * 16 + hi5 * 8]
+                    : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+        }
+        __builtin_amdgcn_s_setprio(1);
+        s[f] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qf, s[f], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
       }
-      __builtin_amdgcn_s_setprio(1);
-      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qf, s, 0, 0, 0);
-      __builtin_amdgcn_s_setprio(0);
     }
 
     // ---- mask + lane-local online softmax (this lane owns query my_q) ----
-    bool full = (kb * FA32_KVBLK + FA32_KVBLK <= Sk) &&
-                (!CAUSAL || kb * FA32_KVBLK + FA32_KVBLK - 1 <= my_q + causal_off);
+    bool full = (kb * KVB + KVB <= Sk) &&
+                (!CAUSAL || kb * KVB + KVB - 1 <= my_q + causal_off);
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      int kvp = kb * FA32_KVBLK + (r & 3) + 8 * (r >> 2) + 4 * hi5;
-      bool dead = !full && ((kvp >= Sk) || (CAUSAL && kvp > my_q + causal_off));
-      s[r] = dead ? -1e30f : s[r] * scale;
-    }
-    float smax = s[0];
+    for (int f = 0; f < KF; ++f)
 #pragma unroll
-    for (int r = 1; r < 16; ++r) smax = fmaxf(smax, s[r]);
+      for (int r = 0; r < 16; ++r) {
+        int kvp = kb * KVB + f * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi5;
+        bool dead = !full && ((kvp >= Sk) || (CAUSAL && kvp > my_q + causal_off));
+        s[f][r] = dead ? -1e30f : s[f][r] * scale;
+      }
+    float smax = s[0][0];
+#pragma unroll
+    for (int f = 0; f < KF; ++f)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) smax = fmaxf(smax, s[f][r]);
     smax = fmaxf(smax, __shfl_xor(smax, 32, WAVE));  // partner combine
-    float m_new = fmaxf(m_run, smax);
-    float rs = __expf(m_run - m_new);
+    if (!DEFER || kb == 0 ||
+        __ballot(smax > m_run + 8.f) != 0ull) {
+      float m_new = fmaxf(m_run, smax);
+      float rs = __expf(m_run - m_new);
+      if (kb > 0 && __ballot(rs < 0.999999f) != 0ull) {
+        l_run *= rs;
+#pragma unroll
+        for (int dt = 0; dt < DT; ++dt)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) acc[dt][r] *= rs;
+      }
+      m_run = m_new;
+    }
     float psum = 0.f;
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      s[r] = __expf(s[r] - m_new);
-      psum += s[r];
-    }
+    for (int f = 0; f < KF; ++f)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        s[f][r] = __expf(s[f][r] - m_run);
+        psum += s[f][r];
+      }
     psum += __shfl_xor(psum, 32, WAVE);
-    l_run = (kb == 0) ? psum : l_run * rs + psum;
-    m_run = m_new;
-    if (kb > 0 && __ballot(rs < 0.999999f) != 0ull) {
-#pragma unroll
-      for (int dt = 0; dt < DT; ++dt)
-#pragma unroll
-        for (int r = 0; r < 16; ++r) acc[dt][r] *= rs;
-    }
+    l_run = (kb == 0) ? psum : l_run + psum;
 
     // ---- P → bf16 B-fragments in registers (cvt_pk + permlane32_swap) ----
     // own regs pack kv pairs; half-exchange composes the 16-kv chunks:
     //   chunk c frag words = [A'(pk01,pk45), A'(pk23,pk67), B'(same), B'(same)]
-    bf16x8 pfrag[2];
+    bf16x8 pfrag[2 * KF];
 #pragma unroll
-    for (int c = 0; c < 2; ++c) {
-      int base = c * 8;
-      int pk01 = (int)cvt_pk_bf16(s[base + 0], s[base + 1]);
-      int pk23 = (int)cvt_pk_bf16(s[base + 2], s[base + 3]);
-      int pk45 = (int)cvt_pk_bf16(s[base + 4], s[base + 5]);
-      int pk67 = (int)cvt_pk_bf16(s[base + 6], s[base + 7]);
+    for (int c = 0; c < 2 * KF; ++c) {
+      float* sv = (float*)&s[c >> 1] + (c & 1) * 8;
+      int pk01 = (int)cvt_pk_bf16(sv[0], sv[1]);
+      int pk23 = (int)cvt_pk_bf16(sv[2], sv[3]);
+      int pk45 = (int)cvt_pk_bf16(sv[4], sv[5]);
+      int pk67 = (int)cvt_pk_bf16(sv[6], sv[7]);
       i32x2 x = __builtin_amdgcn_permlane32_swap(pk01, pk45, false, false);
       i32x2 y = __builtin_amdgcn_permlane32_swap(pk23, pk67, false, false);
       union {
@@ -226,7 +249,7 @@ __global__ __launch_bounds__(FA32_NWAVES * WAVE) void fa32_kernel(
 
     // ---- O^T += V^T · P^T (V^T via tr16 hardware transpose reads) ----
 #pragma unroll
-    for (int c = 0; c < 2; ++c) {
+    for (int c = 0; c < 2 * KF; ++c) {
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
         // lane (within its 16-group: 4a+b) issues the source read for
@@ -280,9 +303,33 @@ extern "C" void fa32_fwd_strided_bf16(
   st.ob = strides[9]; st.oh = strides[10]; st.os = strides[11];
   dim3 grid((Sq + 127) / 128, Hq, B);
   dim3 block(FA32_NWAVES * WAVE);
+  // KVB is chosen per head-dim (A/B-measured, gpurun_out/attn_ab_r2.txt):
+  //   D=64:  KVB=64 wins (447 vs 407 TF — fewer barrier/softmax rounds)
+  //   D=128: KVB=64 LOSES 40% (2x LDS -> 2 blocks/CU occupancy cliff,
+  //          the same mode r1's double-buffer ablation hit); keep 32.
+  // Ablation toggles: MODAL_AMD_FA_KVB=32|64 forces the block size,
+  // MODAL_AMD_FA_NODEFER disables defer-max rescale skipping (+5-10%).
+  static const int kvb_env = [] {
+    const char* e = getenv("MODAL_AMD_FA_KVB");
+    return e ? atoi(e) : 0;
+  }();
+  static const bool defer_env = getenv("MODAL_AMD_FA_NODEFER") == nullptr;
 #define L32(DD, CC)                                                           \
-  hipLaunchKernelGGL((fa32_kernel<DD, CC>), grid, block, 0, stream, Qp, Kp,   \
-                     Vp, Op, B, Hq, Hkv, Sq, Sk, scale, st)
+  do {                                                                        \
+    const int kvb = kvb_env ? kvb_env : (DD == 64 ? 64 : 32);                 \
+    if (kvb == 64 && defer_env)                                               \
+      hipLaunchKernelGGL((fa32_kernel<DD, CC, 64, true>), grid, block, 0,     \
+                         stream, Qp, Kp, Vp, Op, B, Hq, Hkv, Sq, Sk, scale, st); \
+    else if (kvb == 64)                                                       \
+      hipLaunchKernelGGL((fa32_kernel<DD, CC, 64, false>), grid, block, 0,    \
+                         stream, Qp, Kp, Vp, Op, B, Hq, Hkv, Sq, Sk, scale, st); \
+    else if (defer_env)                                                       \
+      hipLaunchKernelGGL((fa32_kernel<DD, CC, 32, true>), grid, block, 0,     \
+                         stream, Qp, Kp, Vp, Op, B, Hq, Hkv, Sq, Sk, scale, st); \
+    else                                                                      \
+      hipLaunchKernelGGL((fa32_kernel<DD, CC, 32, false>), grid, block, 0,    \
+                         stream, Qp, Kp, Vp, Op, B, Hq, Hkv, Sq, Sk, scale, st); \
+  } while (0)
   if (D == 64) {
     if (causal) L32(64, true); else L32(64, false);
   } else if (D == 128) {
