@@ -104,6 +104,17 @@ class Conv3x3(nn.Conv2d):
             return False
         return True
 
+    def forward_gn(self, x, gn, residual=None, upsample: bool = False):
+        """Fused GroupNorm+SiLU -> conv (gn: a GroupNormSiLU module with
+        silu=True).  Falls back to gn(x) then conv off the kernel path."""
+        if self._use_kernel(x) and gn.silu:
+            wr, b = self._packed()
+            return OF.conv3x3_gn(x, wr, b, self.out_channels,
+                                 gn._f32("weight"), gn._f32("bias"),
+                                 gn.groups, gn.eps, residual=residual,
+                                 raw_weight=self.weight, upsample=upsample)
+        return self.forward(gn(x), residual=residual, upsample=upsample)
+
     def forward(self, x, residual=None, upsample: bool = False):
         if self._use_kernel(x):
             wr, b = self._packed()

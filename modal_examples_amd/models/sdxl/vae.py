@@ -27,9 +27,10 @@ class VAEResnet(nn.Module):
         self.skip = Conv1x1(c_in, c_out) if c_in != c_out else nn.Identity()
 
     def forward(self, x):
-        h = self.conv1(self.norm1(x))
-        # conv2 fuses the skip add (K3 epilogue)
-        return self.conv2(self.norm2(h), residual=self.skip(x))
+        # both convs fuse their preceding GroupNorm+SiLU into staging; conv2
+        # additionally fuses the skip add in its epilogue (K3)
+        h = self.conv1.forward_gn(x, self.norm1)
+        return self.conv2.forward_gn(h, self.norm2, residual=self.skip(x))
 
 
 class VAEMidAttention(nn.Module):
@@ -88,7 +89,7 @@ class VAEDecoder(nn.Module):
                 h = blk(h)
             if len(ups):
                 h = ups[0](h, upsample=True)  # fused nearest-2x + conv
-        return self.conv_out(self.norm_out(h))
+        return self.conv_out.forward_gn(h, self.norm_out)
 
 
 class VAEDecoderSmall(VAEDecoder):

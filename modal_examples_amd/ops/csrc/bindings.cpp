@@ -49,7 +49,11 @@ void cfg_euler_dev_bf16(const void*, const void*, const void*, void*,
                         hipStream_t);
 void advance_step(long long*, hipStream_t);
 void conv3x3_bf16(const void*, const void*, const void*, const void*, void*,
-                  int, int, int, int, int, int, int, int, hipStream_t);
+                  const float*, const float*, int, int, int, int, int, int,
+                  int, int, hipStream_t);
+void gn_conv_coeffs_bf16(const void*, float*, const float*, const float*,
+                         float*, float*, int, int, int, long long, int, float,
+                         hipStream_t);
 }
 
 namespace {
@@ -274,7 +278,40 @@ torch::Tensor conv3x3(torch::Tensor x, torch::Tensor wr, torch::Tensor bias,
     resp = residual->data_ptr();
   }
   conv3x3_bf16(x.data_ptr(), wr.data_ptr(), bias.data_ptr(), resp,
-               out.data_ptr(), N, C, H, W, (int)K, C16, Kpad,
+               out.data_ptr(), nullptr, nullptr, N, C, H, W, (int)K, C16,
+               Kpad, upsample ? 1 : 0, cur_stream());
+  return out;
+}
+
+// GroupNorm+SiLU fused INTO the conv's staging read: one stats pass over x,
+// per-(n,c) coefficients, then the conv applies silu(x*sc+sh) while staging.
+torch::Tensor conv3x3_gn(torch::Tensor x, torch::Tensor wr, torch::Tensor bias,
+                         c10::optional<torch::Tensor> residual, int64_t K,
+                         bool upsample, torch::Tensor gamma,
+                         torch::Tensor beta, int64_t groups, double eps) {
+  check_bf16(x, "x");
+  check_bf16(wr, "wr");
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  long long HW = (long long)H * W;
+  int Kpad = wr.size(1), C16 = wr.size(2);
+  auto fopt = x.options().dtype(torch::kFloat32);
+  auto ws = torch::zeros({(long long)N * groups * 2}, fopt);
+  auto scale = torch::empty({(long long)N * C16}, fopt);
+  auto shift = torch::empty({(long long)N * C16}, fopt);
+  gn_conv_coeffs_bf16(x.data_ptr(), ws.data_ptr<float>(),
+                      gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                      scale.data_ptr<float>(), shift.data_ptr<float>(), N, C,
+                      C16, HW, (int)groups, (float)eps, cur_stream());
+  int Ho = upsample ? H * 2 : H, Wo = upsample ? W * 2 : W;
+  auto out = torch::empty({N, K, Ho, Wo}, x.options());
+  const void* resp = nullptr;
+  if (residual.has_value()) {
+    check_bf16(*residual, "residual");
+    resp = residual->data_ptr();
+  }
+  conv3x3_bf16(x.data_ptr(), wr.data_ptr(), bias.data_ptr(), resp,
+               out.data_ptr(), scale.data_ptr<float>(),
+               shift.data_ptr<float>(), N, C, Ho, Wo, (int)K, C16, Kpad,
                upsample ? 1 : 0, cur_stream());
   return out;
 }
@@ -401,6 +438,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw_", &adamw_, "fused AdamW step (K9)");
   m.def("sample_gumbel", &sample_gumbel, "fused sampling (K8)");
   m.def("conv3x3", &conv3x3, "NCHW implicit-GEMM 3x3 conv, fused bias+res (K3)");
+  m.def("conv3x3_gn", &conv3x3_gn, "GroupNorm+SiLU fused into the K3 conv");
   m.def("softmax_fwd", &softmax_fwd);
   m.def("tr16_probe", [](torch::Tensor out) {
     tr16_probe((short*)out.data_ptr(), cur_stream());

@@ -56,12 +56,16 @@ typedef __attribute__((ext_vector_type(16))) float f32x16c;
 // directly (out pixel (y,x) <- src[y/2][x/2]), eliminating the materialized
 // F.interpolate pass before every Upsample conv.  H/W are OUTPUT dims;
 // Hs/Ws the source's.
-template <bool UP>
+// GN: fuse GroupNorm+SiLU into the staging read — silu(x*scale[c]+shift[c])
+// with per-(n,c) coefficients from gn_conv_coeffs_bf16 (norms.hip).  The
+// gn_norm write+read pass over the full tensor disappears.
+template <bool UP, bool GN>
 __global__ __launch_bounds__(CV_NW * WAVE) void conv3x3_kernel(
     const short* __restrict__ in, const short* __restrict__ wr,
     const float* __restrict__ bias, const short* __restrict__ res,
-    short* __restrict__ out, int C, int H, int W, int Hs, int Ws, int K,
-    int C16, int Kpad, int npix_x, int npix, int nk) {
+    short* __restrict__ out, const float* __restrict__ gn_scale,
+    const float* __restrict__ gn_shift, int C, int H, int W, int Hs, int Ws,
+    int K, int C16, int Kpad, int npix_x, int npix, int nk) {
   __shared__ alignas(16) short patch[PATCH_ELEMS];
   __shared__ alignas(16) short wlds[WLDS_ELEMS];
 
@@ -113,10 +117,22 @@ __global__ __launch_bounds__(CV_NW * WAVE) void conv3x3_kernel(
           const long long hw = (long long)Hs * Ws;
           const short* src = in + in_n + (long long)(c0 + oct * 8) * hw +
                              (long long)sy * Ws + sx;
+          // GN coeffs: tiny [N, C16] arrays, L1/L2-broadcast across pixels
+          const float* gsc = GN ? gn_scale + (long long)n * C16 + c0 + oct * 8
+                                : nullptr;
+          const float* gsh = GN ? gn_shift + (long long)n * C16 + c0 + oct * 8
+                                : nullptr;
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
             int c = c0 + oct * 8 + j;
-            v[j] = (c < C) ? src[(long long)j * hw] : (short)0;
+            if (c < C) {
+              if (GN) {
+                float f = bf2f(src[(long long)j * hw]) * gsc[j] + gsh[j];
+                v[j] = f2bf(f / (1.f + __expf(-f)));
+              } else {
+                v[j] = src[(long long)j * hw];
+              }
+            }
           }
         }
       }
@@ -229,6 +245,7 @@ __global__ __launch_bounds__(CV_NW * WAVE) void conv3x3_kernel(
 
 extern "C" void conv3x3_bf16(const void* in, const void* wrepack,
                              const void* bias, const void* residual, void* out,
+                             const float* gn_scale, const float* gn_shift,
                              int N, int C, int H, int W, int K, int C16,
                              int Kpad, int upsample, hipStream_t stream) {
   // H/W are the OUTPUT dims; with upsample the source is H/2 x W/2.
@@ -240,14 +257,12 @@ extern "C" void conv3x3_bf16(const void* in, const void* wrepack,
   const int nk = (K + CV_BK - 1) / CV_BK;
   dim3 grid(npix * nk, N);
   dim3 block(CV_NW * WAVE);
-  if (upsample)
-    hipLaunchKernelGGL(conv3x3_kernel<true>, grid, block, 0, stream,
-                       (const short*)in, (const short*)wrepack,
-                       (const float*)bias, (const short*)residual, (short*)out,
-                       C, H, W, Hs, Ws, K, C16, Kpad, npix_x, npix, nk);
-  else
-    hipLaunchKernelGGL(conv3x3_kernel<false>, grid, block, 0, stream,
-                       (const short*)in, (const short*)wrepack,
-                       (const float*)bias, (const short*)residual, (short*)out,
-                       C, H, W, Hs, Ws, K, C16, Kpad, npix_x, npix, nk);
+  const bool gn = gn_scale != nullptr;
+#define CVL(UPV, GNV)                                                           hipLaunchKernelGGL((conv3x3_kernel<UPV, GNV>), grid, block, 0, stream,                           (const short*)in, (const short*)wrepack,                                      (const float*)bias, (const short*)residual, (short*)out,                      gn_scale, gn_shift, C, H, W, Hs, Ws, K, C16, Kpad,                            npix_x, npix, nk)
+  if (upsample) {
+    if (gn) CVL(true, true); else CVL(true, false);
+  } else {
+    if (gn) CVL(false, true); else CVL(false, false);
+  }
+#undef CVL
 }

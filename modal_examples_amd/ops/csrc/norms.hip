@@ -210,3 +210,46 @@ extern "C" void rmsnorm_bf16(const void* x, void* y, const float* gamma,
   hipLaunchKernelGGL(rmsnorm_kernel, dim3((unsigned)blocks), dim3(256), 0,
                      stream, (const short*)x, (short*)y, gamma, rows, D, eps);
 }
+
+
+// ---- GN stats only + per-(n,c) affine coefficients (conv-fusion path) ----
+// The fused conv (conv3x3.hip GN variant) applies silu(x*scale + shift)
+// during its staging read, so gn_norm's full write+read pass disappears.
+// scale[n*Cpad + c] = gamma[c]*rstd(n,g);  shift = beta[c] - mean*scale.
+
+__global__ __launch_bounds__(256) void gn_coeff_kernel(
+    const float* __restrict__ WS, const float* __restrict__ gamma,
+    const float* __restrict__ beta, float* __restrict__ scale,
+    float* __restrict__ shift, int N, int C, int Cpad, long long HW, int G,
+    float eps) {
+  int i = blockIdx.x * 256 + threadIdx.x;
+  if (i >= N * Cpad) return;
+  int n = i / Cpad, c = i % Cpad;
+  if (c >= C) {
+    scale[i] = 0.f;
+    shift[i] = 0.f;
+    return;
+  }
+  int cpg = C / G, g = c / cpg;
+  long long slab = (long long)cpg * HW;
+  float mean = WS[(n * G + g) * 2] / (float)slab;
+  float var = WS[(n * G + g) * 2 + 1] / (float)slab - mean * mean;
+  float rstd = rsqrtf(var + eps);
+  float sc = gamma[c] * rstd;
+  scale[i] = sc;
+  shift[i] = beta[c] - mean * sc;
+}
+
+extern "C" void gn_conv_coeffs_bf16(const void* x, float* ws,
+                                    const float* gamma, const float* beta,
+                                    float* scale, float* shift, int N, int C,
+                                    int Cpad, long long HW, int G, float eps,
+                                    hipStream_t stream) {
+  int split = 1;
+  while (N * G * split < 1024 && (long long)split * 2048 < HW) split *= 2;
+  hipLaunchKernelGGL(gn_partial_kernel, dim3(N * G * split), dim3(256), 0,
+                     stream, (const short*)x, ws, N, C, HW, G, split);
+  hipLaunchKernelGGL(gn_coeff_kernel, dim3((N * Cpad + 255) / 256), dim3(256),
+                     0, stream, ws, gamma, beta, scale, shift, N, C, Cpad, HW,
+                     G, eps);
+}

@@ -212,6 +212,32 @@ def conv3x3(x, wr, bias, K: int, residual=None, raw_weight=None,
     return out
 
 
+def conv3x3_gn(x, wr, bias, K: int, gamma, beta, groups: int = 32,
+               eps: float = 1e-5, residual=None, raw_weight=None,
+               upsample: bool = False):
+    """GroupNorm+SiLU fused into the K3 conv's staging read: one stats pass
+    over x + the conv — the gn_norm write/read pass over the full tensor
+    disappears."""
+    ext = _ext_for(x, raw_weight if raw_weight is not None else x)
+    if ext is None:
+        assert raw_weight is not None
+        h = torch.nn.functional.group_norm(x.float(), groups, gamma.float(),
+                                           beta.float(), eps)
+        h = torch.nn.functional.silu(h)
+        if upsample:
+            h = torch.nn.functional.interpolate(h, scale_factor=2.0,
+                                                mode="nearest")
+        y = torch.nn.functional.conv2d(h, raw_weight.float(), bias.float(),
+                                       padding=1)
+        if residual is not None:
+            y = y + residual.float()
+        return y.to(x.dtype)
+    return ext.conv3x3_gn(x.contiguous(), wr, bias,
+                          residual.contiguous() if residual is not None else None,
+                          K, upsample, gamma.float().contiguous(),
+                          beta.float().contiguous(), groups, eps)
+
+
 def adamw_step(p, g, m, v, lr, beta1=0.9, beta2=0.999, eps=1e-8, wd=0.01, step=1):
     """Fused AdamW (K9); in-place on p/m/v."""
     ext = _ext_for(p, any_dtype=True)

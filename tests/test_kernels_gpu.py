@@ -333,3 +333,25 @@ def test_conv3x3_fused_upsample():
     want = torch.nn.functional.conv2d(up, w.float(), b, padding=1)
     assert got.shape == want.shape
     _close(got, want, atol=5e-2, rtol=5e-2)
+
+
+@requires_gpu
+@pytest.mark.parametrize("upsample", [False, True])
+def test_conv3x3_gn_fused(upsample):
+    """GN+SiLU-fused conv == group_norm+silu (+interp) + conv2d fp32."""
+    N, C, K, H, W = 2, 64, 32, 16, 32
+    torch.manual_seed(3)
+    x = torch.randn(N, C, H, W, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(K, C, 3, 3, device="cuda", dtype=torch.bfloat16) / math.sqrt(C * 9)
+    b = torch.randn(K, device="cuda", dtype=torch.float32)
+    gamma = torch.rand(C, device="cuda") + 0.5
+    beta = torch.randn(C, device="cuda") * 0.1
+    wr = F.repack_conv3x3_weight(w)
+    got = F.conv3x3_gn(x, wr, b, K, gamma, beta, groups=8, raw_weight=w,
+                       upsample=upsample)
+    h = torch.nn.functional.group_norm(x.float(), 8, gamma, beta, 1e-5)
+    h = torch.nn.functional.silu(h)
+    if upsample:
+        h = torch.nn.functional.interpolate(h, scale_factor=2.0, mode="nearest")
+    want = torch.nn.functional.conv2d(h, w.float(), b, padding=1)
+    _close(got, want, atol=5e-2, rtol=5e-2)
