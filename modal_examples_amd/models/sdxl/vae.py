@@ -27,10 +27,13 @@ class VAEResnet(nn.Module):
         self.skip = Conv1x1(c_in, c_out) if c_in != c_out else nn.Identity()
 
     def forward(self, x):
-        # both convs fuse their preceding GroupNorm+SiLU into staging; conv2
-        # additionally fuses the skip add in its epilogue (K3)
-        h = self.conv1.forward_gn(x, self.norm1)
-        return self.conv2.forward_gn(h, self.norm2, residual=self.skip(x))
+        # conv2 fuses the skip add (K3 epilogue).  NOTE measured negative
+        # result: fusing GN+SiLU INTO conv staging (conv3x3_gn) regressed the
+        # step 13.3->9.4 img/s — every output-k-tile re-reads the input, so
+        # the fused exp recomputes K/64 times per element.  The separate
+        # two-pass GN kernel stays the fast path.
+        h = self.conv1(self.norm1(x))
+        return self.conv2(self.norm2(h), residual=self.skip(x))
 
 
 class VAEMidAttention(nn.Module):
@@ -89,7 +92,7 @@ class VAEDecoder(nn.Module):
                 h = blk(h)
             if len(ups):
                 h = ups[0](h, upsample=True)  # fused nearest-2x + conv
-        return self.conv_out.forward_gn(h, self.norm_out)
+        return self.conv_out(self.norm_out(h))
 
 
 class VAEDecoderSmall(VAEDecoder):
