@@ -29,21 +29,19 @@
 #include "common.h"
 
 #include <cstdio>
+#include <cstdlib>
 
 #define CV_NW 4      // waves per block
-#define CV_ROWS 8    // output rows per block
 #define CV_COLS 32   // output cols per block
 #define CV_BK 64     // output channels per block
 #define CV_CC 16     // input-channel chunk (MFMA K)
 
 typedef __attribute__((ext_vector_type(16))) float f32x16c;
 
-// LDS patch: [CV_ROWS+2][CV_COLS+2][CV_CC] bf16, c innermost.
-#define PATCH_R (CV_ROWS + 2)
+// RPW = output rows per wave (block rows = CV_NW*RPW).  RPW=2 runs 3
+// waves/SIMD; RPW=4 doubles MFMA per staged byte/A-read at 2 waves/SIMD.
+// LDS patch: [rows+2][CV_COLS+2][CV_CC] bf16, c innermost.
 #define PATCH_C (CV_COLS + 2)
-#define PATCH_ELEMS (PATCH_R * PATCH_C * CV_CC)
-#define STAGE_SLOTS (PATCH_R * PATCH_C * 2)             // (row, col, c-oct)
-#define SLOTS_PER_T ((STAGE_SLOTS + CV_NW * WAVE - 1) / (CV_NW * WAVE))
 
 // weight tile in LDS: [9 taps][CV_BK k][CV_CC c], staged once per c-chunk and
 // shared by all 4 waves (per-wave global A-reads were 4x-redundant L2
@@ -59,13 +57,19 @@ typedef __attribute__((ext_vector_type(16))) float f32x16c;
 // GN: fuse GroupNorm+SiLU into the staging read — silu(x*scale[c]+shift[c])
 // with per-(n,c) coefficients from gn_conv_coeffs_bf16 (norms.hip).  The
 // gn_norm write+read pass over the full tensor disappears.
-template <bool UP, bool GN>
+template <bool UP, bool GN, int RPW>
 __global__ __launch_bounds__(CV_NW * WAVE) void conv3x3_kernel(
     const short* __restrict__ in, const short* __restrict__ wr,
     const float* __restrict__ bias, const short* __restrict__ res,
     short* __restrict__ out, const float* __restrict__ gn_scale,
     const float* __restrict__ gn_shift, int C, int H, int W, int Hs, int Ws,
     int K, int C16, int Kpad, int npix_x, int npix, int nk) {
+  constexpr int CV_ROWS = CV_NW * RPW;
+  constexpr int PATCH_R = CV_ROWS + 2;
+  constexpr int PATCH_ELEMS = PATCH_R * PATCH_C * CV_CC;
+  constexpr int STAGE_SLOTS = PATCH_R * PATCH_C * 2;  // (row, col, c-oct)
+  constexpr int SLOTS_PER_T =
+      (STAGE_SLOTS + CV_NW * WAVE - 1) / (CV_NW * WAVE);
   __shared__ alignas(16) short patch[PATCH_ELEMS];
   __shared__ alignas(16) short wlds[WLDS_ELEMS];
 
@@ -181,10 +185,10 @@ __global__ __launch_bounds__(CV_NW * WAVE) void conv3x3_kernel(
     }
   };
 
-  // accumulators: [row-pair rr][k-fragment mf]
-  f32x16c acc[2][2];
+  // accumulators: [wave-row rr][k-fragment mf]
+  f32x16c acc[RPW][2];
 #pragma unroll
-  for (int a = 0; a < 2; ++a)
+  for (int a = 0; a < RPW; ++a)
 #pragma unroll
     for (int b = 0; b < 2; ++b)
 #pragma unroll
@@ -208,8 +212,8 @@ __global__ __launch_bounds__(CV_NW * WAVE) void conv3x3_kernel(
       bf16x8 a0 = *(const bf16x8*)&wlds[(tap * CV_BK + l31) * CV_CC + hi5 * 8];
       bf16x8 a1 = *(const bf16x8*)&wlds[(tap * CV_BK + 32 + l31) * CV_CC + hi5 * 8];
 #pragma unroll
-      for (int rr = 0; rr < 2; ++rr) {
-        const int row = w * 2 + rr;
+      for (int rr = 0; rr < RPW; ++rr) {
+        const int row = w * RPW + rr;
         bf16x8 b = *(const bf16x8*)&patch[((row + dy) * PATCH_C + l31 + dx) *
                                               CV_CC + hi5 * 8];
         __builtin_amdgcn_s_setprio(1);
@@ -224,8 +228,8 @@ __global__ __launch_bounds__(CV_NW * WAVE) void conv3x3_kernel(
   const int x = x0 + l31;
   if (x < W) {
 #pragma unroll
-    for (int rr = 0; rr < 2; ++rr) {
-      const int y = y0 + w * 2 + rr;
+    for (int rr = 0; rr < RPW; ++rr) {
+      const int y = y0 + w * RPW + rr;
       if (y >= H) continue;
 #pragma unroll
       for (int mf = 0; mf < 2; ++mf) {
@@ -253,18 +257,38 @@ extern "C" void conv3x3_bf16(const void* in, const void* wrepack,
   // H/W are the OUTPUT dims; with upsample the source is H/2 x W/2.
   const int Hs = upsample ? H / 2 : H;
   const int Ws = upsample ? W / 2 : W;
+  // rows-per-wave A/B (MODAL_AMD_CONV_RPW=4): doubles MFMA per staged
+  // byte/A-read at 2 waves/SIMD occupancy.
+  static const int rpw = [] {
+    const char* e = getenv("MODAL_AMD_CONV_RPW");
+    return (e && atoi(e) == 4) ? 4 : 2;
+  }();
+  const int rows = CV_NW * rpw;
   const int npix_x = (W + CV_COLS - 1) / CV_COLS;
-  const int npix_y = (H + CV_ROWS - 1) / CV_ROWS;
+  const int npix_y = (H + rows - 1) / rows;
   const int npix = npix_x * npix_y;
   const int nk = (K + CV_BK - 1) / CV_BK;
   dim3 grid(npix * nk, N);
   dim3 block(CV_NW * WAVE);
   const bool gn = gn_scale != nullptr;
-#define CVL(UPV, GNV)                                                           hipLaunchKernelGGL((conv3x3_kernel<UPV, GNV>), grid, block, 0, stream,                           (const short*)in, (const short*)wrepack,                                      (const float*)bias, (const short*)residual, (short*)out,                      gn_scale, gn_shift, C, H, W, Hs, Ws, K, C16, Kpad,                            npix_x, npix, nk)
-  if (upsample) {
-    if (gn) CVL(true, true); else CVL(true, false);
+#define CVL(UPV, GNV, RPWV) \
+  hipLaunchKernelGGL((conv3x3_kernel<UPV, GNV, RPWV>), grid, block, 0, \
+                     stream, (const short*)in, (const short*)wrepack, \
+                     (const float*)bias, (const short*)residual, (short*)out, \
+                     gn_scale, gn_shift, C, H, W, Hs, Ws, K, C16, Kpad, \
+                     npix_x, npix, nk)
+  if (rpw == 4) {
+    if (upsample) {
+      if (gn) CVL(true, true, 4); else CVL(true, false, 4);
+    } else {
+      if (gn) CVL(false, true, 4); else CVL(false, false, 4);
+    }
   } else {
-    if (gn) CVL(false, true); else CVL(false, false);
+    if (upsample) {
+      if (gn) CVL(true, true, 2); else CVL(true, false, 2);
+    } else {
+      if (gn) CVL(false, true, 2); else CVL(false, false, 2);
+    }
   }
 #undef CVL
 }
