@@ -41,7 +41,11 @@ def rocprof_stats_command(cmd: List[str], out_dir: str = "gpurun_out/prof",
                           name: str = "run") -> List[str]:
     """Build a rocprofv3 --stats invocation (kernel time table).  NOTE: never
     combine --pmc with trace domains in one run (node-stability rule)."""
-    return ["rocprofv3", "--stats", "-d", out_dir, "-o", name, "--"] + cmd
+    # ROCm 7.2 rocprofv3: --stats must combine with a tracing type, and the
+    # default output is a rocpd DB whose post-processing takes many minutes —
+    # csv keeps it seconds.
+    return ["rocprofv3", "--kernel-trace", "--stats", "-f", "csv",
+            "-d", out_dir, "-o", name, "--"] + cmd
 
 
 def rocprof_counters_command(cmd: List[str], counters: Optional[List[str]] = None,
