@@ -71,6 +71,7 @@ class Call:
         "t_submit",
         "sticky_key",
         "pool",
+        "on_done",
     )
 
     def __init__(self, method_name, args_blob, is_gen, timeout, retries: RetryPolicy, spawned=False,
@@ -93,6 +94,7 @@ class Call:
         self.t_submit = time.monotonic()
         self.sticky_key = sticky_key
         self.pool = None  # set at submit; lets FunctionCall.cancel reach us
+        self.on_done = None  # optional completion callback (map wait-any)
 
     def wait(self, timeout: Optional[float] = None):
         if not self.event.wait(timeout):
@@ -491,6 +493,11 @@ class Pool:
         if call.gen_q is not None:
             call.gen_q.close()
         call.event.set()
+        if call.on_done is not None:
+            try:
+                call.on_done(call)
+            except Exception:
+                pass
         if call.spawned and self.on_spawned_result is not None:
             try:
                 self.on_spawned_result(call)

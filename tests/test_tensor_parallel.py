@@ -163,3 +163,35 @@ def test_clustered_spawn_gather():
 
     fc = whoami.spawn()
     assert fc.get(timeout=60) == 0
+
+
+def test_all_reduce_smart_matches_ring_both_regimes():
+    """Size-aware all-reduce: the one-shot small-message path and the ring
+    large-message path both produce the exact all_reduce sum (gloo world 4,
+    SURVEY §5.8 algorithm-selection row)."""
+    import modal_examples_amd as modal
+
+    app = modal.App("test-smart-allreduce")
+
+    @app.function(timeout=120)
+    @modal.experimental.clustered(size=4)
+    def reduce_both() -> list:
+        import torch
+        import torch.distributed as dist
+
+        from modal_examples_amd.parallel.collectives import all_reduce_smart
+
+        dist.init_process_group("gloo", init_method="env://")
+        r = dist.get_rank()
+        small = torch.full((64,), float(r + 1))          # one-shot path
+        big = torch.full((64 * 1024,), float(r + 1))     # ring path (256 KiB)
+        all_reduce_smart(small)
+        all_reduce_smart(big)
+        want = float(sum(range(1, 5)))
+        ok = bool(small.eq(want).all()) and bool(big.eq(want).all())
+        dist.barrier()
+        dist.destroy_process_group()
+        return [ok, float(small[0]), float(big[0])]
+
+    out = reduce_both.remote()
+    assert out[0], out
