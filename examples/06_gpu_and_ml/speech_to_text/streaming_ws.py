@@ -56,6 +56,24 @@ def web():
     w = FastAPI()
     transcriber = StreamTranscriber()
 
+    # vanilla-JS mic-streaming page (the kyutai frontend role)
+    from pathlib import Path
+
+    front = Path(__file__).parent / "stt_frontend"
+
+    @w.get("/")
+    async def index():
+        from fastapi.responses import HTMLResponse
+
+        return HTMLResponse((front / "index.html").read_text())
+
+    @w.get("/app.js")
+    async def appjs():
+        from fastapi.responses import Response
+
+        return Response((front / "app.js").read_text(),
+                        media_type="text/javascript")
+
     @w.websocket("/stream")
     async def stream(ws: WebSocket):
         """Protocol: binary frames = float32 PCM @16 kHz; text "flush" ends an
@@ -101,6 +119,8 @@ def self_test():
     root = build_ingress_app(app)
     rng = np.random.default_rng(0)
     with TestClient(root) as client:
+        page = client.get("/stt-ws/")
+        assert page.status_code == 200 and "Streaming transcription" in page.text
         with client.websocket_connect("/stt-ws/stream") as ws:
             for utt in range(2):
                 t0 = time.monotonic()
