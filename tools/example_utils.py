@@ -33,6 +33,8 @@ def parse_frontmatter(path: Path) -> dict:
     import yaml
 
     lines = path.read_text().splitlines()
+    if lines and lines[0].startswith("#!"):
+        lines = lines[1:]  # tolerate a shebang (misc/hello_shebang.py)
     if not lines or lines[0].strip() != "# ---":
         return {}
     block = []
