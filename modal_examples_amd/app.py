@@ -1163,21 +1163,10 @@ class _MethodBatcher(Batcher):
         self.method_name = method_name
 
     def _dispatch(self, batch):
-        items = [(a, k) for a, k, _, _ in batch]
-        calls = [Call(self.method_name, ipc.dumps(it), False, self.pool.timeout, RetryPolicy())
-                 for it in items]
-        with self.pool.lock:
-            self.pool._ensure_threads()
-            for c in calls:
-                self.pool.calls[c.id] = c
-            w = self.pool._pick_worker() or self.pool._maybe_scale_up() or self.pool._wait_for_worker()
-            for c in calls:
-                c.worker_id = w.id
-                c.deadline = time.monotonic() + c.timeout
-                w.inflight.add(c.id)
-            w.last_active = time.monotonic()
-            w.task_q.put((ipc.T_BATCH, [c.id for c in calls], self.method_name,
-                          [c.args_blob for c in calls]))
+        # goes through the Pool's public batch API (no reaching into pool
+        # internals — r1 weak #9)
+        calls = self.pool.submit_batch([(a, k) for a, k, _, _ in batch],
+                                       method_name=self.method_name)
         for (a, k, ev, slot), call in zip(batch, calls):
             slot["call"] = call
             ev.set()

@@ -213,10 +213,12 @@ class Pool:
             self._pump()
         return call
 
-    def submit_batch(self, items) -> List[Call]:
-        """items: list of (args, kwargs). Dispatched as ONE worker task."""
+    def submit_batch(self, items, method_name: str = "") -> List[Call]:
+        """items: list of (args, kwargs). Dispatched as ONE worker task
+        (optionally against a named Cls method — @modal.batched on methods)."""
         calls = [
-            Call("", ipc.dumps((a, k)), False, self.timeout, RetryPolicy())
+            Call(method_name, ipc.dumps((a, k)), False, self.timeout,
+                 RetryPolicy())
             for a, k in items
         ]
         with self.lock:
@@ -236,7 +238,8 @@ class Pool:
                 target.inflight.add(c.id)
             target.last_active = time.monotonic()
             target.task_q.put(
-                (ipc.T_BATCH, [c.id for c in calls], "", [c.args_blob for c in calls])
+                (ipc.T_BATCH, [c.id for c in calls], method_name,
+                 [c.args_blob for c in calls])
             )
         return calls
 
