@@ -24,6 +24,10 @@ def main():
     ap.add_argument("--layers", type=int, default=0, help="0 = full 32")
     ap.add_argument("--no-graph", action="store_true")
     ap.add_argument("--small", action="store_true")
+    ap.add_argument("--ragged", action="store_true",
+                    help="mixed prompt lengths (uniform prompt_len/4 .. "
+                         "prompt_len) — exercises the one-padded-forward "
+                         "ragged prefill")
     args = ap.parse_args()
 
     import torch
@@ -45,10 +49,14 @@ def main():
         torch.cuda.synchronize()
     init_s = time.perf_counter() - t0
 
-    prompt = [(i % 1000) + 10 for i in range(args.prompt_len)]
-    for _ in range(args.requests):
-        eng.add_request(list(prompt), max_new_tokens=args.gen_len,
-                        temperature=0.0)
+    import random
+
+    rng = random.Random(7)
+    for r in range(args.requests):
+        plen = (rng.randint(args.prompt_len // 4, args.prompt_len)
+                if args.ragged else args.prompt_len)
+        prompt = [((i * 31 + r) % 1000) + 10 for i in range(plen)]
+        eng.add_request(prompt, max_new_tokens=args.gen_len, temperature=0.0)
     if device == "cuda":
         torch.cuda.synchronize()
     t0 = time.perf_counter()
@@ -58,12 +66,13 @@ def main():
     dt = time.perf_counter() - t0
 
     out_toks = sum(len(r.out_tokens) for r in eng.finished.values())
-    in_toks = args.requests * args.prompt_len
+    in_toks = sum(len(r.prompt) for r in eng.finished.values())
     print(json.dumps({
         "metric": "llama decode throughput",
         "model": f"llama3-8b-class ({cfg.n_layers}L)",
         "requests": args.requests,
         "prompt_len": args.prompt_len,
+        "ragged": args.ragged,
         "gen_len": args.gen_len,
         "elapsed_s": round(dt, 3),
         "init_s": round(init_s, 2),
