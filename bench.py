@@ -28,7 +28,11 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=3, help="timed generate() calls")
     ap.add_argument("--warmup", type=int, default=1)
-    ap.add_argument("--batch", type=int, default=4, help="images per rank per step")
+    # batch 16 is the measured throughput knee with the r2 hand-conv kernels
+    # (scripts/sweep_batch.py: 13.3 img/s @B4 -> 16.2 @B16, 16.3 @B24,
+    # declining @B32) and the top of the reference's own stated batch range
+    # (text_to_image.py:13 "batch 1-16")
+    ap.add_argument("--batch", type=int, default=16, help="images per rank per step")
     ap.add_argument("--denoise-steps", type=int, default=4)
     ap.add_argument("--latent", type=int, default=128, help="128 → 1024px")
     ap.add_argument("--small", action="store_true", help="test-size model")
