@@ -36,9 +36,14 @@ def all_ranks_broadcast() -> str:
 
     from modal_examples_amd.parallel.cluster import get_cluster_info
 
+    import os
+
     info = get_cluster_info()
     rank, world = info.rank, len(info.container_ips)
-    use_gpu = torch.cuda.is_available()
+    # RCCL only when this rank has its OWN pinned GPU (the runtime sets
+    # HIP_VISIBLE_DEVICES per worker); on a box with fewer GPUs than ranks
+    # the gang runs gloo — two ranks on one device is an RCCL error
+    use_gpu = torch.cuda.is_available() and bool(os.environ.get("HIP_VISIBLE_DEVICES"))
     backend = "nccl" if use_gpu else "gloo"  # "nccl" IS RCCL on ROCm
     dist.init_process_group(backend, init_method="env://")
     dev = "cuda:0" if use_gpu else "cpu"  # each rank sees only its pinned GPU

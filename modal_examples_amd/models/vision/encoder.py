@@ -51,11 +51,12 @@ class ViTBlock(nn.Module):
 
     def forward(self, x):
         B, S, h = x.shape
-        qkv = self.qkv(self.norm1(x.float()).to(x.dtype)).view(
-            B, S, 3, self.heads, self.hd)
+        # LayerNorm runs in the module dtype: affine weights are bf16 after
+        # .to(dtype), and GPU layer_norm rejects float-input/bf16-weight
+        qkv = self.qkv(self.norm1(x)).view(B, S, 3, self.heads, self.hd)
         o = OF.attention_qkv(qkv[:, :, 0], qkv[:, :, 1], qkv[:, :, 2])
         x = x + self.proj(o)
-        return x + self.mlp(self.norm2(x.float()).to(x.dtype))
+        return x + self.mlp(self.norm2(x))
 
 
 class VisionEncoder(nn.Module):
@@ -80,7 +81,7 @@ class VisionEncoder(nn.Module):
         x = x + self.pos[:, : x.shape[1]]
         for blk in self.blocks:
             x = blk(x)
-        e = self.out_proj(self.norm(x.float()).to(x.dtype))
+        e = self.out_proj(self.norm(x))
         return torch.nn.functional.normalize(e.float(), dim=-1)
 
 
