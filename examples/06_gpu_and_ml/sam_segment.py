@@ -46,7 +46,8 @@ class Segmenter:
                    thresh: float = 0.75) -> list:
         """Mask from a point prompt, decoded from the CACHED embedding."""
         e = self.embeddings[image_id]
-        side = self.cfg.image_size // self.cfg.patch
+        side = int(e.shape[0] ** 0.5)  # from the CACHED embedding, not the
+        # config image size — prompts address the embedded image's grid
         prompt_idx = (py // self.cfg.patch) * side + (px // self.cfg.patch)
         sim = (e @ e[prompt_idx]).float()  # embeddings are L2-normalized
         mask = (sim >= thresh).reshape(side, side)

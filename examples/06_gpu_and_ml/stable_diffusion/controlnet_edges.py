@@ -67,7 +67,11 @@ def main():
     r1 = m.generate.remote("a neon sign", edges_a)
     r1_again = m.generate.remote("a neon sign", edges_a)
     r2 = m.generate.remote("a neon sign", edges_b)
-    assert r1 == r1_again, "conditioned generation must be deterministic"
-    assert r1["checksum"] != r2["checksum"], "control map must steer the output"
+    # repeatable up to the GroupNorm stats kernel's float atomics (a few
+    # uint8 rounding flips); a different control map moves the output far more
+    tol = max(256, int(0.001 * r1["checksum"]))
+    assert abs(r1["checksum"] - r1_again["checksum"]) <= tol, (r1, r1_again)
+    diff = abs(r1["checksum"] - r2["checksum"])
+    assert diff > 4 * tol, f"control map must steer the output ({diff} <= {4*tol})"
     print(f"control steers output: {r1['checksum']} vs {r2['checksum']}, "
           f"image {r1['shape']}")
