@@ -69,9 +69,11 @@ def main():
     r2 = m.generate.remote("a neon sign", edges_b)
     # repeatable up to the GroupNorm stats kernel's float atomics (a few
     # uint8 rounding flips); a different control map moves the output far more
-    tol = max(256, int(0.001 * r1["checksum"]))
-    assert abs(r1["checksum"] - r1_again["checksum"]) <= tol, (r1, r1_again)
+    drift = abs(r1["checksum"] - r1_again["checksum"])  # GN float atomics
+    tol = max(1000, int(0.0005 * r1["checksum"]))
+    assert drift <= tol, (r1, r1_again)
     diff = abs(r1["checksum"] - r2["checksum"])
-    assert diff > 4 * tol, f"control map must steer the output ({diff} <= {4*tol})"
+    assert diff > 10 * max(drift, 100), \
+        f"control map must steer the output (diff {diff}, drift {drift})"
     print(f"control steers output: {r1['checksum']} vs {r2['checksum']}, "
           f"image {r1['shape']}")
