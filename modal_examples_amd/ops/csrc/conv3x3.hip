@@ -57,12 +57,7 @@ typedef __attribute__((ext_vector_type(16))) float f32x16c;
 // GN: fuse GroupNorm+SiLU into the staging read — silu(x*scale[c]+shift[c])
 // with per-(n,c) coefficients from gn_conv_coeffs_bf16 (norms.hip).  The
 // gn_norm write+read pass over the full tensor disappears.
-// VS: vectorized staging — each thread LOADS one bf16x8 of 8 consecutive x
-// (16B-aligned span [x0-8, x0+40)) and SCATTERS 8 ds_write_b16 into the
-// c-contiguous patch, instead of 8 scalar global loads + 1 vector LDS
-// write.  Trades VMEM request count (the PMC-flagged staging cost,
-// VALU:MFMA 6:1) for LDS-pipe writes.  A/B via MODAL_AMD_CONV_VSTAGE=1.
-template <bool UP, bool GN, int RPW, bool VS = false>
+template <bool UP, bool GN, int RPW>
 __global__ __launch_bounds__(CV_NW * WAVE) void conv3x3_kernel(
     const short* __restrict__ in, const short* __restrict__ wr,
     const float* __restrict__ bias, const short* __restrict__ res,
@@ -106,86 +101,11 @@ __global__ __launch_bounds__(CV_NW * WAVE) void conv3x3_kernel(
   // ---- staging: slot s = (row, col, c-oct); thread gathers 8 strided
   // channel values (coalesced across lanes: consecutive threads read
   // consecutive x) and writes ONE bf16x8 to the c-contiguous LDS slot.
-  // --- VS staging: slot v = (c, row, col-octet); span of 6 octets covers
-  // patch cols [-1, 34) after the aligned base x0-8.
-  constexpr int VOCT = 6;
-  constexpr int VSLOTS = CV_CC * PATCH_R * VOCT;
-  constexpr int VSLOTS_PER_T = (VSLOTS + CV_NW * WAVE - 1) / (CV_NW * WAVE);
-  bf16x8 vreg[VS ? VSLOTS_PER_T : 1];
-  auto vstage_load = [&](int cc) {
-    const int c0 = cc * CV_CC;
-    // unroll 1: full unroll quadrupled live ranges (194 VGPR + SGPR spill)
-#pragma unroll 1
-    for (int i = 0; i < (VS ? VSLOTS_PER_T : 0); ++i) {
-      int sl = i * CV_NW * WAVE + tid;
-      bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (sl < VSLOTS) {
-        int oct = sl % VOCT;
-        int u = sl / VOCT;
-        int row = u % PATCH_R;
-        int c = c0 + u / PATCH_R;
-        int y = y0 + row - 1;
-        int xb = x0 - 8 + oct * 8;
-        if (c < C && y >= 0 && y < H && xb < W) {
-          const int sy = UP ? (y >> 1) : y;
-          const long long hw = (long long)Hs * Ws;
-          const short* srow = in + in_n + (long long)c * hw +
-                              (long long)sy * Ws;
-          if (!UP && xb >= 0 && xb + 8 <= W) {
-            v = *(const bf16x8*)&srow[xb];
-          } else {
-#pragma unroll
-            for (int k = 0; k < 8; ++k) {
-              int x = xb + k;
-              int sx = UP ? (x >> 1) : x;
-              v[k] = (x >= 0 && x < W) ? srow[sx] : (short)0;
-            }
-          }
-          if (GN) {
-            float gsc = gn_scale[(long long)n * C16 + c];
-            float gsh = gn_shift[(long long)n * C16 + c];
-#pragma unroll
-            for (int k = 0; k < 8; ++k) {
-              int x = xb + k;
-              float f = bf2f(v[k]) * gsc + gsh;
-              f = f / (1.f + __expf(-f));
-              v[k] = (x >= 0 && x < W) ? f2bf(f) : (short)0;
-            }
-          }
-        }
-      }
-      vreg[VS ? i : 0] = v;
-    }
-  };
-  auto vstage_write = [&]() {
-#pragma unroll 1
-    for (int i = 0; i < (VS ? VSLOTS_PER_T : 0); ++i) {
-      int sl = i * CV_NW * WAVE + tid;
-      if (sl < VSLOTS) {
-        int oct = sl % VOCT;
-        int u = sl / VOCT;
-        int row = u % PATCH_R;
-        int cl = u / PATCH_R;  // c - c0
-        bf16x8 v = vreg[VS ? i : 0];
-#pragma unroll
-        for (int k = 0; k < 8; ++k) {
-          int col = oct * 8 + k - 7;  // patch col = x - (x0-1)
-          if (col >= 0 && col < PATCH_C)
-            patch[(row * PATCH_C + col) * CV_CC + cl] = v[k];
-        }
-      }
-    }
-  };
-
-  bf16x8 sreg[VS ? 1 : SLOTS_PER_T];
+  bf16x8 sreg[SLOTS_PER_T];
   auto stage_load = [&](int cc) {
-    if (VS) {
-      vstage_load(cc);
-      return;
-    }
     const int c0 = cc * CV_CC;
 #pragma unroll
-    for (int i = 0; i < (VS ? 0 : SLOTS_PER_T); ++i) {
+    for (int i = 0; i < SLOTS_PER_T; ++i) {
       int s = i * CV_NW * WAVE + tid;
       bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
       if (s < STAGE_SLOTS) {
@@ -222,24 +142,19 @@ __global__ __launch_bounds__(CV_NW * WAVE) void conv3x3_kernel(
           }
         }
       }
-      sreg[VS ? 0 : i] = v;
+      sreg[i] = v;
     }
   };
   auto stage_write = [&]() {
-    if (VS) {
-      vstage_write();
-      return;
-    }
 #pragma unroll
-    for (int i = 0; i < (VS ? 0 : SLOTS_PER_T); ++i) {
+    for (int i = 0; i < SLOTS_PER_T; ++i) {
       int s = i * CV_NW * WAVE + tid;
       if (s < STAGE_SLOTS) {
         int col = s % PATCH_C;
         int u = s / PATCH_C;
         int oct = u & 1;
         int row = u >> 1;
-        *(bf16x8*)&patch[(row * PATCH_C + col) * CV_CC + oct * 8] =
-            sreg[VS ? 0 : i];
+        *(bf16x8*)&patch[(row * PATCH_C + col) * CV_CC + oct * 8] = sreg[i];
       }
     }
   };
@@ -362,20 +277,7 @@ extern "C" void conv3x3_bf16(const void* in, const void* wrepack,
                      (const float*)bias, (const short*)residual, (short*)out, \
                      gn_scale, gn_shift, C, H, W, Hs, Ws, K, C16, Kpad, \
                      npix_x, npix, nk)
-  static const bool vstage = getenv("MODAL_AMD_CONV_VSTAGE") != nullptr;
-#define CVLV(UPV, GNV)                                                        \
-  hipLaunchKernelGGL((conv3x3_kernel<UPV, GNV, 2, true>), grid, block, 0,     \
-                     stream, (const short*)in, (const short*)wrepack,         \
-                     (const float*)bias, (const short*)residual, (short*)out, \
-                     gn_scale, gn_shift, C, H, W, Hs, Ws, K, C16, Kpad,       \
-                     npix_x, npix, nk)
-  if (vstage && rpw == 2) {
-    if (upsample) {
-      if (gn) CVLV(true, true); else CVLV(true, false);
-    } else {
-      if (gn) CVLV(false, true); else CVLV(false, false);
-    }
-  } else if (rpw == 4) {
+  if (rpw == 4) {
     if (upsample) {
       if (gn) CVL(true, true, 4); else CVL(true, false, 4);
     } else {
@@ -388,6 +290,5 @@ extern "C" void conv3x3_bf16(const void* in, const void* wrepack,
       if (gn) CVL(false, true, 2); else CVL(false, false, 2);
     }
   }
-#undef CVLV
 #undef CVL
 }
