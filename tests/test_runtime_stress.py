@@ -55,3 +55,19 @@ def test_spawn_burst():
     calls = [inc.spawn(i) for i in range(150)]
     vals = modal.functions.gather(*calls)
     assert vals == [i + 1 for i in range(150)]
+
+
+def test_unordered_map_large_window_event_driven():
+    """1000 unordered inputs drain correctly through the wait-any event path
+    (no 2ms busy-poll; r1 weak #8)."""
+    import modal_examples_amd as modal
+
+    app = modal.App("test-map-large")
+
+    @app.function(max_containers=4)
+    @modal.concurrent(max_inputs=8)
+    def inc(x: int) -> int:
+        return x + 1
+
+    out = sorted(inc.map(range(1000), order_outputs=False))
+    assert out == list(range(1, 1001))
