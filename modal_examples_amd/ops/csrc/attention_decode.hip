@@ -29,7 +29,11 @@
 #define DEC_WAVES 4
 #define MAX_G 8
 
-template <int D>
+// GT = compile-time GQA group size: register arrays (qf/accv) and the LDS
+// merge buffers are sized by GT, not MAX_G — at Llama's G=4 this halves the
+// VGPR/LDS footprint (205 VGPR/66 KB -> more waves + blocks per CU; decode
+// is HBM-latency-bound so occupancy IS the bandwidth lever, probe_vmcnt_r2).
+template <int D, int GT>
 __global__ __launch_bounds__(DEC_THREADS) void paged_decode_kernel(
     const short* __restrict__ Q,       // [B, Hq, D]
     const short* __restrict__ Kc,      // cache, layout above
@@ -62,9 +66,9 @@ __global__ __launch_bounds__(DEC_THREADS) void paged_decode_kernel(
   const int part = w * ROWS + rg;     // global partial index
 
   // Q fragments: per head g, this lane's 8 bf16 of q (its slot), as f32
-  float qf[MAX_G][8];
+  float qf[GT][8];
 #pragma unroll
-  for (int g = 0; g < MAX_G; ++g) {
+  for (int g = 0; g < GT; ++g) {
     if (g < G) {
       const short* qp = &Q[(((long long)b * Hq) + hkv * G + g) * D + slot * 8];
       bf16x8 q8 = *(const bf16x8*)qp;
@@ -73,9 +77,9 @@ __global__ __launch_bounds__(DEC_THREADS) void paged_decode_kernel(
     }
   }
 
-  float m_run[MAX_G], l_run[MAX_G], accv[MAX_G][8];
+  float m_run[GT], l_run[GT], accv[GT][8];
 #pragma unroll
-  for (int g = 0; g < MAX_G; ++g) {
+  for (int g = 0; g < GT; ++g) {
     m_run[g] = -1e30f;
     l_run[g] = 0.f;
 #pragma unroll
@@ -120,7 +124,7 @@ __global__ __launch_bounds__(DEC_THREADS) void paged_decode_kernel(
     for (int j = 0; j < 8; ++j) kfl[j] = bf2f(k8[j]);
 
 #pragma unroll
-    for (int g = 0; g < MAX_G; ++g) {
+    for (int g = 0; g < GT; ++g) {
       if (g >= G) break;
       float d = 0.f;
 #pragma unroll
@@ -140,12 +144,12 @@ __global__ __launch_bounds__(DEC_THREADS) void paged_decode_kernel(
   }
 
   // ---- merge partials through LDS ----
-  __shared__ float sm[MAX_G][NPART];
-  __shared__ float sl[MAX_G][NPART];
-  __shared__ float sacc[MAX_G][NPART][D];
+  __shared__ float sm[GT][NPART];
+  __shared__ float sl[GT][NPART];
+  __shared__ float sacc[GT][NPART][D];
 
 #pragma unroll
-  for (int g = 0; g < MAX_G; ++g) {
+  for (int g = 0; g < GT; ++g) {
     if (g >= G) break;
     if (slot == 0) {
       sm[g][part] = m_run[g];
@@ -222,19 +226,28 @@ extern "C" void paged_decode_bf16(const void* q, const void* kc, const void* vc,
   dim3 grid(Hkv, B, splits);
   dim3 block(DEC_THREADS);
   float* ws_arg = splits > 1 ? ws : nullptr;
-#define DLAUNCH(DD)                                                          \
-  hipLaunchKernelGGL((paged_decode_kernel<DD>), grid, block, 0, stream,      \
+  const int G = Hq / Hkv;
+#define DLAUNCH(DD, GG)                                                      \
+  hipLaunchKernelGGL((paged_decode_kernel<DD, GG>), grid, block, 0, stream,  \
                      (const short*)q, (const short*)kc, (const short*)vc,    \
                      block_table, seq_lens, (short*)o, ws_arg, B, Hq, Hkv,   \
                      block_size, max_blocks, splits, scale)
+#define DDISPATCH(DD)                                                        \
+  do {                                                                       \
+    if (G == 1) DLAUNCH(DD, 1);                                              \
+    else if (G == 2) DLAUNCH(DD, 2);                                         \
+    else if (G <= 4) DLAUNCH(DD, 4);                                         \
+    else DLAUNCH(DD, 8);                                                     \
+  } while (0)
   if (D == 64) {
-    DLAUNCH(64);
+    DDISPATCH(64);
   } else if (D == 128) {
-    DLAUNCH(128);
+    DDISPATCH(128);
   } else {
     fprintf(stderr, "paged_decode_bf16: unsupported head_dim %d\n", D);
     abort();
   }
+#undef DDISPATCH
 #undef DLAUNCH
   if (splits > 1) {
     hipLaunchKernelGGL(decode_merge_kernel, dim3(B * Hq), dim3(256), 0,
