@@ -303,40 +303,28 @@ extern "C" void fa32_fwd_strided_bf16(
   st.ob = strides[9]; st.oh = strides[10]; st.os = strides[11];
   dim3 grid((Sq + 127) / 128, Hq, B);
   dim3 block(FA32_NWAVES * WAVE);
-  // KVB is chosen per head-dim (A/B-measured, gpurun_out/attn_ab_r2.txt):
+  // KVB is chosen per head-dim (A/B-measured, profiles/attn_ab_r2.txt):
   //   D=64:  KVB=64 wins (447 vs 407 TF — fewer barrier/softmax rounds)
-  //   D=128: KVB=64 LOSES 40% (2x LDS -> 2 blocks/CU occupancy cliff,
-  //          the same mode r1's double-buffer ablation hit); keep 32.
-  // Ablation toggles: MODAL_AMD_FA_KVB=32|64 forces the block size,
-  // MODAL_AMD_FA_NODEFER disables defer-max rescale skipping (+5-10%).
-  static const int kvb_env = [] {
-    const char* e = getenv("MODAL_AMD_FA_KVB");
-    return e ? atoi(e) : 0;
-  }();
+  //   D=128: KVB=64 LOST 40% (causal instantiation lands at 1 wave/SIMD —
+  //          the r1 double-buffer failure mode); only KVB=32 is built.
+  // MODAL_AMD_FA_NODEFER disables defer-max rescale skipping (+5-10% A/B).
   static const bool defer_env = getenv("MODAL_AMD_FA_NODEFER") == nullptr;
-#define L32(DD, CC)                                                           \
+#define L32K(DD, CC, KK)                                                      \
   do {                                                                        \
-    const int kvb = kvb_env ? kvb_env : (DD == 64 ? 64 : 32);                 \
-    if (kvb == 64 && defer_env)                                               \
-      hipLaunchKernelGGL((fa32_kernel<DD, CC, 64, true>), grid, block, 0,     \
-                         stream, Qp, Kp, Vp, Op, B, Hq, Hkv, Sq, Sk, scale, st); \
-    else if (kvb == 64)                                                       \
-      hipLaunchKernelGGL((fa32_kernel<DD, CC, 64, false>), grid, block, 0,    \
-                         stream, Qp, Kp, Vp, Op, B, Hq, Hkv, Sq, Sk, scale, st); \
-    else if (defer_env)                                                       \
-      hipLaunchKernelGGL((fa32_kernel<DD, CC, 32, true>), grid, block, 0,     \
+    if (defer_env)                                                            \
+      hipLaunchKernelGGL((fa32_kernel<DD, CC, KK, true>), grid, block, 0,     \
                          stream, Qp, Kp, Vp, Op, B, Hq, Hkv, Sq, Sk, scale, st); \
     else                                                                      \
-      hipLaunchKernelGGL((fa32_kernel<DD, CC, 32, false>), grid, block, 0,    \
+      hipLaunchKernelGGL((fa32_kernel<DD, CC, KK, false>), grid, block, 0,    \
                          stream, Qp, Kp, Vp, Op, B, Hq, Hkv, Sq, Sk, scale, st); \
   } while (0)
   if (D == 64) {
-    if (causal) L32(64, true); else L32(64, false);
+    if (causal) L32K(64, true, 64); else L32K(64, false, 64);
   } else if (D == 128) {
-    if (causal) L32(128, true); else L32(128, false);
+    if (causal) L32K(128, true, 32); else L32K(128, false, 32);
   } else {
     fprintf(stderr, "fa32: unsupported head_dim %d\n", D);
     abort();
   }
-#undef L32
+#undef L32K
 }
