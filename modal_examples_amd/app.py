@@ -118,8 +118,15 @@ class _Runtime:
         q = store.QueueStore(_NAMED_SPAWN_QUEUE)
         deployments = store.DictStore("__deployments__")
         last_beat = 0.0
+        last_gc = time.monotonic()
         while not self._stop_dispatch:
             now = time.monotonic()
+            if now - last_gc > 300.0:
+                last_gc = now
+                try:
+                    store.gc_results()  # 7-day spawn-result retention
+                except Exception:
+                    pass
             if App._registry and now - last_beat > 5.0:
                 # heartbeat: from_name in other processes only trusts fresh
                 # records, so a dead deploy process stops attracting calls

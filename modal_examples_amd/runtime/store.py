@@ -95,7 +95,35 @@ def get_result(call_id: str):
     if row is None:
         return False, False, None, ""
     status, payload, tb = row
+    if status == "expired":
+        from ..exception import OutputExpiredError
+
+        return True, False, OutputExpiredError(
+            f"result of {call_id} expired (older than the retention window)"), tb
     return True, status == "ok", pickle.loads(payload), tb
+
+
+RESULT_TTL_S = 7 * 86400       # reference durability: "1M inputs, 7 days"
+TOMBSTONE_TTL_S = 30 * 86400   # expired markers linger so get() can explain
+
+
+def gc_results(now: Optional[float] = None) -> int:
+    """Spawn-result retention sweep: payloads older than 7 days become
+    lightweight 'expired' tombstones (FunctionCall.get then raises
+    OutputExpiredError); tombstones older than 30 days are dropped.
+    Returns rows touched.  Called periodically by the runtime daemon."""
+    conn = _DB.get()
+    t = time.time() if now is None else now
+    with _write_txn(conn):
+        cur = conn.execute(
+            "UPDATE results SET status='expired', payload=? "
+            "WHERE status != 'expired' AND ts < ?",
+            (cloudpickle.dumps(None), t - RESULT_TTL_S))
+        n = cur.rowcount
+        cur = conn.execute("DELETE FROM results WHERE status='expired' AND ts < ?",
+                           (t - TOMBSTONE_TTL_S,))
+        n += cur.rowcount
+    return n
 
 
 # ---------------- named Dict ----------------

@@ -113,3 +113,40 @@ def test_worker_sigkill_midcall_retries_on_fresh_worker():
     pid = preemptible.remote(key)
     assert isinstance(pid, int)
     assert pid != d.get(key), "retry must run in a new worker process"
+
+
+def test_spawn_result_retention_gc():
+    """Durable spawn results expire after the 7-day retention window: get()
+    then raises OutputExpiredError; ancient tombstones are dropped
+    (amazon_embeddings.py:17-18 durability contract)."""
+    import time
+
+    import pytest as _pytest
+
+    import modal_examples_amd as modal
+    from modal_examples_amd.exception import OutputExpiredError
+    from modal_examples_amd.runtime import store
+
+    store.put_result("fc-gc-fresh", True, 41)
+    store.put_result("fc-gc-old", True, 42)
+    store.put_result("fc-gc-ancient", True, 43)
+    conn = store._DB.get()
+    with conn:
+        conn.execute("UPDATE results SET ts=? WHERE call_id='fc-gc-old'",
+                     (time.time() - 8 * 86400,))
+        conn.execute("UPDATE results SET ts=? WHERE call_id='fc-gc-ancient'",
+                     (time.time() - 40 * 86400,))
+    touched = store.gc_results()
+    assert touched >= 2
+    found, ok, val, _ = store.get_result("fc-gc-fresh")
+    assert found and ok and val == 41
+    fc = modal.FunctionCall.from_id("fc-gc-old")
+    with _pytest.raises(OutputExpiredError):
+        fc.get(timeout=1)
+    # ancient: second sweep after tombstoning drops the row entirely
+    with conn:
+        conn.execute("UPDATE results SET ts=? WHERE call_id='fc-gc-ancient'",
+                     (time.time() - 40 * 86400,))
+    store.gc_results()
+    found, *_ = store.get_result("fc-gc-ancient")
+    assert not found
