@@ -130,13 +130,30 @@ class Sandbox:
             })
         except Exception:
             pass
+        self._probe = readiness_probe
+        self._detached = False
         if readiness_probe is not None and readiness_probe.kind == "exec":
-            deadline = time.monotonic() + min(60.0, timeout)
-            while time.monotonic() < deadline:
-                p = self.exec(*readiness_probe.args)
-                if p.wait() == 0:
-                    break
-                time.sleep(0.5)
+            self.wait_until_ready(min(60.0, timeout))
+
+    def wait_until_ready(self, timeout: float = 60.0) -> bool:
+        """Poll the readiness probe until it succeeds (sb.wait_until_ready,
+        13_sandboxes usage).  True on ready; False on timeout/no probe."""
+        if self._probe is None or self._probe.kind != "exec":
+            return self.poll() is None
+        deadline = time.monotonic() + timeout
+        while time.monotonic() < deadline:
+            p = self.exec(*self._probe.args)
+            if p.wait() == 0:
+                return True
+            time.sleep(0.5)
+        return False
+
+    def detach(self):
+        """Detach from the app lifecycle: the sandbox (and its timeout reaper)
+        keeps running after the creating context exits (sb.detach,
+        13_sandboxes/opencode_server.py pattern); reattach via from_id."""
+        self._detached = True
+        return self
 
     # --- modal API ---
 

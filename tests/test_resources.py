@@ -459,3 +459,19 @@ def test_bucket_mount_prefix_sync_and_writeback(tmp_path):
             break
         time.sleep(0.3)
     assert c.get("sync-bkt", "data/out.txt") == b"produced"
+
+
+def test_sandbox_wait_until_ready_and_detach(tmp_path):
+    marker = tmp_path / "ready"
+    sb = modal.Sandbox.create(
+        "bash", "-c", f"sleep 0.7; touch {marker}; sleep 30",
+        timeout=60,
+        readiness_probe=modal.Probe.with_exec(["test", "-f", str(marker)]))
+    try:
+        assert sb.wait_until_ready(20) is True
+        assert sb.poll() is None  # still running
+        assert sb.detach() is sb
+        sb2 = modal.Sandbox.from_id(sb.object_id)
+        assert sb2.poll() is None
+    finally:
+        sb.terminate()
