@@ -186,12 +186,48 @@ class WhisperModel(nn.Module):
                 # decode kernel wants [B, H, T, D] contiguous caches
                 c["ck"] = ck.transpose(1, 2).contiguous()
                 c["cv"] = cv.transpose(1, 2).contiguous()
+            elif c.get("persistent"):
+                # graph-captured decode: cross caches are STATIC buffers
+                # refreshed in place per call
+                _, ck, cv = layer.cross_attn.qkv(h, audio)
+                c["ck"].copy_(ck.transpose(1, 2))
+                c["cv"].copy_(cv.transpose(1, 2))
             x = x + layer.cross_attn(layer.ln2(x), ctx=audio)
             x = x + layer.mlp(layer.ln3(x))
         x = self.dec_ln(x[:, -1:])
         return (x @ self.tok_embed.weight.T.to(x.dtype))[:, 0].float()
 
     @torch.no_grad()
+    def decode_step_dev(self, tokens: torch.Tensor, pos_d: torch.Tensor,
+                        caches, audio_len: int):
+        """Graph-capturable single-token step: the position lives ON DEVICE
+        (int64 [1]) so the whole step replays as one hipGraph (K10's role for
+        the Whisper decoder, like the Llama engine's captured decode)."""
+        B = tokens.shape[0]
+        x = self.tok_embed(tokens).unsqueeze(1) + self.pos_embed.index_select(
+            0, pos_d).to(self.tok_embed.weight.dtype)
+        lens_self = (pos_d + 1).to(torch.int32).expand(B).contiguous()
+        lens_cross = torch.full((B,), audio_len, dtype=torch.int32,
+                                device=x.device)
+        for li, layer in enumerate(self.dec_layers):
+            c = caches[li]
+            h = layer.ln1(x)
+            q, k, v = layer.self_attn.qkv(h)  # BSHD [B,1,h,d]
+            c["k"].index_copy_(2, pos_d, k.transpose(1, 2))
+            c["v"].index_copy_(2, pos_d, v.transpose(1, 2))
+            o = OF.paged_decode(q[:, 0].contiguous(), c["k"], c["v"], None,
+                                lens_self)
+            x = x + layer.self_attn.out(o.reshape(B, 1, -1))
+            h2 = layer.ln2(x)
+            q2 = layer.cross_attn.q(h2).view(B, 1, -1)
+            q2 = q2.view(B, layer.cross_attn.h, layer.cross_attn.d)
+            o2 = OF.paged_decode(q2.contiguous(), c["ck"], c["cv"], None,
+                                 lens_cross)
+            x = x + layer.cross_attn.out(o2.reshape(B, 1, -1))
+            x = x + layer.mlp(layer.ln3(x))
+        x = self.dec_ln(x)
+        return (x @ self.tok_embed.weight.T.to(x.dtype))[:, 0].float()
+
     def decode_step(self, tokens: torch.Tensor, pos: int, caches, audio_len: int):
         """Single-token step for all sequences; contiguous caches (K6 kernel)."""
         B = tokens.shape[0]

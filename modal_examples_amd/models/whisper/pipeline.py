@@ -60,6 +60,54 @@ class WhisperPipeline:
             self.model = WhisperModel(self.cfg).to(self.device, dtype)
         self.model.eval()
         self.sot, self.eot = 1, 2  # synthetic special tokens
+        self.use_graph = self.device.type == "cuda"
+        from ...gpu.graphs import GraphLRU
+
+        self._graphs = GraphLRU(3)  # per-batch-size captured decode steps
+
+    def _decode_state(self, B: int):
+        """Persistent per-batch-size decode state + ONE captured step:
+        consume cur @ pos -> logits -> cur := argmax, pos += 1 (all on
+        device; the host loop is replay + one [B] read per token)."""
+        st = self._graphs.get(B)
+        if st is not None:
+            return st
+        cfg = self.cfg
+        H, D = cfg.n_head, cfg.n_state // cfg.n_head
+        dev = self.device
+        caches = [{
+            "persistent": True,
+            "k": torch.zeros(B, H, cfg.n_text_ctx, D, device=dev, dtype=self.dtype),
+            "v": torch.zeros(B, H, cfg.n_text_ctx, D, device=dev, dtype=self.dtype),
+            "ck": torch.zeros(B, H, cfg.n_audio_ctx, D, device=dev, dtype=self.dtype),
+            "cv": torch.zeros(B, H, cfg.n_audio_ctx, D, device=dev, dtype=self.dtype),
+        } for _ in range(cfg.n_text_layer)]
+        st = {
+            "caches": caches,
+            "cur": torch.zeros(B, dtype=torch.long, device=dev),
+            "pos": torch.ones(1, dtype=torch.long, device=dev),
+        }
+
+        def one_step():
+            logits = self.model.decode_step_dev(st["cur"], st["pos"],
+                                                st["caches"], cfg.n_audio_ctx)
+            st["cur"].copy_(logits.argmax(-1))
+            st["pos"].add_(1)
+
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side), torch.no_grad():
+            for _ in range(2):
+                st["pos"].fill_(1)
+                one_step()
+        torch.cuda.current_stream().wait_stream(side)
+        g = torch.cuda.CUDAGraph()
+        st["pos"].fill_(1)
+        with torch.cuda.graph(g), torch.no_grad():
+            one_step()
+        st["graph"] = g
+        self._graphs.put(B, st)
+        return st
 
     @torch.no_grad()
     def transcribe(self, audio_batch: List[torch.Tensor], max_tokens: int = 32
@@ -76,13 +124,18 @@ class WhisperPipeline:
         audio = self.model.encode(mel)
 
         H, D = cfg.n_head, cfg.n_state // cfg.n_head
-        caches = [
-            {
-                "k": torch.zeros(B, H, cfg.n_text_ctx, D, device=self.device, dtype=self.dtype),
-                "v": torch.zeros(B, H, cfg.n_text_ctx, D, device=self.device, dtype=self.dtype),
-            }
-            for _ in range(cfg.n_text_layer)
-        ]
+        if self.use_graph:
+            # hipGraph decode (K10): persistent caches + one captured step
+            st = self._decode_state(B)
+            caches = st["caches"]
+        else:
+            caches = [
+                {
+                    "k": torch.zeros(B, H, cfg.n_text_ctx, D, device=self.device, dtype=self.dtype),
+                    "v": torch.zeros(B, H, cfg.n_text_ctx, D, device=self.device, dtype=self.dtype),
+                }
+                for _ in range(cfg.n_text_layer)
+            ]
         tokens = torch.full((B, 1), self.sot, device=self.device, dtype=torch.long)
         logits = self.model.decode_prefill(tokens, audio, caches)
         outs = [[] for _ in range(B)]
@@ -90,12 +143,20 @@ class WhisperPipeline:
         cur = logits.argmax(-1)
         for i in range(B):
             outs[i].append(int(cur[i]))
+        if self.use_graph:
+            st["cur"].copy_(cur)
+            st["pos"].fill_(1)
         for pos in range(1, min(max_tokens, cfg.n_text_ctx - 1)):
-            logits = self.model.decode_step(cur, pos, caches, cfg.n_audio_ctx)
-            cur = logits.argmax(-1)
+            if self.use_graph:
+                st["graph"].replay()
+                cur_h = st["cur"].cpu()  # ONE [B] read per token
+            else:
+                logits = self.model.decode_step(cur, pos, caches, cfg.n_audio_ctx)
+                cur = logits.argmax(-1)
+                cur_h = cur.cpu()
             for i in range(B):
                 if alive[i]:
-                    t = int(cur[i])
+                    t = int(cur_h[i])
                     outs[i].append(t)
                     if t == self.eot:
                         alive[i] = False
