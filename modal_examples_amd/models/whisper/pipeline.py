@@ -113,8 +113,18 @@ class WhisperPipeline:
     def transcribe(self, audio_batch: List[torch.Tensor], max_tokens: int = 32
                    ) -> List[List[int]]:
         """Batch of mono 16 kHz waveforms → token id sequences (greedy)."""
-        B = len(audio_batch)
+        B_req = len(audio_batch)
         cfg = self.cfg
+        if self.use_graph:
+            # the dynamic batcher produces VARIABLE batch sizes; pad to the
+            # next power of two so graph captures stay bounded (<=7 keys)
+            B = 1
+            while B < B_req:
+                B *= 2
+            audio_batch = list(audio_batch) + [
+                torch.zeros(1600) for _ in range(B - B_req)]
+        else:
+            B = B_req
         maxlen = cfg.n_audio_ctx * 2 * HOP
         padded = torch.zeros(B, maxlen, device=self.device)
         for i, a in enumerate(audio_batch):
@@ -162,7 +172,7 @@ class WhisperPipeline:
                         alive[i] = False
             if not alive.any():
                 break
-        return outs
+        return outs[:B_req]
 
     def transcribe_text(self, audio_batch, max_tokens: int = 32) -> List[str]:
         return [" ".join(f"t{t}" for t in seq if t > 2)
