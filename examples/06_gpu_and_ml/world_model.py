@@ -56,9 +56,9 @@ def train_world_model(shards: list, steps: int = 150) -> dict:
 
     device = "cuda" if torch.cuda.is_available() else "cpu"
     data = [t for shard in shards for t in shard]
-    S = torch.tensor(np.array([d[0] for d in data]), device=device)
-    A = torch.tensor(np.array([d[1] for d in data]), device=device)
-    S2 = torch.tensor(np.array([d[2] for d in data]), device=device)
+    S = torch.tensor(np.array([d[0] for d in data], dtype="float32"), device=device)
+    A = torch.tensor(np.array([d[1] for d in data], dtype="float32"), device=device)
+    S2 = torch.tensor(np.array([d[2] for d in data], dtype="float32"), device=device)
 
     torch.manual_seed(0)
     model = nn.Sequential(
