@@ -368,3 +368,33 @@ def test_ragged_prefill_matches_solo_runs():
             if sid in solo.finished:
                 break
         assert solo.finished[sid].out_tokens == batch_out[rid], p
+
+
+@pytest.mark.parametrize("seed", [1, 2, 3])
+def test_ragged_prefill_random_mixes(seed):
+    """Randomized mixed-length batches: greedy outputs equal solo runs for
+    every prompt regardless of bucketing decisions."""
+    import random
+
+    rng = random.Random(seed)
+    torch.manual_seed(0)
+    cfg = LlamaConfig.small()
+    eng = LlamaEngine(cfg, device="cpu", dtype=torch.float32, max_batch=16,
+                      kv_blocks=128, use_graph=False, seed=3)
+    prompts = [[rng.randrange(5, 200) for _ in range(rng.randrange(1, 24))]
+               for _ in range(rng.randrange(3, 9))]
+    ids = [eng.add_request(p, max_new_tokens=4) for p in prompts]
+    for _ in range(80):
+        eng.step()
+        if len(eng.finished) == len(prompts):
+            break
+    torch.manual_seed(0)
+    solo = LlamaEngine(cfg, device="cpu", dtype=torch.float32, max_batch=16,
+                       kv_blocks=128, use_graph=False, seed=3)
+    for p, rid in zip(prompts, ids):
+        sid = solo.add_request(p, max_new_tokens=4)
+        for _ in range(40):
+            solo.step()
+            if sid in solo.finished:
+                break
+        assert solo.finished[sid].out_tokens == eng.finished[rid].out_tokens, p
