@@ -2,6 +2,7 @@
 equivalence — CPU tier, small config."""
 import math
 
+import pytest
 import torch
 
 from modal_examples_amd.models.whisper.model import WhisperConfig, WhisperModel
@@ -107,3 +108,30 @@ def test_forward_train_grads_flow_and_match_decode_shapes():
         toks2[:, -1] = (toks2[:, -1] + 1) % cfg.vocab_size
         l2 = m.forward_train(mel, toks2)
         assert torch.allclose(logits[:, :-1], l2[:, :-1], atol=1e-4)
+
+
+@pytest.mark.gpu
+def test_graph_decode_matches_eager_gpu():
+    """hipGraph-captured decode == eager decode_step, including the pow-2
+    batch padding path (batch 3 -> padded 4)."""
+    import torch
+
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from modal_examples_amd.models.whisper.model import WhisperConfig
+    from modal_examples_amd.models.whisper.pipeline import WhisperPipeline
+
+    cfg = WhisperConfig.small_test()
+    g = WhisperPipeline(cfg, device="cuda", dtype=torch.bfloat16, seed=11)
+    e = WhisperPipeline(cfg, device="cuda", dtype=torch.bfloat16, seed=11)
+    e.use_graph = False
+    clips = [torch.randn(16000) * 0.3, torch.randn(8000) * 0.2,
+             torch.randn(12000) * 0.25]
+    out_g = g.transcribe(clips, max_tokens=8)
+    out_e = e.transcribe(clips, max_tokens=8)
+    assert len(out_g) == 3
+    same = sum(a == b for a, b in zip(out_g, out_e))
+    assert same >= 2, (out_g, out_e)  # bf16 nondeterminism may flip one argmax
+    # second call reuses the captured graph at a different batch size
+    out2 = g.transcribe(clips[:2], max_tokens=6)
+    assert len(out2) == 2 and g._graphs.evictions == 0
