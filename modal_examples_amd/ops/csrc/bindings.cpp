@@ -109,7 +109,9 @@ torch::Tensor paged_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
   long long s_bound = block_table.has_value()
       ? (long long)max_blocks * block_size : (long long)kc.size(2);
   int splits = 1;
-  while (B * Hkv * splits < 512 && (long long)splits * 256 < s_bound &&
+  // fill the chip for latency-path decodes: allow chunks down to 128 kv
+  // rows (B1/Hkv8/S4096 was 128 workgroups = half the CUs idle)
+  while (B * Hkv * splits < 512 && (long long)splits * 128 < s_bound &&
          splits < 64)
     splits *= 2;
   torch::Tensor ws;
