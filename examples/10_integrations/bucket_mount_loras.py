@@ -73,10 +73,10 @@ def main():
     # publish two adapters to the bucket (the community-hub producer)
     c = S3Client(start_s3_server())
     for name, seed in (("watercolor", 1), ("neon", 2)):
-        torch.manual_seed(seed)
-        net = build_net(torch)
+        net = build_net(torch)  # (re-seeds to 0 internally)
         # nn.Sequential children are named "0"/"2" — target them explicitly
         apply_lora(net, rank=4, targets=("0", "2"))
+        torch.manual_seed(seed)  # per-adapter weights AFTER the base build
         for p in net.parameters():  # give the adapter non-zero weights
             if p.requires_grad and p.dim() == 2:
                 torch.nn.init.normal_(p, std=0.2)
