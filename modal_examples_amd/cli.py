@@ -100,12 +100,54 @@ def cmd_serve(target: str, extra_args):
     print(f"serving {app.name} at {url}")
     for f in list(app.web_endpoints.values()):
         print(f"  → {url}/{getattr(f.raw, '_modal_flags', {}).get('label') or f.name}")
+
+    # hot reload (the reference `modal serve` watches the file and redeploys
+    # on change): poll the target's mtime; on edit, tear down the ingress +
+    # pools, re-import the module, serve the fresh app on the same port.
+    import os as _os
+    import sys as _sys
+
+    path = target.split("::")[0]
+    mtime = _os.path.getmtime(path) if _os.path.exists(path) else 0
+    deadline = time.monotonic() + timeout if timeout else None
     try:
-        if timeout:
-            time.sleep(timeout)
-        else:
-            while True:
-                time.sleep(3600)
+        while deadline is None or time.monotonic() < deadline:
+            time.sleep(0.5)
+            try:
+                m = _os.path.getmtime(path)
+            except OSError:
+                continue
+            if m != mtime:
+                mtime = m
+                print(f"⟳ {path} changed — reloading")
+                stop_serving()
+                # wait for the old uvicorn to release the port (rebind race)
+                import socket as _socket
+
+                for _ in range(50):
+                    s_ = _socket.socket()
+                    try:
+                        s_.bind(("127.0.0.1", ns.port))
+                        s_.close()
+                        break
+                    except OSError:
+                        s_.close()
+                        time.sleep(0.1)
+                from .app import _Runtime
+
+                _Runtime.reset()
+                for name in [n for n, mod_ in list(_sys.modules.items())
+                             if getattr(mod_, "__file__", None)
+                             and _os.path.abspath(str(getattr(mod_, "__file__")))
+                             == _os.path.abspath(path)]:
+                    del _sys.modules[name]
+                try:
+                    mod = load_module(target)
+                    app = find_app(mod)
+                    url = serve(app, port=ns.port, block=False)
+                    print(f"re-serving {app.name} at {url}")
+                except Exception as e:  # noqa: BLE001 — keep watching
+                    print(f"reload failed (fix the file and save again): {e}")
     except KeyboardInterrupt:
         pass
     finally:

@@ -97,3 +97,64 @@ def test_serve_binds_and_answers(cli_env):
     finally:
         if proc.poll() is None:
             proc.terminate()
+
+
+def test_serve_hot_reload(tmp_path):
+    """`modal serve` watches the target file: editing it swaps the served
+    app in place (the reference serve behavior)."""
+    import subprocess
+    import sys
+    import time
+    import urllib.request
+    from pathlib import Path
+
+    src_v1 = '''
+import modal_examples_amd as modal
+app = modal.App("reload-demo")
+
+@app.function()
+@modal.fastapi_endpoint(method="GET", label="greet")
+def greet():
+    return {"version": 1}
+'''
+    f = tmp_path / "serve_mod.py"
+    f.write_text(src_v1)
+    repo = str(Path(__file__).resolve().parent.parent)
+    import os
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
+    port = "18931"
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "modal_examples_amd", "serve", str(f),
+         "--port", port, "--timeout", "40"],
+        cwd=repo, env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        text=True)
+    try:
+        def fetch(timeout=20):
+            deadline = time.time() + timeout
+            while time.time() < deadline:
+                try:
+                    with urllib.request.urlopen(
+                            f"http://127.0.0.1:{port}/greet", timeout=2) as r:
+                        import json
+
+                        return json.loads(r.read())
+                except Exception:
+                    time.sleep(0.3)
+            return None
+
+        assert fetch() == {"version": 1}
+        time.sleep(0.5)
+        f.write_text(src_v1.replace('"version": 1', '"version": 2'))
+        deadline = time.time() + 25
+        got = None
+        while time.time() < deadline:
+            got = fetch(5)
+            if got == {"version": 2}:
+                break
+            time.sleep(0.5)
+        assert got == {"version": 2}, got
+    finally:
+        proc.terminate()
+        proc.wait(timeout=10)
