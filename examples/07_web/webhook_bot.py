@@ -38,8 +38,11 @@ def handle_command(channel: str, command: str, arg: str) -> str:
 @app.function()
 @modal.fastapi_endpoint(method="POST", label="webhook")
 def webhook(channel: str = "", command: str = "", arg: str = ""):
-    """Platforms demand an ACK within seconds: spawn and return."""
-    call = handle_command.spawn(channel, command, arg)
+    """Platforms demand an ACK within seconds: spawn and return.  The
+    endpoint runs inside a worker container, so it hands off BY NAME (the
+    store-backed dispatcher pattern, 09_job_queues/pipeline_orchestration)."""
+    bot = modal.Function.from_name("example-webhook-bot", "handle_command")
+    call = bot.spawn(channel, command, arg)
     return {"ack": True, "job": call.object_id}
 
 
