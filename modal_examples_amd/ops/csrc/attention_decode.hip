@@ -24,16 +24,31 @@
 #include "common.h"
 
 #include <cstdio>
+#include <type_traits>
 
 #define DEC_THREADS 256
 #define DEC_WAVES 4
 #define MAX_G 8
 
+typedef __attribute__((ext_vector_type(2))) unsigned int u32x2d;
+
+// OCP e4m3 -> f32 via the gfx950 HW converter (4 fp8 per dword, constant
+// lane select required by the builtin).
+DEV_INLINE void fp8x4_to_f32(unsigned w, float* out4) {
+  out4[0] = __builtin_amdgcn_cvt_f32_fp8(w, 0);
+  out4[1] = __builtin_amdgcn_cvt_f32_fp8(w, 1);
+  out4[2] = __builtin_amdgcn_cvt_f32_fp8(w, 2);
+  out4[3] = __builtin_amdgcn_cvt_f32_fp8(w, 3);
+}
+
 // GT = compile-time GQA group size: register arrays (qf/accv) and the LDS
 // merge buffers are sized by GT, not MAX_G — at Llama's G=4 this halves the
 // VGPR/LDS footprint (205 VGPR/66 KB -> more waves + blocks per CU; decode
 // is HBM-latency-bound so occupancy IS the bandwidth lever, probe_vmcnt_r2).
-template <int D, int GT>
+// FP8: KV cache stored as OCP e4m3 (the vllm_low_latency FP8 role) — halves
+// KV bytes per token; each lane loads 8 fp8 (8 B) instead of 8 bf16 (16 B)
+// and converts through the HW fp8 pipe.
+template <int D, int GT, bool FP8>
 __global__ __launch_bounds__(DEC_THREADS) void paged_decode_kernel(
     const short* __restrict__ Q,       // [B, Hq, D]
     const short* __restrict__ Kc,      // cache, layout above
@@ -97,31 +112,53 @@ __global__ __launch_bounds__(DEC_THREADS) void paged_decode_kernel(
     return (((long long)b * Hkv + hkv) * (long long)block_size + kv) * D;
   };
   // depth-2 pipeline with NAMED registers (a runtime-indexed ring spills to
-  // scratch — measured 3.5x slower; guide rule #20)
-  bf16x8 kA = bf16x8{0, 0, 0, 0, 0, 0, 0, 0}, vA = kA, kB = kA, vB = kA;
+  // scratch — measured 3.5x slower; guide rule #20).  kv_t is 16 B of bf16
+  // or 8 B of fp8 — 8 elements per lane either way.
+  typedef typename std::conditional<FP8, u32x2d, bf16x8>::type kv_t;
+  const unsigned char* Kc8 = (const unsigned char*)Kc;
+  const unsigned char* Vc8 = (const unsigned char*)Vc;
+  auto load_kv = [&](const short* base16, const unsigned char* base8,
+                     long long elem_off) -> kv_t {
+    if (FP8) {
+      return *(const kv_t*)&base8[elem_off];
+    }
+    return *(const kv_t*)&base16[elem_off];
+  };
+  auto to_f32 = [&](kv_t r, float* out8) {
+    if (FP8) {
+      u32x2d w = *(u32x2d*)&r;
+      fp8x4_to_f32(w[0], out8);
+      fp8x4_to_f32(w[1], out8 + 4);
+    } else {
+      bf16x8 b = *(bf16x8*)&r;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) out8[j] = bf2f(b[j]);
+    }
+  };
+  kv_t kA = {}, vA = {}, kB = {}, vB = {};
   int kv0 = kv_lo + part;
   if (kv0 < S) {
-    long long o0 = row_addr(kv0);
-    kA = *(const bf16x8*)&Kc[o0 + slot * 8];
-    vA = *(const bf16x8*)&Vc[o0 + slot * 8];
+    long long o0 = row_addr(kv0) + slot * 8;
+    kA = load_kv(Kc, Kc8, o0);
+    vA = load_kv(Vc, Vc8, o0);
   }
   if (kv0 + NPART < S) {
-    long long o1 = row_addr(kv0 + NPART);
-    kB = *(const bf16x8*)&Kc[o1 + slot * 8];
-    vB = *(const bf16x8*)&Vc[o1 + slot * 8];
+    long long o1 = row_addr(kv0 + NPART) + slot * 8;
+    kB = load_kv(Kc, Kc8, o1);
+    vB = load_kv(Vc, Vc8, o1);
   }
   for (int kv = kv0; kv < S; kv += NPART) {
-    bf16x8 k8 = kA, v8 = vA;
+    kv_t k8 = kA, v8 = vA;
     kA = kB;
     vA = vB;
     if (kv + 2 * NPART < S) {
-      long long offn = row_addr(kv + 2 * NPART);
-      kB = *(const bf16x8*)&Kc[offn + slot * 8];
-      vB = *(const bf16x8*)&Vc[offn + slot * 8];
+      long long offn = row_addr(kv + 2 * NPART) + slot * 8;
+      kB = load_kv(Kc, Kc8, offn);
+      vB = load_kv(Vc, Vc8, offn);
     }
-    float kfl[8];
-#pragma unroll
-    for (int j = 0; j < 8; ++j) kfl[j] = bf2f(k8[j]);
+    float kfl[8], vfl[8];
+    to_f32(k8, kfl);
+    to_f32(v8, vfl);
 
 #pragma unroll
     for (int g = 0; g < GT; ++g) {
@@ -139,7 +176,7 @@ __global__ __launch_bounds__(DEC_THREADS) void paged_decode_kernel(
       l_run[g] = l_run[g] * rs + p;
       m_run[g] = m_new;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) accv[g][j] = accv[g][j] * rs + p * bf2f(v8[j]);
+      for (int j = 0; j < 8; ++j) accv[g][j] = accv[g][j] * rs + p * vfl[j];
     }
   }
 
@@ -216,7 +253,7 @@ extern "C" void paged_decode_bf16(const void* q, const void* kc, const void* vc,
                                   const int* block_table, const int* seq_lens,
                                   void* o, float* ws, int B, int Hq, int Hkv,
                                   int D, int block_size, int max_blocks,
-                                  int splits, float scale,
+                                  int splits, float scale, int kv_fp8,
                                   hipStream_t stream) {
   if (Hq / Hkv > MAX_G) {
     fprintf(stderr, "paged_decode_bf16: GQA group %d > %d\n", Hq / Hkv, MAX_G);
@@ -227,17 +264,22 @@ extern "C" void paged_decode_bf16(const void* q, const void* kc, const void* vc,
   dim3 block(DEC_THREADS);
   float* ws_arg = splits > 1 ? ws : nullptr;
   const int G = Hq / Hkv;
-#define DLAUNCH(DD, GG)                                                      \
-  hipLaunchKernelGGL((paged_decode_kernel<DD, GG>), grid, block, 0, stream,  \
-                     (const short*)q, (const short*)kc, (const short*)vc,    \
-                     block_table, seq_lens, (short*)o, ws_arg, B, Hq, Hkv,   \
-                     block_size, max_blocks, splits, scale)
+#define DLAUNCH(DD, GG, F8)                                                  \
+  hipLaunchKernelGGL((paged_decode_kernel<DD, GG, F8>), grid, block, 0,      \
+                     stream, (const short*)q, (const short*)kc,              \
+                     (const short*)vc, block_table, seq_lens, (short*)o,     \
+                     ws_arg, B, Hq, Hkv, block_size, max_blocks, splits,     \
+                     scale)
+#define DGROUP(DD, F8)                                                      \
+  do {                                                                       \
+    if (G == 1) DLAUNCH(DD, 1, F8);                                          \
+    else if (G == 2) DLAUNCH(DD, 2, F8);                                     \
+    else if (G <= 4) DLAUNCH(DD, 4, F8);                                     \
+    else DLAUNCH(DD, 8, F8);                                                 \
+  } while (0)
 #define DDISPATCH(DD)                                                        \
   do {                                                                       \
-    if (G == 1) DLAUNCH(DD, 1);                                              \
-    else if (G == 2) DLAUNCH(DD, 2);                                         \
-    else if (G <= 4) DLAUNCH(DD, 4);                                         \
-    else DLAUNCH(DD, 8);                                                     \
+    if (kv_fp8) DGROUP(DD, true); else DGROUP(DD, false);                    \
   } while (0)
   if (D == 64) {
     DDISPATCH(64);
@@ -248,6 +290,7 @@ extern "C" void paged_decode_bf16(const void* q, const void* kc, const void* vc,
     abort();
   }
 #undef DDISPATCH
+#undef DGROUP
 #undef DLAUNCH
   if (splits > 1) {
     hipLaunchKernelGGL(decode_merge_kernel, dim3(B * Hq), dim3(256), 0,

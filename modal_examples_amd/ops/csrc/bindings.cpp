@@ -21,7 +21,7 @@ void fa_fwd_strided_bf16(const void*, const void*, const void*, void*, int,
                          const long long*, hipStream_t);
 void paged_decode_bf16(const void*, const void*, const void*, const int*,
                        const int*, void*, float*, int, int, int, int, int,
-                       int, int, float, hipStream_t);
+                       int, int, float, int, hipStream_t);
 void groupnorm_silu_bf16(const void*, void*, float*, const float*,
                          const float*, int, int, long long, int, float, int,
                          hipStream_t);
@@ -90,8 +90,13 @@ torch::Tensor paged_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
                            torch::Tensor seq_lens, int64_t block_size,
                            double scale) {
   check_bf16(q, "q");
-  check_bf16(kc, "k_cache");
-  check_bf16(vc, "v_cache");
+  // KV cache: bf16, or OCP e4m3 fp8 (the vllm_low_latency FP8-serving role
+  // — halves KV bytes per decoded token)
+  const bool kv_fp8 = kc.scalar_type() == torch::kFloat8_e4m3fn;
+  TORCH_CHECK(kv_fp8 || kc.scalar_type() == torch::kBFloat16,
+              "k_cache must be bf16 or float8_e4m3fn");
+  TORCH_CHECK(vc.scalar_type() == kc.scalar_type(), "kv cache dtype mismatch");
+  TORCH_CHECK(kc.is_contiguous() && vc.is_contiguous());
   int B = q.size(0), Hq = q.size(1), D = q.size(2);
   int Hkv, max_blocks = 0;
   const int* bt_ptr = nullptr;
@@ -124,7 +129,7 @@ torch::Tensor paged_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
   paged_decode_bf16(q.data_ptr(), kc.data_ptr(), vc.data_ptr(), bt_ptr,
                     seq_lens.data_ptr<int>(), o.data_ptr(), ws_ptr, B, Hq,
                     Hkv, D, (int)block_size, max_blocks, splits, (float)scale,
-                    cur_stream());
+                    kv_fp8 ? 1 : 0, cur_stream());
   return o;
 }
 

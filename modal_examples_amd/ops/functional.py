@@ -76,10 +76,14 @@ def attention_qkv(q, k, v, causal: bool = False, scale: Optional[float] = None):
 
 def paged_decode(q, k_cache, v_cache, block_table, seq_lens, block_size: int = 0,
                  scale: Optional[float] = None):
-    """Decode attention over a (paged) KV cache (K6). q [B,Hq,D]."""
+    """Decode attention over a (paged) KV cache (K6). q [B,Hq,D].
+    Caches may be bf16 or float8_e4m3fn (fp8 halves KV bytes/token)."""
     scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
     ext = _ext_for(q)
     if ext is None:
+        if k_cache.dtype not in (torch.float32, torch.bfloat16, torch.float16):
+            k_cache = k_cache.to(torch.float32)
+            v_cache = v_cache.to(torch.float32)
         return ref.paged_decode_ref(q, k_cache, v_cache, block_table, seq_lens,
                                     block_size, scale)
     return ext.paged_decode(q.contiguous(), k_cache, v_cache, block_table,

@@ -24,6 +24,7 @@ def main():
     ap.add_argument("--layers", type=int, default=0, help="0 = full 32")
     ap.add_argument("--no-graph", action="store_true")
     ap.add_argument("--small", action="store_true")
+    ap.add_argument("--kv-dtype", choices=("bf16", "fp8"), default="bf16")
     ap.add_argument("--ragged", action="store_true",
                     help="mixed prompt lengths (uniform prompt_len/4 .. "
                          "prompt_len) — exercises the one-padded-forward "
@@ -43,7 +44,7 @@ def main():
     t0 = time.perf_counter()
     eng = LlamaEngine(cfg, device=device, dtype=dtype,
                       use_graph=not args.no_graph and device == "cuda",
-                      max_batch=args.requests)
+                      max_batch=args.requests, kv_dtype=args.kv_dtype)
     eng.warmup()  # decode-graph capture is cold-start work
     if device == "cuda":
         torch.cuda.synchronize()
@@ -79,6 +80,7 @@ def main():
         "input_tok_per_s": round(in_toks / dt, 1),
         "output_tok_per_s": round(out_toks / dt, 1),
         "kv_blocks": eng.num_blocks,
+        "kv_dtype": args.kv_dtype,
         "hipgraph": eng.use_graph,
     }), flush=True)
 

@@ -355,3 +355,20 @@ def test_conv3x3_gn_fused(upsample):
         h = torch.nn.functional.interpolate(h, scale_factor=2.0, mode="nearest")
     want = torch.nn.functional.conv2d(h, w.float(), b, padding=1)
     _close(got, want, atol=5e-2, rtol=5e-2)
+
+
+@requires_gpu
+@pytest.mark.parametrize("D", [64, 128])
+def test_paged_decode_fp8_kv(D):
+    """fp8 (e4m3) KV-cache decode vs the fp32 reference over the same
+    quantized values — the kernel's HW fp8 conversion must match torch's."""
+    B, Hq, Hkv, S = 4, 8, 2, 256
+    torch.manual_seed(0)
+    q = torch.randn(B, Hq, D, device="cuda", dtype=torch.bfloat16)
+    k8 = (torch.randn(B, Hkv, S, D, device="cuda") * 0.5).to(torch.float8_e4m3fn)
+    v8 = (torch.randn(B, Hkv, S, D, device="cuda") * 0.5).to(torch.float8_e4m3fn)
+    lens = torch.full((B,), S, dtype=torch.int32, device="cuda")
+    got = F.paged_decode(q, k8.contiguous(), v8.contiguous(), None, lens)
+    want = ref.paged_decode_ref(q.float(), k8.float(), v8.float(), None,
+                                lens, S, 1.0 / math.sqrt(D))
+    _close(got, want, atol=3e-2, rtol=3e-2)
