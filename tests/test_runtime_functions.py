@@ -385,3 +385,20 @@ def test_function_call_cancel_after_done_keeps_result():
     assert call.get(timeout=30) == 7
     call.cancel()  # no-op after completion
     assert call.get(timeout=5) == 7
+
+
+def test_spawn_aio_and_gather_aio():
+    """`.spawn.aio` + async gather (reference async patterns, 08_advanced)."""
+    import asyncio
+
+    appa = modal.App("test-spawn-aio")
+
+    @appa.function()
+    def plus1(x: int) -> int:
+        return x + 1
+
+    async def go():
+        fcs = await asyncio.gather(*(plus1.spawn.aio(i) for i in range(4)))
+        return await asyncio.gather(*(fc.get_aio(30) for fc in fcs))
+
+    assert asyncio.run(go()) == [1, 2, 3, 4]
