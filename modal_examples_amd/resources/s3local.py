@@ -39,14 +39,25 @@ def make_s3_app():
         key = urllib.parse.unquote(parts[1]) if len(parts) > 1 else ""
         root = bucket_root() / bucket
 
+        def path_ok() -> bool:
+            # reject traversal: the resolved object path must stay inside
+            # the bucket root (and bucket names must be plain)
+            if "/" in bucket or ".." in bucket or bucket.startswith("."):
+                return False
+            if not key:
+                return True
+            full = (root / key).resolve()
+            return str(full).startswith(str(root.resolve()) + "/") or \
+                full == root.resolve()
+
         async def respond(status, body=b"", ctype="application/xml"):
             await send({"type": "http.response.start", "status": status,
                         "headers": [(b"content-type", ctype.encode()),
                                     (b"content-length", str(len(body)).encode())]})
             await send({"type": "http.response.body", "body": body})
 
-        if not bucket:
-            await respond(400, b"<Error><Code>NoBucket</Code></Error>")
+        if not bucket or not path_ok():
+            await respond(400, b"<Error><Code>InvalidKey</Code></Error>")
             return
 
         if method == "GET" and (not key or "list-type" in qs):

@@ -475,3 +475,23 @@ def test_sandbox_wait_until_ready_and_detach(tmp_path):
         assert sb2.poll() is None
     finally:
         sb.terminate()
+
+
+def test_s3_endpoint_rejects_path_traversal():
+    import urllib.error
+    import urllib.request
+
+    from modal_examples_amd.resources.s3local import S3Client, start_s3_server
+
+    ep = start_s3_server()
+    c = S3Client(ep)
+    c.put("trav-bkt", "safe.txt", b"ok")
+    for bad in ("trav-bkt/../../escape.txt", "..%2F..%2Fescape"):
+        req = urllib.request.Request(f"{ep}/{bad}", data=b"x", method="PUT")
+        try:
+            urllib.request.urlopen(req, timeout=5)
+            raised = False
+        except urllib.error.HTTPError as e:
+            raised = e.code == 400
+        assert raised, bad
+    assert c.get("trav-bkt", "safe.txt") == b"ok"
