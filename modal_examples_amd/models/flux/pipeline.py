@@ -52,6 +52,10 @@ class FluxPipeline:
         self.model.eval()
         self.vae.eval()
         self._graphs = GraphLRU(graph_cache)
+        import threading
+
+        self._graph_lock = threading.Lock()  # shared graph buffers: one
+        # generate at a time per pipeline (concurrent callers serialize)
         self.image_size = latent_size * 8
 
     def encode(self, prompts: List[str]):
@@ -127,14 +131,15 @@ class FluxPipeline:
         ctx, pooled = self.encode(prompts)
         sigmas = flow_sigmas(steps)
         if self.use_graph:
-            st = self._get_graph(batch, steps)
-            st["x"].copy_(x)
-            st["ctx"].copy_(ctx)
-            st["pooled"].copy_(pooled)
-            st["step"].zero_()
-            for _ in range(steps):
-                st["graph"].replay()
-            x = st["x"].clone()
+            with self._graph_lock:
+                st = self._get_graph(batch, steps)
+                st["x"].copy_(x)
+                st["ctx"].copy_(ctx)
+                st["pooled"].copy_(pooled)
+                st["step"].zero_()
+                for _ in range(steps):
+                    st["graph"].replay()
+                x = st["x"].clone()
         else:
             x = self._denoise_eager(x, ctx, pooled, sigmas)
         if not decode:

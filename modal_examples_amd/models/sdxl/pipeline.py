@@ -77,6 +77,9 @@ class SDXLPipeline:
         self.unet.eval()
         self.vae.eval()
         self._graphs = GraphLRU(4)  # (batch, steps, cfg_on) -> graph state, LRU-bounded
+        import threading
+
+        self._graph_lock = threading.Lock()  # graph buffers are shared state
         self.image_size = latent_size * 8
 
     # -------------------------------------------------- conditioning
@@ -175,14 +178,15 @@ class SDXLPipeline:
         ctx, add = self.encode(prompts)
 
         if self.use_graph and guidance <= 1.0:
-            st = self._get_graph(batch, steps, guidance)
-            st["x"].copy_(x)
-            st["ctx"].copy_(ctx)
-            st["add"].copy_(add)
-            st["step"].zero_()
-            for _ in range(steps):
-                st["graph"].replay()
-            x = st["x"].clone()
+            with self._graph_lock:
+                st = self._get_graph(batch, steps, guidance)
+                st["x"].copy_(x)
+                st["ctx"].copy_(ctx)
+                st["add"].copy_(add)
+                st["step"].zero_()
+                for _ in range(steps):
+                    st["graph"].replay()
+                x = st["x"].clone()
         else:
             x = self._denoise_eager(x, ctx, add, sigmas, timesteps, guidance)
 

@@ -61,6 +61,10 @@ class WhisperPipeline:
         self.model.eval()
         self.sot, self.eot = 1, 2  # synthetic special tokens
         self.use_graph = self.device.type == "cuda"
+        import threading
+
+        self._graph_lock = threading.Lock()  # graph state is per-pipeline;
+        # serialize transcribe under @modal.concurrent callers
         from ...gpu.graphs import GraphLRU
 
         self._graphs = GraphLRU(3)  # per-batch-size captured decode steps
@@ -113,6 +117,12 @@ class WhisperPipeline:
     def transcribe(self, audio_batch: List[torch.Tensor], max_tokens: int = 32
                    ) -> List[List[int]]:
         """Batch of mono 16 kHz waveforms → token id sequences (greedy)."""
+        if self.use_graph:
+            with self._graph_lock:
+                return self._transcribe_inner(audio_batch, max_tokens)
+        return self._transcribe_inner(audio_batch, max_tokens)
+
+    def _transcribe_inner(self, audio_batch, max_tokens):
         B_req = len(audio_batch)
         cfg = self.cfg
         if self.use_graph:
