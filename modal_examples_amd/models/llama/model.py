@@ -71,9 +71,8 @@ class LlamaBlock(nn.Module):
                 v.view(B, S, self.nkv, self.hd))
 
     def ffn(self, x):
-        gu = self.gate_up(x)
-        g, u = gu[..., : self.ffn_dim], gu[..., self.ffn_dim:]
-        return self.down(OF.silu_mul(g.contiguous(), u.contiguous()))
+        # fused kernel reads both halves of the gate_up output in place
+        return self.down(OF.glu_fused(self.gate_up(x), gelu=False))
 
 
 class LlamaModel(nn.Module):

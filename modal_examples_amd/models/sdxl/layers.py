@@ -186,9 +186,8 @@ class GEGLUFeedForward(nn.Module):
         self.inner = inner
 
     def forward(self, x):
-        ab = self.proj_in(x)
-        a, b = ab[..., : self.inner], ab[..., self.inner:]
-        return self.proj_out(OF.geglu(a.contiguous(), b.contiguous()))
+        # fused kernel reads both GEGLU halves of the projection in place
+        return self.proj_out(OF.glu_fused(self.proj_in(x), gelu=True))
 
 
 class TransformerBlock(nn.Module):

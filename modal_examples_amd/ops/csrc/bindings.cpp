@@ -33,6 +33,8 @@ void cfg_euler_bf16(const void*, const void*, const void*, void*, float, float,
                     long long, hipStream_t);
 void silu_mul_bf16(const void*, const void*, void*, long long, hipStream_t);
 void geglu_bf16(const void*, const void*, void*, long long, hipStream_t);
+void glu_fused_bf16(const void*, void*, long long, long long, int,
+                    hipStream_t);
 void add_bf16(const void*, const void*, void*, long long, hipStream_t);
 void rope_bf16(void*, const float*, const float*, long long, int, int, int,
                long long, long long, long long, const int*, hipStream_t);
@@ -224,6 +226,22 @@ torch::Tensor geglu(torch::Tensor a, torch::Tensor b) {
   check_bf16(a, "a");
   auto y = torch::empty_like(a);
   geglu_bf16(a.data_ptr(), b.data_ptr(), y.data_ptr(), a.numel(), cur_stream());
+  return y;
+}
+
+// act(src[..., :I]) * src[..., I:] without slicing copies — src is the
+// fused gate_up / GEGLU projection output [rows, 2I]
+torch::Tensor glu_fused(torch::Tensor src, bool gelu) {
+  check_bf16(src, "src");
+  long long inner = src.size(-1) / 2;
+  TORCH_CHECK(src.size(-1) % 2 == 0 && inner % 8 == 0,
+              "inner dim must be even and 8-aligned");
+  long long rows = src.numel() / (2 * inner);
+  auto sizes = src.sizes().vec();
+  sizes.back() = inner;
+  auto y = torch::empty(sizes, src.options());
+  glu_fused_bf16(src.data_ptr(), y.data_ptr(), rows, inner, gelu ? 1 : 0,
+                 cur_stream());
   return y;
 }
 
@@ -440,6 +458,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cfg_euler", &cfg_euler, "fused CFG + Euler step (K4)");
   m.def("silu_mul", &silu_mul);
   m.def("geglu", &geglu);
+  m.def("glu_fused", &glu_fused, "in-place-sliced silu/gelu-mul over a fused [.,2I] projection");
   m.def("add_residual", &add_residual);
   m.def("rope_", &rope_, "in-place RoPE with host cos/sin tables");
   m.def("adamw_", &adamw_, "fused AdamW step (K9)");

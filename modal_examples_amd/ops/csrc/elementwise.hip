@@ -301,3 +301,66 @@ extern "C" void cfg_euler_dev_bf16(const void* xt, const void* eps_c,
 extern "C" void advance_step(long long* step, hipStream_t stream) {
   hipLaunchKernelGGL(advance_step_kernel, dim3(1), dim3(64), 0, stream, step);
 }
+
+// ---------------------------------------------------------------- fused-
+// projection activations: the model computes ONE GEMM producing [rows, 2*I]
+// (gate_up / GEGLU's ab); these kernels read the two halves IN PLACE, so the
+// two .contiguous() slice copies per FFN disappear.
+//   y[row, c] = act(src[row*2I + c]) * src[row*2I + I + c]
+
+template <bool GELU>
+__global__ __launch_bounds__(EW_BLOCK) void glu_fused_kernel(
+    const short* __restrict__ Src, short* __restrict__ Y, long long rows,
+    long long inner) {
+  const float k0 = 0.7978845608028654f, k1 = 0.044715f;
+  long long n = rows * inner;
+  long long stride = (long long)gridDim.x * EW_BLOCK * 8;
+  for (long long i = ((long long)blockIdx.x * EW_BLOCK + threadIdx.x) * 8;
+       i + 8 <= n; i += stride) {
+    long long row = i / inner, c = i - row * inner;  // 8-aligned within a row
+    const short* base = Src + row * 2 * inner + c;
+    bf16x8 a = *(const bf16x8*)base;
+    bf16x8 b = *(const bf16x8*)(base + inner);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bf2f(a[j]);
+      float act;
+      if (GELU) {
+        float t = tanhf(k0 * (f + k1 * f * f * f));
+        act = 0.5f * f * (1.f + t);
+      } else {
+        act = f / (1.f + __expf(-f));
+      }
+      o[j] = f2bf(act * bf2f(b[j]));
+    }
+    *(bf16x8*)&Y[i] = o;
+  }
+  long long full = (n / 8) * 8;
+  for (long long i = full + (long long)blockIdx.x * EW_BLOCK + threadIdx.x;
+       i < n; i += (long long)gridDim.x * EW_BLOCK) {
+    long long row = i / inner, c = i - row * inner;
+    float f = bf2f(Src[row * 2 * inner + c]);
+    float bb = bf2f(Src[row * 2 * inner + inner + c]);
+    float act;
+    if (GELU) {
+      float t = tanhf(k0 * (f + k1 * f * f * f));
+      act = 0.5f * f * (1.f + t);
+    } else {
+      act = f / (1.f + __expf(-f));
+    }
+    Y[i] = f2bf(act * bb);
+  }
+}
+
+extern "C" void glu_fused_bf16(const void* src, void* y, long long rows,
+                               long long inner, int gelu,
+                               hipStream_t stream) {
+  int grid = elementwise_grid(rows * inner, EW_BLOCK);
+  if (gelu)
+    hipLaunchKernelGGL((glu_fused_kernel<true>), dim3(grid), dim3(EW_BLOCK),
+                       0, stream, (const short*)src, (short*)y, rows, inner);
+  else
+    hipLaunchKernelGGL((glu_fused_kernel<false>), dim3(grid), dim3(EW_BLOCK),
+                       0, stream, (const short*)src, (short*)y, rows, inner);
+}

@@ -138,6 +138,18 @@ def geglu(a, b):
     return ext.geglu(a.contiguous(), b.contiguous())
 
 
+def glu_fused(src, gelu: bool = False):
+    """act(src[..., :I]) * src[..., I:] reading the fused projection output
+    in place (no slice copies).  gelu=True -> GEGLU (SDXL FF), else SwiGLU
+    (Llama gate_up)."""
+    inner = src.shape[-1] // 2
+    ext = _ext_for(src)
+    if ext is None or inner % 8 != 0:
+        a, b = src[..., :inner], src[..., inner:]
+        return (ref.geglu_ref(a, b) if gelu else ref.silu_mul_ref(a, b))
+    return ext.glu_fused(src.contiguous(), gelu)
+
+
 def add_residual(a, b):
     ext = _ext_for(a, b)
     if ext is None:

@@ -372,3 +372,13 @@ def test_paged_decode_fp8_kv(D):
     want = ref.paged_decode_ref(q.float(), k8.float(), v8.float(), None,
                                 lens, S, 1.0 / math.sqrt(D))
     _close(got, want, atol=3e-2, rtol=3e-2)
+
+
+@requires_gpu
+@pytest.mark.parametrize("gelu", [False, True])
+def test_glu_fused_matches_sliced(gelu):
+    src = torch.randn(4, 37, 256, device="cuda", dtype=torch.bfloat16)
+    got = F.glu_fused(src, gelu=gelu)
+    a, b = src[..., :128].contiguous(), src[..., 128:].contiguous()
+    want = F.geglu(a, b) if gelu else F.silu_mul(a, b)
+    _close(got, want, atol=1e-2, rtol=1e-2)
