@@ -1,0 +1,113 @@
+"""Fast weight loading: safetensors layout, mmap + pinned staged H2D.
+
+The fresh-process half of the snapshot story (reference:
+06_gpu_and_ml/llm-serving/sglang_snapshot.py:176-218 restores engine weights
+from host residence; gpu_snapshot.py:41-53).  `WeightSnapshot` (snapshot.py)
+covers the in-process warm path; this module covers the cold path: weights
+saved once in safetensors layout, restored by a fresh process WITHOUT pickle
+deserialization — the file is mmapped (page-cache reads, zero parse), bytes
+stream through two pinned staging buffers whose host-memcpy and H2D copies
+overlap on a side stream, and every tensor is carved as a VIEW of one device
+blob (one allocation, no per-tensor cudaMalloc).
+
+torch.load on the same weights measures ~3 GB/s (pickle+zip walk); this path
+is bound by min(page-cache memcpy, pinned H2D) instead.
+"""
+from __future__ import annotations
+
+import json
+import os
+from typing import Dict, Tuple
+
+import torch
+
+_DTYPES = {
+    "BF16": torch.bfloat16, "F16": torch.float16, "F32": torch.float32,
+    "F64": torch.float64, "I64": torch.int64, "I32": torch.int32,
+    "I16": torch.int16, "I8": torch.int8, "U8": torch.uint8,
+    "BOOL": torch.bool, "F8_E4M3": torch.float8_e4m3fn,
+}
+_NAMES = {v: k for k, v in _DTYPES.items()}
+
+
+def save_file(state: Dict[str, torch.Tensor], path: str) -> int:
+    """Write a state dict in safetensors layout. Returns payload bytes."""
+    header: Dict[str, dict] = {}
+    off = 0
+    tensors = []
+    for name, t in state.items():
+        t = t.detach().contiguous().cpu()
+        n = t.numel() * t.element_size()
+        header[name] = {"dtype": _NAMES[t.dtype], "shape": list(t.shape),
+                        "data_offsets": [off, off + n]}
+        tensors.append(t)
+        off += n
+    hdr = json.dumps(header).encode()
+    hdr += b" " * ((8 - len(hdr) % 8) % 8)  # keep payload 8-aligned
+    tmp = path + ".tmp"
+    with open(tmp, "wb") as f:
+        f.write(len(hdr).to_bytes(8, "little"))
+        f.write(hdr)
+        for t in tensors:
+            # uint8 reinterpret works for every dtype incl. bf16/fp8
+            f.write(t.view(torch.uint8).view(-1).numpy().tobytes())
+    os.replace(tmp, path)
+    return off
+
+
+def _parse(path: str) -> Tuple[dict, int, int]:
+    with open(path, "rb") as f:
+        n = int.from_bytes(f.read(8), "little")
+        header = json.loads(f.read(n))
+    header.pop("__metadata__", None)
+    return header, 8 + n, os.path.getsize(path)
+
+
+def _file_bytes(path: str, size: int) -> torch.Tensor:
+    """The whole file as a uint8 CPU tensor over a private mmap."""
+    st = torch.UntypedStorage.from_file(path, False, size)
+    t = torch.empty(0, dtype=torch.uint8)
+    t.set_(st)
+    return t
+
+
+def _carve(blob: torch.Tensor, header: dict) -> Dict[str, torch.Tensor]:
+    """blob = the payload bytes; offsets in the header are payload-relative."""
+    out = {}
+    for name, m in header.items():
+        s, e = m["data_offsets"]
+        out[name] = blob[s:e].view(_DTYPES[m["dtype"]]).view(m["shape"])
+    return out
+
+
+def load_file(path: str, device="cpu", staging_mb: int = 128) -> Dict[str, torch.Tensor]:
+    """Load a safetensors-layout file to `device`.
+
+    CPU: zero-copy views over the mmap.  GPU: one device blob filled by
+    double-buffered pinned staging (host memcpy of chunk i overlaps the H2D
+    of chunk i-1), tensors carved as views.
+    """
+    header, data_off, size = _parse(path)
+    cpu = _file_bytes(path, size)
+    dev = torch.device(device)
+    if dev.type != "cuda":
+        return _carve(cpu[data_off:], header)
+    payload = size - data_off
+    blob = torch.empty(payload, dtype=torch.uint8, device=dev)
+    chunk = staging_mb << 20
+    pinned = [torch.empty(min(chunk, payload), dtype=torch.uint8,
+                          pin_memory=True) for _ in range(2)]
+    events = [torch.cuda.Event(), torch.cuda.Event()]
+    stream = torch.cuda.Stream(dev)
+    src = cpu[data_off:]
+    for i, s in enumerate(range(0, payload, chunk)):
+        e = min(s + chunk, payload)
+        buf, ev = pinned[i % 2], events[i % 2]
+        if i >= 2:
+            ev.synchronize()  # buf's previous H2D must be done before reuse
+        buf[:e - s].copy_(src[s:e])  # page-cache -> pinned (host memcpy)
+        with torch.cuda.stream(stream):
+            blob[s:e].copy_(buf[:e - s], non_blocking=True)
+            ev.record(stream)
+    stream.synchronize()
+    return _carve(blob, header)

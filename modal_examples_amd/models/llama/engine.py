@@ -53,7 +53,7 @@ class LlamaEngine:
                  dtype=torch.bfloat16, max_batch: int = 64,
                  kv_blocks: Optional[int] = None, use_graph: bool = True,
                  eos_id: int = 2, seed: int = 0, top_p: float = 1.0,
-                 kv_dtype: str = "bf16"):
+                 kv_dtype: str = "bf16", init_weights: bool = True):
         self.cfg = cfg or LlamaConfig.llama3_8b()
         self.device = torch.device(device)
         self.dtype = dtype
@@ -68,8 +68,16 @@ class LlamaEngine:
         self.kv_dtype = (torch.float8_e4m3fn if kv_dtype == "fp8"
                          else torch.bfloat16)
         torch.manual_seed(seed)
-        with torch.device(self.device):
-            self.model = LlamaModel(self.cfg).to(self.device, dtype)
+        if init_weights:
+            with torch.device(self.device):
+                self.model = LlamaModel(self.cfg).to(self.device, dtype)
+        else:
+            # cold-restore path: build on meta (no init compute), materialize
+            # empty, then the caller load_state_dict(assign=True)s real
+            # weights (gpu/fastload.py blob views)
+            with torch.device("meta"):
+                self.model = LlamaModel(self.cfg)
+            self.model = self.model.to_empty(device=self.device).to(dtype)
         self.model.eval()
         c = self.cfg
         if kv_blocks is None:

@@ -84,7 +84,9 @@ class LlamaModel(nn.Module):
         self.norm = RMSNormK(cfg.dim, cfg.norm_eps)
         self.lm_head = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
         # NOT buffers: module .to(bf16) must not downcast the f32 trig tables
-        cos, sin = OF.rope_tables(cfg.max_seq, cfg.head_dim, cfg.rope_base)
+        # (forced onto cpu so meta-device construction still gets real tables)
+        with torch.device("cpu"):
+            cos, sin = OF.rope_tables(cfg.max_seq, cfg.head_dim, cfg.rope_base)
         self._rope_cpu = (cos, sin)
         self._rope_cache = {}
 

@@ -27,20 +27,22 @@ def prepare(model: str, shm: str) -> dict:
 
     os.makedirs(shm, exist_ok=True)
     meta = {"model": model}
+    from modal_examples_amd.gpu import fastload
+
     if model == "sdxl":
         from modal_examples_amd.models.sdxl.pipeline import SDXLPipeline
 
         pipe = SDXLPipeline(device="cpu", dtype=torch.bfloat16)
-        torch.save(pipe.unet.state_dict(), f"{shm}/unet.pt")
-        torch.save(pipe.vae.state_dict(), f"{shm}/vae.pt")
+        fastload.save_file(dict(pipe.unet.state_dict()), f"{shm}/unet.safetensors")
+        fastload.save_file(dict(pipe.vae.state_dict()), f"{shm}/vae.safetensors")
         meta["bytes"] = sum(os.path.getsize(f"{shm}/{f}")
-                            for f in ("unet.pt", "vae.pt"))
+                            for f in ("unet.safetensors", "vae.safetensors"))
     else:
         from modal_examples_amd.models.llama.model import LlamaConfig, LlamaModel
 
         m = LlamaModel(LlamaConfig.llama3_8b()).to(torch.bfloat16)
-        torch.save(m.state_dict(), f"{shm}/llama.pt")
-        meta["bytes"] = os.path.getsize(f"{shm}/llama.pt")
+        fastload.save_file(dict(m.state_dict()), f"{shm}/llama.safetensors")
+        meta["bytes"] = os.path.getsize(f"{shm}/llama.safetensors")
     # touch into page cache
     for f in os.listdir(shm):
         with open(os.path.join(shm, f), "rb") as fh:
@@ -58,10 +60,13 @@ def child(model: str, shm: str) -> None:
     if model == "sdxl":
         from modal_examples_amd.models.sdxl.pipeline import SDXLPipeline
 
+        from modal_examples_amd.gpu import fastload
+
         t0 = time.perf_counter()
-        # straight-to-device load; model built EMPTY on device (meta init)
-        sd_u = torch.load(f"{shm}/unet.pt", map_location=dev)
-        sd_v = torch.load(f"{shm}/vae.pt", map_location=dev)
+        # mmap + pinned-staged blob load (gpu/fastload.py); model built
+        # EMPTY on device (meta init), weights assigned as blob views
+        sd_u = fastload.load_file(f"{shm}/unet.safetensors", device=dev)
+        sd_v = fastload.load_file(f"{shm}/vae.safetensors", device=dev)
         pipe = SDXLPipeline(device=dev, dtype=torch.bfloat16,
                             init_weights=False)
         pipe.unet.load_state_dict(sd_u, assign=True)
@@ -81,11 +86,14 @@ def child(model: str, shm: str) -> None:
         from modal_examples_amd.models.llama.model import LlamaConfig
         from modal_examples_amd.models.llama.server import LLMServer
 
+        from modal_examples_amd.gpu import fastload
+
         t0 = time.perf_counter()
-        sd = torch.load(f"{shm}/llama.pt", map_location="cpu")
+        sd = fastload.load_file(f"{shm}/llama.safetensors", device=dev)
         eng = LlamaEngine(LlamaConfig.llama3_8b(), device=dev,
-                          dtype=torch.bfloat16, use_graph=(dev == "cuda"))
-        eng.model.load_state_dict(sd)
+                          dtype=torch.bfloat16, use_graph=(dev == "cuda"),
+                          init_weights=False)
+        eng.model.load_state_dict(sd, assign=True)
         if dev == "cuda":
             torch.cuda.synchronize()
         t_restore = time.perf_counter() - t0

@@ -1,0 +1,92 @@
+"""Fast weight loader: safetensors-layout roundtrip + library interop."""
+import pytest
+import torch
+
+from modal_examples_amd.gpu import fastload
+
+
+def _state():
+    g = torch.Generator().manual_seed(3)
+    return {
+        "w": torch.randn(64, 32, generator=g).to(torch.bfloat16),
+        "b": torch.randn(64, generator=g),
+        "idx": torch.arange(10, dtype=torch.int64),
+        "flag": torch.tensor([True, False]),
+    }
+
+
+def test_roundtrip_cpu(tmp_path):
+    p = str(tmp_path / "w.safetensors")
+    state = _state()
+    payload = fastload.save_file(state, p)
+    assert payload == sum(t.numel() * t.element_size() for t in state.values())
+    out = fastload.load_file(p)
+    assert set(out) == set(state)
+    for k in state:
+        assert out[k].dtype == state[k].dtype and out[k].shape == state[k].shape
+        assert torch.equal(out[k], state[k]), k
+
+
+def test_safetensors_library_reads_our_layout(tmp_path):
+    """The layout IS safetensors: the reference library must read it back."""
+    st = pytest.importorskip("safetensors.torch")
+    p = str(tmp_path / "w.safetensors")
+    state = _state()
+    fastload.save_file(state, p)
+    theirs = st.load_file(p)
+    for k in state:
+        assert torch.equal(theirs[k], state[k]), k
+
+
+def test_we_read_safetensors_library_files(tmp_path):
+    st = pytest.importorskip("safetensors.torch")
+    p = str(tmp_path / "lib.safetensors")
+    state = {k: v.contiguous() for k, v in _state().items()}
+    st.save_file(state, p)
+    ours = fastload.load_file(p)
+    for k in state:
+        assert torch.equal(ours[k], state[k]), k
+
+
+def test_module_roundtrip_assign(tmp_path):
+    p = str(tmp_path / "m.safetensors")
+    m = torch.nn.Sequential(torch.nn.Linear(16, 32), torch.nn.Linear(32, 4))
+    m = m.to(torch.bfloat16)
+    fastload.save_file(dict(m.state_dict()), p)
+    m2 = torch.nn.Sequential(torch.nn.Linear(16, 32), torch.nn.Linear(32, 4))
+    m2 = m2.to(torch.bfloat16)
+    m2.load_state_dict(fastload.load_file(p), assign=True)
+    x = torch.randn(2, 16, dtype=torch.bfloat16)
+    assert torch.equal(m(x), m2(x))
+
+
+@pytest.mark.gpu
+def test_load_to_gpu_matches(tmp_path):
+    p = str(tmp_path / "g.safetensors")
+    state = {"a": torch.randn(3000, 257).to(torch.bfloat16),
+             "b": torch.randn(5).float()}
+    fastload.save_file(state, p)
+    # small staging forces the double-buffer wraparound path
+    out = fastload.load_file(p, device="cuda:0", staging_mb=1)
+    for k in state:
+        assert out[k].is_cuda
+        assert torch.equal(out[k].cpu(), state[k]), k
+
+
+def test_engine_meta_init_restore_matches():
+    """LlamaEngine(init_weights=False) + fastload assign == source model."""
+    import tempfile
+
+    from modal_examples_amd.models.llama.engine import LlamaEngine
+    from modal_examples_amd.models.llama.model import LlamaConfig, LlamaModel
+
+    cfg = LlamaConfig.small()
+    src = LlamaModel(cfg).to(torch.bfloat16)
+    p = tempfile.mkdtemp() + "/w.safetensors"
+    fastload.save_file(dict(src.state_dict()), p)
+    eng = LlamaEngine(cfg, device="cpu", dtype=torch.bfloat16,
+                      use_graph=False, init_weights=False)
+    eng.model.load_state_dict(fastload.load_file(p), assign=True)
+    toks = torch.randint(0, cfg.vocab_size, (2, 12),
+                         generator=torch.Generator().manual_seed(0))
+    assert torch.equal(src.prefill(toks), eng.model.prefill(toks))

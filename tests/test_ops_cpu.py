@@ -104,3 +104,16 @@ def test_functional_cpu_paths():
     assert F.sample(torch.tensor([[0.0, 100.0, 0.0]]), temperature=0.0).item() == 1
     t = F.sample(torch.tensor([[0.0, 100.0, 0.0]]), temperature=1.0, seed=7)
     assert t.item() == 1
+
+
+def test_glu_fused_cpu_fallback():
+    """glu_fused CPU path == manual slice + act(a)*b for both activations."""
+    src = torch.randn(3, 5, 2 * 24, dtype=torch.bfloat16)
+    a, b = src[..., :24], src[..., 24:]
+    swiglu = F.glu_fused(src, gelu=False).float()
+    want = (torch.nn.functional.silu(a.float()) * b.float())
+    assert torch.allclose(swiglu, want, atol=2e-2)
+    geglu = F.glu_fused(src, gelu=True).float()
+    want = (torch.nn.functional.gelu(a.float(), approximate="tanh") * b.float())
+    assert torch.allclose(geglu, want, atol=2e-2)
+    assert swiglu.shape == (3, 5, 24)
