@@ -60,8 +60,14 @@ class GpuPeer:
         # --- media: frames arrive on the negotiated channel
         n = 0
         while n < max_frames:
-            msg = signaling.get(partition=f"{session_id}:media-in", timeout=10)
-            if msg is None or msg.get("type") == "bye":
+            import queue as _q
+
+            try:  # Queue.get(timeout=) raises Empty (reference semantics)
+                msg = signaling.get(partition=f"{session_id}:media-in",
+                                    timeout=10)
+            except _q.Empty:
+                break
+            if msg.get("type") == "bye":
                 break
             frame = torch.as_tensor(np.asarray(msg["frame"], dtype="float32"))
             with torch.no_grad():
