@@ -80,13 +80,27 @@ def _carve(blob: torch.Tensor, header: dict) -> Dict[str, torch.Tensor]:
     return out
 
 
-def load_file(path: str, device="cpu", staging_mb: int = 128) -> Dict[str, torch.Tensor]:
+def _memcpy_mt(dst: torch.Tensor, src: torch.Tensor, pool, t: int) -> None:
+    """Parallel host memcpy: torch copy_ releases the GIL, so slicing the
+    range across a thread pool multiplies page-cache read bandwidth."""
+    n = dst.numel()
+    step = (n + t - 1) // t
+    futs = [pool.submit(lambda a, b: dst[a:b].copy_(src[a:b]), s,
+                        min(s + step, n)) for s in range(0, n, step)]
+    for f in futs:
+        f.result()
+
+
+def load_file(path: str, device="cpu", staging_mb: int = 256,
+              threads: int = 4) -> Dict[str, torch.Tensor]:
     """Load a safetensors-layout file to `device`.
 
     CPU: zero-copy views over the mmap.  GPU: one device blob filled by
-    double-buffered pinned staging (host memcpy of chunk i overlaps the H2D
-    of chunk i-1), tensors carved as views.
+    double-buffered pinned staging (multi-threaded host memcpy of chunk i
+    overlaps the H2D of chunk i-1), tensors carved as views.
     """
+    from concurrent.futures import ThreadPoolExecutor
+
     header, data_off, size = _parse(path)
     cpu = _file_bytes(path, size)
     dev = torch.device(device)
@@ -100,14 +114,15 @@ def load_file(path: str, device="cpu", staging_mb: int = 128) -> Dict[str, torch
     events = [torch.cuda.Event(), torch.cuda.Event()]
     stream = torch.cuda.Stream(dev)
     src = cpu[data_off:]
-    for i, s in enumerate(range(0, payload, chunk)):
-        e = min(s + chunk, payload)
-        buf, ev = pinned[i % 2], events[i % 2]
-        if i >= 2:
-            ev.synchronize()  # buf's previous H2D must be done before reuse
-        buf[:e - s].copy_(src[s:e])  # page-cache -> pinned (host memcpy)
-        with torch.cuda.stream(stream):
-            blob[s:e].copy_(buf[:e - s], non_blocking=True)
-            ev.record(stream)
+    with ThreadPoolExecutor(max_workers=threads) as pool:
+        for i, s in enumerate(range(0, payload, chunk)):
+            e = min(s + chunk, payload)
+            buf, ev = pinned[i % 2], events[i % 2]
+            if i >= 2:
+                ev.synchronize()  # buf's previous H2D must finish before reuse
+            _memcpy_mt(buf[:e - s], src[s:e], pool, threads)
+            with torch.cuda.stream(stream):
+                blob[s:e].copy_(buf[:e - s], non_blocking=True)
+                ev.record(stream)
     stream.synchronize()
     return _carve(blob, header)

@@ -72,12 +72,18 @@ class LlamaEngine:
             with torch.device(self.device):
                 self.model = LlamaModel(self.cfg).to(self.device, dtype)
         else:
-            # cold-restore path: build on meta (no init compute), materialize
-            # empty, then the caller load_state_dict(assign=True)s real
-            # weights (gpu/fastload.py blob views)
-            with torch.device("meta"):
-                self.model = LlamaModel(self.cfg)
-            self.model = self.model.to_empty(device=self.device).to(dtype)
+            # cold-restore path: build on meta (no init compute) directly in
+            # the target dtype so to_empty materializes ONCE with no cast;
+            # the caller load_state_dict(assign=True)s real weights
+            # (gpu/fastload.py blob views)
+            prev = torch.get_default_dtype()
+            try:
+                torch.set_default_dtype(dtype)
+                with torch.device("meta"):
+                    self.model = LlamaModel(self.cfg)
+            finally:
+                torch.set_default_dtype(prev)
+            self.model = self.model.to_empty(device=self.device)
         self.model.eval()
         c = self.cfg
         if kv_blocks is None:

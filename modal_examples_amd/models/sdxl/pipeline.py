@@ -67,13 +67,19 @@ class SDXLPipeline:
                 self.unet = UNetXL(self.cfg).to(self.device, dtype)
                 self.vae = vae_cls().to(self.device, dtype)
         else:
-            # cold-restore path: build on meta (no init compute), materialize
-            # EMPTY on device — load_state_dict fills it (bench_cold.py)
-            with torch.device("meta"):
-                self.unet = UNetXL(self.cfg)
-                self.vae = vae_cls()
-            self.unet = self.unet.to_empty(device=self.device).to(dtype)
-            self.vae = self.vae.to_empty(device=self.device).to(dtype)
+            # cold-restore path: build on meta (no init compute) directly in
+            # the target dtype — to_empty materializes ONCE, no f32->bf16
+            # cast pass; load_state_dict fills it (bench_cold.py)
+            prev = torch.get_default_dtype()
+            try:
+                torch.set_default_dtype(dtype)
+                with torch.device("meta"):
+                    self.unet = UNetXL(self.cfg)
+                    self.vae = vae_cls()
+            finally:
+                torch.set_default_dtype(prev)
+            self.unet = self.unet.to_empty(device=self.device)
+            self.vae = self.vae.to_empty(device=self.device)
         self.unet.eval()
         self.vae.eval()
         self._graphs = GraphLRU(4)  # (batch, steps, cfg_on) -> graph state, LRU-bounded
