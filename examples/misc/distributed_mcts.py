@@ -1,7 +1,7 @@
 # ---
 # cmd: ["python", "-m", "modal_examples_amd", "run", "examples/misc/distributed_mcts.py"]
 # ---
-# # Distributed Monte-Carlo tree search
+# # Distributed Monte-Carlo tree search (misc/distributed_mcts_reasoner.py role)
 #
 # The search tree lives in the client; rollouts fan out with `.map` (one
 # batch of leaves per wave).  The toy domain: choose digits to approach a

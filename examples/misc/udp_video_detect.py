@@ -67,7 +67,10 @@ def main(frames: int = 12):
 
     import numpy as np
 
-    session = "sess-1"
+    import uuid
+
+    session = f"sess-{uuid.uuid4().hex[:8]}"  # unique per run: a stale peer
+    # from an interrupted previous run must not re-register our key
     rendezvous.delete(session)
     call = detector_peer.spawn(session)
     # rendezvous: wait for the peer's punched endpoint

@@ -303,8 +303,16 @@ class Pool:
     def reap_idle(self, force: bool = False):
         now = time.monotonic()
         with self.lock:
+            if self.pending and not force:
+                return  # queued inputs need every worker we have
             for w in list(self.workers.values()):
                 if w.inflight or w.shutting_down:
+                    continue
+                if not w.ready and not force:
+                    # still starting (@enter running): idle time is measured
+                    # from READY, never from spawn — a sub-second
+                    # scaledown_window must not reap a booting container
+                    # (liveness: kill-on-boot + respawn would loop forever)
                     continue
                 idle = now - w.last_active
                 if force or idle > self.scaledown_window:

@@ -196,3 +196,23 @@ def test_exit_hook_runs_on_scaledown():
         time.sleep(0.05)
     assert d.get("ran") is True, "exit hook did not run at worker shutdown"
     modal.Dict.delete("exit-hook-proof")
+
+
+def test_subsecond_scaledown_does_not_reap_booting_worker():
+    """Liveness: scaledown_window shorter than @enter must not kill the
+    booting container (reap-on-boot + respawn looped forever; idle time
+    counts from READY, and queued inputs pin the pool)."""
+    app = modal.App("test-slow-enter-scaledown")
+
+    @app.cls(scaledown_window=0.2)
+    class Slow:
+        @modal.enter()
+        def boot(self):
+            time.sleep(1.5)  # >> scaledown_window
+            self.ok = 41
+
+        @modal.method()
+        def get(self):
+            return self.ok + 1
+
+    assert Slow().get.remote() == 42
