@@ -80,19 +80,25 @@ def _carve(blob: torch.Tensor, header: dict) -> Dict[str, torch.Tensor]:
     return out
 
 
-def _memcpy_mt(dst: torch.Tensor, src: torch.Tensor, pool, t: int) -> None:
-    """Parallel host memcpy: torch copy_ releases the GIL, so slicing the
-    range across a thread pool multiplies page-cache read bandwidth."""
-    n = dst.numel()
+def _pread_mt(fd: int, dst_mv: memoryview, file_off: int, n: int, pool,
+              t: int) -> None:
+    """Parallel pread into pinned memory: os.preadv releases the GIL and the
+    kernel copies straight out of the page cache — no mmap minor-faults
+    (4 M faults for 16 GB is what bounds an mmap+memcpy loader)."""
     step = (n + t - 1) // t
-    futs = [pool.submit(lambda a, b: dst[a:b].copy_(src[a:b]), s,
-                        min(s + step, n)) for s in range(0, n, step)]
+
+    def one(a: int, b: int) -> None:
+        got = os.preadv(fd, [dst_mv[a:b]], file_off + a)
+        if got != b - a:
+            raise IOError(f"short read at {file_off + a}: {got} != {b - a}")
+
+    futs = [pool.submit(one, s, min(s + step, n)) for s in range(0, n, step)]
     for f in futs:
         f.result()
 
 
 def load_file(path: str, device="cpu", staging_mb: int = 256,
-              threads: int = 4) -> Dict[str, torch.Tensor]:
+              threads: int = 8) -> Dict[str, torch.Tensor]:
     """Load a safetensors-layout file to `device`.
 
     CPU: zero-copy views over the mmap.  GPU: one device blob filled by
@@ -102,27 +108,30 @@ def load_file(path: str, device="cpu", staging_mb: int = 256,
     from concurrent.futures import ThreadPoolExecutor
 
     header, data_off, size = _parse(path)
-    cpu = _file_bytes(path, size)
     dev = torch.device(device)
     if dev.type != "cuda":
-        return _carve(cpu[data_off:], header)
+        return _carve(_file_bytes(path, size)[data_off:], header)
     payload = size - data_off
     blob = torch.empty(payload, dtype=torch.uint8, device=dev)
     chunk = staging_mb << 20
     pinned = [torch.empty(min(chunk, payload), dtype=torch.uint8,
                           pin_memory=True) for _ in range(2)]
+    views = [memoryview(p.numpy()) for p in pinned]
     events = [torch.cuda.Event(), torch.cuda.Event()]
     stream = torch.cuda.Stream(dev)
-    src = cpu[data_off:]
-    with ThreadPoolExecutor(max_workers=threads) as pool:
-        for i, s in enumerate(range(0, payload, chunk)):
-            e = min(s + chunk, payload)
-            buf, ev = pinned[i % 2], events[i % 2]
-            if i >= 2:
-                ev.synchronize()  # buf's previous H2D must finish before reuse
-            _memcpy_mt(buf[:e - s], src[s:e], pool, threads)
-            with torch.cuda.stream(stream):
-                blob[s:e].copy_(buf[:e - s], non_blocking=True)
-                ev.record(stream)
-    stream.synchronize()
+    fd = os.open(path, os.O_RDONLY)
+    try:
+        with ThreadPoolExecutor(max_workers=threads) as pool:
+            for i, s in enumerate(range(0, payload, chunk)):
+                e = min(s + chunk, payload)
+                buf, mv, ev = pinned[i % 2], views[i % 2], events[i % 2]
+                if i >= 2:
+                    ev.synchronize()  # previous H2D out of buf must finish
+                _pread_mt(fd, mv, data_off + s, e - s, pool, threads)
+                with torch.cuda.stream(stream):
+                    blob[s:e].copy_(buf[:e - s], non_blocking=True)
+                    ev.record(stream)
+        stream.synchronize()
+    finally:
+        os.close(fd)
     return _carve(blob, header)

@@ -53,6 +53,7 @@ def prepare(model: str, shm: str) -> dict:
 
 def child(model: str, shm: str) -> None:
     t_start = time.perf_counter()
+    t_load = 0.0  # file->device portion of restore (llama path reports it)
     import torch
 
     t_import = time.perf_counter()
@@ -90,6 +91,9 @@ def child(model: str, shm: str) -> None:
 
         t0 = time.perf_counter()
         sd = fastload.load_file(f"{shm}/llama.safetensors", device=dev)
+        if dev == "cuda":
+            torch.cuda.synchronize()
+        t_load = time.perf_counter() - t0
         eng = LlamaEngine(LlamaConfig.llama3_8b(), device=dev,
                           dtype=torch.bfloat16, use_graph=(dev == "cuda"),
                           init_weights=False)
@@ -108,6 +112,7 @@ def child(model: str, shm: str) -> None:
         "cold_start_s": round(total, 3),
         "import_s": round(t_import - t_start, 3),
         "restore_s": round(t_restore, 3),
+        "load_s": round(t_load, 3),
         "restore_gb_s": round(nbytes / t_restore / 1e9, 2),
         "first_call_s": round(t_first, 3),
         "weight_gb": round(nbytes / 1e9, 2)}), flush=True)
