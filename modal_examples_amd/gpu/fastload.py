@@ -1,17 +1,18 @@
-"""Fast weight loading: safetensors layout, mmap + pinned staged H2D.
+"""Fast weight loading: safetensors layout, threaded preadv + pinned staged H2D.
 
 The fresh-process half of the snapshot story (reference:
 06_gpu_and_ml/llm-serving/sglang_snapshot.py:176-218 restores engine weights
 from host residence; gpu_snapshot.py:41-53).  `WeightSnapshot` (snapshot.py)
 covers the in-process warm path; this module covers the cold path: weights
 saved once in safetensors layout, restored by a fresh process WITHOUT pickle
-deserialization — the file is mmapped (page-cache reads, zero parse), bytes
-stream through two pinned staging buffers whose host-memcpy and H2D copies
-overlap on a side stream, and every tensor is carved as a VIEW of one device
-blob (one allocation, no per-tensor cudaMalloc).
+deserialization: parallel os.preadv calls copy file bytes straight from the
+page cache into two pinned staging buffers (no mmap minor-faults — 4M faults
+for 16 GB is what bounds mmap loaders), each chunk's read overlaps the
+previous chunk's H2D on a side stream, and every tensor is carved as a VIEW
+of one device blob (one allocation, no per-tensor cudaMalloc).
 
 torch.load on the same weights measures ~3 GB/s (pickle+zip walk); this path
-is bound by min(page-cache memcpy, pinned H2D) instead.
+is bound by min(parallel page-cache read, pinned H2D) instead.
 """
 from __future__ import annotations
 
@@ -102,8 +103,8 @@ def load_file(path: str, device="cpu", staging_mb: int = 256,
     """Load a safetensors-layout file to `device`.
 
     CPU: zero-copy views over the mmap.  GPU: one device blob filled by
-    double-buffered pinned staging (multi-threaded host memcpy of chunk i
-    overlaps the H2D of chunk i-1), tensors carved as views.
+    double-buffered pinned staging (threaded preadv of chunk i overlaps the
+    H2D of chunk i-1), tensors carved as views.
     """
     from concurrent.futures import ThreadPoolExecutor
 
