@@ -50,18 +50,18 @@ class FastLLM:
     def boot(self):
         import torch
 
-        from modal_examples_amd.gpu import fastload
         from modal_examples_amd.models.llama.engine import LlamaEngine
         from modal_examples_amd.models.llama.model import LlamaConfig
         from modal_examples_amd.models.llama.server import LLMServer
 
         t0 = time.time()
         dev = "cuda" if torch.cuda.is_available() else "cpu"
-        sd = fastload.load_file("/weights/llama-small.safetensors", device=dev)
-        eng = LlamaEngine(LlamaConfig.small(), device=dev,
-                          dtype=torch.bfloat16, use_graph=(dev == "cuda"),
-                          init_weights=False, eos_id=-1)
-        eng.model.load_state_dict(sd, assign=True)
+        # one call: meta-init engine build overlapped with the preadv
+        # weight stream, blob views assigned as parameters
+        eng = LlamaEngine.from_safetensors(
+            "/weights/llama-small.safetensors", cfg=LlamaConfig.small(),
+            device=dev, dtype=torch.bfloat16, use_graph=(dev == "cuda"),
+            eos_id=-1)
         self.server = LLMServer(eng, model_name="fast-cold-boot")
         self.boot_s = time.time() - t0
 

@@ -121,6 +121,45 @@ class LlamaEngine:
         self.temps_d = torch.zeros(max_batch, dtype=torch.float32, device=dev)
         self.logits_d = None
 
+    # ------------------------------------------------ cold boot
+
+    def save_safetensors(self, path: str) -> int:
+        """Bake the weights for `from_safetensors` (the snapshot-build step;
+        reference role: sglang_snapshot.py:176-218 warm/sleep/wake)."""
+        from ...gpu import fastload
+
+        return fastload.save_file(dict(self.model.state_dict()), path)
+
+    @classmethod
+    def from_safetensors(cls, path: str, cfg: Optional[LlamaConfig] = None,
+                         device: str = "cuda", **kw) -> "LlamaEngine":
+        """Cold-boot an engine from baked weights: the KV-pool/empty-model
+        allocation (the hipMalloc long pole) runs on a thread WHILE the
+        weight file streams through pinned staging (gpu/fastload.py), then
+        the blob views are assigned as parameters.  Measured: 16 GB Llama-8B
+        end-to-end boot p50 6.4 s, best 2.9 s (profiles/final_cold_llama4)."""
+        import threading
+
+        from ...gpu import fastload
+
+        box: dict = {}
+
+        def build():
+            try:
+                box["eng"] = cls(cfg, device=device, init_weights=False, **kw)
+            except BaseException as e:  # re-raised on the caller thread
+                box["exc"] = e
+
+        th = threading.Thread(target=build)
+        th.start()
+        sd = fastload.load_file(path, device=device)
+        th.join()
+        if "exc" in box:
+            raise box["exc"]
+        eng = box["eng"]
+        eng.model.load_state_dict(sd, assign=True)
+        return eng
+
     # ------------------------------------------------ request lifecycle
 
     def add_request(self, prompt: List[int], max_new_tokens: int = 64,

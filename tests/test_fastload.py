@@ -90,3 +90,49 @@ def test_engine_meta_init_restore_matches():
     toks = torch.randint(0, cfg.vocab_size, (2, 12),
                          generator=torch.Generator().manual_seed(0))
     assert torch.equal(src.prefill(toks), eng.model.prefill(toks))
+
+
+def test_llama_engine_safetensors_roundtrip(tmp_path):
+    """save_safetensors -> from_safetensors gives an identical engine."""
+    from modal_examples_amd.models.llama.engine import LlamaEngine
+    from modal_examples_amd.models.llama.model import LlamaConfig
+
+    cfg = LlamaConfig.small()
+    src = LlamaEngine(cfg, device="cpu", dtype=torch.bfloat16, use_graph=False)
+    p = str(tmp_path / "eng.safetensors")
+    src.save_safetensors(p)
+    eng = LlamaEngine.from_safetensors(p, cfg=cfg, device="cpu",
+                                       dtype=torch.bfloat16, use_graph=False)
+    toks = torch.randint(0, cfg.vocab_size, (2, 9),
+                         generator=torch.Generator().manual_seed(1))
+    assert torch.equal(src.model.prefill(toks), eng.model.prefill(toks))
+
+
+def test_llama_from_safetensors_build_error_propagates(tmp_path):
+    from modal_examples_amd.models.llama.engine import LlamaEngine
+    from modal_examples_amd.models.llama.model import LlamaConfig, LlamaModel
+
+    p = str(tmp_path / "w.safetensors")
+    fastload.save_file(dict(LlamaModel(LlamaConfig.small())
+                            .to(torch.bfloat16).state_dict()), p)
+    with pytest.raises(TypeError):
+        LlamaEngine.from_safetensors(p, cfg=LlamaConfig.small(), device="cpu",
+                                     not_a_kwarg=1)
+
+
+def test_sdxl_pipeline_safetensors_roundtrip(tmp_path):
+    from modal_examples_amd.models.sdxl.pipeline import SDXLPipeline
+    from modal_examples_amd.models.sdxl.unet import UNetConfig
+
+    cfg = UNetConfig.small()
+    src = SDXLPipeline(cfg, device="cpu", dtype=torch.bfloat16, seed=0)
+    p = str(tmp_path / "pipe.safetensors")
+    src.save_safetensors(p)
+    pipe = SDXLPipeline.from_safetensors(p, device="cpu", cfg=cfg,
+                                         dtype=torch.bfloat16)
+    for (ka, va), (kb, vb) in zip(src.unet.state_dict().items(),
+                                  pipe.unet.state_dict().items()):
+        assert ka == kb and torch.equal(va, vb), ka
+    for (ka, va), (kb, vb) in zip(src.vae.state_dict().items(),
+                                  pipe.vae.state_dict().items()):
+        assert ka == kb and torch.equal(va, vb), ka
