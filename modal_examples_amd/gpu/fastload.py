@@ -51,7 +51,8 @@ def save_file(state: Dict[str, torch.Tensor], path: str) -> int:
         f.write(hdr)
         for t in tensors:
             # uint8 reinterpret works for every dtype incl. bf16/fp8
-            f.write(t.view(torch.uint8).view(-1).numpy().tobytes())
+            # (0-dim tensors must be lifted to 1-d before the dtype view)
+            f.write(t.reshape(-1).view(torch.uint8).numpy().tobytes())
     os.replace(tmp, path)
     return off
 

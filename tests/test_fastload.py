@@ -136,3 +136,18 @@ def test_sdxl_pipeline_safetensors_roundtrip(tmp_path):
     for (ka, va), (kb, vb) in zip(src.vae.state_dict().items(),
                                   pipe.vae.state_dict().items()):
         assert ka == kb and torch.equal(va, vb), ka
+
+
+def test_fp8_and_scalar_roundtrip(tmp_path):
+    """fp8 e4m3 payloads (the KV-cache dtype) and 0-dim tensors survive."""
+    p = str(tmp_path / "odd.safetensors")
+    state = {
+        "kv": torch.randn(4, 8).to(torch.float8_e4m3fn),
+        "scalar": torch.tensor(3.5),
+        "empty": torch.zeros(0, 7, dtype=torch.bfloat16),
+    }
+    fastload.save_file(state, p)
+    out = fastload.load_file(p)
+    assert torch.equal(out["kv"].view(torch.uint8), state["kv"].view(torch.uint8))
+    assert out["scalar"].item() == 3.5 and out["scalar"].shape == ()
+    assert out["empty"].shape == (0, 7) and out["empty"].dtype == torch.bfloat16
