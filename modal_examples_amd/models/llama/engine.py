@@ -53,8 +53,16 @@ class LlamaEngine:
                  dtype=torch.bfloat16, max_batch: int = 64,
                  kv_blocks: Optional[int] = None, use_graph: bool = True,
                  eos_id: int = 2, seed: int = 0, top_p: float = 1.0,
-                 kv_dtype: str = "bf16", init_weights: bool = True):
+                 kv_dtype: str = "bf16", init_weights: bool = True,
+                 tp=None):
+        """tp: optional parallel.tp.TPGroup — head-sharded tensor parallelism
+        (vllm_inference.py:180 --tensor-parallel-size role).  Every rank runs
+        the same engine loop on identical requests; the KV cache holds only
+        this rank's kv-head shard, logits come out identical on all ranks
+        (replicated lm_head after the in-model all-reduces), so seeded
+        sampling stays rank-consistent with no extra broadcast."""
         self.cfg = cfg or LlamaConfig.llama3_8b()
+        self.tp = tp
         self.device = torch.device(device)
         self.dtype = dtype
         self.max_batch = max_batch
@@ -70,7 +78,7 @@ class LlamaEngine:
         torch.manual_seed(seed)
         if init_weights:
             with torch.device(self.device):
-                self.model = LlamaModel(self.cfg).to(self.device, dtype)
+                self.model = LlamaModel(self.cfg, tp=tp).to(self.device, dtype)
         else:
             # cold-restore path: build on meta (no init compute) directly in
             # the target dtype so to_empty materializes ONCE with no cast;
@@ -80,7 +88,7 @@ class LlamaEngine:
             try:
                 torch.set_default_dtype(dtype)
                 with torch.device("meta"):
-                    self.model = LlamaModel(self.cfg)
+                    self.model = LlamaModel(self.cfg, tp=tp)
             finally:
                 torch.set_default_dtype(prev)
             self.model = self.model.to_empty(device=self.device)
@@ -90,7 +98,8 @@ class LlamaEngine:
             if self.device.type == "cuda":
                 free, _ = torch.cuda.mem_get_info(self.device)
                 kv_elt = 1 if self.kv_dtype == torch.float8_e4m3fn else 2
-                per_block = c.n_layers * c.n_kv_heads * BLOCK * c.head_dim * 2 * kv_elt
+                per_block = (c.n_layers * self.model.blocks[0].nkv * BLOCK
+                             * c.head_dim * 2 * kv_elt)
                 kv_blocks = max(64, int(free * 0.6 / per_block))
             else:
                 kv_blocks = 256
@@ -98,7 +107,8 @@ class LlamaEngine:
         self.max_blocks_per_seq = (c.max_seq + BLOCK - 1) // BLOCK
         cache_dt = self.kv_dtype if self.device.type == "cuda" else dtype
         self.cache_k = torch.zeros(
-            c.n_layers, kv_blocks, c.n_kv_heads, BLOCK, c.head_dim,
+            c.n_layers, kv_blocks, self.model.blocks[0].nkv, BLOCK,
+            c.head_dim,
             device=self.device, dtype=cache_dt)
         self.cache_v = torch.zeros_like(self.cache_k)
         self.free_blocks = list(range(kv_blocks - 1, 0, -1))  # block 0 = pad

@@ -46,18 +46,27 @@ class LlamaConfig:
 
 
 class LlamaBlock(nn.Module):
-    def __init__(self, cfg: LlamaConfig):
+    def __init__(self, cfg: LlamaConfig, tp_world: int = 1):
+        """tp_world>1: Megatron head-sharded block (the --tensor-parallel-size
+        role of vllm_inference.py:180 / trtllm tensor_parallel_size).  qkv and
+        gate_up are column-sharded (this rank's q/kv-head and gate/up slices,
+        NO comm in forward), o_proj and down are row-sharded — the model adds
+        ONE all-reduce after each (two collectives per block over xGMI)."""
         super().__init__()
         d, hd = cfg.dim, cfg.head_dim
-        self.nq, self.nkv = cfg.n_heads, cfg.n_kv_heads
+        assert cfg.n_heads % tp_world == 0 and cfg.n_kv_heads % tp_world == 0 \
+            and cfg.ffn_dim % tp_world == 0, \
+            f"tp={tp_world} must divide heads {cfg.n_heads}/{cfg.n_kv_heads}" \
+            f" and ffn {cfg.ffn_dim}"
+        self.nq, self.nkv = cfg.n_heads // tp_world, cfg.n_kv_heads // tp_world
         self.attn_norm = RMSNormK(d, cfg.norm_eps)
         self.qkv = nn.Linear(d, (self.nq + 2 * self.nkv) * hd, bias=False)
         self.o_proj = nn.Linear(self.nq * hd, d, bias=False)
         self.ffn_norm = RMSNormK(d, cfg.norm_eps)
-        self.gate_up = nn.Linear(d, 2 * cfg.ffn_dim, bias=False)
-        self.down = nn.Linear(cfg.ffn_dim, d, bias=False)
+        self.gate_up = nn.Linear(d, 2 * cfg.ffn_dim // tp_world, bias=False)
+        self.down = nn.Linear(cfg.ffn_dim // tp_world, d, bias=False)
         self.hd = hd
-        self.ffn_dim = cfg.ffn_dim
+        self.ffn_dim = cfg.ffn_dim // tp_world
 
     def project_qkv(self, x):
         """x [B,S,d] → BSHD views of the fused projection (zero copies):
@@ -76,11 +85,18 @@ class LlamaBlock(nn.Module):
 
 
 class LlamaModel(nn.Module):
-    def __init__(self, cfg: LlamaConfig):
+    def __init__(self, cfg: LlamaConfig, tp=None):
+        """tp: optional parallel.tp.TPGroup — shards every block across the
+        group (embed/norms/lm_head replicated; activations stay replicated,
+        so logits are identical on every rank and sampling needs no extra
+        broadcast)."""
         super().__init__()
         self.cfg = cfg
+        self.tp = tp if (tp is not None and tp.world > 1) else None
+        w = self.tp.world if self.tp else 1
         self.embed = nn.Embedding(cfg.vocab_size, cfg.dim)
-        self.blocks = nn.ModuleList([LlamaBlock(cfg) for _ in range(cfg.n_layers)])
+        self.blocks = nn.ModuleList(
+            [LlamaBlock(cfg, tp_world=w) for _ in range(cfg.n_layers)])
         self.norm = RMSNormK(cfg.dim, cfg.norm_eps)
         self.lm_head = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
         # NOT buffers: module .to(bf16) must not downcast the f32 trig tables
@@ -89,6 +105,11 @@ class LlamaModel(nn.Module):
             cos, sin = OF.rope_tables(cfg.max_seq, cfg.head_dim, cfg.rope_base)
         self._rope_cpu = (cos, sin)
         self._rope_cache = {}
+
+    def _ar(self, t):
+        """Sum-reduce a row-parallel partial across the TP group (no-op
+        single-rank)."""
+        return self.tp.all_reduce(t) if self.tp is not None else t
 
     def _rope_tables(self, device) -> tuple:
         key = str(device)
@@ -112,9 +133,9 @@ class LlamaModel(nn.Module):
             q = OF.rope(q.transpose(1, 2), rc, rs, inplace=gpu).transpose(1, 2)
             k = OF.rope(k.transpose(1, 2), rc, rs, inplace=gpu).transpose(1, 2)
             if kv_writer is not None:
-                kv_writer(li, k, v)  # [B,S,nkv,hd]
-            x = x + blk.o_proj(OF.attention_qkv(q, k, v, causal=True))
-            x = x + blk.ffn(blk.ffn_norm(x))
+                kv_writer(li, k, v)  # [B,S,nkv_local,hd]
+            x = x + self._ar(blk.o_proj(OF.attention_qkv(q, k, v, causal=True)))
+            x = x + self._ar(blk.ffn(blk.ffn_norm(x)))
         if last_pos is None:
             x = self.norm(x[:, -1:])
         else:
@@ -137,9 +158,46 @@ class LlamaModel(nn.Module):
                         inplace=gpu).transpose(1, 2)
             k = OF.rope(k.transpose(1, 2), rc, rs, positions=positions,
                         inplace=gpu).transpose(1, 2)
-            kv_append(li, k, v)  # [B,1,nkv,hd]
+            kv_append(li, k, v)  # [B,1,nkv_local,hd]
             o = kv_attend(li, q[:, 0].contiguous() if not q[:, 0].is_contiguous() else q[:, 0])
-            x = x + blk.o_proj(o.reshape(o.shape[0], 1, -1))
-            x = x + blk.ffn(blk.ffn_norm(x))
+            x = x + self._ar(blk.o_proj(o.reshape(o.shape[0], 1, -1)))
+            x = x + self._ar(blk.ffn(blk.ffn_norm(x)))
         x = self.norm(x[:, -1:])
         return self.lm_head(x)[:, 0].float()
+
+
+def shard_llama_state(state: dict, cfg: LlamaConfig, rank: int,
+                      world: int) -> dict:
+    """Slice a FULL single-GPU state dict into rank's TP shard.
+
+    Matches LlamaBlock(tp_world=world): per block, qkv/gate_up rows are this
+    rank's q/kv-head and gate/up slices, o_proj/down columns are the
+    matching input slices; embed/norms/lm_head replicate.  Used to TP-ify a
+    baked checkpoint (fastload file) without materializing N copies.
+    """
+    hd = cfg.head_dim
+    nq_l, nkv_l = cfg.n_heads // world, cfg.n_kv_heads // world
+    ffn_l = cfg.ffn_dim // world
+    q_off, k_off = 0, cfg.n_heads * hd
+    v_off = k_off + cfg.n_kv_heads * hd
+    out = {}
+    for k, v in state.items():
+        if k.endswith(".qkv.weight"):
+            rows = torch.cat([
+                v[q_off + rank * nq_l * hd: q_off + (rank + 1) * nq_l * hd],
+                v[k_off + rank * nkv_l * hd: k_off + (rank + 1) * nkv_l * hd],
+                v[v_off + rank * nkv_l * hd: v_off + (rank + 1) * nkv_l * hd],
+            ])
+            out[k] = rows
+        elif k.endswith(".o_proj.weight"):
+            out[k] = v[:, rank * nq_l * hd: (rank + 1) * nq_l * hd].contiguous()
+        elif k.endswith(".gate_up.weight"):
+            out[k] = torch.cat([
+                v[rank * ffn_l: (rank + 1) * ffn_l],
+                v[cfg.ffn_dim + rank * ffn_l: cfg.ffn_dim + (rank + 1) * ffn_l],
+            ])
+        elif k.endswith(".down.weight"):
+            out[k] = v[:, rank * ffn_l: (rank + 1) * ffn_l].contiguous()
+        else:
+            out[k] = v
+    return out

@@ -30,7 +30,14 @@ class TPGroup:
 
     def all_reduce(self, t: torch.Tensor) -> torch.Tensor:
         if self.world > 1:
-            dist.all_reduce(t, group=self.group)
+            if t.dtype is torch.bfloat16 and not t.is_cuda:
+                # gloo (the CPU test backend) has no bf16 reduction; RCCL
+                # reduces bf16 natively on GPU
+                t32 = t.float()
+                dist.all_reduce(t32, group=self.group)
+                t.copy_(t32)
+            else:
+                dist.all_reduce(t, group=self.group)
         return t
 
     def all_gather_cat(self, t: torch.Tensor, dim: int = -1) -> torch.Tensor:
