@@ -49,8 +49,26 @@ y = down_s(torch.nn.functional.gelu(up_s(x), approximate="tanh"))
 
 full = sum(p.numel() for p in (*up.parameters(), *down.parameters()))
 shard = sum(p.numel() for m in (up_s, down_s) for p in m.parameters())
+
+# ---- act 2: the whole serving ENGINE under TP (engine `tp=` option):
+# head-sharded blocks, per-rank KV-head cache shard, rank-identical
+# logits -> the continuous-batching loop needs no extra broadcast
+from modal_examples_amd.models.llama.engine import LlamaEngine
+from modal_examples_amd.models.llama.model import (LlamaConfig, LlamaModel,
+                                                   shard_llama_state)
+
+cfg = LlamaConfig.small()
+torch.manual_seed(0)
+full_state = dict(LlamaModel(cfg).to(torch.bfloat16).state_dict())
+eng = LlamaEngine(cfg, device=device, dtype=torch.bfloat16,
+                  use_graph=(device != "cpu"), eos_id=-1, tp=tp)
+eng.model.load_state_dict(shard_llama_state(full_state, cfg, rank, world))
+eng.add_request(list(range(10, 22)), max_new_tokens=6, temperature=0.0)
+eng.run_until_done(max_steps=100)
+toks = eng.finished[1].out_tokens
+
 out = {"rank": rank, "max_err": float((y - ref).abs().max()),
-       "shard_frac": shard / full}
+       "shard_frac": shard / full, "engine_tokens": toks}
 print("TPRESULT " + json.dumps(out), flush=True)
 dist.barrier()
 dist.destroy_process_group()
@@ -89,5 +107,9 @@ def main(world: int = 2):
               f"params held = {r['shard_frac']:.2%} of full")
         assert r["max_err"] < 1e-4
         assert abs(r["shard_frac"] - 1 / world) < 0.05
+    # TP engine: every rank decoded the SAME tokens (no divergence)
+    tok_sets = {tuple(results[rank]["engine_tokens"]) for rank in range(world)}
+    assert len(tok_sets) == 1 and len(results[0]["engine_tokens"]) == 6
     print(f"TP={world}: exact full-model output from 1/{world} shards, "
-          "one all-reduce per block")
+          "one all-reduce per block; TP engine decoded "
+          f"{results[0]['engine_tokens']} identically on every rank")
