@@ -45,6 +45,10 @@ from .resources.dict_queue import Dict, Queue  # noqa: F401
 from .resources.image import Image  # noqa: F401
 from .resources.secret import Secret  # noqa: F401
 from .resources.volume import CloudBucketMount, Volume  # noqa: F401
+
+# modal-API alias: NetworkFileSystem is the older shared-writable-FS object;
+# Volume covers its semantics here (same commit/reload-free shared dir)
+NetworkFileSystem = Volume
 from .runtime.sandbox import ContainerProcess, Probe, Sandbox, Tunnel  # noqa: F401
 from .web.ingress import (  # noqa: F401
     asgi_app,
