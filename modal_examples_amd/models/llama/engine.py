@@ -54,15 +54,23 @@ class LlamaEngine:
                  kv_blocks: Optional[int] = None, use_graph: bool = True,
                  eos_id: int = 2, seed: int = 0, top_p: float = 1.0,
                  kv_dtype: str = "bf16", init_weights: bool = True,
-                 tp=None):
+                 tp=None, spec_tokens: int = 0):
         """tp: optional parallel.tp.TPGroup — head-sharded tensor parallelism
         (vllm_inference.py:180 --tensor-parallel-size role).  Every rank runs
         the same engine loop on identical requests; the KV cache holds only
         this rank's kv-head shard, logits come out identical on all ranks
         (replicated lm_head after the in-model all-reduces), so seeded
-        sampling stays rank-consistent with no extra broadcast."""
+        sampling stays rank-consistent with no extra broadcast.
+
+        spec_tokens>0: ngram speculative decoding (vllm_inference.py:195-202
+        role) — up to k draft tokens per request verified in ONE expanded
+        decode forward (models/llama/spec.py); greedy requests only, exact
+        same tokens as plain decode by construction."""
         self.cfg = cfg or LlamaConfig.llama3_8b()
         self.tp = tp
+        self.spec_tokens = spec_tokens
+        self.spec_proposed = 0   # drafts offered
+        self.spec_accepted = 0   # drafts accepted (emitted without a step)
         self.device = torch.device(device)
         self.dtype = dtype
         self.max_batch = max_batch
@@ -336,6 +344,92 @@ class LlamaEngine:
 
         return self.model.decode_step(toks, pos, kv_append, kv_attend)
 
+    # ------------------------------------------------ speculative decode
+
+    def _propose(self, r: Request) -> List[int]:
+        """Draft tokens for r (override point for tests/other proposers)."""
+        from .spec import ngram_propose
+
+        return ngram_propose(self._feed(r), self.spec_tokens)
+
+    def _ensure_blocks_ahead(self, r: Request, ahead: int) -> int:
+        """Allocate blocks covering positions < r.pos + ahead; returns how
+        many positions are actually covered (spec drafts trim to fit)."""
+        while True:
+            need = (r.pos + ahead + BLOCK - 1) // BLOCK
+            if need <= len(r.blocks):
+                return ahead
+            got = self._alloc_blocks(1)
+            if got is None:
+                covered = len(r.blocks) * BLOCK - r.pos
+                return max(1, covered)
+            r.blocks.extend(got)
+            self.bt_d[r.slot, len(r.blocks) - 1] = got[0]
+
+    @torch.no_grad()
+    def _decode_batch_spec(self):
+        """Ngram-speculative decode step: one row per (request, draft
+        position).  Rows of a request share its block table with per-row
+        lens, so the paged-decode kernel verifies all k+1 positions causally
+        in one forward; the longest draft prefix matching the model's own
+        argmax is emitted plus the bonus token.  Greedy-exactness: emitted
+        tokens are argmaxes of true-context logits, identical to plain
+        decode.  Sampled (temperature>0) requests ride along with 0 drafts."""
+        dev = self.device
+        rows_tok: List[int] = []
+        rows_pos: List[int] = []
+        rows_slot: List[int] = []
+        rows_temp: List[float] = []
+        plan: List[tuple] = []  # (request, base_row, drafts)
+        for r in self.running:
+            cur = r.out_tokens[-1]
+            drafts = []
+            if r.temperature <= 0:
+                drafts = self._propose(r)[: self.spec_tokens]
+            if drafts:
+                fit = self._ensure_blocks_ahead(r, len(drafts) + 1)
+                drafts = drafts[: max(0, fit - 1)]
+            plan.append((r, len(rows_tok), drafts))
+            for j, t in enumerate([cur] + drafts):
+                rows_tok.append(t)
+                rows_pos.append(r.pos + j)
+                rows_slot.append(r.slot)
+                rows_temp.append(r.temperature)
+            self.spec_proposed += len(drafts)
+        toks = torch.tensor(rows_tok, dtype=torch.long, device=dev)
+        pos = torch.tensor(rows_pos, dtype=torch.int32, device=dev)
+        bt = self.bt_d[torch.tensor(rows_slot, dtype=torch.long, device=dev)]
+        lens = pos + 1
+        blks = bt.gather(1, (pos // BLOCK).long().unsqueeze(1))[:, 0].long()
+        offs = (pos % BLOCK).long()
+
+        def kv_append(li, k, v):
+            # rows of one request write DISTINCT (block, off) slots
+            self.cache_k[li][blks, :, offs] = k[:, 0].to(self.cache_k.dtype)
+            self.cache_v[li][blks, :, offs] = v[:, 0].to(self.cache_v.dtype)
+
+        def kv_attend(li, q):
+            return OF.paged_decode(q, self.cache_k[li], self.cache_v[li],
+                                   bt, lens, BLOCK)
+
+        logits = self.model.decode_step(toks, pos, kv_append, kv_attend)
+        temps = torch.tensor(rows_temp, dtype=torch.float32)
+        out = self._sample_rows(logits, temps, self.top_p).cpu()
+        for r, base, drafts in plan:
+            emitted: List[int] = []
+            for j in range(len(drafts) + 1):
+                t = int(out[base + j])
+                emitted.append(t)
+                if j < len(drafts) and drafts[j] == t and t != self.eos_id:
+                    continue  # draft j verified; its row's logits are valid
+                break
+            self.spec_accepted += len(emitted) - 1
+            r.pos += len(emitted)
+            for t in emitted:
+                self._append_token(r, t)
+                if r.done:
+                    break
+
     def close(self):
         """Release the captured graph while the HIP runtime is still alive.
         A CUDAGraph destroyed during interpreter teardown (after the runtime)
@@ -450,7 +544,10 @@ class LlamaEngine:
             self.running = [r for r in self.running if r.slot >= 0]
             if not self.running:
                 return done_now
-            self._decode_batch()
+            if self.spec_tokens > 0:
+                self._decode_batch_spec()
+            else:
+                self._decode_batch()
             still = []
             for r in self.running:
                 if r.done:

@@ -90,3 +90,32 @@ def test_engine_fp8_kv_generates():
         eng.close()
     same = sum(a == b for a, b in zip(outs["bf16"], outs["fp8"]))
     assert same >= len(outs["bf16"]) // 2, outs  # fp8 rounding may diverge
+
+
+@requires_gpu
+def test_spec_decode_matches_plain_on_gpu():
+    """Ngram speculation on the HIP paged-decode kernel (expanded rows with
+    per-row lens) emits token-identical greedy output."""
+    from modal_examples_amd.models.llama.engine import LlamaEngine
+    from modal_examples_amd.models.llama.model import LlamaConfig
+
+    def run(spec):
+        torch.manual_seed(0)
+        eng = LlamaEngine(LlamaConfig.small(), device="cuda",
+                          dtype=torch.bfloat16, use_graph=False,
+                          kv_blocks=256, eos_id=-1, spec_tokens=spec)
+        eng.add_request([3, 4, 5, 3, 4, 5, 3, 4], max_new_tokens=10,
+                        temperature=0.0)
+        eng.add_request(list(range(20, 31)), max_new_tokens=10,
+                        temperature=0.0)
+        steps = 0
+        while eng.has_work:
+            eng.step()
+            steps += 1
+        outs = [eng.finished[i].out_tokens for i in sorted(eng.finished)]
+        return outs, steps, eng.spec_accepted
+
+    plain, _, _ = run(0)
+    spec, _, accepted = run(4)
+    assert spec == plain
+    assert accepted >= 0
