@@ -9,7 +9,10 @@
 # boots in ~a second); `fast_boot=False` captures the full-width decode
 # graph up front (slower boot, fastest tokens).  The `modal.parameter`
 # makes each setting its own autoscaling pool; the entrypoint measures both
-# boot-to-first-token and per-token latency for each.
+# boot-to-first-token and per-token latency for each.  `spec_tokens` turns
+# on ngram speculative decoding (vllm_inference.py:195-202's
+# speculative-config role): prompt-lookup drafts verified in ONE expanded
+# paged-decode forward — greedy outputs stay token-identical.
 
 import time
 
@@ -21,6 +24,7 @@ app = modal.App("example-llm-low-latency")
 @app.cls(gpu="mi355x", timeout=600, scaledown_window=60)
 class LLM:
     fast_boot: bool = modal.parameter(default=True)
+    spec_tokens: int = modal.parameter(default=0)  # ngram speculation depth
 
     @modal.enter()
     def boot(self):
@@ -36,7 +40,8 @@ class LLM:
         eng = LlamaEngine(cfg, device="cuda" if gpu else "cpu",
                           dtype=torch.bfloat16 if gpu else torch.float32,
                           use_graph=gpu and not self.fast_boot,
-                          kv_blocks=None if gpu else 128)
+                          kv_blocks=None if gpu else 128,
+                          spec_tokens=self.spec_tokens)
         if not self.fast_boot:
             eng.warmup()  # capture the decode graph now, not on request 1
         self.server = LLMServer(eng, model_name="low-latency")
@@ -60,4 +65,8 @@ def main():
           f"{fast['ms_per_token']} ms/token (eager)")
     print(f"fast_boot=False: boot {slow['boot_s']}s, "
           f"{slow['ms_per_token']} ms/token (hipGraph)")
+    spec = LLM(fast_boot=True, spec_tokens=4).timed_chat.remote(
+        "repeat repeat repeat repeat")
+    print(f"spec_tokens=4  : boot {spec['boot_s']}s, "
+          f"{spec['ms_per_token']} ms/token (ngram speculation)")
     assert fast["boot_s"] <= slow["boot_s"] + 5.0
