@@ -40,6 +40,7 @@ class Request:
     out_tokens: List[int] = field(default_factory=list)
     blocks: List[int] = field(default_factory=list)
     pos: int = 0  # tokens stored in cache
+    pf_done: int = 0  # prefilled tokens (chunked-prefill progress)
     slot: int = -1
     done: bool = False
     error: Optional[str] = None
@@ -54,7 +55,8 @@ class LlamaEngine:
                  kv_blocks: Optional[int] = None, use_graph: bool = True,
                  eos_id: int = 2, seed: int = 0, top_p: float = 1.0,
                  kv_dtype: str = "bf16", init_weights: bool = True,
-                 tp=None, spec_tokens: int = 0):
+                 tp=None, spec_tokens: int = 0,
+                 chunked_prefill: int = 0):
         """tp: optional parallel.tp.TPGroup — head-sharded tensor parallelism
         (vllm_inference.py:180 --tensor-parallel-size role).  Every rank runs
         the same engine loop on identical requests; the KV cache holds only
@@ -65,12 +67,20 @@ class LlamaEngine:
         spec_tokens>0: ngram speculative decoding (vllm_inference.py:195-202
         role) — up to k draft tokens per request verified in ONE expanded
         decode forward (models/llama/spec.py); greedy requests only, exact
-        same tokens as plain decode by construction."""
+        same tokens as plain decode by construction.
+
+        chunked_prefill>0: prompts longer than this prefill one chunk per
+        engine step (the --chunked-prefill-size role of deepseek_v4.py:102 /
+        very_large_models.py:169) so a long prompt cannot stall decode of
+        the running batch; chunks attend the cached prefix via per-row lens
+        on the paged decode kernel."""
         self.cfg = cfg or LlamaConfig.llama3_8b()
         self.tp = tp
         self.spec_tokens = spec_tokens
         self.spec_proposed = 0   # drafts offered
         self.spec_accepted = 0   # drafts accepted (emitted without a step)
+        self.chunked_prefill = chunked_prefill
+        self.prefilling: List[Request] = []  # long prompts mid-chunk
         self.device = torch.device(device)
         self.dtype = dtype
         self.max_batch = max_batch
@@ -210,6 +220,7 @@ class LlamaEngine:
     def _release(self, r: Request):
         self.free_blocks.extend(r.blocks)
         r.blocks = []
+        r.pf_done = 0
         if r.slot >= 0:
             i = r.slot
             self._slots[i] = None
@@ -343,6 +354,61 @@ class LlamaEngine:
                                    bt, lens, BLOCK)
 
         return self.model.decode_step(toks, pos, kv_append, kv_attend)
+
+    # ------------------------------------------------ chunked prefill
+
+    @torch.no_grad()
+    def _prefill_chunk(self, r: Request) -> bool:
+        """Advance one chunk of r's prompt into the cache; True when the
+        prefill completed (first token sampled, slot armed for decode).
+
+        Rows pos..pos+C-1 share r's block table with per-row lens, so the
+        paged decode kernel gives each chunk token exactly-causal attention
+        over the cached prefix + earlier in-chunk tokens.  bt_d/active_d
+        stay zero until completion so graph-mode decode keeps treating the
+        slot as inactive (its pad writes go to block 0)."""
+        dev = self.device
+        feed = self._feed(r)
+        start = r.pf_done
+        end = min(start + self.chunked_prefill, len(feed))
+        toks = torch.tensor(feed[start:end], dtype=torch.long, device=dev)
+        pos = torch.arange(start, end, dtype=torch.int32, device=dev)
+        bt_row = torch.zeros(self.max_blocks_per_seq, dtype=torch.int32,
+                             device=dev)
+        bt_row[: len(r.blocks)] = torch.tensor(r.blocks, dtype=torch.int32,
+                                               device=dev)
+        bt = bt_row.unsqueeze(0).expand(end - start, -1)
+        lens = pos + 1
+        blks = bt_row[(pos // BLOCK).long()].long()
+        offs = (pos % BLOCK).long()
+
+        def kv_append(li, k, v):
+            self.cache_k[li][blks, :, offs] = k[:, 0].to(self.cache_k.dtype)
+            self.cache_v[li][blks, :, offs] = v[:, 0].to(self.cache_v.dtype)
+
+        def kv_attend(li, q):
+            return OF.paged_decode(q, self.cache_k[li], self.cache_v[li],
+                                   bt, lens, BLOCK)
+
+        logits = self.model.decode_step(toks, pos, kv_append, kv_attend)
+        r.pf_done = end
+        r.pos = end
+        if end < len(feed):
+            return False
+        temps = torch.tensor([r.temperature])
+        first = self._sample_rows(logits[-1:], temps, self.top_p).cpu()
+        self._append_token(r, int(first[0]))
+        if r.t_first_token is None:
+            r.t_first_token = time.monotonic()
+        slot = r.slot
+        self.toks_d[slot] = int(r.out_tokens[-1])
+        self.pos_d[slot] = r.pos
+        self.lens_d[slot] = r.pos + 1
+        self.bt_d[slot, : len(r.blocks)] = torch.tensor(
+            r.blocks, device=dev, dtype=torch.int32)
+        self.active_d[slot] = 1
+        self.temps_d[slot] = r.temperature
+        return True
 
     # ------------------------------------------------ speculative decode
 
@@ -509,6 +575,14 @@ class LlamaEngine:
                 break  # no KV blocks / slots free — keep waiting
             self.waiting.pop(0)
             admitted.append(r)
+        # long prompts go to the chunked-prefill lane (one chunk per step,
+        # interleaved with decode) instead of a monolithic forward
+        if self.chunked_prefill > 0:
+            longs = [r for r in admitted
+                     if len(self._feed(r)) > self.chunked_prefill]
+            if longs:
+                admitted = [r for r in admitted if r not in longs]
+                self.prefilling.extend(longs)
         # ragged prefill: sort by length and bucket so right-padding waste
         # stays <= ~30% — one forward per bucket, mixed lengths welcome
         admitted.sort(key=lambda r: len(self._feed(r)))
@@ -532,6 +606,16 @@ class LlamaEngine:
                     self.running.append(r)
 
         done_now = []
+        # advance each mid-prefill prompt by ONE chunk (bounds the stall any
+        # single long prompt can impose on the running batch)
+        for r in list(self.prefilling):
+            if self._prefill_chunk(r):
+                self.prefilling.remove(r)
+                if r.done:
+                    self._retire(r)
+                    done_now.append(r)
+                else:
+                    self.running.append(r)
         if self.running:
             for r in self.running:
                 if not self._ensure_blocks(r):
@@ -576,4 +660,4 @@ class LlamaEngine:
 
     @property
     def has_work(self) -> bool:
-        return bool(self.waiting or self.running)
+        return bool(self.waiting or self.running or self.prefilling)

@@ -119,3 +119,25 @@ def test_spec_decode_matches_plain_on_gpu():
     spec, _, accepted = run(4)
     assert spec == plain
     assert accepted >= 0
+
+
+@requires_gpu
+def test_chunked_prefill_matches_on_gpu():
+    """Chunked prefill through the HIP paged-decode kernel == monolithic
+    FA prefill, token for token."""
+    from modal_examples_amd.models.llama.engine import LlamaEngine
+    from modal_examples_amd.models.llama.model import LlamaConfig
+
+    def run(chunk):
+        torch.manual_seed(0)
+        eng = LlamaEngine(LlamaConfig.small(), device="cuda",
+                          dtype=torch.bfloat16, use_graph=False,
+                          kv_blocks=256, eos_id=-1, chunked_prefill=chunk)
+        g = torch.Generator().manual_seed(2)
+        eng.add_request(torch.randint(0, 1024, (37,), generator=g).tolist(),
+                        max_new_tokens=6, temperature=0.0)
+        while eng.has_work:
+            eng.step()
+        return eng.finished[1].out_tokens
+
+    assert run(8) == run(0)
