@@ -377,7 +377,8 @@ class LlamaEngine:
                              device=dev)
         bt_row[: len(r.blocks)] = torch.tensor(r.blocks, dtype=torch.int32,
                                                device=dev)
-        bt = bt_row.unsqueeze(0).expand(end - start, -1)
+        # materialized (not expand()ed): the HIP kernel reads bt row-strided
+        bt = bt_row.unsqueeze(0).repeat(end - start, 1)
         lens = pos + 1
         blks = bt_row[(pos // BLOCK).long()].long()
         offs = (pos % BLOCK).long()

@@ -86,8 +86,11 @@ def paged_decode(q, k_cache, v_cache, block_table, seq_lens, block_size: int = 0
             v_cache = v_cache.to(torch.float32)
         return ref.paged_decode_ref(q, k_cache, v_cache, block_table, seq_lens,
                                     block_size, scale)
-    return ext.paged_decode(q.contiguous(), k_cache, v_cache, block_table,
-                            seq_lens.int(), block_size, scale)
+    # .contiguous() also materializes expand()ed (stride-0) block tables —
+    # the kernel indexes bt + row*stride and would read out of bounds
+    return ext.paged_decode(q.contiguous(), k_cache, v_cache,
+                            block_table.contiguous(), seq_lens.int(),
+                            block_size, scale)
 
 
 def groupnorm_silu(x, gamma, beta, groups: int = 32, eps: float = 1e-5,
