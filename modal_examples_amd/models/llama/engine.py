@@ -56,7 +56,8 @@ class LlamaEngine:
                  eos_id: int = 2, seed: int = 0, top_p: float = 1.0,
                  kv_dtype: str = "bf16", init_weights: bool = True,
                  tp=None, spec_tokens: int = 0,
-                 chunked_prefill: int = 0):
+                 chunked_prefill: int = 0,
+                 prefix_cache: bool = False):
         """tp: optional parallel.tp.TPGroup — head-sharded tensor parallelism
         (vllm_inference.py:180 --tensor-parallel-size role).  Every rank runs
         the same engine loop on identical requests; the KV cache holds only
@@ -73,7 +74,14 @@ class LlamaEngine:
         engine step (the --chunked-prefill-size role of deepseek_v4.py:102 /
         very_large_models.py:169) so a long prompt cannot stall decode of
         the running batch; chunks attend the cached prefix via per-row lens
-        on the paged decode kernel."""
+        on the paged decode kernel.
+
+        prefix_cache=True: block-level KV prefix caching (the RadixAttention
+        /unified-radix-tree role of inkling_small.py:99 and vLLM's automatic
+        prefix caching) — FULL prompt blocks are chain-hashed; a request
+        whose prompt prefix matches cached blocks shares them (refcounted,
+        never written) and prefills only the suffix.  Shared system prompts
+        cost their KV compute once."""
         self.cfg = cfg or LlamaConfig.llama3_8b()
         self.tp = tp
         self.spec_tokens = spec_tokens
@@ -81,6 +89,13 @@ class LlamaEngine:
         self.spec_accepted = 0   # drafts accepted (emitted without a step)
         self.chunked_prefill = chunked_prefill
         self.prefilling: List[Request] = []  # long prompts mid-chunk
+        self.prefix_cache = prefix_cache
+        self._pc_map: Dict = {}    # chain-hash -> block id
+        self._pc_hash: Dict = {}   # block id -> chain-hash (cached blocks)
+        self._pc_refs: Dict = {}   # block id -> live references
+        self._pc_lru: Dict = {}    # zero-ref cached blocks, LRU order
+        self.prefix_hit_tokens = 0
+        self.prefix_lookup_tokens = 0
         self.device = torch.device(device)
         self.dtype = dtype
         self.max_batch = max_batch
@@ -205,9 +220,53 @@ class LlamaEngine:
         return r.req_id
 
     def _alloc_blocks(self, n: int) -> Optional[List[int]]:
+        if len(self.free_blocks) < n and self._pc_lru:
+            self._pc_evict(n - len(self.free_blocks))
         if len(self.free_blocks) < n:
             return None
         return [self.free_blocks.pop() for _ in range(n)]
+
+    # ---- block-level prefix cache (refcounted, full prompt blocks only) ----
+
+    def _pc_chain_hashes(self, feed: List[int]) -> List[int]:
+        hs, h = [], 0
+        for b in range(len(feed) // BLOCK):
+            h = hash((h, tuple(feed[b * BLOCK:(b + 1) * BLOCK])))
+            hs.append(h)
+        return hs
+
+    def _pc_release_block(self, bid: int):
+        n = self._pc_refs.get(bid, 0) - 1
+        if n <= 0:
+            self._pc_refs.pop(bid, None)
+            self._pc_lru[bid] = True  # evictable (insertion order = LRU)
+        else:
+            self._pc_refs[bid] = n
+
+    def _pc_evict(self, n: int) -> int:
+        freed = 0
+        while freed < n and self._pc_lru:
+            bid = next(iter(self._pc_lru))
+            del self._pc_lru[bid]
+            h = self._pc_hash.pop(bid, None)
+            if h is not None:
+                self._pc_map.pop(h, None)
+            self.free_blocks.append(bid)
+            freed += 1
+        return freed
+
+    def _pc_register(self, r: Request):
+        """Cache r's freshly-computed FULL blocks (called at prefill end)."""
+        if not self.prefix_cache:
+            return
+        cached = self._feed(r)[: r.pos]
+        for i, h in enumerate(self._pc_chain_hashes(cached)):
+            bid = r.blocks[i]
+            if bid in self._pc_hash or h in self._pc_map:
+                continue  # shared-in block, or same content cached elsewhere
+            self._pc_map[h] = bid
+            self._pc_hash[bid] = h
+            self._pc_refs[bid] = self._pc_refs.get(bid, 0) + 1
 
     def _take_slot(self, r: Request) -> bool:
         for i, s in enumerate(self._slots):
@@ -218,7 +277,11 @@ class LlamaEngine:
         return False
 
     def _release(self, r: Request):
-        self.free_blocks.extend(r.blocks)
+        for bid in r.blocks:
+            if bid in self._pc_hash:
+                self._pc_release_block(bid)
+            else:
+                self.free_blocks.append(bid)
         r.blocks = []
         r.pf_done = 0
         if r.slot >= 0:
@@ -280,6 +343,7 @@ class LlamaEngine:
             self.bt_d[slot, : len(r.blocks)] = bt
             self.active_d[slot] = 1
             self.temps_d[slot] = r.temperature
+            self._pc_register(r)
 
     @staticmethod
     def _feed(r: Request) -> List[int]:
@@ -288,17 +352,37 @@ class LlamaEngine:
         return r.prompt + r.out_tokens
 
     def _admit(self, r: Request) -> bool:
-        """Reserve blocks + a slot (no compute)."""
-        L = len(self._feed(r))
+        """Reserve blocks + a slot (no compute).  With prefix_cache on,
+        leading prompt blocks whose chain hash is cached are SHARED
+        (refcounted before any allocation so eviction can't race them) and
+        the request prefills only its suffix."""
+        feed = self._feed(r)
+        L = len(feed)
+        shared: List[int] = []
+        if self.prefix_cache:
+            hs = self._pc_chain_hashes(feed)
+            if hs and len(hs) * BLOCK == L:
+                hs = hs[:-1]  # always compute >= 1 token (the logits source)
+            for h in hs:
+                bid = self._pc_map.get(h)
+                if bid is None:
+                    break
+                shared.append(bid)
+                self._pc_refs[bid] = self._pc_refs.get(bid, 0) + 1
+                self._pc_lru.pop(bid, None)
+            self.prefix_lookup_tokens += L
         # +1 spare block of headroom, but never beyond the per-slot table
         nblk = min((L + BLOCK) // BLOCK + 1, self.max_blocks_per_seq)
-        blocks = self._alloc_blocks(nblk)
-        if blocks is None:
+        fresh = self._alloc_blocks(nblk - len(shared))
+        if fresh is None or not self._take_slot(r):
+            if fresh is not None:
+                self.free_blocks.extend(fresh)
+            for bid in shared:
+                self._pc_release_block(bid)
             return False
-        if not self._take_slot(r):
-            self.free_blocks.extend(blocks)
-            return False
-        r.blocks = blocks
+        r.blocks = shared + fresh
+        r.pf_done = len(shared) * BLOCK  # suffix-only prefill start
+        self.prefix_hit_tokens += r.pf_done
         return True
 
     def _append_token(self, r: Request, tok: int):
@@ -370,7 +454,8 @@ class LlamaEngine:
         dev = self.device
         feed = self._feed(r)
         start = r.pf_done
-        end = min(start + self.chunked_prefill, len(feed))
+        step_w = self.chunked_prefill or (len(feed) - start)
+        end = min(start + step_w, len(feed))
         toks = torch.tensor(feed[start:end], dtype=torch.long, device=dev)
         pos = torch.arange(start, end, dtype=torch.int32, device=dev)
         bt_row = torch.zeros(self.max_blocks_per_seq, dtype=torch.int32,
@@ -409,6 +494,7 @@ class LlamaEngine:
             r.blocks, device=dev, dtype=torch.int32)
         self.active_d[slot] = 1
         self.temps_d[slot] = r.temperature
+        self._pc_register(r)
         return True
 
     # ------------------------------------------------ speculative decode
@@ -577,13 +663,14 @@ class LlamaEngine:
             self.waiting.pop(0)
             admitted.append(r)
         # long prompts go to the chunked-prefill lane (one chunk per step,
-        # interleaved with decode) instead of a monolithic forward
-        if self.chunked_prefill > 0:
-            longs = [r for r in admitted
-                     if len(self._feed(r)) > self.chunked_prefill]
-            if longs:
-                admitted = [r for r in admitted if r not in longs]
-                self.prefilling.extend(longs)
+        # interleaved with decode) instead of a monolithic forward; prefix-
+        # cache hits (pf_done>0) use the same lane to prefill ONLY the suffix
+        longs = [r for r in admitted if r.pf_done > 0
+                 or (self.chunked_prefill > 0
+                     and len(self._feed(r)) > self.chunked_prefill)]
+        if longs:
+            admitted = [r for r in admitted if r not in longs]
+            self.prefilling.extend(longs)
         # ragged prefill: sort by length and bucket so right-padding waste
         # stays <= ~30% — one forward per bucket, mixed lengths welcome
         admitted.sort(key=lambda r: len(self._feed(r)))

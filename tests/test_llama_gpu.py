@@ -141,3 +141,31 @@ def test_chunked_prefill_matches_on_gpu():
         return eng.finished[1].out_tokens
 
     assert run(8) == run(0)
+
+
+@requires_gpu
+def test_prefix_cache_matches_on_gpu():
+    """Prefix-cached suffix prefill through the HIP kernels == cold run."""
+    from modal_examples_amd.models.llama.engine import BLOCK, LlamaEngine
+    from modal_examples_amd.models.llama.model import LlamaConfig
+
+    def run(pc):
+        torch.manual_seed(0)
+        eng = LlamaEngine(LlamaConfig.small(), device="cuda",
+                          dtype=torch.bfloat16, use_graph=False,
+                          kv_blocks=256, eos_id=-1, prefix_cache=pc)
+        g = torch.Generator().manual_seed(6)
+        system = torch.randint(0, 1024, (3 * BLOCK,), generator=g).tolist()
+        outs = []
+        for tail in ([7, 8, 9], [11, 12]):
+            rid = eng.add_request(system + tail, max_new_tokens=6,
+                                  temperature=0.0)
+            while eng.has_work:
+                eng.step()
+            outs.append(eng.finished[rid].out_tokens)
+        return outs, eng.prefix_hit_tokens
+
+    cold, _ = run(False)
+    warm, hits = run(True)
+    assert warm == cold
+    assert hits == 3 * BLOCK

@@ -1,0 +1,99 @@
+"""Block-level KV prefix caching (RadixAttention/automatic-prefix-caching
+role, inkling_small.py:99): shared prompt prefixes reuse cached blocks,
+prefill touches only the suffix, outputs stay exactly equal."""
+import torch
+
+from modal_examples_amd.models.llama.engine import BLOCK, LlamaEngine
+from modal_examples_amd.models.llama.model import LlamaConfig
+
+
+def _mk(pc=False, blocks=256):
+    return LlamaEngine(LlamaConfig.small(), device="cpu",
+                       dtype=torch.bfloat16, use_graph=False, eos_id=-1,
+                       seed=0, prefix_cache=pc, kv_blocks=blocks)
+
+
+def _run_one(eng, prompt, n=5):
+    rid = eng.add_request(prompt, max_new_tokens=n, temperature=0.0)
+    while eng.has_work:
+        eng.step()
+    return eng.finished[rid].out_tokens
+
+
+def test_prefix_hit_reuses_blocks_and_matches():
+    g = torch.Generator().manual_seed(6)
+    system = torch.randint(0, 1024, (3 * BLOCK,), generator=g).tolist()
+    q1 = system + [7, 8, 9]
+    q2 = system + [11, 12, 13, 14]
+
+    plain = _mk(False)
+    want1, want2 = _run_one(plain, q1), _run_one(plain, q2)
+
+    eng = _mk(True)
+    assert _run_one(eng, q1) == want1
+    assert eng.prefix_hit_tokens == 0  # first request is the cold fill
+    assert _run_one(eng, q2) == want2
+    assert eng.prefix_hit_tokens == 3 * BLOCK  # whole system prompt reused
+
+
+def test_prefix_cache_exact_multiple_of_block():
+    """L % BLOCK == 0: the last block is recomputed (logits need >=1 token),
+    earlier blocks still shared; output identical."""
+    g = torch.Generator().manual_seed(8)
+    p = torch.randint(0, 1024, (2 * BLOCK,), generator=g).tolist()
+    want = _run_one(_mk(False), p)
+    eng = _mk(True)
+    assert _run_one(eng, p) == want
+    assert _run_one(eng, p) == want
+    assert eng.prefix_hit_tokens == BLOCK  # blocks-1 shared on the rerun
+
+
+def test_shared_blocks_are_actually_shared():
+    g = torch.Generator().manual_seed(9)
+    system = torch.randint(0, 1024, (2 * BLOCK,), generator=g).tolist()
+    eng = _mk(True)
+    r1 = eng.add_request(system + [1], max_new_tokens=30, temperature=0.0)
+    eng.step()  # r1 prefills and registers its blocks
+    r2 = eng.add_request(system + [2], max_new_tokens=30, temperature=0.0)
+    eng.step()
+    a = eng.finished.get(r1) or next(r for r in eng.running if r.req_id == r1)
+    b = eng.finished.get(r2) or next(r for r in eng.running if r.req_id == r2)
+    assert a.blocks[:2] == b.blocks[:2]  # same physical blocks
+    assert a.blocks[2:] != b.blocks[2:]
+    while eng.has_work:
+        eng.step()
+
+
+def test_eviction_recycles_cached_blocks_under_pressure():
+    """With a tiny pool, unreferenced cached blocks are evicted so new
+    requests still run (and produce the right tokens)."""
+    g = torch.Generator().manual_seed(10)
+    eng = _mk(True, blocks=16)
+    plain = _mk(False, blocks=16)
+    for i in range(6):
+        p = torch.randint(0, 1024, (2 * BLOCK + i,), generator=g).tolist()
+        g2 = torch.Generator().manual_seed(10)
+        torch.randint(0, 1024, (2 * BLOCK + i,), generator=g2)
+        want = _run_one(plain, p, n=3)
+        assert _run_one(eng, p, n=3) == want, i
+
+
+def test_all_engine_features_compose():
+    """prefix_cache + chunked_prefill + spec_tokens together still produce
+    exactly the plain engine's greedy output."""
+    g = torch.Generator().manual_seed(12)
+    system = torch.randint(0, 1024, (2 * BLOCK + 5,), generator=g).tolist()
+    prompts = [system + [3, 4, 5, 3, 4, 5], system + [9], system]
+    plain = _mk(False)
+    want = []
+    for p in prompts:
+        want.append(_run_one(plain, p, n=6))
+    eng = LlamaEngine(LlamaConfig.small(), device="cpu",
+                      dtype=torch.bfloat16, use_graph=False, eos_id=-1,
+                      seed=0, prefix_cache=True, chunked_prefill=8,
+                      spec_tokens=3, kv_blocks=256)
+    got = []
+    for p in prompts:
+        got.append(_run_one(eng, p, n=6))
+    assert got == want
+    assert eng.prefix_hit_tokens > 0
