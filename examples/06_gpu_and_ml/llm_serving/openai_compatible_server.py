@@ -26,12 +26,21 @@ class LLMService:
         from modal_examples_amd.models.llama.model import LlamaConfig
         from modal_examples_amd.models.llama.server import LLMServer
 
+        import os
+
         gpu = torch.cuda.is_available()
         cfg = LlamaConfig.llama3_8b() if gpu else LlamaConfig.small()
+        # serving knobs (the vLLM/sglang flag roles): ngram speculation,
+        # chunked prefill, prefix caching — all exact-output-preserving
         eng = LlamaEngine(cfg, device="cuda" if gpu else "cpu",
                           dtype=torch.bfloat16 if gpu else torch.float32,
                           use_graph=gpu,
-                          kv_blocks=None if gpu else 128)
+                          kv_blocks=None if gpu else 128,
+                          spec_tokens=int(os.environ.get("SPEC_TOKENS", "0")),
+                          chunked_prefill=int(
+                              os.environ.get("CHUNKED_PREFILL", "2048")),
+                          prefix_cache=os.environ.get(
+                              "PREFIX_CACHE", "1") == "1")
         self.server = LLMServer(eng, model_name="llama-3-8b-mi355x")
 
     @modal.web_server(port=PORT, startup_timeout=300)
