@@ -200,6 +200,13 @@ class LlamaEngine:
         if "exc" in box:
             raise box["exc"]
         eng = box["eng"]
+        if eng.tp is not None and eng.tp.world > 1:
+            # baked checkpoints are FULL: carve this rank's TP shard from
+            # the loaded blob (device-side slicing; shards stay on device)
+            from .model import shard_llama_state
+
+            sd = shard_llama_state(sd, eng.cfg, eng.tp.rank, eng.tp.world)
+            sd = {k: v.contiguous() for k, v in sd.items()}
         eng.model.load_state_dict(sd, assign=True)
         return eng
 

@@ -97,3 +97,35 @@ def test_all_engine_features_compose():
         got.append(_run_one(eng, p, n=6))
     assert got == want
     assert eng.prefix_hit_tokens > 0
+
+
+def test_stress_pressure_mixed_features_all_requests_finish():
+    """Allocator soak: tiny pool, all features on, preemption + eviction +
+    sharing interleave — every request must finish with the right length
+    and the pool must be fully recovered at the end."""
+    g = torch.Generator().manual_seed(13)
+    eng = LlamaEngine(LlamaConfig.small(), device="cpu",
+                      dtype=torch.bfloat16, use_graph=False, eos_id=-1,
+                      seed=0, prefix_cache=True, chunked_prefill=8,
+                      spec_tokens=2, kv_blocks=24, max_batch=4)
+    base = torch.randint(0, 1024, (BLOCK,), generator=g).tolist()
+    want_len = {}
+    for i in range(10):
+        tail = torch.randint(0, 1024, (1 + i % 5,), generator=g).tolist()
+        n = 3 + i % 4
+        rid = eng.add_request(base + tail, max_new_tokens=n,
+                              temperature=0.0 if i % 3 else 0.7)
+        want_len[rid] = n
+    steps = 0
+    while eng.has_work:
+        eng.step()
+        steps += 1
+        assert steps < 2000, "scheduler livelock"
+    assert len(eng.finished) == 10
+    for rid, n in want_len.items():
+        r = eng.finished[rid]
+        assert r.error is None and len(r.out_tokens) == n, (rid, r.error)
+    # every block is either free or held by the prefix cache — none leaked
+    cached = len(eng._pc_hash)
+    assert len(eng.free_blocks) + cached == eng.num_blocks - 1  # -1 pad
+    assert set(eng.free_blocks).isdisjoint(eng._pc_hash.keys())

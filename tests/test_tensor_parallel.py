@@ -316,3 +316,57 @@ def test_llama_tp_engine_world2_continuous_batching():
             assert p.exitcode == 0, f"tp engine worker exit {p.exitcode}"
         assert results["match-0"] is True and results["match-1"] is True
         assert results["toks-0"] == results["toks-1"]
+
+
+def _tp_coldboot_worker(rank, world, port, results):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world), "LOCAL_RANK": str(rank),
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    import tempfile
+
+    import torch.distributed as dist
+
+    from modal_examples_amd.gpu import fastload
+    from modal_examples_amd.models.llama.engine import LlamaEngine
+    from modal_examples_amd.models.llama.model import LlamaConfig, LlamaModel
+    from modal_examples_amd.parallel.tp import TPGroup
+
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        cfg = LlamaConfig.small()
+        torch.manual_seed(0)
+        full = LlamaModel(cfg).to(torch.bfloat16)
+        path = os.path.join(tempfile.gettempdir(), f"tpboot-{port}.safetensors")
+        if rank == 0:
+            fastload.save_file(dict(full.state_dict()), path)
+        dist.barrier()
+        eng = LlamaEngine.from_safetensors(
+            path, cfg=cfg, device="cpu", dtype=torch.bfloat16,
+            use_graph=False, tp=TPGroup())
+        toks = torch.randint(0, cfg.vocab_size, (1, 8),
+                             generator=torch.Generator().manual_seed(2))
+        results[f"err-{rank}"] = float(
+            (full.prefill(toks) - eng.model.prefill(toks)).abs().max())
+    finally:
+        dist.destroy_process_group()
+
+
+def test_tp_cold_boot_shards_baked_checkpoint():
+    """from_safetensors(tp=...) carves each rank's shard from the FULL
+    baked file; restored TP engine matches the full model."""
+    import multiprocessing as mp
+
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as mgr:
+        results = mgr.dict()
+        port = _free_port()
+        procs = [ctx.Process(target=_tp_coldboot_worker,
+                             args=(r, 2, port, results)) for r in range(2)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=240)
+            assert p.exitcode == 0, f"tp coldboot worker exit {p.exitcode}"
+        for r in range(2):
+            assert results[f"err-{r}"] < 0.05, results[f"err-{r}"]
