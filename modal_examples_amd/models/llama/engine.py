@@ -57,7 +57,8 @@ class LlamaEngine:
                  kv_dtype: str = "bf16", init_weights: bool = True,
                  tp=None, spec_tokens: int = 0,
                  chunked_prefill: int = 0,
-                 prefix_cache: bool = False):
+                 prefix_cache: bool = False,
+                 gpu_mem_util: float = 0.6):
         """tp: optional parallel.tp.TPGroup — head-sharded tensor parallelism
         (vllm_inference.py:180 --tensor-parallel-size role).  Every rank runs
         the same engine loop on identical requests; the KV cache holds only
@@ -81,7 +82,11 @@ class LlamaEngine:
         prefix caching) — FULL prompt blocks are chain-hashed; a request
         whose prompt prefix matches cached blocks shares them (refcounted,
         never written) and prefills only the suffix.  Shared system prompts
-        cost their KV compute once."""
+        cost their KV compute once.
+
+        gpu_mem_util: fraction of free HBM the KV pool claims when
+        kv_blocks is unset (--gpu-memory-utilization / --mem-fraction-static
+        role, lfm_snapshot.py:316 / deepseek_v4_flash.py:255)."""
         self.cfg = cfg or LlamaConfig.llama3_8b()
         self.tp = tp
         self.spec_tokens = spec_tokens
@@ -133,7 +138,7 @@ class LlamaEngine:
                 kv_elt = 1 if self.kv_dtype == torch.float8_e4m3fn else 2
                 per_block = (c.n_layers * self.model.blocks[0].nkv * BLOCK
                              * c.head_dim * 2 * kv_elt)
-                kv_blocks = max(64, int(free * 0.6 / per_block))
+                kv_blocks = max(64, int(free * gpu_mem_util / per_block))
             else:
                 kv_blocks = 256
         self.num_blocks = kv_blocks
