@@ -161,6 +161,12 @@ def create_openai_app(server: LLMServer):
         M.inc("llm_preemptions_total", 0)  # ensure series exists
         if eng.preemptions:
             M.observe("llm_preemptions", float(eng.preemptions))
+        # engine-feature counters (vLLM exposes the same families)
+        M.observe("llm_spec_tokens_proposed", float(eng.spec_proposed))
+        M.observe("llm_spec_tokens_accepted", float(eng.spec_accepted))
+        M.observe("llm_prefix_cache_hit_tokens", float(eng.prefix_hit_tokens))
+        M.observe("llm_prefix_cache_lookup_tokens",
+                  float(eng.prefix_lookup_tokens))
         return PlainTextResponse(M.render_prometheus())
 
     @app.get("/v1/models")
