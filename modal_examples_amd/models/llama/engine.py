@@ -661,7 +661,8 @@ class LlamaEngine:
             r = self.waiting[0]
             if not self._admit(r):
                 need = (len(self._feed(r)) + BLOCK) // BLOCK + 1
-                if not self.running and not admitted and need > self.num_blocks - 1:
+                if (not self.running and not admitted and not self.prefilling
+                        and need > self.num_blocks - 1):
                     # can NEVER fit even in an empty cache: fail it instead of
                     # livelocking the scheduler
                     self.waiting.pop(0)
