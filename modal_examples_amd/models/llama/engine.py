@@ -744,6 +744,14 @@ class LlamaEngine:
         return done_now
 
     def _retire(self, r: Request):
+        if r.error is None and r.blocks:
+            # multi-turn reuse (radix-tree behavior): the finished request's
+            # full blocks — prompt AND generated tokens — enter the prefix
+            # cache, so a follow-up turn whose prompt replays this
+            # conversation prefills only its new text.  Registration
+            # happens BEFORE release so the blocks convert to cached
+            # (refcounted) instead of returning to the free pool.
+            self._pc_register(r)
         self._release(r)
         self.finished[r.req_id] = r
 

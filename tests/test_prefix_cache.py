@@ -129,3 +129,26 @@ def test_stress_pressure_mixed_features_all_requests_finish():
     cached = len(eng._pc_hash)
     assert len(eng.free_blocks) + cached == eng.num_blocks - 1  # -1 pad
     assert set(eng.free_blocks).isdisjoint(eng._pc_hash.keys())
+
+
+def test_multi_turn_conversation_reuse():
+    """Turn 2's prompt replays turn 1 (prompt + generated answer): the
+    conversation's blocks are cached at retire, so turn 2 prefills only the
+    new user text — and produces exactly the cold engine's tokens."""
+    g = torch.Generator().manual_seed(14)
+    turn1 = torch.randint(0, 1024, (2 * BLOCK,), generator=g).tolist()
+
+    def conversation(eng):
+        a1 = _run_one(eng, turn1, n=BLOCK)  # answer spans a full block
+        turn2 = turn1 + a1 + [101, 102, 103]
+        a2 = _run_one(eng, turn2, n=4)
+        return a1, a2
+
+    want = conversation(_mk(False))
+    eng = _mk(True)
+    got = conversation(eng)
+    assert got == want
+    # turn 1's cache holds prompt(32) + 15 generated tokens (the final
+    # emitted token's KV is never appended), so exactly the 2 full prompt
+    # blocks are reusable — and both hit
+    assert eng.prefix_hit_tokens == 2 * BLOCK, eng.prefix_hit_tokens
