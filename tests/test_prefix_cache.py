@@ -152,3 +152,34 @@ def test_multi_turn_conversation_reuse():
     # emitted token's KV is never appended), so exactly the 2 full prompt
     # blocks are reusable — and both hit
     assert eng.prefix_hit_tokens == 2 * BLOCK, eng.prefix_hit_tokens
+
+
+def test_differential_fuzz_exact_features_vs_plain():
+    """Randomized differential soak: prefix caching reuses bit-identical
+    KV and speculation verifies with the decode kernel itself, so ANY
+    greedy request mix through spec+prefix must match plain exactly.
+    (Chunked prefill is excluded here by design: its chunks sum attention
+    in a different bf16 order than the monolithic pass — same property as
+    vLLM's chunked prefill — so argmax ties may resolve differently; its
+    own tests pin exactness on fixed seeds.)"""
+    for seed in range(5):
+        g = torch.Generator().manual_seed(100 + seed)
+        prompts = []
+        base = torch.randint(0, 1024, (int(torch.randint(4, 40, (1,),
+                             generator=g)),), generator=g).tolist()
+        for i in range(4):
+            if int(torch.randint(0, 2, (1,), generator=g)) and prompts:
+                p = list(prompts[-1][: len(prompts[-1]) // 2]) + \
+                    torch.randint(0, 1024, (3,), generator=g).tolist()
+            else:
+                p = base + torch.randint(
+                    0, 1024, (int(torch.randint(1, 20, (1,), generator=g)),),
+                    generator=g).tolist()
+            prompts.append(p)
+        plain = _mk(False)
+        rich = LlamaEngine(LlamaConfig.small(), device="cpu",
+                           dtype=torch.bfloat16, use_graph=False, eos_id=-1,
+                           seed=0, prefix_cache=True,
+                           spec_tokens=3, kv_blocks=64, max_batch=3)
+        for p in prompts:
+            assert _run_one(rich, p, n=5) == _run_one(plain, p, n=5), seed
