@@ -41,6 +41,8 @@ class Request:
     blocks: List[int] = field(default_factory=list)
     pos: int = 0  # tokens stored in cache
     pf_done: int = 0  # prefilled tokens (chunked-prefill progress)
+    presence_penalty: float = 0.0
+    frequency_penalty: float = 0.0
     slot: int = -1
     done: bool = False
     error: Optional[str] = None
@@ -218,7 +220,9 @@ class LlamaEngine:
     # ------------------------------------------------ request lifecycle
 
     def add_request(self, prompt: List[int], max_new_tokens: int = 64,
-                    temperature: float = 0.0, stream_cb=None) -> int:
+                    temperature: float = 0.0, stream_cb=None,
+                    presence_penalty: float = 0.0,
+                    frequency_penalty: float = 0.0) -> int:
         limit = self.cfg.max_seq
         if len(prompt) >= limit:
             prompt = prompt[-(limit - 1):]  # keep the most recent context
@@ -226,7 +230,8 @@ class LlamaEngine:
             # a request past max_seq would overflow the per-slot block table
             max_new_tokens = max(1, limit - len(prompt))
         r = Request(self._next_id, list(prompt), max_new_tokens, temperature,
-                    stream_cb=stream_cb)
+                    stream_cb=stream_cb, presence_penalty=presence_penalty,
+                    frequency_penalty=frequency_penalty)
         self._next_id += 1
         self.waiting.append(r)
         return r.req_id
@@ -429,6 +434,24 @@ class LlamaEngine:
                             seed=self.seed ^ (0x5EED + self._step_count))
         return torch.where(temps <= 0, greedy, sampled.to(greedy.device))
 
+    def _apply_penalties(self, logits: torch.Tensor, rows) -> None:
+        """OpenAI presence/frequency penalties over the OUTPUT tokens so
+        far (vLLM semantics); `rows` maps each logits row to its Request
+        (None rows untouched).  Host loop touches only penalized rows."""
+        for i, r in enumerate(rows):
+            if r is None or not r.out_tokens or (
+                    r.presence_penalty == 0 and r.frequency_penalty == 0):
+                continue
+            counts: Dict[int, int] = {}
+            for t in r.out_tokens:
+                counts[t] = counts.get(t, 0) + 1
+            idx = torch.tensor(list(counts), dtype=torch.long,
+                               device=logits.device)
+            cnt = torch.tensor(list(counts.values()), dtype=torch.float32,
+                               device=logits.device)
+            logits[i, idx] -= (r.frequency_penalty * cnt
+                               + r.presence_penalty)
+
     # ------------------------------------------------ decode
 
     def _run_decode(self, limit: Optional[int] = None):
@@ -549,7 +572,10 @@ class LlamaEngine:
         for r in self.running:
             cur = r.out_tokens[-1]
             drafts = []
-            if r.temperature <= 0:
+            if (r.temperature <= 0 and r.presence_penalty == 0
+                    and r.frequency_penalty == 0):
+                # penalties evolve WITHIN an accepted run, so penalized
+                # requests decode unspeculated
                 drafts = self._propose(r)[: self.spec_tokens]
             if drafts:
                 fit = self._ensure_blocks_ahead(r, len(drafts) + 1)
@@ -578,6 +604,11 @@ class LlamaEngine:
                                    bt, lens, BLOCK)
 
         logits = self.model.decode_step(toks, pos, kv_append, kv_attend)
+        row_reqs = [None] * len(rows_tok)
+        for r, base, drafts in plan:
+            if r.presence_penalty or r.frequency_penalty:
+                row_reqs[base] = r  # penalized requests have exactly 1 row
+        self._apply_penalties(logits, row_reqs)
         temps = torch.tensor(rows_temp, dtype=torch.float32)
         out = self._sample_rows(logits, temps, self.top_p).cpu()
         for r, base, drafts in plan:
@@ -629,6 +660,11 @@ class LlamaEngine:
         else:
             lim = max(r.slot for r in self.running) + 1
             logits = self._run_decode(lim)
+        row_reqs = [self._slots[i] for i in range(lim)]
+        if any(r is not None and (r.presence_penalty or r.frequency_penalty)
+               for r in row_reqs):
+            logits = logits.clone() if logits is self.logits_d else logits
+            self._apply_penalties(logits, row_reqs)
         new_toks = self._sample_rows(logits, self.temps_d[:lim], self.top_p)
         # advance device state without host staging
         self.toks_d[:lim].copy_(new_toks.long())

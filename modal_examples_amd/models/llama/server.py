@@ -66,11 +66,14 @@ class LLMServer:
                     ev.set()
 
     def submit(self, prompt: str, max_tokens: int = 64, temperature: float = 0.0,
-               stream_cb=None) -> int:
+               stream_cb=None, presence_penalty: float = 0.0,
+               frequency_penalty: float = 0.0) -> int:
         ids = self.tok.encode(prompt)
         with self._lock:
-            rid = self.engine.add_request(ids, max_tokens, temperature,
-                                          stream_cb=stream_cb)
+            rid = self.engine.add_request(
+                ids, max_tokens, temperature, stream_cb=stream_cb,
+                presence_penalty=presence_penalty,
+                frequency_penalty=frequency_penalty)
             self._events[rid] = threading.Event()
         self._wake.set()
         return rid
@@ -100,8 +103,12 @@ class LLMServer:
         return text[:cut], "stop"
 
     def generate(self, prompt: str, max_tokens: int = 64,
-                 temperature: float = 0.0, stop=None) -> str:
-        rid = self.submit(prompt, max_tokens, temperature)
+                 temperature: float = 0.0, stop=None,
+                 presence_penalty: float = 0.0,
+                 frequency_penalty: float = 0.0) -> str:
+        rid = self.submit(prompt, max_tokens, temperature,
+                          presence_penalty=presence_penalty,
+                          frequency_penalty=frequency_penalty)
         r = self.wait(rid)
         text, _ = self.apply_stop(self.tok.decode(r.out_tokens), stop)
         return text
@@ -175,14 +182,18 @@ def create_openai_app(server: LLMServer):
                 "data": [{"id": server.model_name, "object": "model"}]}
 
     async def _run(prompt: str, max_tokens: int, temperature: float,
-                   stream: bool, chat: bool, stop=None):
+                   stream: bool, chat: bool, stop=None,
+                   presence_penalty: float = 0.0,
+                   frequency_penalty: float = 0.0):
         created = int(time.time())
         rid_str = f"cmpl-{uuid.uuid4().hex[:12]}"
         if not stream:
             loop = asyncio.get_running_loop()
             text = await loop.run_in_executor(
-                None, lambda: server.generate(prompt, max_tokens, temperature,
-                                              stop=stop))
+                None, lambda: server.generate(
+                    prompt, max_tokens, temperature, stop=stop,
+                    presence_penalty=presence_penalty,
+                    frequency_penalty=frequency_penalty))
             usage = {"prompt_tokens": len(server.tok.encode(prompt)),
                      "completion_tokens": len(text.split()),
                      "total_tokens": len(server.tok.encode(prompt)) + len(text.split())}
@@ -258,7 +269,11 @@ def create_openai_app(server: LLMServer):
         return await _run(body.get("prompt", ""), int(body.get("max_tokens", 64)),
                           float(body.get("temperature", 0.0)),
                           bool(body.get("stream", False)), chat=False,
-                          stop=body.get("stop"))
+                          stop=body.get("stop"),
+                          presence_penalty=float(
+                              body.get("presence_penalty", 0) or 0),
+                          frequency_penalty=float(
+                              body.get("frequency_penalty", 0) or 0))
 
     @app.post("/v1/chat/completions")
     async def chat_completions(body: dict):
@@ -267,7 +282,11 @@ def create_openai_app(server: LLMServer):
         return await _run(prompt, int(body.get("max_tokens", 64)),
                           float(body.get("temperature", 0.0)),
                           bool(body.get("stream", False)), chat=True,
-                          stop=body.get("stop"))
+                          stop=body.get("stop"),
+                          presence_penalty=float(
+                              body.get("presence_penalty", 0) or 0),
+                          frequency_penalty=float(
+                              body.get("frequency_penalty", 0) or 0))
 
     return app
 

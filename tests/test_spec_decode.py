@@ -85,3 +85,43 @@ def test_spec_mixed_temperature_batch():
     # greedy request must match a plain engine run of the same prompt
     want, _ = _greedy(_mk(0), [[3, 4, 5, 3, 4, 5, 3, 4]], n=6)
     assert outs[0] == want[0]
+
+
+def test_frequency_penalty_prevents_repeats():
+    """A huge frequency penalty makes every greedy output token distinct;
+    penalty=0 reproduces the plain run exactly."""
+    prompt = [5, 6, 7, 8]
+    base, _ = _greedy(_mk(0), [prompt], n=10)
+
+    eng = _mk(0)
+    eng.add_request(prompt, max_new_tokens=10, temperature=0.0,
+                    frequency_penalty=1e9)
+    while eng.has_work:
+        eng.step()
+    toks = eng.finished[1].out_tokens
+    assert len(set(toks)) == len(toks) == 10
+    assert toks[0] == base[0][0]  # first token unaffected (no output yet)
+
+    eng0 = _mk(0)
+    eng0.add_request(prompt, max_new_tokens=10, temperature=0.0,
+                     presence_penalty=0.0, frequency_penalty=0.0)
+    while eng0.has_work:
+        eng0.step()
+    assert eng0.finished[1].out_tokens == base[0]
+
+
+def test_penalized_requests_skip_speculation_but_match():
+    """Penalties disable drafts (they evolve within a run) — output equals
+    an unspeculated penalized engine."""
+    def run(spec):
+        eng = _mk(spec)
+        eng.add_request([3, 4, 5, 3, 4, 5], max_new_tokens=8,
+                        temperature=0.0, presence_penalty=2.0)
+        while eng.has_work:
+            eng.step()
+        return eng.finished[1].out_tokens, eng.spec_proposed
+
+    plain, _ = run(0)
+    spec, proposed = run(4)
+    assert spec == plain
+    assert proposed == 0  # no drafts were even offered
