@@ -71,3 +71,45 @@ def test_unordered_map_large_window_event_driven():
 
     out = sorted(inc.map(range(1000), order_outputs=False))
     assert out == list(range(1, 1001))
+
+
+def test_scaledown_churn_under_intermittent_load():
+    """Aggressive scaledown + bursts arriving between idle gaps: every
+    burst must complete even while the reaper is retiring workers (the
+    reap-on-boot liveness class of bug)."""
+    import time
+
+    app = modal.App("stress-churn")
+
+    @app.function(scaledown_window=0.15, max_containers=3)
+    def work(x: int) -> int:
+        time.sleep(0.02)
+        return x * 2
+
+    for burst in range(4):
+        got = sorted(work.map(range(8)))
+        assert got == [x * 2 for x in range(8)]
+        time.sleep(0.4)  # let the reaper take everything down between bursts
+
+
+def test_cancel_does_not_poison_the_pool():
+    """Cancel an in-flight call (tears its container down), then the pool
+    must still serve fresh calls promptly."""
+    import time
+
+    app = modal.App("stress-cancel-recover")
+
+    @app.function(max_containers=2)
+    def slow_or_fast(x: int) -> int:
+        import time as t
+
+        if x < 0:
+            t.sleep(30)
+        return x + 1
+
+    fc = slow_or_fast.spawn(-1)
+    time.sleep(0.5)  # let it start executing
+    fc.cancel()
+    t0 = time.monotonic()
+    assert slow_or_fast.remote(41) == 42
+    assert time.monotonic() - t0 < 20, "pool did not recover after cancel"
