@@ -34,7 +34,8 @@ def flow_sigmas(steps: int, shift: float = 1.0) -> torch.Tensor:
 class FluxPipeline:
     def __init__(self, cfg: Optional[MMDiTConfig] = None, device: str = "cuda",
                  dtype=torch.bfloat16, latent_size: int = 128,
-                 use_graph: bool = True, seed: int = 0, graph_cache: int = 4):
+                 use_graph: bool = True, seed: int = 0, graph_cache: int = 4,
+                 init_weights: bool = True):
         self.cfg = cfg or MMDiTConfig.schnell()
         self.device = torch.device(device)
         self.dtype = dtype
@@ -45,10 +46,24 @@ class FluxPipeline:
 
             kernel_cache.restore()
         torch.manual_seed(seed)
-        with torch.device(self.device):
-            self.model = MMDiT(self.cfg).to(self.device, dtype)
-            vae_cls = VAEDecoder if self.cfg.hidden >= 1024 else VAEDecoderSmall
-            self.vae = vae_cls().to(self.device, dtype)
+        vae_cls = VAEDecoder if self.cfg.hidden >= 1024 else VAEDecoderSmall
+        if init_weights:
+            with torch.device(self.device):
+                self.model = MMDiT(self.cfg).to(self.device, dtype)
+                self.vae = vae_cls().to(self.device, dtype)
+        else:
+            # cold-restore: meta-build in target dtype, weights assigned by
+            # from_safetensors (gpu/fastload.py blob views)
+            prev = torch.get_default_dtype()
+            try:
+                torch.set_default_dtype(dtype)
+                with torch.device("meta"):
+                    self.model = MMDiT(self.cfg)
+                    self.vae = vae_cls()
+            finally:
+                torch.set_default_dtype(prev)
+            self.model = self.model.to_empty(device=self.device)
+            self.vae = self.vae.to_empty(device=self.device)
         self.model.eval()
         self.vae.eval()
         self._graphs = GraphLRU(graph_cache)
@@ -151,3 +166,30 @@ class FluxPipeline:
     def param_count(self) -> int:
         return sum(p.numel() for p in self.model.parameters()) + sum(
             p.numel() for p in self.vae.parameters())
+
+    # ------------------------------------------------ cold boot
+
+    def save_safetensors(self, path: str) -> int:
+        """Bake mmdit+vae into one safetensors-layout file (the compile-
+        cache Volume role of flux.py:246-272 — here the bakeable artifact is
+        the weights; graphs capture in seconds at boot)."""
+        from ...gpu import fastload
+
+        state = {f"model.{k}": v for k, v in self.model.state_dict().items()}
+        state.update({f"vae.{k}": v for k, v in self.vae.state_dict().items()})
+        return fastload.save_file(state, path)
+
+    @classmethod
+    def from_safetensors(cls, path: str, device: str = "cuda",
+                         **kw) -> "FluxPipeline":
+        from ...gpu import fastload
+
+        pipe = cls(device=device, init_weights=False, **kw)
+        sd = fastload.load_file(path, device=device)
+        pipe.model.load_state_dict(
+            {k[6:]: v for k, v in sd.items() if k.startswith("model.")},
+            assign=True)
+        pipe.vae.load_state_dict(
+            {k[4:]: v for k, v in sd.items() if k.startswith("vae.")},
+            assign=True)
+        return pipe

@@ -51,13 +51,24 @@ def log_mel_spectrogram(audio: torch.Tensor, n_mels: int, n_frames: int) -> torc
 
 class WhisperPipeline:
     def __init__(self, cfg: Optional[WhisperConfig] = None, device: str = "cuda",
-                 dtype=torch.bfloat16, seed: int = 0):
+                 dtype=torch.bfloat16, seed: int = 0, init_weights: bool = True):
         self.cfg = cfg or WhisperConfig.large_v3()
         self.device = torch.device(device)
         self.dtype = dtype
         torch.manual_seed(seed)
-        with torch.device(self.device):
-            self.model = WhisperModel(self.cfg).to(self.device, dtype)
+        if init_weights:
+            with torch.device(self.device):
+                self.model = WhisperModel(self.cfg).to(self.device, dtype)
+        else:
+            # cold-restore: meta-build in target dtype (from_safetensors)
+            prev = torch.get_default_dtype()
+            try:
+                torch.set_default_dtype(dtype)
+                with torch.device("meta"):
+                    self.model = WhisperModel(self.cfg)
+            finally:
+                torch.set_default_dtype(prev)
+            self.model = self.model.to_empty(device=self.device)
         self.model.eval()
         self.sot, self.eot = 1, 2  # synthetic special tokens
         self.use_graph = self.device.type == "cuda"
@@ -187,3 +198,21 @@ class WhisperPipeline:
     def transcribe_text(self, audio_batch, max_tokens: int = 32) -> List[str]:
         return [" ".join(f"t{t}" for t in seq if t > 2)
                 for seq in self.transcribe(audio_batch, max_tokens)]
+
+    # ------------------------------------------------ cold boot
+
+    def save_safetensors(self, path: str) -> int:
+        """Bake the model for `from_safetensors` cold boot."""
+        from ...gpu import fastload
+
+        return fastload.save_file(dict(self.model.state_dict()), path)
+
+    @classmethod
+    def from_safetensors(cls, path: str, device: str = "cuda",
+                         **kw) -> "WhisperPipeline":
+        from ...gpu import fastload
+
+        pipe = cls(device=device, init_weights=False, **kw)
+        pipe.model.load_state_dict(fastload.load_file(path, device=device),
+                                   assign=True)
+        return pipe

@@ -151,3 +151,33 @@ def test_fp8_and_scalar_roundtrip(tmp_path):
     assert torch.equal(out["kv"].view(torch.uint8), state["kv"].view(torch.uint8))
     assert out["scalar"].item() == 3.5 and out["scalar"].shape == ()
     assert out["empty"].shape == (0, 7) and out["empty"].dtype == torch.bfloat16
+
+
+def test_flux_and_whisper_safetensors_roundtrip(tmp_path):
+    """Every flagship pipeline cold-boots from one baked file."""
+    from modal_examples_amd.models.flux.mmdit import MMDiTConfig
+    from modal_examples_amd.models.flux.pipeline import FluxPipeline
+    from modal_examples_amd.models.whisper.model import WhisperConfig
+    from modal_examples_amd.models.whisper.pipeline import WhisperPipeline
+
+    fcfg = MMDiTConfig.small() if hasattr(MMDiTConfig, "small") else None
+    if fcfg is not None:
+        src = FluxPipeline(fcfg, device="cpu", use_graph=False, latent_size=8)
+        p = str(tmp_path / "flux.safetensors")
+        src.save_safetensors(p)
+        pipe = FluxPipeline.from_safetensors(
+            p, device="cpu", cfg=fcfg, use_graph=False, latent_size=8)
+        for (ka, va), (kb, vb) in zip(src.model.state_dict().items(),
+                                      pipe.model.state_dict().items()):
+            assert ka == kb and torch.equal(va, vb), ka
+
+    wcfg = WhisperConfig.small_test()
+    if wcfg is not None:
+        src = WhisperPipeline(wcfg, device="cpu")
+        p = str(tmp_path / "wh.safetensors")
+        src.save_safetensors(p)
+        pipe = WhisperPipeline.from_safetensors(p, device="cpu", cfg=wcfg)
+        for (ka, va), (kb, vb) in zip(src.model.state_dict().items(),
+                                      pipe.model.state_dict().items()):
+            assert ka == kb and torch.equal(va, vb), ka
+    assert fcfg is not None or wcfg is not None, "no small configs found"
