@@ -87,10 +87,11 @@ def paged_decode(q, k_cache, v_cache, block_table, seq_lens, block_size: int = 0
         return ref.paged_decode_ref(q, k_cache, v_cache, block_table, seq_lens,
                                     block_size, scale)
     # .contiguous() also materializes expand()ed (stride-0) block tables —
-    # the kernel indexes bt + row*stride and would read out of bounds
-    return ext.paged_decode(q.contiguous(), k_cache, v_cache,
-                            block_table.contiguous(), seq_lens.int(),
-                            block_size, scale)
+    # the kernel indexes bt + row*stride and would read out of bounds.
+    # block_table=None = contiguous (non-paged) cache mode.
+    bt = block_table.contiguous() if block_table is not None else None
+    return ext.paged_decode(q.contiguous(), k_cache, v_cache, bt,
+                            seq_lens.int(), block_size, scale)
 
 
 def groupnorm_silu(x, gamma, beta, groups: int = 32, eps: float = 1e-5,

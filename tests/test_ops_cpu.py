@@ -117,3 +117,25 @@ def test_glu_fused_cpu_fallback():
     want = (torch.nn.functional.gelu(a.float(), approximate="tanh") * b.float())
     assert torch.allclose(geglu, want, atol=2e-2)
     assert swiglu.shape == (3, 5, 24)
+
+
+def test_paged_decode_wrapper_preserves_none_block_table(monkeypatch):
+    """Contiguous (non-paged) mode passes block_table=None through the
+    extension wrapper — the stride-0-hardening .contiguous() must not be
+    applied to None (GPU-tier regression found on hardware)."""
+    seen = {}
+
+    class FakeExt:
+        @staticmethod
+        def paged_decode(q, k, v, bt, lens, bs, scale):
+            seen["bt"] = bt
+            return torch.zeros(q.shape[0], q.shape[1], q.shape[2])
+
+    monkeypatch.setattr(F, "_ext_for", lambda *a: FakeExt())
+    q = torch.randn(2, 4, 64, dtype=torch.bfloat16)
+    kc = torch.randn(2, 4, 16, 64, dtype=torch.bfloat16)
+    F.paged_decode(q, kc, kc, None, torch.tensor([5, 3]))
+    assert seen["bt"] is None
+    bt = torch.zeros(1, 4, dtype=torch.int32).expand(2, -1)
+    F.paged_decode(q, kc, kc, bt, torch.tensor([5, 3]))
+    assert seen["bt"].is_contiguous() and seen["bt"].shape == (2, 4)
