@@ -26,3 +26,13 @@ def test_bench_help_and_defaults():
     assert r.returncode == 0
     for flag in ("--gpus", "--steps", "--warmup"):
         assert flag in r.stdout
+
+
+def test_typecheck_tier_passes():
+    """tools/typecheck.py (internal/typecheck.py role) is green."""
+    import subprocess
+    import sys
+
+    r = subprocess.run([sys.executable, "tools/typecheck.py"],
+                       capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stdout + r.stderr
