@@ -129,6 +129,14 @@ def test_openai_routes():
                 r2 = await c.post("/v1/completions", json={
                     "prompt": "abc def", "max_tokens": 3})
                 assert r2.json()["object"] == "text_completion"
+                # OpenAI penalties (reference client sends both:
+                # openai_compatible/client.py:24-27): a huge frequency
+                # penalty forbids repeats in the completion
+                r3 = await c.post("/v1/completions", json={
+                    "prompt": "abc def", "max_tokens": 8,
+                    "frequency_penalty": 1e9})
+                toks = r3.json()["choices"][0]["text"].split()
+                assert len(set(toks)) == len(toks)
 
         asyncio.run(go())
     finally:
