@@ -1,4 +1,5 @@
 # ---
+# deploy: true
 # cmd: ["python", "-m", "modal_examples_amd", "run", "examples/07_web/badges.py"]
 # ---
 # # Dynamic SVG badge endpoint (07_web/badges.py role)

@@ -1,4 +1,5 @@
 # ---
+# deploy: true
 # cmd: ["python", "-m", "modal_examples_amd", "serve", "examples/07_web/basic_web.py", "--timeout", "3"]
 # ---
 # # Web endpoints

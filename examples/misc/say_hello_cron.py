@@ -1,4 +1,5 @@
 # ---
+# deploy: true
 # cmd: ["python", "-m", "modal_examples_amd", "run", "examples/misc/say_hello_cron.py"]
 # ---
 # # A deployed cron greeter (misc/say_hello_cron.py role).

@@ -36,3 +36,15 @@ def test_typecheck_tier_passes():
     r = subprocess.run([sys.executable, "tools/typecheck.py"],
                        capture_output=True, text=True, timeout=120)
     assert r.returncode == 0, r.stdout + r.stderr
+
+
+def test_deploy_all_cd_tier():
+    """tools/deploy_all.py (cd.yml / internal/deploy.py role) deploys every
+    `deploy: true` example."""
+    import subprocess
+    import sys
+
+    r = subprocess.run([sys.executable, "tools/deploy_all.py"],
+                       capture_output=True, text=True, timeout=280)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "deployed 3/3" in r.stdout, r.stdout
