@@ -125,3 +125,30 @@ def test_penalized_requests_skip_speculation_but_match():
     spec, proposed = run(4)
     assert spec == plain
     assert proposed == 0  # no drafts were even offered
+
+
+def test_ngram_propose_properties():
+    """Property sweep: any proposal is a verbatim continuation of an
+    earlier occurrence of the current suffix, length-capped at k."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    @settings(max_examples=200, deadline=None)
+    @given(st.lists(st.integers(0, 7), max_size=40), st.integers(1, 6))
+    def prop(ctx, k):
+        out = ngram_propose(ctx, k)
+        assert len(out) <= k
+        if out:
+            # some suffix n-gram of ctx occurs earlier, followed by `out`
+            found = False
+            for n in range(1, 4):
+                if len(ctx) < n + 1:
+                    continue
+                tail = ctx[-n:]
+                for i in range(len(ctx) - n):
+                    if ctx[i:i + n] == tail and \
+                            ctx[i + n:i + n + len(out)] == out:
+                        found = True
+            assert found, (ctx, k, out)
+
+    prop()
