@@ -25,6 +25,11 @@ def main():
     ap.add_argument("--no-graph", action="store_true")
     ap.add_argument("--small", action="store_true")
     ap.add_argument("--kv-dtype", choices=("bf16", "fp8"), default="bf16")
+    ap.add_argument("--spec", type=int, default=0,
+                    help="ngram speculative tokens (eager decode path)")
+    ap.add_argument("--prefix-cache", action="store_true")
+    ap.add_argument("--chunked", type=int, default=0,
+                    help="chunked-prefill size (0 = monolithic)")
     ap.add_argument("--ragged", action="store_true",
                     help="mixed prompt lengths (uniform prompt_len/4 .. "
                          "prompt_len) — exercises the one-padded-forward "
@@ -43,8 +48,12 @@ def main():
     dtype = torch.bfloat16 if device == "cuda" else torch.float32
     t0 = time.perf_counter()
     eng = LlamaEngine(cfg, device=device, dtype=dtype,
-                      use_graph=not args.no_graph and device == "cuda",
-                      max_batch=args.requests, kv_dtype=args.kv_dtype)
+                      use_graph=(not args.no_graph and device == "cuda"
+                                 and args.spec == 0),
+                      max_batch=args.requests, kv_dtype=args.kv_dtype,
+                      spec_tokens=args.spec,
+                      prefix_cache=args.prefix_cache,
+                      chunked_prefill=args.chunked)
     eng.warmup()  # decode-graph capture is cold-start work
     if device == "cuda":
         torch.cuda.synchronize()
@@ -82,6 +91,9 @@ def main():
         "kv_blocks": eng.num_blocks,
         "kv_dtype": args.kv_dtype,
         "hipgraph": eng.use_graph,
+        "spec": {"proposed": eng.spec_proposed, "accepted": eng.spec_accepted}
+        if args.spec else None,
+        "prefix_hit_tokens": eng.prefix_hit_tokens if args.prefix_cache else None,
     }), flush=True)
 
 
