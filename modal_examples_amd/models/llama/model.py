@@ -33,6 +33,8 @@ class LlamaConfig:
     rope_base: float = 500000.0
     norm_eps: float = 1e-5
     max_seq: int = 8192
+    n_experts: int = 0   # >0: MoE FFN (Mixtral/DeepSeek-class)
+    experts_per_tok: int = 2
 
     @staticmethod
     def llama3_8b() -> "LlamaConfig":
@@ -43,6 +45,12 @@ class LlamaConfig:
         return LlamaConfig(vocab_size=1024, dim=256, n_layers=2, n_heads=4,
                            n_kv_heads=2, head_dim=64, ffn_dim=512,
                            rope_base=10000.0, max_seq=512)
+
+    @staticmethod
+    def moe_small() -> "LlamaConfig":
+        return LlamaConfig(vocab_size=1024, dim=256, n_layers=2, n_heads=4,
+                           n_kv_heads=2, head_dim=64, ffn_dim=256,
+                           rope_base=10000.0, max_seq=512, n_experts=4)
 
 
 class LlamaBlock(nn.Module):
@@ -63,8 +71,16 @@ class LlamaBlock(nn.Module):
         self.qkv = nn.Linear(d, (self.nq + 2 * self.nkv) * hd, bias=False)
         self.o_proj = nn.Linear(self.nq * hd, d, bias=False)
         self.ffn_norm = RMSNormK(d, cfg.norm_eps)
-        self.gate_up = nn.Linear(d, 2 * cfg.ffn_dim // tp_world, bias=False)
-        self.down = nn.Linear(cfg.ffn_dim // tp_world, d, bias=False)
+        if cfg.n_experts > 0:
+            assert tp_world == 1, "MoE + TP not supported (EP is a non-goal)"
+            self.moe = MoEFFN(d, cfg.ffn_dim, cfg.n_experts,
+                              cfg.experts_per_tok)
+            self.gate_up = self.down = None
+        else:
+            self.moe = None
+            self.gate_up = nn.Linear(d, 2 * cfg.ffn_dim // tp_world,
+                                     bias=False)
+            self.down = nn.Linear(cfg.ffn_dim // tp_world, d, bias=False)
         self.hd = hd
         self.ffn_dim = cfg.ffn_dim // tp_world
 
@@ -80,6 +96,8 @@ class LlamaBlock(nn.Module):
                 v.view(B, S, self.nkv, self.hd))
 
     def ffn(self, x):
+        if self.moe is not None:
+            return self.moe(x)
         # fused kernel reads both halves of the gate_up output in place
         return self.down(OF.glu_fused(self.gate_up(x), gelu=False))
 
@@ -201,3 +219,40 @@ def shard_llama_state(state: dict, cfg: LlamaConfig, rank: int,
         else:
             out[k] = v
     return out
+
+
+class MoEFFN(nn.Module):
+    """Top-k mixture-of-experts FFN (the MoE model class the reference
+    serves: deepseek_v4.py, gpt_oss_inference.py, misc/trtllm_deepseek.py —
+    Mixtral/DeepSeek-style routed experts).
+
+    Router scores per token; top-k experts run their SwiGLU on the tokens
+    routed to them (token-dispatch loop: E is small, every expert GEMM is a
+    dense hipBLASLt call over its token group; glu_fused reads both halves
+    of the fused gate_up in place)."""
+
+    def __init__(self, dim: int, ffn_dim: int, n_experts: int, top_k: int):
+        super().__init__()
+        self.router = nn.Linear(dim, n_experts, bias=False)
+        self.gate_up = nn.Parameter(
+            torch.empty(n_experts, 2 * ffn_dim, dim).normal_(std=0.02))
+        self.down = nn.Parameter(
+            torch.empty(n_experts, dim, ffn_dim).normal_(std=0.02))
+        self.n_experts, self.top_k = n_experts, top_k
+
+    def forward(self, x):
+        from ...ops import functional as OF
+
+        B, S, d = x.shape
+        flat = x.reshape(-1, d)
+        scores = self.router(flat).float()                  # [N, E]
+        w, idx = scores.topk(self.top_k, dim=-1)            # [N, k]
+        w = torch.softmax(w, dim=-1).to(x.dtype)
+        out = torch.zeros_like(flat)
+        for e in idx.unique().tolist():
+            hit = (idx == e)                                # [N, k]
+            rows = hit.any(-1).nonzero(as_tuple=True)[0]
+            contrib = (w * hit).sum(-1)[rows].unsqueeze(-1)  # routed weight
+            h = OF.glu_fused(flat[rows] @ self.gate_up[e].T, gelu=False)
+            out[rows] += contrib * (h @ self.down[e].T)
+        return out.view(B, S, d)
