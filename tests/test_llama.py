@@ -406,3 +406,33 @@ def test_ragged_prefill_random_mixes(seed):
             if sid in solo.finished:
                 break
         assert solo.finished[sid].out_tokens == eng.finished[rid].out_tokens, p
+
+
+def test_server_concurrent_submitters_with_features():
+    """16 threads submit against one server (prefix cache + speculation
+    on): every request completes, and identical greedy prompts produce
+    identical outputs regardless of interleaving."""
+    import concurrent.futures
+
+    from modal_examples_amd.models.llama.model import LlamaConfig
+
+    eng = LlamaEngine(LlamaConfig.small(), device="cpu",
+                      dtype=torch.bfloat16, use_graph=False, eos_id=-1,
+                      seed=0, prefix_cache=True, spec_tokens=3,
+                      kv_blocks=256, max_batch=8)
+    srv = LLMServer(eng, "conc-test")
+    try:
+        def one(i):
+            prompt = "shared system preamble " * 3 + f"q{i % 4}"
+            return (i % 4, srv.generate(prompt, max_tokens=6))
+
+        with concurrent.futures.ThreadPoolExecutor(16) as ex:
+            results = list(ex.map(one, range(32)))
+        by_prompt = {}
+        for key, text in results:
+            by_prompt.setdefault(key, set()).add(text)
+        assert len(results) == 32
+        for key, texts in by_prompt.items():
+            assert len(texts) == 1, f"prompt {key} diverged: {texts}"
+    finally:
+        srv.shutdown()
