@@ -106,7 +106,10 @@ class LlamaEngine:
         self.device = torch.device(device)
         self.dtype = dtype
         self.max_batch = max_batch
-        self.use_graph = use_graph and self.device.type == "cuda"
+        # spec mode decodes through the expanded eager path; a captured
+        # full-width decode graph would never replay — don't pay its capture
+        self.use_graph = (use_graph and self.device.type == "cuda"
+                          and spec_tokens == 0)
         self.eos_id = eos_id
         self.top_p = top_p
         self.seed = seed
