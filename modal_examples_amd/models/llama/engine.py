@@ -43,6 +43,7 @@ class Request:
     pf_done: int = 0  # prefilled tokens (chunked-prefill progress)
     presence_penalty: float = 0.0
     frequency_penalty: float = 0.0
+    stop_seqs: Optional[List[List[int]]] = None  # token-level early stop
     slot: int = -1
     done: bool = False
     error: Optional[str] = None
@@ -225,7 +226,8 @@ class LlamaEngine:
     def add_request(self, prompt: List[int], max_new_tokens: int = 64,
                     temperature: float = 0.0, stream_cb=None,
                     presence_penalty: float = 0.0,
-                    frequency_penalty: float = 0.0) -> int:
+                    frequency_penalty: float = 0.0,
+                    stop_seqs: Optional[List[List[int]]] = None) -> int:
         limit = self.cfg.max_seq
         if len(prompt) >= limit:
             prompt = prompt[-(limit - 1):]  # keep the most recent context
@@ -234,7 +236,8 @@ class LlamaEngine:
             max_new_tokens = max(1, limit - len(prompt))
         r = Request(self._next_id, list(prompt), max_new_tokens, temperature,
                     stream_cb=stream_cb, presence_penalty=presence_penalty,
-                    frequency_penalty=frequency_penalty)
+                    frequency_penalty=frequency_penalty,
+                    stop_seqs=[list(q) for q in stop_seqs or [] if q])
         self._next_id += 1
         self.waiting.append(r)
         return r.req_id
@@ -414,6 +417,13 @@ class LlamaEngine:
                 pass
         if tok == self.eos_id or len(r.out_tokens) >= r.max_new_tokens:
             r.done = True
+        elif r.stop_seqs:
+            # vLLM-style server-side stop: end generation the moment the
+            # output ends with any stop token sequence (no wasted decode)
+            for q in r.stop_seqs:
+                if len(r.out_tokens) >= len(q) and                         r.out_tokens[-len(q):] == q:
+                    r.done = True
+                    break
 
     # ------------------------------------------------ sampling
 

@@ -28,6 +28,9 @@ class SyntheticTokenizer:
     def encode(self, text: str) -> List[int]:
         ids = [self.bos]
         for w in text.split():
+            if w[:1] == "t" and w[1:].isdigit() and int(w[1:]) < self.vocab_size:
+                ids.append(int(w[1:]))  # decode() round-trips: t<id> -> id
+                continue
             h = int.from_bytes(hashlib.md5(w.encode()).digest()[:4], "little")
             ids.append(10 + h % (self.vocab_size - 10))
         return ids
@@ -67,13 +70,16 @@ class LLMServer:
 
     def submit(self, prompt: str, max_tokens: int = 64, temperature: float = 0.0,
                stream_cb=None, presence_penalty: float = 0.0,
-               frequency_penalty: float = 0.0) -> int:
+               frequency_penalty: float = 0.0, stop=None) -> int:
         ids = self.tok.encode(prompt)
+        seqs = [stop] if isinstance(stop, str) else list(stop or [])
+        stop_ids = [self.tok.encode(q)[1:] for q in seqs]  # drop BOS
         with self._lock:
             rid = self.engine.add_request(
                 ids, max_tokens, temperature, stream_cb=stream_cb,
                 presence_penalty=presence_penalty,
-                frequency_penalty=frequency_penalty)
+                frequency_penalty=frequency_penalty,
+                stop_seqs=[q for q in stop_ids if q])
             self._events[rid] = threading.Event()
         self._wake.set()
         return rid
@@ -108,9 +114,17 @@ class LLMServer:
                  frequency_penalty: float = 0.0) -> str:
         rid = self.submit(prompt, max_tokens, temperature,
                           presence_penalty=presence_penalty,
-                          frequency_penalty=frequency_penalty)
+                          frequency_penalty=frequency_penalty, stop=stop)
         r = self.wait(rid)
-        text, _ = self.apply_stop(self.tok.decode(r.out_tokens), stop)
+        toks = list(r.out_tokens)
+        # engine-side early stop leaves the matched stop tokens at the
+        # tail — strip them (OpenAI: stop sequence is not returned)
+        for q in ([stop] if isinstance(stop, str) else list(stop or [])):
+            qi = self.tok.encode(q)[1:]
+            if qi and len(toks) >= len(qi) and toks[-len(qi):] == qi:
+                toks = toks[: -len(qi)]
+                break
+        text, _ = self.apply_stop(self.tok.decode(toks), stop)
         return text
 
     def shutdown(self):

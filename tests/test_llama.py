@@ -287,7 +287,7 @@ def test_stop_sequences_truncate():
         assert len(words) >= 3
         stop_word = words[2]
         cut = srv.generate("stop test prompt", max_tokens=8, stop=stop_word)
-        assert cut == " ".join(words[:2]) + " "
+        assert cut.split() == words[:2]
         assert stop_word not in cut
 
         api = create_openai_app(srv)
@@ -434,5 +434,27 @@ def test_server_concurrent_submitters_with_features():
         assert len(results) == 32
         for key, texts in by_prompt.items():
             assert len(texts) == 1, f"prompt {key} diverged: {texts}"
+    finally:
+        srv.shutdown()
+
+
+def test_engine_stops_at_stop_sequence_without_wasted_decode():
+    """Token-level stop (vLLM server-side stop role): generation ends the
+    step the stop sequence appears; the stop tokens are stripped."""
+    from modal_examples_amd.models.llama.model import LlamaConfig
+
+    eng = LlamaEngine(LlamaConfig.small(), device="cpu",
+                      dtype=torch.bfloat16, use_graph=False, eos_id=-1,
+                      seed=0)
+    srv = LLMServer(eng, "stop-test")
+    try:
+        # discover what greedy emits, then stop at its 3rd token
+        full = srv.generate("abc def", max_tokens=8)
+        words = full.split()
+        assert len(words) == 8
+        stopped = srv.generate("abc def", max_tokens=8, stop=words[2])
+        assert stopped.split() == words[:2]
+        r = eng.finished[max(eng.finished)]
+        assert len(r.out_tokens) == 3  # stopped mid-generation, not post-hoc
     finally:
         srv.shutdown()
