@@ -64,6 +64,11 @@ class LLMServer:
             with self._lock:
                 done = self.engine.step()
             for r in done:
+                if r.stream_cb is not None:
+                    try:
+                        r.stream_cb(None)  # end-of-stream sentinel
+                    except Exception:
+                        pass
                 ev = self._events.pop(r.req_id, None)
                 if ev is not None:
                     ev.set()
@@ -232,7 +237,8 @@ def create_openai_app(server: LLMServer):
         def cb(tok_id):
             loop.call_soon_threadsafe(q.put_nowait, tok_id)
 
-        rid = server.submit(prompt, max_tokens, temperature, stream_cb=cb)
+        rid = server.submit(prompt, max_tokens, temperature, stream_cb=cb,
+                            stop=stop)  # engine-side early stop
 
         async def gen():
             sent = 0
@@ -242,6 +248,8 @@ def create_openai_app(server: LLMServer):
                     tok_id = await asyncio.wait_for(q.get(), timeout=120)
                 except asyncio.TimeoutError:
                     break
+                if tok_id is None:
+                    break  # request finished (eos / stop / max_tokens)
                 sent += 1
                 piece = f"t{tok_id} "
                 if stop:
