@@ -44,6 +44,8 @@ class Request:
     presence_penalty: float = 0.0
     frequency_penalty: float = 0.0
     stop_seqs: Optional[List[List[int]]] = None  # token-level early stop
+    want_logprobs: bool = False
+    out_logprobs: List[float] = field(default_factory=list)
     slot: int = -1
     done: bool = False
     error: Optional[str] = None
@@ -227,7 +229,8 @@ class LlamaEngine:
                     temperature: float = 0.0, stream_cb=None,
                     presence_penalty: float = 0.0,
                     frequency_penalty: float = 0.0,
-                    stop_seqs: Optional[List[List[int]]] = None) -> int:
+                    stop_seqs: Optional[List[List[int]]] = None,
+                    logprobs: bool = False) -> int:
         limit = self.cfg.max_seq
         if len(prompt) >= limit:
             prompt = prompt[-(limit - 1):]  # keep the most recent context
@@ -237,7 +240,8 @@ class LlamaEngine:
         r = Request(self._next_id, list(prompt), max_new_tokens, temperature,
                     stream_cb=stream_cb, presence_penalty=presence_penalty,
                     frequency_penalty=frequency_penalty,
-                    stop_seqs=[list(q) for q in stop_seqs or [] if q])
+                    stop_seqs=[list(q) for q in stop_seqs or [] if q],
+                    want_logprobs=logprobs)
         self._next_id += 1
         self.waiting.append(r)
         return r.req_id
@@ -352,10 +356,14 @@ class LlamaEngine:
         last_pos = torch.tensor([x - 1 for x in lens], device=self.device)
         logits = self.model.prefill(toks, kv_writer, last_pos=last_pos)
         temps = torch.tensor([r.temperature for r in group])
-        first = self._sample_rows(logits, temps, self.top_p).cpu()
+        first_d = self._sample_rows(logits, temps, self.top_p)
+        lps = (self._logprobs_of(logits, first_d)
+               if any(r.want_logprobs for r in group) else None)
+        first = first_d.cpu()
         for i, r in enumerate(group):
             r.pos = lens[i]
-            self._append_token(r, int(first[i]))
+            self._append_token(r, int(first[i]),
+                               lps[i] if lps is not None else None)
             if r.t_first_token is None:
                 r.t_first_token = time.monotonic()
             slot = r.slot
@@ -408,7 +416,15 @@ class LlamaEngine:
         self.prefix_hit_tokens += r.pf_done
         return True
 
-    def _append_token(self, r: Request, tok: int):
+    @staticmethod
+    def _logprobs_of(logits: torch.Tensor, toks: torch.Tensor) -> torch.Tensor:
+        """log P(chosen token) per row (OpenAI `logprobs` field)."""
+        lp = torch.log_softmax(logits.float(), dim=-1)
+        return lp.gather(1, toks.long().view(-1, 1))[:, 0].cpu()
+
+    def _append_token(self, r: Request, tok: int, lp: Optional[float] = None):
+        if r.want_logprobs and lp is not None:
+            r.out_logprobs.append(float(lp))
         r.out_tokens.append(tok)
         if r.stream_cb is not None:
             try:
@@ -530,8 +546,10 @@ class LlamaEngine:
         if end < len(feed):
             return False
         temps = torch.tensor([r.temperature])
-        first = self._sample_rows(logits[-1:], temps, self.top_p).cpu()
-        self._append_token(r, int(first[0]))
+        first_d = self._sample_rows(logits[-1:], temps, self.top_p)
+        lp = (self._logprobs_of(logits[-1:], first_d)[0]
+              if r.want_logprobs else None)
+        self._append_token(r, int(first_d.cpu()[0]), lp)
         if r.t_first_token is None:
             r.t_first_token = time.monotonic()
         slot = r.slot
@@ -623,7 +641,10 @@ class LlamaEngine:
                 row_reqs[base] = r  # penalized requests have exactly 1 row
         self._apply_penalties(logits, row_reqs)
         temps = torch.tensor(rows_temp, dtype=torch.float32)
-        out = self._sample_rows(logits, temps, self.top_p).cpu()
+        out_d = self._sample_rows(logits, temps, self.top_p)
+        all_lps = (self._logprobs_of(logits, out_d)
+                   if any(r.want_logprobs for r in self.running) else None)
+        out = out_d.cpu()
         for r, base, drafts in plan:
             emitted: List[int] = []
             for j in range(len(drafts) + 1):
@@ -634,8 +655,9 @@ class LlamaEngine:
                 break
             self.spec_accepted += len(emitted) - 1
             r.pos += len(emitted)
-            for t in emitted:
-                self._append_token(r, t)
+            for j, t in enumerate(emitted):
+                self._append_token(
+                    r, t, all_lps[base + j] if all_lps is not None else None)
                 if r.done:
                     break
 
@@ -679,6 +701,8 @@ class LlamaEngine:
             logits = logits.clone() if logits is self.logits_d else logits
             self._apply_penalties(logits, row_reqs)
         new_toks = self._sample_rows(logits, self.temps_d[:lim], self.top_p)
+        lps = (self._logprobs_of(logits, new_toks)
+               if any(r.want_logprobs for r in self.running) else None)
         # advance device state without host staging
         self.toks_d[:lim].copy_(new_toks.long())
         self.pos_d.add_(self.active_d)
@@ -686,7 +710,8 @@ class LlamaEngine:
         toks_host = new_toks.cpu()  # the one D2H sync per step
         for r in list(self.running):
             r.pos += 1
-            self._append_token(r, int(toks_host[r.slot]))
+            self._append_token(r, int(toks_host[r.slot]),
+                               lps[r.slot] if lps is not None else None)
 
     def _ensure_blocks(self, r: Request) -> bool:
         need = (r.pos + 1 + BLOCK - 1) // BLOCK

@@ -75,7 +75,8 @@ class LLMServer:
 
     def submit(self, prompt: str, max_tokens: int = 64, temperature: float = 0.0,
                stream_cb=None, presence_penalty: float = 0.0,
-               frequency_penalty: float = 0.0, stop=None) -> int:
+               frequency_penalty: float = 0.0, stop=None,
+               logprobs: bool = False) -> int:
         ids = self.tok.encode(prompt)
         seqs = [stop] if isinstance(stop, str) else list(stop or [])
         stop_ids = [self.tok.encode(q)[1:] for q in seqs]  # drop BOS
@@ -84,7 +85,7 @@ class LLMServer:
                 ids, max_tokens, temperature, stream_cb=stream_cb,
                 presence_penalty=presence_penalty,
                 frequency_penalty=frequency_penalty,
-                stop_seqs=[q for q in stop_ids if q])
+                stop_seqs=[q for q in stop_ids if q], logprobs=logprobs)
             self._events[rid] = threading.Event()
         self._wake.set()
         return rid
@@ -288,6 +289,28 @@ def create_openai_app(server: LLMServer):
 
     @app.post("/v1/completions")
     async def completions(body: dict):
+        if body.get("logprobs") and not body.get("stream"):
+            # eval-harness surface (vLLM serves the same field)
+            loop = asyncio.get_running_loop()
+
+            def run_lp():
+                rid = server.submit(body.get("prompt", ""),
+                                    int(body.get("max_tokens", 64)),
+                                    float(body.get("temperature", 0.0)),
+                                    logprobs=True)
+                return server.wait(rid)
+
+            r = await loop.run_in_executor(None, run_lp)
+            toks = [f"t{t}" for t in r.out_tokens]
+            return JSONResponse({
+                "id": f"cmpl-{uuid.uuid4().hex[:12]}",
+                "object": "text_completion", "model": server.model_name,
+                "choices": [{"index": 0,
+                             "text": server.tok.decode(r.out_tokens),
+                             "logprobs": {"tokens": toks,
+                                          "token_logprobs": r.out_logprobs},
+                             "finish_reason": "stop"}],
+            })
         return await _run(body.get("prompt", ""), int(body.get("max_tokens", 64)),
                           float(body.get("temperature", 0.0)),
                           bool(body.get("stream", False)), chat=False,

@@ -458,3 +458,35 @@ def test_engine_stops_at_stop_sequence_without_wasted_decode():
         assert len(r.out_tokens) == 3  # stopped mid-generation, not post-hoc
     finally:
         srv.shutdown()
+
+
+def test_logprobs_returned_and_sane():
+    """`logprobs` (eval-harness surface): one logprob per emitted token,
+    all finite and <= 0; greedy token is the argmax so its logprob is the
+    row max."""
+    import asyncio
+
+    import httpx
+
+    from modal_examples_amd.models.llama.model import LlamaConfig
+
+    eng = LlamaEngine(LlamaConfig.small(), device="cpu",
+                      dtype=torch.bfloat16, use_graph=False, eos_id=-1,
+                      seed=0, spec_tokens=3)
+    srv = LLMServer(eng, "lp-test")
+    app = create_openai_app(srv)
+    try:
+        async def go():
+            tr = httpx.ASGITransport(app=app)
+            async with httpx.AsyncClient(transport=tr,
+                                         base_url="http://t") as c:
+                r = await c.post("/v1/completions", json={
+                    "prompt": "alpha beta alpha beta", "max_tokens": 6,
+                    "logprobs": True})
+                return r.json()["choices"][0]["logprobs"]
+
+        lp = asyncio.run(go())
+        assert len(lp["tokens"]) == 6 == len(lp["token_logprobs"])
+        assert all(v <= 0 and v == v for v in lp["token_logprobs"])
+    finally:
+        srv.shutdown()
