@@ -289,6 +289,27 @@ def create_openai_app(server: LLMServer):
 
     @app.post("/v1/completions")
     async def completions(body: dict):
+        n = int(body.get("n", 1) or 1)
+        if n > 1 and not body.get("stream"):
+            # OpenAI `n`: independent sampled completions (per-row gumbel
+            # noise differs inside a batch, so parallel submits diverge)
+            loop = asyncio.get_running_loop()
+
+            def run_n():
+                rids = [server.submit(body.get("prompt", ""),
+                                      int(body.get("max_tokens", 64)),
+                                      float(body.get("temperature", 1.0)))
+                        for _ in range(n)]
+                return [server.wait(rid) for rid in rids]
+
+            rs = await loop.run_in_executor(None, run_n)
+            return JSONResponse({
+                "id": f"cmpl-{uuid.uuid4().hex[:12]}",
+                "object": "text_completion", "model": server.model_name,
+                "choices": [
+                    {"index": i, "text": server.tok.decode(r.out_tokens),
+                     "finish_reason": "stop"} for i, r in enumerate(rs)],
+            })
         if body.get("logprobs") and not body.get("stream"):
             # eval-harness surface (vLLM serves the same field)
             loop = asyncio.get_running_loop()

@@ -490,3 +490,33 @@ def test_logprobs_returned_and_sane():
         assert all(v <= 0 and v == v for v in lp["token_logprobs"])
     finally:
         srv.shutdown()
+
+
+def test_n_completions_diverge_when_sampled():
+    """OpenAI `n`: parallel sampled completions come back as n choices and
+    (at temperature 1) are not all identical."""
+    import asyncio
+
+    import httpx
+
+    from modal_examples_amd.models.llama.model import LlamaConfig
+
+    eng = LlamaEngine(LlamaConfig.small(), device="cpu",
+                      dtype=torch.bfloat16, use_graph=False, eos_id=-1,
+                      seed=0)
+    srv = LLMServer(eng, "n-test")
+    app = create_openai_app(srv)
+    try:
+        async def go():
+            tr = httpx.ASGITransport(app=app)
+            async with httpx.AsyncClient(transport=tr,
+                                         base_url="http://t") as c:
+                r = await c.post("/v1/completions", json={
+                    "prompt": "pick one", "max_tokens": 8, "n": 4,
+                    "temperature": 1.0})
+                return [ch["text"] for ch in r.json()["choices"]]
+
+        texts = asyncio.run(go())
+        assert len(texts) == 4 and len(set(texts)) >= 2
+    finally:
+        srv.shutdown()
