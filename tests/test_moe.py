@@ -75,3 +75,27 @@ def test_moe_cold_boot_roundtrip(tmp_path):
     toks = torch.randint(0, cfg.vocab_size, (1, 7),
                          generator=torch.Generator().manual_seed(3))
     assert torch.equal(src.model.prefill(toks), eng.model.prefill(toks))
+
+
+def test_moe_differential_fuzz_exact_features():
+    """Randomized greedy mixes through MoE + spec + prefix == plain MoE."""
+    def run(prompts, **kw):
+        eng = LlamaEngine(LlamaConfig.moe_small(), device="cpu",
+                          dtype=torch.bfloat16, use_graph=False, eos_id=-1,
+                          seed=0, kv_blocks=96, max_batch=3, **kw)
+        outs = []
+        for p in prompts:
+            rid = eng.add_request(p, max_new_tokens=4, temperature=0.0)
+            while eng.has_work:
+                eng.step()
+            outs.append(eng.finished[rid].out_tokens)
+        return outs
+
+    for seed in range(3):
+        g = torch.Generator().manual_seed(300 + seed)
+        base = torch.randint(0, 1024, (20,), generator=g).tolist()
+        prompts = [base + torch.randint(0, 1024, (1 + i,),
+                                        generator=g).tolist()
+                   for i in range(3)]
+        assert run(prompts, spec_tokens=3, prefix_cache=True) == \
+            run(prompts), seed
