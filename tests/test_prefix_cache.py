@@ -183,3 +183,29 @@ def test_differential_fuzz_exact_features_vs_plain():
                            spec_tokens=3, kv_blocks=64, max_batch=3)
         for p in prompts:
             assert _run_one(rich, p, n=5) == _run_one(plain, p, n=5), seed
+
+
+def test_mixed_admission_ragged_group_plus_prefix_hits():
+    """One step admits BOTH a prefix-hit request (chunk lane, suffix only)
+    and fresh ragged requests (group lane) — outputs match plain engine."""
+    g = torch.Generator().manual_seed(21)
+    system = torch.randint(0, 1024, (2 * BLOCK,), generator=g).tolist()
+    fresh_a = torch.randint(0, 1024, (7,), generator=g).tolist()
+    fresh_b = torch.randint(0, 1024, (12,), generator=g).tolist()
+
+    def run(pc):
+        eng = _mk(pc)
+        # warm the cache with the system prompt
+        r0 = eng.add_request(system + [1], max_new_tokens=3, temperature=0.0)
+        while eng.has_work:
+            eng.step()
+        # same step: one hit + two fresh ragged
+        r1 = eng.add_request(system + [2, 3], max_new_tokens=4,
+                             temperature=0.0)
+        r2 = eng.add_request(fresh_a, max_new_tokens=4, temperature=0.0)
+        r3 = eng.add_request(fresh_b, max_new_tokens=4, temperature=0.0)
+        while eng.has_work:
+            eng.step()
+        return [eng.finished[i].out_tokens for i in (r0, r1, r2, r3)]
+
+    assert run(True) == run(False)
