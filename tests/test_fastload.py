@@ -181,3 +181,27 @@ def test_flux_and_whisper_safetensors_roundtrip(tmp_path):
                                       pipe.model.state_dict().items()):
             assert ka == kb and torch.equal(va, vb), ka
     assert fcfg is not None or wcfg is not None, "no small configs found"
+
+
+def test_unicode_and_long_keys(tmp_path):
+    """Header JSON handles unicode and very long parameter names."""
+    p = str(tmp_path / "u.safetensors")
+    state = {
+        "ünïcødé.wéïght": torch.randn(3, 3),
+        ("blocks." + "x" * 300 + ".weight"): torch.randn(2),
+    }
+    fastload.save_file(state, p)
+    out = fastload.load_file(p)
+    for k in state:
+        assert torch.equal(out[k], state[k])
+
+
+def test_corrupt_header_raises_cleanly(tmp_path):
+    p = str(tmp_path / "bad.safetensors")
+    with open(p, "wb") as f:
+        f.write((1 << 20).to_bytes(8, "little"))  # header len > file size
+        f.write(b"{not json")
+    import json
+
+    with pytest.raises((json.JSONDecodeError, ValueError, OSError)):
+        fastload.load_file(p)
