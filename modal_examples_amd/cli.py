@@ -349,6 +349,10 @@ def main(argv=None):
         cmd_secret(rest)
     elif cmd == "app":
         cmd_app(rest)
+    elif cmd in ("--version", "version"):
+        from . import __version__
+
+        print(f"modal_examples_amd {__version__} (MI355X / ROCm)")
     else:
         raise SystemExit(
             f"unknown command {cmd!r}; use run/serve/deploy/shell/app/"
