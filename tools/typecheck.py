@@ -5,8 +5,7 @@ parallel type checking with py.typed opt-in; check.yml:10-36 ruff lint).
 This image has no mypy/ruff wheels, so the tier degrades to what the stdlib
 proves: every file must compile, and an AST pass enforces the lint rules the
 repo actually relies on (no bare excepts, no tab characters).  py.typed
-packages are
-listed so a mypy-equipped environment can run
+packages are listed so a mypy-equipped environment can run
 `mypy $(python tools/typecheck.py --list-typed)` directly.
 
 Usage: python tools/typecheck.py [--list-typed]
@@ -19,6 +18,7 @@ import sys
 from pathlib import Path
 
 REPO = Path(__file__).resolve().parent.parent
+
 
 def typed_packages() -> list:
     return sorted(str(p.parent) for p in REPO.rglob("py.typed")
