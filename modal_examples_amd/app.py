@@ -26,7 +26,6 @@ import time
 from dataclasses import dataclass, field
 from typing import Any, Callable, Dict, List, Optional
 
-from . import config
 from .exception import InvalidError, NotFoundError
 from .gpu.device_pool import DevicePool, parse_gpu
 from .runtime import ipc, store

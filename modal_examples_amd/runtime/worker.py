@@ -16,7 +16,6 @@ from __future__ import annotations
 import os
 from pathlib import Path
 import queue as _queue
-import sys
 import threading
 import traceback
 from concurrent.futures import ThreadPoolExecutor
