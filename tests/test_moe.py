@@ -99,3 +99,21 @@ def test_moe_differential_fuzz_exact_features():
                    for i in range(3)]
         assert run(prompts, spec_tokens=3, prefix_cache=True) == \
             run(prompts), seed
+
+
+def test_invalid_parallel_configs_fail_loudly():
+    import pytest as _pt
+
+    class FakeTP:
+        world, rank = 2, 0
+
+    # MoE + TP is a documented non-goal: loud assert, not silent wrongness
+    with _pt.raises(AssertionError):
+        LlamaModel(LlamaConfig.moe_small(), tp=FakeTP())
+
+    class FakeTP3:
+        world, rank = 3, 0
+
+    # tp must divide heads/ffn
+    with _pt.raises(AssertionError):
+        LlamaModel(LlamaConfig.small(), tp=FakeTP3())
