@@ -158,3 +158,13 @@ def greet():
     finally:
         proc.terminate()
         proc.wait(timeout=10)
+
+
+def test_version_flag(cli_env):
+    import subprocess
+    import sys
+
+    r = subprocess.run([sys.executable, "-m", "modal_examples_amd",
+                        "--version"], capture_output=True, text=True,
+                       timeout=60, env=cli_env)
+    assert r.returncode == 0 and "modal_examples_amd" in r.stdout
