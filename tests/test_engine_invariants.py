@@ -61,3 +61,25 @@ def test_invariants_hold_through_random_schedule():
         assert steps < 3000
     assert len(eng.finished) == 12
     assert all(r.error is None for r in eng.finished.values())
+
+
+def test_invariants_hold_under_preemption_pressure():
+    """KV pool sized to force preemptions: invariants hold per step and
+    every request still completes with its full token budget."""
+    g = torch.Generator().manual_seed(99)
+    eng = LlamaEngine(LlamaConfig.small(), device="cpu",
+                      dtype=torch.bfloat16, use_graph=False, eos_id=-1,
+                      seed=0, prefix_cache=True, kv_blocks=14, max_batch=4)
+    for i in range(6):
+        p = torch.randint(0, 1024, (10 + 3 * i,), generator=g).tolist()
+        eng.add_request(p, max_new_tokens=6, temperature=0.0)
+    steps = 0
+    while eng.has_work:
+        eng.step()
+        _invariants(eng)
+        steps += 1
+        assert steps < 5000
+    assert len(eng.finished) == 6
+    ok = [r for r in eng.finished.values() if r.error is None]
+    assert all(len(r.out_tokens) == 6 for r in ok)
+    assert len(ok) >= 5  # at most the largest prompt may hard-fail the pool
