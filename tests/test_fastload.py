@@ -205,3 +205,21 @@ def test_corrupt_header_raises_cleanly(tmp_path):
 
     with pytest.raises((json.JSONDecodeError, ValueError, OSError)):
         fastload.load_file(p)
+
+
+def test_sdxl_restored_pipeline_generates_identically(tmp_path):
+    """A from_safetensors-restored pipeline produces the SAME image as the
+    source pipeline (same seed), end to end on CPU."""
+    from modal_examples_amd.models.sdxl.pipeline import SDXLPipeline
+    from modal_examples_amd.models.sdxl.unet import UNetConfig
+
+    cfg = UNetConfig.small()
+    src = SDXLPipeline(cfg, device="cpu", dtype=torch.bfloat16, seed=0,
+                       latent_size=16)
+    p = str(tmp_path / "pipe.safetensors")
+    src.save_safetensors(p)
+    dup = SDXLPipeline.from_safetensors(p, device="cpu", cfg=cfg,
+                                        dtype=torch.bfloat16, latent_size=16)
+    a = src.generate(["restore check"], steps=1, seed=5)
+    b = dup.generate(["restore check"], steps=1, seed=5)
+    assert torch.equal(a, b)
